@@ -1,0 +1,120 @@
+"""Engine checkpoint save/load in the reference on-disk layout.
+
+Parity: reference `runtime/engine.py:4749` (save_checkpoint),
+`:4271` (load_checkpoint), name scheme `_get_zero_ckpt_prefix:4195`
+(`zero_pp_rank_{dp}_mp_rank_{mp:02}_optim_states.pt`,
+`mp_rank_{mp:02}_model_states.pt`, `latest` tag file).
+"""
+import os
+
+import torch
+
+from .. import comm as dist
+from ..utils.logging import log_dist, logger
+
+VERSION = "0.1.0-mi355x"
+
+
+def _tag(engine, tag):
+    return tag if tag is not None else f"global_step{engine.global_steps}"
+
+
+def _model_states_name(mp_rank=0):
+    return f"mp_rank_{mp_rank:02d}_model_states.pt"
+
+
+def _zero_ckpt_name(dp_rank, mp_rank=0):
+    return f"zero_pp_rank_{dp_rank}_mp_rank_{mp_rank:02d}_optim_states.pt"
+
+
+def save_checkpoint(engine, save_dir, tag=None, client_state=None,
+                    save_latest=True, exclude_frozen_parameters=False):
+    tag = _tag(engine, tag)
+    ckpt_dir = os.path.join(save_dir, str(tag))
+    os.makedirs(ckpt_dir, exist_ok=True)
+    dp_rank = engine.get_data_parallel_rank()
+    is_zero = hasattr(engine.optimizer, "state_dict") and \
+        not isinstance(engine.optimizer, torch.optim.Optimizer)
+
+    # model states: rank 0 of each DP group (mp_rank 0 — no TP yet)
+    if dp_rank == 0:
+        state = {
+            "module": engine.module_state_dict(
+                exclude_frozen_parameters=exclude_frozen_parameters),
+            "buffer_names": [n for n, _ in engine.module.named_buffers()],
+            "optimizer": None if is_zero else (
+                engine.optimizer.state_dict()
+                if hasattr(engine.optimizer, "state_dict") else None),
+            "lr_scheduler": (engine.lr_scheduler.state_dict()
+                             if engine.lr_scheduler is not None else None),
+            "ds_config": engine.config.to_dict(),
+            "ds_version": VERSION,
+            "global_steps": engine.global_steps,
+            "global_samples": engine.global_samples,
+            "skipped_steps": engine.skipped_steps,
+            "micro_steps": engine.micro_steps,
+        }
+        if client_state:
+            state.update(client_state)
+        torch.save(state, os.path.join(ckpt_dir, _model_states_name()))
+
+    # zero shards: every dp rank
+    if is_zero:
+        zstate = {"optimizer_state_dict": engine.optimizer.state_dict(),
+                  "ds_version": VERSION}
+        torch.save(zstate, os.path.join(ckpt_dir, _zero_ckpt_name(dp_rank)))
+
+    if dist.is_initialized():
+        dist.barrier()
+    if save_latest and dist.get_rank() == 0:
+        with open(os.path.join(save_dir, "latest"), "w") as f:
+            f.write(str(tag))
+    log_dist(f"saved checkpoint {ckpt_dir}", ranks=[0])
+    return True
+
+
+def load_checkpoint(engine, load_dir, tag=None, load_module_strict=True,
+                    load_optimizer_states=True, load_lr_scheduler_states=True,
+                    load_module_only=False):
+    if tag is None:
+        latest = os.path.join(load_dir, "latest")
+        if not os.path.exists(latest):
+            logger.warning(f"no 'latest' file in {load_dir}")
+            return None, {}
+        with open(latest) as f:
+            tag = f.read().strip()
+    ckpt_dir = os.path.join(load_dir, str(tag))
+    model_file = os.path.join(ckpt_dir, _model_states_name())
+    state = torch.load(model_file, map_location="cpu", weights_only=False)
+
+    is_zero = hasattr(engine.optimizer, "load_state_dict") and \
+        not isinstance(engine.optimizer, torch.optim.Optimizer)
+
+    if engine.zero_optimization_stage() != 3:
+        engine.load_module_state_dict(state["module"],
+                                      strict=load_module_strict)
+
+    if not load_module_only:
+        if is_zero:
+            dp_rank = engine.get_data_parallel_rank()
+            zfile = os.path.join(ckpt_dir, _zero_ckpt_name(dp_rank))
+            zstate = torch.load(zfile, map_location="cpu", weights_only=False)
+            engine.optimizer.load_state_dict(
+                zstate["optimizer_state_dict"],
+                load_optimizer_states=load_optimizer_states)
+        elif load_optimizer_states and state.get("optimizer") is not None \
+                and hasattr(engine.optimizer, "load_state_dict"):
+            engine.optimizer.load_state_dict(state["optimizer"])
+        if load_lr_scheduler_states and engine.lr_scheduler is not None \
+                and state.get("lr_scheduler") is not None:
+            engine.lr_scheduler.load_state_dict(state["lr_scheduler"])
+        engine.global_steps = state.get("global_steps", 0)
+        engine.global_samples = state.get("global_samples", 0)
+        engine.skipped_steps = state.get("skipped_steps", 0)
+        engine.micro_steps = state.get("micro_steps", 0)
+
+    client_state = {k: v for k, v in state.items()
+                    if k not in ("module", "optimizer", "lr_scheduler",
+                                 "buffer_names", "ds_config", "ds_version")}
+    log_dist(f"loaded checkpoint {ckpt_dir}", ranks=[0])
+    return ckpt_dir, client_state

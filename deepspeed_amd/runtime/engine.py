@@ -1,0 +1,452 @@
+"""DeepSpeedEngine — the training engine.
+
+Parity: reference `deepspeed/runtime/engine.py:252` (DeepSpeedEngine):
+`forward:2801`, `backward:3205`, `step:3405`, `save_checkpoint:4749`,
+`load_checkpoint:4271`, `_broadcast_model:1715`, `allreduce_gradients:2891`.
+
+MI355X-native: one process per GPU over RCCL; dtype default bf16; ZeRO
+optimizers own the grad comm (reduce-scatter on a side stream); stage-0 fp32
+falls back to bucketed allreduce at the GAS boundary.
+"""
+import os
+from typing import Callable, Optional
+
+import torch
+
+from .. import comm as dist
+from ..comm import groups
+from ..config import DeepSpeedConfig
+from ..ops.adam import FusedAdam
+from ..utils.logging import log_dist, logger
+from ..utils.timer import SynchronizedWallClockTimer, ThroughputTimer
+from . import lr_schedules
+from .loss_scaler import LossScalerBase
+from .utils import (CheckOverflow, get_grad_norm, see_memory_usage,
+                    clip_tensors_by_global_norm)
+
+MEMORY_OPT_ALLREDUCE_SIZE = int(5e8)
+
+
+class DeepSpeedEngine(torch.nn.Module):
+    def __init__(self,
+                 args=None,
+                 model=None,
+                 optimizer=None,
+                 model_parameters=None,
+                 training_data=None,
+                 lr_scheduler=None,
+                 mpu=None,
+                 dist_init_required=None,
+                 collate_fn=None,
+                 config=None,
+                 config_class: Optional[DeepSpeedConfig] = None,
+                 dont_change_device=False):
+        super().__init__()
+        self.module = model
+        self.client_optimizer = optimizer
+        self.client_lr_scheduler = lr_scheduler
+        self.training_data = training_data
+        self.collate_fn = collate_fn
+        self.mpu = mpu
+        self._config = config_class or DeepSpeedConfig(
+            config, world_size=dist.get_world_size())
+
+        self.global_steps = 0
+        self.global_samples = 0
+        self.micro_steps = 0
+        self.skipped_steps = 0
+        self._is_gradient_accumulation_boundary = None
+
+        self.device = (torch.device("cuda", dist.get_local_rank())
+                       if torch.cuda.is_available() else torch.device("cpu"))
+        self.timers = SynchronizedWallClockTimer()
+        self.tput_timer = ThroughputTimer(
+            batch_size=self.train_batch_size(), start_step=2)
+
+        if mpu is not None:
+            groups.set_mpu(mpu)
+        dist.configure(self._config)
+
+        sp = self._config.ulysses.sequence_parallel_size
+        if sp > 1 and not groups.sequence_parallel_is_initialized():
+            groups.initialize_sequence_parallel(sp)
+
+        self.dp_group = groups.get_sequence_data_parallel_group() \
+            if sp > 1 else groups.get_data_parallel_group()
+        self.dp_world_size = dist.get_world_size(self.dp_group)
+
+        self._configure_distributed_model()
+
+        self.monitor = self._configure_monitor()
+        self.optimizer = None
+        self.lr_scheduler = None
+        self.basic_optimizer = None
+        if optimizer is not None or self._config.optimizer is not None \
+                or model_parameters is not None:
+            self._configure_optimizer(optimizer, model_parameters)
+            self._configure_lr_scheduler(lr_scheduler)
+
+        self.training_dataloader = (self.deepspeed_io(training_data)
+                                    if training_data is not None else None)
+
+        self.losses = None
+        self.flops_profiler = None
+        if self._config.flops_profiler.enabled:
+            from ..profiling.flops_profiler import FlopsProfiler
+            self.flops_profiler = FlopsProfiler(self.module)
+
+        see_memory_usage("engine init done",
+                         force=self._config.memory_breakdown)
+
+    # ------------------------------------------------------------------ cfg
+    def train_batch_size(self):
+        return self._config.train_batch_size
+
+    def train_micro_batch_size_per_gpu(self):
+        return self._config.train_micro_batch_size_per_gpu
+
+    def gradient_accumulation_steps(self):
+        return self._config.gradient_accumulation_steps
+
+    def zero_optimization_stage(self):
+        return self._config.zero_config.stage
+
+    def zero_optimization(self):
+        return self._config.zero_config.stage > 0
+
+    def fp16_enabled(self):
+        return self._config.fp16.enabled
+
+    def bfloat16_enabled(self):
+        return self._config.bf16.enabled
+
+    def gradient_clipping(self):
+        return self._config.gradient_clipping
+
+    def steps_per_print(self):
+        return self._config.steps_per_print
+
+    @property
+    def config(self):
+        return self._config
+
+    @property
+    def global_rank(self):
+        return dist.get_rank()
+
+    @property
+    def world_size(self):
+        return dist.get_world_size()
+
+    @property
+    def local_rank(self):
+        return dist.get_local_rank()
+
+    def get_data_parallel_rank(self):
+        return dist.get_rank(self.dp_group)
+
+    # ------------------------------------------------------------ dist model
+    def _configure_distributed_model(self):
+        dtype = self._config.dtype
+        if self.zero_optimization_stage() == 3:
+            # stage-3: params may already be sharded (zero.Init); leave
+            # placement to the stage-3 machinery.
+            from .zero.stage3_params import module_is_sharded
+            if not module_is_sharded(self.module):
+                self.module.to(dtype)
+        else:
+            self.module.to(dtype)
+            self.module.to(self.device)
+            self._broadcast_model()
+        self.module.train()
+
+    def _broadcast_model(self):
+        """Sync initial weights from DP rank 0 (ref engine.py:1715)."""
+        if self.dp_world_size <= 1:
+            return
+        src_rank = dist.get_global_rank(self.dp_group, 0) \
+            if self.dp_group is not None else 0
+        for p in self.module.parameters():
+            if torch.is_tensor(p):
+                dist.broadcast(p.data, src_rank, group=self.dp_group)
+        for b in self.module.buffers():
+            if torch.is_tensor(b) and b.numel() > 0:
+                dist.broadcast(b.data, src_rank, group=self.dp_group)
+
+    def _configure_monitor(self):
+        from ..monitor.monitor import MonitorMaster
+        return MonitorMaster(self._config)
+
+    # -------------------------------------------------------------- optimizer
+    def _configure_basic_optimizer(self, model_parameters):
+        cfg = self._config.optimizer
+        if cfg is None:
+            return FusedAdam(model_parameters, lr=1e-3)
+        name = cfg.type.lower()
+        params = dict(cfg.params)
+        params.pop("torch_adam", None)
+        if name in ("adam", "adamw", "fusedadam"):
+            params.setdefault("adam_w_mode", name != "adam")
+            return FusedAdam(model_parameters, **params)
+        if name == "cpuadam" or name == "deepspeedcpuadam":
+            from ..ops.cpu_adam import DeepSpeedCPUAdam
+            return DeepSpeedCPUAdam(model_parameters, **params)
+        if name == "lion":
+            from ..ops.lion import FusedLion
+            return FusedLion(model_parameters, **params)
+        if name == "sgd":
+            return torch.optim.SGD(model_parameters, **params)
+        raise ValueError(f"unknown optimizer type {cfg.type}")
+
+    def _configure_optimizer(self, client_optimizer, model_parameters):
+        if client_optimizer is not None:
+            if isinstance(client_optimizer, Callable):
+                basic = client_optimizer(self.module.parameters())
+            else:
+                basic = client_optimizer
+        else:
+            if model_parameters is None:
+                model_parameters = list(self.module.parameters())
+            basic = self._configure_basic_optimizer(model_parameters)
+        self.basic_optimizer = basic
+
+        stage = self.zero_optimization_stage()
+        dtype = self._config.dtype
+        zc = self._config.zero_config
+
+        if stage == 3:
+            from .zero.stage3 import ZeroStage3Optimizer
+            self.optimizer = ZeroStage3Optimizer(
+                basic,
+                module=self.module,
+                engine=self,
+                dp_process_group=self.dp_group,
+                reduce_bucket_size=zc.reduce_bucket_size,
+                prefetch_bucket_size=zc.prefetch_bucket_size,
+                param_persistence_threshold=zc.param_persistence_threshold,
+                max_live_parameters=zc.max_live_parameters,
+                sub_group_size=zc.sub_group_size,
+                overlap_comm=zc.overlap_comm,
+                offload_optimizer=zc.offload_optimizer,
+                clip_grad=self.gradient_clipping(),
+                static_loss_scale=self._static_loss_scale(),
+                dynamic_loss_scale=self._dynamic_loss_scale(),
+                dynamic_loss_args=self._dynamic_loss_args(),
+                dtype=dtype,
+                gradient_accumulation_steps=self.gradient_accumulation_steps())
+        elif stage >= 1 or dtype in (torch.float16, torch.bfloat16):
+            from .zero.stage_1_and_2 import ZeroStage12Optimizer
+            if stage == 0:
+                log_dist("dtype is 16-bit with ZeRO disabled: using "
+                         "stage-1 partitioned fp32-master optimizer "
+                         "(identical numerics, lower memory)", ranks=[0])
+            self.optimizer = ZeroStage12Optimizer(
+                basic,
+                engine=self,
+                stage=max(stage, 1),
+                dp_process_group=self.dp_group,
+                reduce_bucket_size=zc.reduce_bucket_size,
+                allgather_bucket_size=zc.allgather_bucket_size,
+                overlap_comm=zc.overlap_comm,
+                clip_grad=self.gradient_clipping(),
+                static_loss_scale=self._static_loss_scale(),
+                dynamic_loss_scale=self._dynamic_loss_scale(),
+                dynamic_loss_args=self._dynamic_loss_args(),
+                dtype=dtype,
+                gradient_accumulation_steps=self.gradient_accumulation_steps())
+        else:
+            self.optimizer = basic
+        log_dist(f"optimizer: {type(self.optimizer).__name__} "
+                 f"(basic {type(basic).__name__}, stage {stage})", ranks=[0])
+
+    def _static_loss_scale(self):
+        ls = self._config.fp16.loss_scale
+        return ls if ls > 0 else 1.0
+
+    def _dynamic_loss_scale(self):
+        return self._config.fp16.enabled and self._config.fp16.loss_scale == 0
+
+    def _dynamic_loss_args(self):
+        c = self._config.fp16
+        if not self._dynamic_loss_scale():
+            return None
+        return dict(init_scale=2**c.initial_scale_power,
+                    scale_window=c.loss_scale_window,
+                    min_scale=c.min_loss_scale,
+                    delayed_shift=c.hysteresis,
+                    consecutive_hysteresis=c.consecutive_hysteresis)
+
+    def _configure_lr_scheduler(self, client_scheduler):
+        if client_scheduler is not None:
+            if isinstance(client_scheduler, Callable):
+                self.lr_scheduler = client_scheduler(self.optimizer)
+            else:
+                self.lr_scheduler = client_scheduler
+        elif self._config.scheduler is not None:
+            cls = lr_schedules.SCHEDULES[self._config.scheduler.type]
+            self.lr_scheduler = cls(self.optimizer,
+                                    **self._config.scheduler.params)
+
+    # ------------------------------------------------------------- data
+    def deepspeed_io(self, dataset, batch_size=None, route=None):
+        from .dataloader import DeepSpeedDataLoader
+        return DeepSpeedDataLoader(
+            dataset,
+            batch_size=batch_size or self.train_micro_batch_size_per_gpu(),
+            data_parallel_world_size=self.dp_world_size,
+            data_parallel_rank=dist.get_rank(self.dp_group),
+            collate_fn=self.collate_fn)
+
+    # ------------------------------------------------------------- train
+    def is_gradient_accumulation_boundary(self):
+        if self._is_gradient_accumulation_boundary is not None:
+            return self._is_gradient_accumulation_boundary
+        return (self.micro_steps + 1) % self.gradient_accumulation_steps() == 0
+
+    def set_gradient_accumulation_boundary(self, is_boundary):
+        self._is_gradient_accumulation_boundary = is_boundary
+
+    def forward(self, *inputs, **kwargs):
+        if self.flops_profiler is not None and \
+                self.global_steps == self._config.flops_profiler.profile_step:
+            self.flops_profiler.start_profile(ignore_list=None)
+        if self.wall_clock_breakdown():
+            self.timers("forward").start()
+        loss = self.module(*inputs, **kwargs)
+        if self.wall_clock_breakdown():
+            self.timers("forward").stop()
+        return loss
+
+    def wall_clock_breakdown(self):
+        return self._config.wall_clock_breakdown
+
+    def backward(self, loss, retain_graph=False, scale_wrt_gas=True):
+        if self.gradient_accumulation_steps() > 1 and scale_wrt_gas:
+            loss = loss / self.gradient_accumulation_steps()
+        if self.wall_clock_breakdown():
+            self.timers("backward").start()
+        if hasattr(self.optimizer, "backward"):
+            self.optimizer.backward(loss, retain_graph=retain_graph)
+        else:
+            loss.backward(retain_graph=retain_graph)
+            if self.is_gradient_accumulation_boundary():
+                self.allreduce_gradients()
+        if self.wall_clock_breakdown():
+            self.timers("backward").stop()
+        self.micro_steps += 1
+        self.global_samples += self.train_micro_batch_size_per_gpu()
+        return loss
+
+    def allreduce_gradients(self, bucket_size=MEMORY_OPT_ALLREDUCE_SIZE):
+        """ZeRO-0 fp32 fallback: bucketed allreduce (ref engine.py:3784)."""
+        if self.dp_world_size <= 1:
+            return
+        grads = [p.grad for p in self.module.parameters()
+                 if p.grad is not None]
+        bucket, numel = [], 0
+        for g in grads:
+            bucket.append(g)
+            numel += g.numel()
+            if numel >= bucket_size:
+                self._allreduce_bucket(bucket)
+                bucket, numel = [], 0
+        if bucket:
+            self._allreduce_bucket(bucket)
+
+    def _allreduce_bucket(self, bucket):
+        from .utils import flatten_dense_tensors, unflatten_dense_tensors
+        flat = flatten_dense_tensors(bucket)
+        flat.div_(self.dp_world_size)
+        dist.all_reduce(flat, group=self.dp_group)
+        for buf, synced in zip(bucket, unflatten_dense_tensors(flat, bucket)):
+            buf.copy_(synced)
+
+    def step(self, lr_kwargs=None):
+        if self.wall_clock_breakdown():
+            self.timers("step").start()
+        if self.is_gradient_accumulation_boundary():
+            self._take_model_step(lr_kwargs)
+        if self.wall_clock_breakdown():
+            self.timers("step").stop()
+        if self.flops_profiler is not None and \
+                self.global_steps == self._config.flops_profiler.profile_step + 1:
+            self.flops_profiler.print_model_profile(
+                profile_step=self.global_steps,
+                output_file=self._config.flops_profiler.output_file)
+            self.flops_profiler.end_profile()
+            self.flops_profiler = None
+
+    def _take_model_step(self, lr_kwargs=None):
+        self.optimizer.step()
+        overflow = getattr(self.optimizer, "overflow", False)
+        if not isinstance(self.optimizer, torch.optim.Optimizer):
+            pass
+        else:
+            self.optimizer.zero_grad(set_to_none=True)
+        if overflow:
+            self.skipped_steps += 1
+        else:
+            if self.lr_scheduler is not None:
+                self.lr_scheduler.step(**(lr_kwargs or {}))
+        self.global_steps += 1
+        if self.global_steps % self.steps_per_print() == 0:
+            self._report_progress()
+
+    def _report_progress(self):
+        lr = [g["lr"] for g in self.optimizer.param_groups] \
+            if hasattr(self.optimizer, "param_groups") else []
+        loss_scale = getattr(self.optimizer, "loss_scale", 1.0)
+        norm = getattr(self.optimizer, "get_global_grad_norm", lambda: 0.0)()
+        log_dist(f"step={self.global_steps}, skipped={self.skipped_steps}, "
+                 f"lr={lr}, scale={loss_scale}, grad_norm={norm:.4f}",
+                 ranks=[0])
+        self.monitor.write_events([
+            ("Train/lr", lr[0] if lr else 0.0, self.global_steps),
+        ])
+
+    def train(self, mode=True):
+        self.module.train(mode)
+        return self
+
+    def eval(self):
+        self.module.eval()
+        return self
+
+    def zero_grad(self, set_to_none=True):
+        if hasattr(self.optimizer, "zero_grad"):
+            self.optimizer.zero_grad(set_to_none=set_to_none)
+
+    # ---------------------------------------------------------- checkpoint
+    def save_checkpoint(self, save_dir, tag=None, client_state=None,
+                        save_latest=True, exclude_frozen_parameters=False):
+        from .checkpointing import save_checkpoint as _save
+        return _save(self, save_dir, tag=tag, client_state=client_state,
+                     save_latest=save_latest,
+                     exclude_frozen_parameters=exclude_frozen_parameters)
+
+    def load_checkpoint(self, load_dir, tag=None, load_module_strict=True,
+                        load_optimizer_states=True, load_lr_scheduler_states=True,
+                        load_module_only=False):
+        from .checkpointing import load_checkpoint as _load
+        return _load(self, load_dir, tag=tag,
+                     load_module_strict=load_module_strict,
+                     load_optimizer_states=load_optimizer_states,
+                     load_lr_scheduler_states=load_lr_scheduler_states,
+                     load_module_only=load_module_only)
+
+    def module_state_dict(self, exclude_frozen_parameters=False):
+        return self.module.state_dict()
+
+    def load_module_state_dict(self, state_dict, strict=True):
+        self.module.load_state_dict(state_dict, strict=strict)
+
+    # ------------------------------------------------------------- misc
+    def get_lr(self):
+        return [g["lr"] for g in self.optimizer.param_groups]
+
+    def get_global_grad_norm(self):
+        return getattr(self.optimizer, "get_global_grad_norm", lambda: 0.0)()
+
+    def destroy(self):
+        if hasattr(self.optimizer, "destroy"):
+            self.optimizer.destroy()

@@ -1,0 +1,470 @@
+"""ZeRO stage 3: partitioned parameters + gradients + optimizer states.
+
+Parity: reference `runtime/zero/stage3.py:149` (DeepSpeedZeroOptimizer_Stage3),
+`runtime/zero/parameter_offload.py:130` (hook installer),
+`runtime/zero/partitioned_param_coordinator.py:73` (prefetch coordinator).
+
+MI355X-first redesign (not a translation):
+- Per-param contiguous all-gather inside an RCCL coalescing group (see
+  stage3_params.py) — zero reassembly copies.
+- Gradients reduce-scatter per-param (coalesced) into flat fp32 shard
+  accumulators every micro-step; full grads freed immediately.
+- Optimizer states live in flat fp32 "sub-group" slabs sized for cache
+  residency; the fused HIP AdamW walks them as one multi-tensor launch.
+- Prefetch: module-order trace recorded on step 0, replayed with an
+  element-budget lookahead (prefetch_bucket_size) so all-gathers overlap
+  compute on the same RCCL stream without CPU-side stalls.
+"""
+from collections import OrderedDict
+
+import torch
+
+from ... import comm as dist
+from ...utils.logging import log_dist, logger
+from ..loss_scaler import CreateLossScaler
+from .stage3_params import (ZeroParamStatus, all_gather_params,
+                            convert_to_zero_param, free_param, is_zero_param)
+
+ALIGN = 64
+
+
+def _pad_to(n, m):
+    return (n + m - 1) // m * m
+
+
+class SubGroup:
+    """Flat fp32 master + grad accumulator over a run of param shards."""
+
+    __slots__ = ("params", "offsets", "master32", "grad32", "group_idx",
+                 "numel")
+
+    def __init__(self, params, offsets, numel, group_idx, device):
+        self.params = params
+        self.offsets = offsets
+        self.numel = numel
+        self.group_idx = group_idx
+        self.master32 = torch.empty(numel, dtype=torch.float32, device=device)
+        for p in params:
+            off = offsets[p]
+            self.master32[off:off + p.ds_tensor.numel()].copy_(
+                p.ds_tensor.float())
+        self.master32 = self.master32.detach().requires_grad_(True)
+        self.grad32 = torch.zeros(numel, dtype=torch.float32, device=device)
+
+    def grad_shard_view(self, p):
+        off = self.offsets[p]
+        return self.grad32[off:off + p.ds_tensor.numel()]
+
+    def copy_master_to_shards(self):
+        for p in self.params:
+            off = self.offsets[p]
+            p.ds_tensor.copy_(
+                self.master32[off:off + p.ds_tensor.numel()].detach())
+
+
+class ZeroStage3Optimizer:
+    def __init__(self,
+                 init_optimizer,
+                 module,
+                 engine=None,
+                 dp_process_group=None,
+                 reduce_bucket_size=int(5e8),
+                 prefetch_bucket_size=int(5e7),
+                 param_persistence_threshold=int(1e5),
+                 max_live_parameters=int(1e9),
+                 sub_group_size=int(1e9),
+                 overlap_comm=True,
+                 offload_optimizer=None,
+                 clip_grad=0.0,
+                 static_loss_scale=1.0,
+                 dynamic_loss_scale=False,
+                 dynamic_loss_args=None,
+                 dtype=torch.bfloat16,
+                 gradient_accumulation_steps=1):
+        self.optimizer = init_optimizer
+        self.module = module
+        self.dp_group = dp_process_group
+        self.world = dist.get_world_size(self.dp_group)
+        self.rank = dist.get_rank(self.dp_group)
+        self.dtype = dtype
+        self.clip_grad = clip_grad
+        self.reduce_bucket_size = int(reduce_bucket_size)
+        self.prefetch_bucket_size = int(prefetch_bucket_size)
+        self.persist_threshold = int(param_persistence_threshold)
+        self.sub_group_size = int(sub_group_size)
+        self.gradient_accumulation_steps = gradient_accumulation_steps
+        self.micro_step = 0
+        self.overlap_comm = overlap_comm
+
+        self.device = (torch.device("cuda", torch.cuda.current_device())
+                       if torch.cuda.is_available() else torch.device("cpu"))
+        self.offload_optimizer = (offload_optimizer is not None
+                                  and getattr(offload_optimizer, "device", "none") == "cpu")
+
+        self.loss_scaler = CreateLossScaler(dtype, static_loss_scale,
+                                            dynamic_loss_scale,
+                                            dynamic_loss_args)
+        self.overflow = False
+
+        self._shard_module_params()
+        self._build_sub_groups()
+        self._gather_persistent_params()
+        self._install_module_hooks()
+        self._install_grad_hooks()
+
+        # reduction state
+        self._ipg_params = []
+        self._ipg_numel = 0
+        self._reduce_works = []
+
+        # trace / prefetch state
+        self._trace = []            # module order recorded on first forward
+        self._trace_complete = False
+        self._trace_pos = 0
+        self._inflight = {}         # module -> AllGatherHandle
+
+        self._global_grad_norm = 0.0
+        log_dist(
+            f"ZeRO-3: {sum(p.ds_numel for p in self._all_params)/1e9:.2f}B "
+            f"params, {len(self.sub_groups)} sub-groups, world {self.world}",
+            ranks=[0])
+
+    # ---------------------------------------------------------------- setup
+    def _shard_module_params(self):
+        """Convert any unsharded params; move buffers to device."""
+        params = list(self.module.parameters())
+        src = (dist.get_global_rank(self.dp_group, 0)
+               if self.dp_group is not None and self.world > 1 else 0)
+        for p in params:
+            if not is_zero_param(p):
+                p.data = p.data.to(self.device, self.dtype)
+                if self.world > 1:
+                    dist.broadcast(p.data, src, group=self.dp_group)
+                convert_to_zero_param(p, self.dp_group, self.device,
+                                      self.dtype, self.persist_threshold)
+        for b in self.module.buffers():
+            b.data = b.data.to(self.device)
+        # dedupe (tied weights appear once in .parameters() already)
+        self._all_params = [p for p in self.module.parameters()]
+        for i, p in enumerate(self._all_params):
+            p.ds_param_index = i
+
+    def _build_sub_groups(self):
+        """Pack param shards into flat fp32 sub-groups per optimizer group."""
+        param_to_group = {}
+        for gi, g in enumerate(self.optimizer.param_groups):
+            for p in g["params"]:
+                param_to_group[p] = gi
+        self.sub_groups = []
+        for gi in range(len(self.optimizer.param_groups)):
+            gparams = [p for p in self._all_params
+                       if param_to_group.get(p, 0) == gi and p.requires_grad]
+            cur, offsets, numel = [], OrderedDict(), 0
+            for p in gparams:
+                if numel >= self.sub_group_size and cur:
+                    self.sub_groups.append(
+                        SubGroup(cur, offsets, numel, gi, self.device))
+                    cur, offsets, numel = [], OrderedDict(), 0
+                offsets[p] = numel
+                cur.append(p)
+                numel += p.ds_tensor.numel()
+            if cur:
+                self.sub_groups.append(
+                    SubGroup(cur, offsets, numel, gi, self.device))
+        self.param_to_subgroup = {}
+        for sg in self.sub_groups:
+            for p in sg.params:
+                self.param_to_subgroup[p] = sg
+        for gi, g in enumerate(self.optimizer.param_groups):
+            g["params"] = [sg.master32 for sg in self.sub_groups
+                           if sg.group_idx == gi]
+
+    def _gather_persistent_params(self):
+        persist = [p for p in self._all_params if p.ds_persist]
+        if persist:
+            all_gather_params(persist, self.dp_group, async_op=False).wait()
+
+    # ------------------------------------------------------------- hooks
+    def _install_module_hooks(self):
+        """fetch/release hooks on every module owning direct params."""
+        self._module_hooks = []
+        for mod in self.module.modules():
+            direct = [p for p in mod.parameters(recurse=False)
+                      if is_zero_param(p)]
+            if not direct:
+                continue
+            mod._ds_direct_params = direct
+            self._module_hooks.append(mod.register_forward_pre_hook(
+                self._pre_forward_hook))
+            self._module_hooks.append(mod.register_forward_hook(
+                self._post_forward_hook))
+            self._module_hooks.append(mod.register_full_backward_pre_hook(
+                self._pre_backward_hook))
+            self._module_hooks.append(mod.register_full_backward_hook(
+                self._post_backward_hook))
+        # root hook: reset trace replay position each step
+        self._module_hooks.append(self.module.register_forward_pre_hook(
+            self._root_pre_forward))
+
+    def _root_pre_forward(self, mod, inputs):
+        self._trace_pos = 0
+        if self._trace and not self._trace_complete:
+            self._trace_complete = True
+
+    # -- forward path
+    def _pre_forward_hook(self, mod, inputs):
+        self.fetch_sub_module(mod, forward=True)
+
+    def _post_forward_hook(self, mod, inputs, output):
+        self.release_sub_module(mod)
+
+    # -- backward path
+    def _pre_backward_hook(self, mod, grad_output):
+        self.fetch_sub_module(mod, forward=False)
+
+    def _post_backward_hook(self, mod, grad_input, grad_output):
+        # NOTE: for modules whose inputs don't require grad (the first layer)
+        # this hook fires BEFORE the module's internal backward nodes run, so
+        # it must not free data. Actual freeing happens per-param in
+        # _on_grad_ready (post-accumulate-grad == module backward complete).
+        for p in mod._ds_direct_params:
+            p.ds_active_sub_modules.discard(id(mod))
+
+    def fetch_sub_module(self, mod, forward=True):
+        params = mod._ds_direct_params
+        for p in params:
+            p.ds_active_sub_modules.add(id(mod))
+        # record trace on first iteration
+        if forward and not self._trace_complete:
+            self._trace.append(mod)
+        h = self._inflight.pop(mod, None)
+        if h is not None:
+            h.wait()
+        need = [p for p in params
+                if p.ds_status == ZeroParamStatus.NOT_AVAILABLE]
+        if need:
+            all_gather_params(need, self.dp_group).wait()
+        # publish data for params gathered by a shared handle
+        for p in params:
+            assert p.ds_status == ZeroParamStatus.AVAILABLE, \
+                f"param {p.ds_id} not available after fetch"
+        if forward and self._trace_complete:
+            self._prefetch(forward=True)
+
+    def _prefetch(self, forward=True):
+        """Launch lookahead all-gathers along the recorded trace."""
+        budget = self.prefetch_bucket_size
+        # advance trace_pos to current position lazily
+        pos = self._trace_pos
+        n = len(self._trace)
+        launched = 0
+        while pos < n and budget > 0:
+            mod = self._trace[pos]
+            pos += 1
+            if mod in self._inflight:
+                continue
+            need = [p for p in mod._ds_direct_params
+                    if p.ds_status == ZeroParamStatus.NOT_AVAILABLE]
+            if not need:
+                continue
+            budget -= sum(p.ds_numel for p in need)
+            self._inflight[mod] = all_gather_params(need, self.dp_group)
+            launched += 1
+            if launched >= 8:
+                break
+        self._trace_pos = min(pos, n)
+
+    def release_sub_module(self, mod):
+        for p in mod._ds_direct_params:
+            p.ds_active_sub_modules.discard(id(mod))
+            if not p.ds_active_sub_modules and not p.ds_persist:
+                free_param(p)
+
+    def _install_grad_hooks(self):
+        self._grad_hooks = []
+        for p in self._all_params:
+            if p.requires_grad:
+                self._grad_hooks.append(
+                    p.register_post_accumulate_grad_hook(self._on_grad_ready))
+
+    # --------------------------------------------------------- grad reduce
+    def _on_grad_ready(self, p):
+        self._ipg_params.append(p)
+        self._ipg_numel += p.ds_numel
+        # grad finalized => this param's backward consumers have run; safe to
+        # drop the gathered weight now (backward-release point).
+        if not p.ds_active_sub_modules and not p.ds_persist:
+            free_param(p)
+        if self._ipg_numel >= self.reduce_bucket_size:
+            self._flush_ipg()
+
+    def _flush_ipg(self):
+        if not self._ipg_params:
+            return
+        params = self._ipg_params
+        self._ipg_params = []
+        self._ipg_numel = 0
+        world = self.world
+        use_coalescing = (torch.cuda.is_available() and world > 1
+                          and len(params) > 1)
+        shards = []
+        if world == 1:
+            for p in params:
+                sg = self.param_to_subgroup[p]
+                sg.grad_shard_view(p)[:p.ds_numel].add_(
+                    p.grad.reshape(-1).float())
+                p.grad = None
+            return
+        inputs = []
+        for p in params:
+            shard_numel = p.ds_tensor.numel()
+            padded = torch.zeros(shard_numel * world, dtype=p.grad.dtype,
+                                 device=p.grad.device)
+            padded[:p.ds_numel].copy_(p.grad.reshape(-1))
+            padded.div_(world)
+            inputs.append(padded)
+            shards.append(torch.empty(shard_numel, dtype=p.grad.dtype,
+                                      device=p.grad.device))
+            p.grad = None
+        if use_coalescing:
+            from torch.distributed.distributed_c10d import _coalescing_manager
+            with _coalescing_manager(self.dp_group, self.device,
+                                     async_ops=True) as cm:
+                for out, inp in zip(shards, inputs):
+                    dist.reduce_scatter_tensor(out, inp, group=self.dp_group)
+            cm.wait()
+        else:
+            for out, inp in zip(shards, inputs):
+                dist.reduce_scatter_tensor(out, inp, group=self.dp_group)
+        for p, shard in zip(params, shards):
+            sg = self.param_to_subgroup[p]
+            sg.grad_shard_view(p).add_(shard.float())
+
+    # -------------------------------------------------------------- train
+    def backward(self, loss, retain_graph=False):
+        self.micro_step += 1
+        self.loss_scaler.backward(loss.float(), retain_graph=retain_graph)
+        self._flush_ipg()
+
+    def is_gradient_accumulation_boundary(self):
+        return self.micro_step % self.gradient_accumulation_steps == 0
+
+    def has_overflow(self):
+        found = False
+        for sg in self.sub_groups:
+            s = sg.grad32.sum()
+            if torch.isinf(s) or torch.isnan(s):
+                found = True
+                break
+        if dist.is_initialized():
+            t = torch.tensor([1.0 if found else 0.0], device=self.device)
+            dist.all_reduce(t, op=dist.ReduceOp.MAX, group=self.dp_group)
+            found = bool(t.item())
+        return found
+
+    def _unscale_and_clip(self):
+        scale = self.loss_scaler.loss_scale
+        combined = scale
+        if self.clip_grad > 0.0:
+            total_sq = torch.zeros(1, dtype=torch.float64, device=self.device)
+            for sg in self.sub_groups:
+                total_sq += sg.grad32.double().pow(2).sum()
+            if dist.is_initialized():
+                dist.all_reduce(total_sq, group=self.dp_group)
+            norm = (total_sq.sqrt().item()**1.0) / scale
+            self._global_grad_norm = norm
+            clip = norm / self.clip_grad
+            if clip > 1.0:
+                combined = scale * clip
+        if combined != 1.0:
+            for sg in self.sub_groups:
+                sg.grad32.mul_(1.0 / combined)
+
+    def step(self, closure=None):
+        assert closure is None
+        self._flush_ipg()
+
+        if self.dtype == torch.float16:
+            self.overflow = self.has_overflow()
+            self.loss_scaler.update_scale(self.overflow)
+            if self.overflow:
+                log_dist("OVERFLOW: skipping step, new scale "
+                         f"{self.loss_scaler.loss_scale}", ranks=[0])
+                self._clear_grads()
+                return
+
+        self._unscale_and_clip()
+
+        for sg in self.sub_groups:
+            sg.master32.grad = sg.grad32
+        self.optimizer.step()
+        for sg in self.sub_groups:
+            sg.master32.grad = None
+            sg.copy_master_to_shards()
+        self._clear_grads()
+        self._refresh_persistent_params()
+
+    def _refresh_persistent_params(self):
+        persist = [p for p in self._all_params if p.ds_persist]
+        for p in persist:
+            p.ds_status = ZeroParamStatus.NOT_AVAILABLE
+        if persist:
+            all_gather_params(persist, self.dp_group, async_op=False).wait()
+
+    def _clear_grads(self):
+        for sg in self.sub_groups:
+            sg.grad32.zero_()
+
+    def zero_grad(self, set_to_none=True):
+        pass
+
+    # -------------------------------------------------------------- state
+    @property
+    def loss_scale(self):
+        return self.loss_scaler.loss_scale
+
+    def get_global_grad_norm(self):
+        return self._global_grad_norm
+
+    @property
+    def param_groups(self):
+        return self.optimizer.param_groups
+
+    def state_dict(self):
+        return {
+            "loss_scaler": self.loss_scaler,
+            "base_optimizer_state": self.optimizer.state_dict(),
+            "fp32_flat_groups": [sg.master32 for sg in self.sub_groups],
+            "param_shapes": self._param_shapes(),
+            "zero_stage": 3,
+            "partition_count": self.world,
+        }
+
+    def _param_shapes(self):
+        """name -> full shape, in module order (for zero_to_fp32)."""
+        shapes = OrderedDict()
+        pmap = {id(p): n for n, p in self.module.named_parameters()}
+        for sg in self.sub_groups:
+            for p in sg.params:
+                shapes[pmap.get(id(p), f"param_{p.ds_id}")] = {
+                    "shape": p.ds_shape, "numel": p.ds_numel,
+                    "shard_numel": p.ds_tensor.numel(),
+                    "subgroup_offset": sg.offsets[p]}
+        return shapes
+
+    def load_state_dict(self, sd, load_optimizer_states=True):
+        if "loss_scaler" in sd:
+            self.loss_scaler = sd["loss_scaler"]
+        if load_optimizer_states and "base_optimizer_state" in sd:
+            self.optimizer.load_state_dict(sd["base_optimizer_state"])
+        saved = sd.get("fp32_flat_groups", [])
+        assert len(saved) == len(self.sub_groups)
+        for sg, s in zip(self.sub_groups, saved):
+            sg.master32.data.copy_(s.data.to(sg.master32.device))
+            sg.copy_master_to_shards()
+        self._refresh_persistent_params()
+
+    def destroy(self):
+        for h in self._grad_hooks + self._module_hooks:
+            h.remove()
+        self._grad_hooks, self._module_hooks = [], []

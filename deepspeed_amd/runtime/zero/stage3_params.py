@@ -1,0 +1,267 @@
+"""ZeRO-3 parameter sharding: shard state machine + coalesced all-gather.
+
+Parity: reference `runtime/zero/partition_parameters.py:940` (zero.Init),
+`:236` (ZeroParamStatus), `:1512` (all_gather_coalesced), `:1750`
+(_partition_param).
+
+MI355X-first redesign: every parameter keeps a 1/world `ds_tensor` shard on
+device; gathering a module's parameters issues one `all_gather_into_tensor`
+PER PARAM inside an RCCL coalescing group (single fused launch, and each
+param's output buffer IS its full contiguous tensor — no interleaved-span
+reassembly copies, which matters at 8 TB/s HBM). On gloo (CPU tests) the
+calls run back-to-back instead.
+"""
+import math
+from enum import Enum
+
+import torch
+
+from ... import comm as dist
+from ...utils.logging import log_dist
+
+ALIGN = 64
+
+
+class ZeroParamStatus(Enum):
+    NOT_AVAILABLE = 1
+    INFLIGHT = 2
+    AVAILABLE = 3
+
+
+_next_param_id = [0]
+
+
+def is_zero_param(p):
+    return hasattr(p, "ds_id")
+
+
+def module_is_sharded(module):
+    return any(is_zero_param(p) for p in module.parameters())
+
+
+def _pad_to(numel, multiple):
+    return (numel + multiple - 1) // multiple * multiple
+
+
+def _supports_coalescing(group):
+    if not torch.cuda.is_available():
+        return False
+    try:
+        import torch.distributed as td
+        backend = td.get_backend(group) if group is not None else td.get_backend()
+        return "nccl" in str(backend)
+    except Exception:
+        return False
+
+
+def convert_to_zero_param(p, dp_group, device, dtype,
+                          persist_threshold=int(1e5)):
+    """Shard one parameter in place: keep 1/world on this rank."""
+    if is_zero_param(p):
+        return p
+    world = dist.get_world_size(dp_group)
+    rank = dist.get_rank(dp_group)
+    p.ds_id = _next_param_id[0]
+    _next_param_id[0] += 1
+    p.ds_shape = tuple(p.shape)
+    p.ds_numel = p.numel()
+    numel_padded = _pad_to(p.ds_numel, world * ALIGN)
+    shard_numel = numel_padded // world
+    flat = p.data.to(dtype).reshape(-1)
+    start = rank * shard_numel
+    end = min(start + shard_numel, p.ds_numel)
+    ds_tensor = torch.zeros(shard_numel, dtype=dtype, device=device)
+    if end > start:
+        ds_tensor[:end - start].copy_(flat[start:end])
+    p.ds_tensor = ds_tensor
+    p.ds_persist = p.ds_numel <= persist_threshold
+    p.ds_status = ZeroParamStatus.NOT_AVAILABLE
+    p.ds_active_sub_modules = set()
+    p.ds_full_buffer = None
+    p.data = torch.empty(0, dtype=dtype, device=device)
+    return p
+
+
+def free_param(p):
+    """Drop the gathered full tensor; shard stays."""
+    if p.ds_status == ZeroParamStatus.AVAILABLE and not p.ds_persist:
+        p.data = torch.empty(0, dtype=p.ds_tensor.dtype,
+                             device=p.ds_tensor.device)
+        p.ds_full_buffer = None
+        p.ds_status = ZeroParamStatus.NOT_AVAILABLE
+
+
+class AllGatherHandle:
+    """Waits on an in-flight coalesced gather and publishes p.data."""
+
+    def __init__(self, params, works, buffers, group):
+        self.params = params
+        self.works = works  # list of work objs or a coalescing-manager
+        self.buffers = buffers
+        self.group = group
+        self.complete = False
+
+    def wait(self):
+        if self.complete:
+            return
+        for w in self.works:
+            if w is not None:
+                w.wait()
+        for p, buf in zip(self.params, self.buffers):
+            p.data = buf.narrow(0, 0, p.ds_numel).view(p.ds_shape)
+            p.ds_full_buffer = buf
+            p.ds_status = ZeroParamStatus.AVAILABLE
+        self.complete = True
+
+
+def all_gather_params(params, dp_group, async_op=True):
+    """Launch coalesced all-gathers for NOT_AVAILABLE params.
+
+    Returns an AllGatherHandle (already complete when nothing to do).
+    Must be called identically on all ranks of the group.
+    """
+    todo = [p for p in params if p.ds_status == ZeroParamStatus.NOT_AVAILABLE]
+    for p in todo:
+        p.ds_status = ZeroParamStatus.INFLIGHT
+    if not todo:
+        return AllGatherHandle([], [], [], dp_group)
+    world = dist.get_world_size(dp_group)
+    buffers = []
+    works = []
+    if world == 1:
+        for p in todo:
+            buffers.append(p.ds_tensor)
+        h = AllGatherHandle(todo, [], buffers, dp_group)
+        h.wait()
+        return h
+    use_coalescing = _supports_coalescing(dp_group) and len(todo) > 1
+    if use_coalescing:
+        from torch.distributed.distributed_c10d import _coalescing_manager
+        device = todo[0].ds_tensor.device
+        with _coalescing_manager(dp_group, device, async_ops=True) as cm:
+            for p in todo:
+                buf = torch.empty(p.ds_tensor.numel() * world,
+                                  dtype=p.ds_tensor.dtype, device=device)
+                dist.all_gather_into_tensor(buf, p.ds_tensor, group=dp_group)
+                buffers.append(buf)
+        works = [cm]
+    else:
+        for p in todo:
+            buf = torch.empty(p.ds_tensor.numel() * world,
+                              dtype=p.ds_tensor.dtype,
+                              device=p.ds_tensor.device)
+            w = dist.all_gather_into_tensor(buf, p.ds_tensor, group=dp_group,
+                                            async_op=async_op)
+            buffers.append(buf)
+            works.append(w)
+    handle = AllGatherHandle(todo, works, buffers, dp_group)
+    if not async_op:
+        handle.wait()
+    return handle
+
+
+class Init:
+    """Context manager: parameters created inside are sharded at module
+    construction (ref partition_parameters.py:940).
+
+    Implementation: patches nn.Module.register_parameter so each fresh
+    parameter is converted immediately after registration — construction of
+    an N-param model never holds more than one full parameter per rank.
+    """
+
+    def __init__(self, module=None, data_parallel_group=None, dtype=None,
+                 config_dict_or_path=None, enabled=True, device=None,
+                 param_persistence_threshold=int(1e5)):
+        self.enabled = enabled
+        self.dp_group = data_parallel_group
+        self.dtype = dtype or torch.bfloat16
+        self.device = device or (
+            torch.device("cuda", torch.cuda.current_device())
+            if torch.cuda.is_available() else torch.device("cpu"))
+        self.persist_threshold = param_persistence_threshold
+        self._orig_register = None
+        if config_dict_or_path is not None:
+            from ...config import DeepSpeedConfig
+            cfg = DeepSpeedConfig(config_dict_or_path,
+                                  world_size=dist.get_world_size())
+            self.dtype = cfg.dtype
+            self.persist_threshold = cfg.zero_config.param_persistence_threshold
+        if module is not None:
+            self._convert_module(module)
+
+    def _convert_module(self, module):
+        for p in module.parameters():
+            convert_to_zero_param(p, self.dp_group, self.device, self.dtype,
+                                  self.persist_threshold)
+        for b in module.buffers():
+            b.data = b.data.to(self.device)
+
+    def __enter__(self):
+        if not self.enabled:
+            return self
+        if not dist.is_initialized():
+            dist.init_distributed()
+        init = self
+
+        self._orig_register = torch.nn.Module.register_parameter
+
+        def wrapped_register(mod, name, param):
+            init._orig_register(mod, name, param)
+            if param is not None and not is_zero_param(param):
+                convert_to_zero_param(param, init.dp_group, init.device,
+                                      init.dtype, init.persist_threshold)
+
+        torch.nn.Module.register_parameter = wrapped_register
+        return self
+
+    def __exit__(self, *exc):
+        if self._orig_register is not None:
+            torch.nn.Module.register_parameter = self._orig_register
+            self._orig_register = None
+        return False
+
+
+class GatheredParameters:
+    """Temporarily gather sharded params (ref zero.GatheredParameters)."""
+
+    def __init__(self, params, modifier_rank=None, enabled=True, fwd_module=None):
+        if isinstance(params, torch.nn.Parameter):
+            params = [params]
+        self.params = [p for p in params if is_zero_param(p)]
+        self.enabled = enabled and len(self.params) > 0
+        self.modifier_rank = modifier_rank
+        self.dp_group = None
+
+    def __enter__(self):
+        if not self.enabled:
+            return
+        h = all_gather_params(self.params, self.dp_group)
+        h.wait()
+
+    def __exit__(self, *exc):
+        if not self.enabled:
+            return False
+        if self.modifier_rank is not None:
+            # push modifications back into shards from the modifier rank
+            for p in self.params:
+                src = dist.get_global_rank(self.dp_group, self.modifier_rank) \
+                    if self.dp_group is not None else self.modifier_rank
+                dist.broadcast(p.data, src, group=self.dp_group)
+        for p in self.params:
+            _repartition(p)
+        return False
+
+
+def _repartition(p):
+    """Write p.data back into the shard, then free."""
+    world = p.ds_tensor.numel()
+    rank = dist.get_rank()  # world group assumption for default dp
+    shard_numel = p.ds_tensor.numel()
+    flat = p.data.reshape(-1)
+    start = rank * shard_numel
+    end = min(start + shard_numel, p.ds_numel)
+    if end > start:
+        p.ds_tensor[:end - start].copy_(flat[start:end])
+    p.ds_status = ZeroParamStatus.AVAILABLE
+    if not p.ds_persist:
+        free_param(p)
