@@ -1,0 +1,168 @@
+"""Llama-family causal LM, MI355X-native.
+
+Built directly on the deepspeed_amd op set (HIP RMSNorm / RoPE / SwiGLU /
+fused cross-entropy; hipBLASLt GEMMs via F.linear). Used by bench.py for the
+BASELINE config "Llama-3 8B ZeRO-3 bf16".
+"""
+from dataclasses import dataclass, field
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..ops.attention import flash_attention
+from ..ops.functional import (apply_rope, build_rope_cache,
+                              fused_cross_entropy, rms_norm, swiglu)
+
+
+@dataclass
+class LlamaConfig:
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_hidden_layers: int = 32
+    num_attention_heads: int = 32
+    num_key_value_heads: int = 8
+    vocab_size: int = 128256
+    max_position_embeddings: int = 8192
+    rope_theta: float = 500000.0
+    rms_norm_eps: float = 1e-5
+    tie_word_embeddings: bool = False
+    initializer_range: float = 0.02
+    activation_checkpointing: bool = True
+
+    @property
+    def head_dim(self):
+        return self.hidden_size // self.num_attention_heads
+
+
+LLAMA_CONFIGS = {
+    "llama3-8b": LlamaConfig(),
+    "llama3-70b": LlamaConfig(hidden_size=8192, intermediate_size=28672,
+                              num_hidden_layers=80, num_attention_heads=64,
+                              num_key_value_heads=8),
+    "llama-tiny": LlamaConfig(hidden_size=256, intermediate_size=688,
+                              num_hidden_layers=4, num_attention_heads=8,
+                              num_key_value_heads=4, vocab_size=2048,
+                              max_position_embeddings=512,
+                              activation_checkpointing=False),
+    "llama-small": LlamaConfig(hidden_size=1024, intermediate_size=2816,
+                               num_hidden_layers=8, num_attention_heads=16,
+                               num_key_value_heads=4, vocab_size=32000,
+                               max_position_embeddings=4096),
+}
+
+
+class LlamaRMSNorm(nn.Module):
+    def __init__(self, hidden_size, eps=1e-5):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(hidden_size))
+        self.eps = eps
+
+    def forward(self, x):
+        return rms_norm(x, self.weight, self.eps)
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        h, d = cfg.hidden_size, cfg.head_dim
+        self.q_proj = nn.Linear(h, cfg.num_attention_heads * d, bias=False)
+        self.k_proj = nn.Linear(h, cfg.num_key_value_heads * d, bias=False)
+        self.v_proj = nn.Linear(h, cfg.num_key_value_heads * d, bias=False)
+        self.o_proj = nn.Linear(cfg.num_attention_heads * d, h, bias=False)
+
+    def forward(self, x, cos, sin):
+        B, S, _ = x.shape
+        d = self.cfg.head_dim
+        q = self.q_proj(x).view(B, S, -1, d)
+        k = self.k_proj(x).view(B, S, -1, d)
+        v = self.v_proj(x).view(B, S, -1, d)
+        q = apply_rope(q, cos, sin)
+        k = apply_rope(k, cos, sin)
+        o = flash_attention(q, k, v, causal=True)
+        return self.o_proj(o.reshape(B, S, -1))
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.gate_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size,
+                                   bias=False)
+        self.up_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size,
+                                 bias=False)
+        self.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size,
+                                   bias=False)
+
+    def forward(self, x):
+        return self.down_proj(swiglu(self.gate_proj(x), self.up_proj(x)))
+
+
+class LlamaDecoderLayer(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.input_layernorm = LlamaRMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
+        self.self_attn = LlamaAttention(cfg)
+        self.post_attention_layernorm = LlamaRMSNorm(cfg.hidden_size,
+                                                     cfg.rms_norm_eps)
+        self.mlp = LlamaMLP(cfg)
+
+    def forward(self, x, cos, sin):
+        x = x + self.self_attn(self.input_layernorm(x), cos, sin)
+        x = x + self.mlp(self.post_attention_layernorm(x))
+        return x
+
+
+class LlamaModel(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+        self.layers = nn.ModuleList(
+            [LlamaDecoderLayer(cfg) for _ in range(cfg.num_hidden_layers)])
+        self.norm = LlamaRMSNorm(cfg.hidden_size, cfg.rms_norm_eps)
+        cos, sin = build_rope_cache(cfg.max_position_embeddings, cfg.head_dim,
+                                    cfg.rope_theta)
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+
+    def forward(self, input_ids):
+        x = self.embed_tokens(input_ids)
+        S = input_ids.shape[1]
+        cos = self.rope_cos[:S]
+        sin = self.rope_sin[:S]
+        for layer in self.layers:
+            if self.cfg.activation_checkpointing and self.training:
+                x = torch.utils.checkpoint.checkpoint(
+                    layer, x, cos, sin, use_reentrant=False)
+            else:
+                x = layer(x, cos, sin)
+        return self.norm(x)
+
+
+class LlamaForCausalLM(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.model = LlamaModel(cfg)
+        self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False)
+        if cfg.tie_word_embeddings:
+            self.lm_head.weight = self.model.embed_tokens.weight
+        self.apply(self._init_weights)
+
+    def _init_weights(self, module):
+        std = self.cfg.initializer_range
+        if isinstance(module, nn.Linear):
+            module.weight.data.normal_(0.0, std)
+            if module.bias is not None:
+                module.bias.data.zero_()
+        elif isinstance(module, nn.Embedding):
+            module.weight.data.normal_(0.0, std)
+
+    def forward(self, input_ids, labels=None):
+        h = self.model(input_ids)
+        logits = self.lm_head(h)
+        if labels is None:
+            return logits
+        # next-token prediction: shift
+        return fused_cross_entropy(logits[:, :-1, :], labels[:, 1:])

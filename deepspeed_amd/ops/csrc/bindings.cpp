@@ -1,0 +1,55 @@
+// pybind bindings for the deepspeed_amd gfx950 kernel set.
+#include <torch/extension.h>
+
+void multi_tensor_adam(std::vector<at::Tensor> params,
+                       std::vector<at::Tensor> grads,
+                       std::vector<at::Tensor> exp_avgs,
+                       std::vector<at::Tensor> exp_avg_sqs, double lr,
+                       double beta1, double beta2, double eps, long step,
+                       long adamw_mode, long bias_correction,
+                       double weight_decay, std::vector<at::Tensor> out16);
+void multi_tensor_lion(std::vector<at::Tensor> params,
+                       std::vector<at::Tensor> grads,
+                       std::vector<at::Tensor> exp_avgs, double lr,
+                       double beta1, double beta2, double weight_decay,
+                       std::vector<at::Tensor> out16);
+std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps);
+std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
+                                    at::Tensor rstd);
+std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w,
+                                      c10::optional<at::Tensor> b, double eps);
+std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
+                                      at::Tensor w, at::Tensor mean,
+                                      at::Tensor rstd);
+void rope_inplace(at::Tensor t, at::Tensor cos, at::Tensor sin, long pos0,
+                  bool backward);
+at::Tensor swiglu_fwd(at::Tensor g, at::Tensor u);
+std::vector<at::Tensor> swiglu_bwd(at::Tensor dy, at::Tensor g, at::Tensor u);
+std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits,
+                                          at::Tensor targets,
+                                          long ignore_index);
+at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets,
+                             at::Tensor lse, at::Tensor dloss,
+                             long ignore_index);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("multi_tensor_adam", &multi_tensor_adam, "fused AdamW (gfx950)",
+        py::arg("params"), py::arg("grads"), py::arg("exp_avgs"),
+        py::arg("exp_avg_sqs"), py::arg("lr"), py::arg("beta1"),
+        py::arg("beta2"), py::arg("eps"), py::arg("step"),
+        py::arg("adamw_mode"), py::arg("bias_correction"),
+        py::arg("weight_decay"), py::arg("out16") = std::vector<at::Tensor>());
+  m.def("multi_tensor_lion", &multi_tensor_lion, "fused Lion (gfx950)",
+        py::arg("params"), py::arg("grads"), py::arg("exp_avgs"),
+        py::arg("lr"), py::arg("beta1"), py::arg("beta2"),
+        py::arg("weight_decay"), py::arg("out16") = std::vector<at::Tensor>());
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("rope_inplace", &rope_inplace);
+  m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("cross_entropy_fwd", &cross_entropy_fwd);
+  m.def("cross_entropy_bwd", &cross_entropy_bwd);
+}

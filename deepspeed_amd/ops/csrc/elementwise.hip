@@ -1,0 +1,148 @@
+// RoPE + SwiGLU elementwise kernels for MI355X (gfx950).
+//
+// Role parity: reference csrc/transformer/inference/csrc/apply_rotary_pos_emb.cu
+// and gelu.cu (gated activations). MI355X-native: memory-bound, host-side
+// precomputed cos/sin table (guide Appendix B: on-device trig turns these
+// VALU-bound), vectorized bf16 IO, grid-stride.
+#include <torch/extension.h>
+
+#include "common.h"
+
+// q/k layout: [B, S, H, D] bf16 contiguous; cos/sin: [S, D/2] fp32.
+// Llama "rotate_half" pairing: (d, d + D/2).
+__global__ void rope_kernel(short* __restrict__ t,
+                            const float* __restrict__ cs,  // [S, D/2] cos
+                            const float* __restrict__ sn,  // [S, D/2] sin
+                            long long total_pairs, int S, int H, int D,
+                            int pos0, int backward) {
+  int half = D / 2;
+  long long i0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = i0; i < total_pairs; i += stride) {
+    int d = (int)(i % half);
+    long long rem = i / half;
+    int h = (int)(rem % H);
+    long long rem2 = rem / H;
+    int s = (int)(rem2 % S);
+    long long b = rem2 / S;
+    long long base = ((b * S + s) * (long long)H + h) * D;
+    float c = cs[(long long)(s) * half + d];
+    float sv = sn[(long long)(s) * half + d];
+    if (backward) sv = -sv;
+    float x1 = bf2f(t[base + d]);
+    float x2 = bf2f(t[base + d + half]);
+    t[base + d] = f2bf(x1 * c - x2 * sv);
+    t[base + d + half] = f2bf(x2 * c + x1 * sv);
+  }
+}
+
+void rope_inplace(at::Tensor t, at::Tensor cos, at::Tensor sin, long pos0,
+                  bool backward) {
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16 && t.is_contiguous());
+  TORCH_CHECK(t.dim() == 4, "rope expects [B,S,H,D]");
+  int B = t.size(0), S = t.size(1), H = t.size(2), D = t.size(3);
+  long long total_pairs = (long long)B * S * H * (D / 2);
+  auto stream = c10::hip::getCurrentHIPStream();
+  int block = 256;
+  int grid = grid_for(total_pairs, block);
+  hipLaunchKernelGGL(rope_kernel, dim3(grid), dim3(block), 0, stream.stream(),
+                     reinterpret_cast<short*>(t.data_ptr()),
+                     cos.data_ptr<float>(), sin.data_ptr<float>(), total_pairs,
+                     S, H, D, (int)pos0, backward ? 1 : 0);
+  HIP_CHECK_KERNEL();
+}
+
+// -------- SwiGLU: y = silu(g) * u; packed gu = [..., 2*I] (g|u) ----------
+__global__ void swiglu_fwd_kernel(const short* __restrict__ g,
+                                  const short* __restrict__ u,
+                                  short* __restrict__ y, long long n) {
+  long long i0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  long long n8 = n / 8;
+  for (long long i = i0; i < n8; i += stride) {
+    bf16x8 gv = reinterpret_cast<const bf16x8*>(g)[i];
+    bf16x8 uv = reinterpret_cast<const bf16x8*>(u)[i];
+    bf16x8 o;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      float gf = bf2f(gv.v[k]);
+      float s = gf / (1.f + __expf(-gf));
+      o.v[k] = f2bf(s * bf2f(uv.v[k]));
+    }
+    reinterpret_cast<bf16x8*>(y)[i] = o;
+  }
+  for (long long i = n8 * 8 + i0; i < n; i += stride) {
+    float gf = bf2f(g[i]);
+    float s = gf / (1.f + __expf(-gf));
+    y[i] = f2bf(s * bf2f(u[i]));
+  }
+}
+
+__global__ void swiglu_bwd_kernel(const short* __restrict__ dy,
+                                  const short* __restrict__ g,
+                                  const short* __restrict__ u,
+                                  short* __restrict__ dg,
+                                  short* __restrict__ du, long long n) {
+  long long i0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  long long n8 = n / 8;
+  for (long long i = i0; i < n8; i += stride) {
+    bf16x8 dv = reinterpret_cast<const bf16x8*>(dy)[i];
+    bf16x8 gv = reinterpret_cast<const bf16x8*>(g)[i];
+    bf16x8 uv = reinterpret_cast<const bf16x8*>(u)[i];
+    bf16x8 og, ou;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      float gf = bf2f(gv.v[k]);
+      float d = bf2f(dv.v[k]);
+      float sig = 1.f / (1.f + __expf(-gf));
+      float silu = gf * sig;
+      og.v[k] = f2bf(d * bf2f(uv.v[k]) * (sig + silu * (1.f - sig)));
+      ou.v[k] = f2bf(d * silu);
+    }
+    reinterpret_cast<bf16x8*>(dg)[i] = og;
+    reinterpret_cast<bf16x8*>(du)[i] = ou;
+  }
+  for (long long i = n8 * 8 + i0; i < n; i += stride) {
+    float gf = bf2f(g[i]);
+    float d = bf2f(dy[i]);
+    float sig = 1.f / (1.f + __expf(-gf));
+    float silu = gf * sig;
+    dg[i] = f2bf(d * bf2f(u[i]) * (sig + silu * (1.f - sig)));
+    du[i] = f2bf(d * silu);
+  }
+}
+
+at::Tensor swiglu_fwd(at::Tensor g, at::Tensor u) {
+  TORCH_CHECK(g.scalar_type() == at::kBFloat16 && g.is_contiguous());
+  auto y = at::empty_like(g);
+  long long n = g.numel();
+  auto stream = c10::hip::getCurrentHIPStream();
+  int block = 256;
+  int grid = grid_for(n / 8 + 1, block);
+  hipLaunchKernelGGL(swiglu_fwd_kernel, dim3(grid), dim3(block), 0,
+                     stream.stream(),
+                     reinterpret_cast<const short*>(g.data_ptr()),
+                     reinterpret_cast<const short*>(u.data_ptr()),
+                     reinterpret_cast<short*>(y.data_ptr()), n);
+  HIP_CHECK_KERNEL();
+  return y;
+}
+
+std::vector<at::Tensor> swiglu_bwd(at::Tensor dy, at::Tensor g, at::Tensor u) {
+  auto dg = at::empty_like(g);
+  auto du = at::empty_like(u);
+  long long n = g.numel();
+  auto stream = c10::hip::getCurrentHIPStream();
+  int block = 256;
+  int grid = grid_for(n / 8 + 1, block);
+  hipLaunchKernelGGL(swiglu_bwd_kernel, dim3(grid), dim3(block), 0,
+                     stream.stream(),
+                     reinterpret_cast<const short*>(dy.data_ptr()),
+                     reinterpret_cast<const short*>(g.data_ptr()),
+                     reinterpret_cast<const short*>(u.data_ptr()),
+                     reinterpret_cast<short*>(dg.data_ptr()),
+                     reinterpret_cast<short*>(du.data_ptr()), n);
+  HIP_CHECK_KERNEL();
+  return {dg, du};
+}

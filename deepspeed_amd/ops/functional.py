@@ -1,0 +1,237 @@
+"""Autograd wrappers for the gfx950 kernel set.
+
+GPU path: hand-written HIP kernels (required — no silent eager fallback).
+CPU path (unit tests / no-GPU container): equivalent fp32 torch math.
+"""
+import torch
+
+from .loader import get_ext
+
+
+# --------------------------------------------------------------- RMSNorm
+class _RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        if x.is_cuda:
+            ext = get_ext(required=True)
+            y, rstd = ext.rmsnorm_fwd(x.contiguous(), weight.contiguous(), eps)
+        else:
+            x32 = x.float()
+            rstd = torch.rsqrt(x32.pow(2).mean(-1, keepdim=True) + eps)
+            y = (x32 * rstd * weight.float()).to(x.dtype)
+            rstd = rstd.reshape(-1)
+        ctx.save_for_backward(x, weight, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, rstd = ctx.saved_tensors
+        if x.is_cuda:
+            ext = get_ext(required=True)
+            dx, dw = ext.rmsnorm_bwd(dy.contiguous(), x, weight, rstd)
+        else:
+            H = x.shape[-1]
+            x32 = x.float()
+            dy32 = dy.float()
+            rs = rstd.reshape(*x.shape[:-1], 1).float()
+            w32 = weight.float()
+            g = dy32 * w32
+            dot = (g * x32).sum(-1, keepdim=True)
+            dx = (rs * (g - x32 * dot * rs * rs / H)).to(x.dtype)
+            dw = (dy32 * x32 * rs).reshape(-1, H).sum(0)
+        return dx, dw.to(weight.dtype), None
+
+
+def rms_norm(x, weight, eps=1e-5):
+    return _RMSNormFn.apply(x, weight, eps)
+
+
+# -------------------------------------------------------------- LayerNorm
+class _LayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        if x.is_cuda:
+            ext = get_ext(required=True)
+            y, mean, rstd = ext.layernorm_fwd(x.contiguous(),
+                                              weight.contiguous(),
+                                              bias.contiguous()
+                                              if bias is not None else None,
+                                              eps)
+        else:
+            x32 = x.float()
+            mean = x32.mean(-1, keepdim=True)
+            var = x32.var(-1, unbiased=False, keepdim=True)
+            rstd = torch.rsqrt(var + eps)
+            xhat = (x32 - mean) * rstd
+            y = (xhat * weight.float() +
+                 (bias.float() if bias is not None else 0)).to(x.dtype)
+            mean = mean.reshape(-1)
+            rstd = rstd.reshape(-1)
+        ctx.save_for_backward(x, weight, mean, rstd)
+        ctx.has_bias = bias is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, mean, rstd = ctx.saved_tensors
+        if x.is_cuda:
+            ext = get_ext(required=True)
+            dx, dw, db = ext.layernorm_bwd(dy.contiguous(), x, weight, mean,
+                                           rstd)
+        else:
+            H = x.shape[-1]
+            x32 = x.float()
+            dy32 = dy.float()
+            mu = mean.reshape(*x.shape[:-1], 1)
+            rs = rstd.reshape(*x.shape[:-1], 1)
+            xhat = (x32 - mu) * rs
+            g = dy32 * weight.float()
+            c1 = g.mean(-1, keepdim=True)
+            c2 = (g * xhat).mean(-1, keepdim=True)
+            dx = ((g - c1 - xhat * c2) * rs).to(x.dtype)
+            dw = (dy32 * xhat).reshape(-1, H).sum(0)
+            db = dy32.reshape(-1, H).sum(0)
+        return (dx, dw.to(weight.dtype),
+                db.to(weight.dtype) if ctx.has_bias else None, None)
+
+
+def layer_norm(x, weight, bias=None, eps=1e-5):
+    return _LayerNormFn.apply(x, weight, bias, eps)
+
+
+# ------------------------------------------------------------------ RoPE
+class _RoPEFn(torch.autograd.Function):
+    """Rotates [B,S,H,D] with precomputed cos/sin [S, D/2] (rotate-half)."""
+
+    @staticmethod
+    def forward(ctx, t, cos, sin):
+        ctx.save_for_backward(cos, sin)
+        if t.is_cuda:
+            ext = get_ext(required=True)
+            out = t.contiguous().clone()
+            ext.rope_inplace(out, cos, sin, 0, False)
+        else:
+            out = _rope_torch(t, cos, sin, sign=1.0)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        cos, sin = ctx.saved_tensors
+        if dy.is_cuda:
+            ext = get_ext(required=True)
+            dx = dy.contiguous().clone()
+            ext.rope_inplace(dx, cos, sin, 0, True)
+        else:
+            dx = _rope_torch(dy, cos, sin, sign=-1.0)
+        return dx, None, None
+
+
+def _rope_torch(t, cos, sin, sign):
+    B, S, H, D = t.shape
+    half = D // 2
+    t32 = t.float()
+    x1 = t32[..., :half]
+    x2 = t32[..., half:]
+    c = cos[:S].reshape(1, S, 1, half)
+    s = sin[:S].reshape(1, S, 1, half) * sign
+    o1 = x1 * c - x2 * s
+    o2 = x2 * c + x1 * s
+    return torch.cat([o1, o2], dim=-1).to(t.dtype)
+
+
+def apply_rope(t, cos, sin):
+    return _RoPEFn.apply(t, cos, sin)
+
+
+def build_rope_cache(seq_len, head_dim, base=500000.0, device="cpu",
+                     scaling=None):
+    """Host-precomputed cos/sin table (guide Appendix B: no device trig)."""
+    half = head_dim // 2
+    inv_freq = 1.0 / (base ** (torch.arange(0, half, dtype=torch.float32,
+                                            device=device) / half))
+    pos = torch.arange(seq_len, dtype=torch.float32, device=device)
+    freqs = torch.outer(pos, inv_freq)
+    return freqs.cos().contiguous(), freqs.sin().contiguous()
+
+
+# ---------------------------------------------------------------- SwiGLU
+class _SwiGLUFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, g, u):
+        ctx.save_for_backward(g, u)
+        if g.is_cuda:
+            ext = get_ext(required=True)
+            return ext.swiglu_fwd(g.contiguous(), u.contiguous())
+        g32 = g.float()
+        return (torch.nn.functional.silu(g32) * u.float()).to(g.dtype)
+
+    @staticmethod
+    def backward(ctx, dy):
+        g, u = ctx.saved_tensors
+        if g.is_cuda:
+            ext = get_ext(required=True)
+            dg, du = ext.swiglu_bwd(dy.contiguous(), g.contiguous(),
+                                    u.contiguous())
+            return dg, du
+        g32, u32, dy32 = g.float(), u.float(), dy.float()
+        sig = torch.sigmoid(g32)
+        silu = g32 * sig
+        dg = (dy32 * u32 * (sig + silu * (1 - sig))).to(g.dtype)
+        du = (dy32 * silu).to(u.dtype)
+        return dg, du
+
+
+def swiglu(g, u):
+    return _SwiGLUFn.apply(g, u)
+
+
+# ---------------------------------------------------------- CrossEntropy
+class _FusedCrossEntropyFn(torch.autograd.Function):
+    """Mean CE over non-ignored rows; logits [N, V] 16-bit, targets [N]."""
+
+    @staticmethod
+    def forward(ctx, logits, targets, ignore_index):
+        if logits.is_cuda:
+            ext = get_ext(required=True)
+            loss, lse = ext.cross_entropy_fwd(logits.contiguous(), targets,
+                                              ignore_index)
+        else:
+            l32 = logits.float()
+            lse = torch.logsumexp(l32, dim=-1)
+            picked = l32.gather(
+                1, targets.clamp(min=0).unsqueeze(1)).squeeze(1)
+            loss = lse - picked
+            loss = torch.where(targets == ignore_index,
+                               torch.zeros_like(loss), loss)
+        valid = (targets != ignore_index)
+        n_valid = valid.sum().clamp(min=1)
+        ctx.save_for_backward(logits, targets, lse, n_valid)
+        ctx.ignore_index = ignore_index
+        return loss.sum() / n_valid.to(loss.dtype)
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logits, targets, lse, n_valid = ctx.saved_tensors
+        scale = (dloss.float() / n_valid.float())
+        dloss_rows = scale.expand(logits.shape[0]).contiguous()
+        if logits.is_cuda:
+            ext = get_ext(required=True)
+            dlogits = ext.cross_entropy_bwd(logits, targets, lse, dloss_rows,
+                                            ctx.ignore_index)
+        else:
+            p = torch.softmax(logits.float(), dim=-1)
+            onehot = torch.zeros_like(p)
+            valid = targets != ctx.ignore_index
+            idx = targets.clamp(min=0)
+            onehot.scatter_(1, idx.unsqueeze(1), 1.0)
+            g = (p - onehot) * dloss_rows.unsqueeze(1)
+            g = torch.where(valid.unsqueeze(1), g, torch.zeros_like(g))
+            dlogits = g.to(logits.dtype)
+        return dlogits, None, None
+
+
+def fused_cross_entropy(logits, targets, ignore_index=-100):
+    """logits [..., V] (bf16 on GPU), targets [...] int64 -> scalar mean."""
+    V = logits.shape[-1]
+    return _FusedCrossEntropyFn.apply(logits.reshape(-1, V),
+                                      targets.reshape(-1), ignore_index)

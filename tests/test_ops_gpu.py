@@ -1,0 +1,132 @@
+"""HIP kernel numerics vs plain fp32 torch references (needs MI355X)."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from deepspeed_amd.ops import functional as Fops
+    from deepspeed_amd.ops.loader import get_ext
+
+
+def _dev():
+    return torch.device("cuda:0")
+
+
+@pytest.mark.parametrize("shape", [(8, 128, 4096), (3, 1, 8192), (2, 7, 1000)])
+def test_rmsnorm_gpu(shape):
+    torch.manual_seed(0)
+    x = torch.randn(*shape, device=_dev(), dtype=torch.bfloat16,
+                    requires_grad=True)
+    w = torch.randn(shape[-1], device=_dev(), dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = Fops.rms_norm(x, w, 1e-5)
+    x32 = x.detach().float().requires_grad_(True)
+    w32 = w.detach().float().requires_grad_(True)
+    ref = x32 * torch.rsqrt(x32.pow(2).mean(-1, keepdim=True) + 1e-5) * w32
+    assert torch.allclose(y.float(), ref, atol=3e-2, rtol=3e-2)
+    g = torch.randn_like(y)
+    y.backward(g)
+    ref.backward(g.float())
+    assert torch.allclose(x.grad.float(), x32.grad, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(w.grad.float(), w32.grad, atol=5e-1, rtol=2e-2)
+
+
+def test_rope_gpu():
+    torch.manual_seed(0)
+    B, S, H, D = 2, 256, 8, 128
+    cos, sin = Fops.build_rope_cache(S, D, base=500000.0, device=_dev())
+    t = torch.randn(B, S, H, D, device=_dev(), dtype=torch.bfloat16,
+                    requires_grad=True)
+    out = Fops.apply_rope(t, cos, sin)
+    ref = Fops._rope_torch(t.detach().float(), cos, sin, sign=1.0)
+    assert torch.allclose(out.float(), ref, atol=2e-2, rtol=2e-2)
+    g = torch.randn_like(out)
+    out.backward(g)
+    gref = Fops._rope_torch(g.float(), cos, sin, sign=-1.0)
+    assert torch.allclose(t.grad.float(), gref, atol=2e-2, rtol=2e-2)
+
+
+def test_swiglu_gpu():
+    torch.manual_seed(0)
+    g = torch.randn(4, 1024, 14336, device=_dev(), dtype=torch.bfloat16,
+                    requires_grad=True)
+    u = torch.randn_like(g).requires_grad_(True)
+    y = Fops.swiglu(g, u)
+    g32 = g.detach().float().requires_grad_(True)
+    u32 = u.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.silu(g32) * u32
+    assert torch.allclose(y.float(), ref, atol=5e-2, rtol=5e-2)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    ref.backward(dy.float())
+    assert torch.allclose(g.grad.float(), g32.grad, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(u.grad.float(), u32.grad, atol=5e-2, rtol=5e-2)
+
+
+@pytest.mark.parametrize("V", [128256, 1000])
+def test_cross_entropy_gpu(V):
+    torch.manual_seed(0)
+    N = 512
+    logits = torch.randn(N, V, device=_dev(), dtype=torch.bfloat16,
+                         requires_grad=True) * 4
+    targets = torch.randint(0, V, (N,), device=_dev())
+    targets[5] = -100
+    loss = Fops.fused_cross_entropy(logits, targets)
+    l32 = logits.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(l32, targets, ignore_index=-100)
+    assert torch.allclose(loss, ref, atol=2e-2, rtol=1e-2), \
+        f"{loss.item()} vs {ref.item()}"
+    loss.backward()
+    ref.backward()
+    assert torch.allclose(logits.grad.float(), l32.grad, atol=2e-3, rtol=5e-2)
+
+
+def test_fused_adam_gpu_vs_torch():
+    torch.manual_seed(0)
+    from deepspeed_amd.ops.adam import FusedAdam
+    p1 = torch.randn(123457, device=_dev(), requires_grad=True)
+    p2 = p1.detach().clone().requires_grad_(True)
+    o1 = FusedAdam([p1], lr=1e-2, weight_decay=0.01)
+    o2 = torch.optim.AdamW([p2], lr=1e-2, weight_decay=0.01)
+    for i in range(5):
+        g = torch.randn(123457, device=_dev())
+        p1.grad = g.clone()
+        p2.grad = g.clone()
+        o1.step()
+        o2.step()
+    assert torch.allclose(p1, p2, atol=1e-5), (p1 - p2).abs().max()
+
+
+def test_adam_fused_bf16_out():
+    ext = get_ext(required=True)
+    torch.manual_seed(0)
+    n = 7777
+    p = torch.randn(n, device=_dev(), dtype=torch.float32)
+    g = torch.randn(n, device=_dev(), dtype=torch.float32)
+    m = torch.zeros(n, device=_dev(), dtype=torch.float32)
+    v = torch.zeros(n, device=_dev(), dtype=torch.float32)
+    out16 = torch.empty(n, device=_dev(), dtype=torch.bfloat16)
+    ext.multi_tensor_adam([p], [g], [m], [v], 1e-3, 0.9, 0.999, 1e-8, 1, 1,
+                          1, 0.0, [out16])
+    assert torch.allclose(out16.float(), p, atol=1e-2, rtol=1e-2)
+
+
+def test_layernorm_gpu():
+    torch.manual_seed(0)
+    x = torch.randn(512, 4096, device=_dev(), dtype=torch.bfloat16,
+                    requires_grad=True)
+    w = torch.randn(4096, device=_dev(), dtype=torch.bfloat16,
+                    requires_grad=True)
+    b = torch.randn(4096, device=_dev(), dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = Fops.layer_norm(x, w, b, 1e-5)
+    x32, w32, b32 = (t.detach().float().requires_grad_(True)
+                     for t in (x, w, b))
+    ref = torch.nn.functional.layer_norm(x32, (4096,), w32, b32, 1e-5)
+    assert torch.allclose(y.float(), ref, atol=5e-2, rtol=5e-2)
+    g = torch.randn_like(y)
+    y.backward(g)
+    ref.backward(g.float())
+    assert torch.allclose(x.grad.float(), x32.grad, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(w.grad.float(), w32.grad, atol=5e-1, rtol=2e-2)
