@@ -105,6 +105,9 @@ class _RoPEFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, t, cos, sin):
+        # engine-wide .to(bf16) casts buffers; the table must stay fp32
+        cos = cos.float()
+        sin = sin.float()
         ctx.save_for_backward(cos, sin)
         if t.is_cuda:
             ext = get_ext(required=True)
