@@ -68,8 +68,8 @@ def test_swiglu_gpu():
 def test_cross_entropy_gpu(V):
     torch.manual_seed(0)
     N = 512
-    logits = torch.randn(N, V, device=_dev(), dtype=torch.bfloat16,
-                         requires_grad=True) * 4
+    logits = (torch.randn(N, V, device=_dev(), dtype=torch.bfloat16) *
+              4).requires_grad_(True)
     targets = torch.randint(0, V, (N,), device=_dev())
     targets[5] = -100
     loss = Fops.fused_cross_entropy(logits, targets)
