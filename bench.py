@@ -26,6 +26,9 @@ def parse_args():
     p.add_argument("--micro-batch", type=int, default=1)
     p.add_argument("--grad-accum", type=int, default=1)
     p.add_argument("--zero-stage", type=int, default=3)
+    p.add_argument("--act-ckpt", action="store_true",
+                   help="enable activation checkpointing (default off: "
+                   "288 GB HBM3E fits full activations at these configs)")
     p.add_argument("--local_rank", type=int, default=-1)
     return p.parse_args()
 
@@ -48,6 +51,7 @@ def main():
         torch.cuda.set_device(device)
 
     cfg = LLAMA_CONFIGS[args.model]
+    cfg.activation_checkpointing = args.act_ckpt
     torch.manual_seed(1234 + rank)
     t0 = time.time()
     # build directly on device: 8B bf16 = 16 GB, fits trivially in 288 GB
@@ -73,10 +77,16 @@ def main():
 
     S = args.seq_len
     B = args.micro_batch
-    data = torch.randint(0, cfg.vocab_size, (B, S), device=device)
+    # rotating synthetic batches: fresh tokens each micro-step (no caching)
+    n_bufs = 8
+    bufs = [torch.randint(0, cfg.vocab_size, (B, S), device=device)
+            for _ in range(n_bufs)]
+    step_idx = [0]
 
     def one_step():
         for _ in range(args.grad_accum):
+            data = bufs[step_idx[0] % n_bufs]
+            step_idx[0] += 1
             loss = engine(data, labels=data)
             engine.backward(loss)
             engine.step()
