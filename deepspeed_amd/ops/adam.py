@@ -25,6 +25,16 @@ class FusedAdam(torch.optim.Optimizer):
         super().__init__(params, defaults)
         self.adam_w_mode = 1 if adam_w_mode else 0
         self.set_grad_none = set_grad_none
+        # ZeRO fusion hooks: grad scale folded into the kernel (no extra
+        # pass over grads) and bf16 shard writeback fused into the update.
+        self._grad_scale = 1.0
+        self._out16 = {}  # param -> flat bf16 tensor (same numel)
+
+    def set_grad_scale(self, scale):
+        self._grad_scale = float(scale)
+
+    def set_fused_out16(self, mapping):
+        self._out16 = dict(mapping)
 
     def zero_grad(self, set_to_none=True):
         if self.set_grad_none or set_to_none:
@@ -66,14 +76,24 @@ class FusedAdam(torch.optim.Optimizer):
 
             if params[0].is_cuda:
                 ext = get_ext(required=True)
+                out16 = []
+                if self._out16 and all(p in self._out16 for p in params):
+                    out16 = [self._out16[p] for p in params]
                 ext.multi_tensor_adam(
                     params, grads_, exp_avgs, exp_avg_sqs, group["lr"], beta1,
                     beta2, group["eps"], step, self.adam_w_mode,
-                    bias_correction, group["weight_decay"])
+                    bias_correction, group["weight_decay"], out16,
+                    self._grad_scale)
             else:
+                if self._grad_scale != 1.0:
+                    for g in grads_:
+                        g.mul_(self._grad_scale)
                 self._cpu_step(params, grads_, exp_avgs, exp_avg_sqs,
                                group["lr"], beta1, beta2, group["eps"], step,
                                bias_correction, group["weight_decay"])
+                for p in params:
+                    if p in self._out16:
+                        self._out16[p].copy_(p.detach().reshape(-1))
         return loss
 
     def _cpu_step(self, params, grads, exp_avgs, exp_avg_sqs, lr, beta1,

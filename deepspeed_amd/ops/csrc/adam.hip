@@ -16,7 +16,7 @@ __global__ void adam_kernel_f32(float* __restrict__ p,
                                 short* __restrict__ out16,  // optional bf16 out
                                 long long n, float lr, float beta1, float beta2,
                                 float eps, float bc1, float bc2, int adamw,
-                                float wd) {
+                                float wd, float gscale) {
   const float step_size = lr / bc1;
   const float bc2_sqrt = sqrtf(bc2);
   long long i0 = (long long)(blockIdx.x) * blockDim.x + threadIdx.x;
@@ -32,7 +32,7 @@ __global__ void adam_kernel_f32(float* __restrict__ p,
     short o16[4];
 #pragma unroll
     for (int k = 0; k < 4; ++k) {
-      float gk = gv.v[k];
+      float gk = gv.v[k] * gscale;
       float pk = pv.v[k];
       if (adamw) {
         pk *= (1.f - lr * wd);
@@ -58,7 +58,7 @@ __global__ void adam_kernel_f32(float* __restrict__ p,
   }
   // tail
   for (long long i = n4 * 4 + i0; i < n; i += stride) {
-    float gk = g[i], pk = p[i];
+    float gk = g[i] * gscale, pk = p[i];
     if (adamw) {
       pk *= (1.f - lr * wd);
     } else if (wd != 0.f) {
@@ -82,7 +82,8 @@ void multi_tensor_adam(std::vector<at::Tensor> params,
                        double beta1, double beta2, double eps, long step,
                        long adamw_mode, long bias_correction,
                        double weight_decay,
-                       std::vector<at::Tensor> out16 /* may be empty */) {
+                       std::vector<at::Tensor> out16 /* may be empty */,
+                       double grad_scale) {
   float bc1 = 1.f, bc2 = 1.f;
   if (bias_correction) {
     bc1 = 1.f - powf((float)beta1, (float)step);
@@ -107,7 +108,8 @@ void multi_tensor_adam(std::vector<at::Tensor> params,
                        exp_avgs[t].data_ptr<float>(),
                        exp_avg_sqs[t].data_ptr<float>(), o16, n, (float)lr,
                        (float)beta1, (float)beta2, (float)eps, bc1, bc2,
-                       (int)adamw_mode, (float)weight_decay);
+                       (int)adamw_mode, (float)weight_decay,
+                       (float)grad_scale);
     HIP_CHECK_KERNEL();
   }
 }

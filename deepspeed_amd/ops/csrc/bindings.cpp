@@ -7,7 +7,10 @@ void multi_tensor_adam(std::vector<at::Tensor> params,
                        std::vector<at::Tensor> exp_avg_sqs, double lr,
                        double beta1, double beta2, double eps, long step,
                        long adamw_mode, long bias_correction,
-                       double weight_decay, std::vector<at::Tensor> out16);
+                       double weight_decay, std::vector<at::Tensor> out16,
+                       double grad_scale);
+void accum_bf16_to_f32(at::Tensor dst, at::Tensor src, double scale);
+at::Tensor l2norm_sq(std::vector<at::Tensor> tensors);
 void multi_tensor_lion(std::vector<at::Tensor> params,
                        std::vector<at::Tensor> grads,
                        std::vector<at::Tensor> exp_avgs, double lr,
@@ -38,7 +41,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("exp_avg_sqs"), py::arg("lr"), py::arg("beta1"),
         py::arg("beta2"), py::arg("eps"), py::arg("step"),
         py::arg("adamw_mode"), py::arg("bias_correction"),
-        py::arg("weight_decay"), py::arg("out16") = std::vector<at::Tensor>());
+        py::arg("weight_decay"), py::arg("out16") = std::vector<at::Tensor>(),
+        py::arg("grad_scale") = 1.0);
+  m.def("accum_bf16_to_f32", &accum_bf16_to_f32, py::arg("dst"),
+        py::arg("src"), py::arg("scale") = 1.0);
+  m.def("l2norm_sq", &l2norm_sq);
   m.def("multi_tensor_lion", &multi_tensor_lion, "fused Lion (gfx950)",
         py::arg("params"), py::arg("grads"), py::arg("exp_avgs"),
         py::arg("lr"), py::arg("beta1"), py::arg("beta2"),

@@ -33,21 +33,29 @@ def _pad_to(n, m):
 
 
 class SubGroup:
-    """Flat fp32 master + grad accumulator over a run of param shards."""
+    """Flat fp32 master + flat bf16 shard slab + grad accumulator.
 
-    __slots__ = ("params", "offsets", "master32", "grad32", "group_idx",
-                 "numel")
+    The 16-bit shards (`p.ds_tensor`) are re-pointed to views of one flat
+    bf16 buffer so the fused Adam kernel writes updated bf16 shards in the
+    same pass as the fp32 update (no separate cast+copy over 16 GB)."""
+
+    __slots__ = ("params", "offsets", "master32", "grad32", "flat16",
+                 "group_idx", "numel")
 
     def __init__(self, params, offsets, numel, group_idx, device):
         self.params = params
         self.offsets = offsets
         self.numel = numel
         self.group_idx = group_idx
+        dtype16 = params[0].ds_tensor.dtype if params else torch.bfloat16
+        self.flat16 = torch.empty(numel, dtype=dtype16, device=device)
         self.master32 = torch.empty(numel, dtype=torch.float32, device=device)
         for p in params:
             off = offsets[p]
-            self.master32[off:off + p.ds_tensor.numel()].copy_(
-                p.ds_tensor.float())
+            n = p.ds_tensor.numel()
+            self.flat16[off:off + n].copy_(p.ds_tensor)
+            p.ds_tensor = self.flat16[off:off + n]
+            self.master32[off:off + n].copy_(p.ds_tensor.float())
         self.master32 = self.master32.detach().requires_grad_(True)
         self.grad32 = torch.zeros(numel, dtype=torch.float32, device=device)
 
@@ -56,10 +64,7 @@ class SubGroup:
         return self.grad32[off:off + p.ds_tensor.numel()]
 
     def copy_master_to_shards(self):
-        for p in self.params:
-            off = self.offsets[p]
-            p.ds_tensor.copy_(
-                self.master32[off:off + p.ds_tensor.numel()].detach())
+        self.flat16.copy_(self.master32.detach())
 
 
 class ZeroStage3Optimizer:
@@ -311,34 +316,54 @@ class ZeroStage3Optimizer:
         if world == 1:
             for p in params:
                 sg = self.param_to_subgroup[p]
-                sg.grad_shard_view(p)[:p.ds_numel].add_(
-                    p.grad.reshape(-1).float())
+                dst = sg.grad_shard_view(p)[:p.ds_numel]
+                src = p.grad.reshape(-1)
+                if p.grad.is_cuda and src.dtype == torch.bfloat16:
+                    from ...ops.loader import get_ext
+                    get_ext(required=True).accum_bf16_to_f32(dst, src, 1.0)
+                else:
+                    dst.add_(src.float())
                 p.grad = None
             return
+        from .stage_1_and_2 import _avg_op
         inputs = []
         for p in params:
             shard_numel = p.ds_tensor.numel()
-            padded = torch.zeros(shard_numel * world, dtype=p.grad.dtype,
+            padded = torch.empty(shard_numel * world, dtype=p.grad.dtype,
                                  device=p.grad.device)
             padded[:p.ds_numel].copy_(p.grad.reshape(-1))
-            padded.div_(world)
+            if padded.numel() > p.ds_numel:
+                padded[p.ds_numel:].zero_()
             inputs.append(padded)
             shards.append(torch.empty(shard_numel, dtype=p.grad.dtype,
                                       device=p.grad.device))
             p.grad = None
         if use_coalescing:
             from torch.distributed.distributed_c10d import _coalescing_manager
+            op = _avg_op(world, inputs[0]) if inputs[0].is_cuda else None
+            if op is None:
+                for inp in inputs:
+                    inp.div_(world)
+                op = dist.ReduceOp.SUM
             with _coalescing_manager(self.dp_group, self.device,
                                      async_ops=True) as cm:
                 for out, inp in zip(shards, inputs):
-                    dist.reduce_scatter_tensor(out, inp, group=self.dp_group)
+                    dist.reduce_scatter_tensor(out, inp, op=op,
+                                               group=self.dp_group)
             cm.wait()
         else:
             for out, inp in zip(shards, inputs):
-                dist.reduce_scatter_tensor(out, inp, group=self.dp_group)
+                op = _avg_op(world, inp)
+                dist.reduce_scatter_tensor(out, inp, op=op,
+                                           group=self.dp_group)
         for p, shard in zip(params, shards):
             sg = self.param_to_subgroup[p]
-            sg.grad_shard_view(p).add_(shard.float())
+            dst = sg.grad_shard_view(p)
+            if shard.is_cuda and shard.dtype == torch.bfloat16:
+                from ...ops.loader import get_ext
+                get_ext(required=True).accum_bf16_to_f32(dst, shard, 1.0)
+            else:
+                dst.add_(shard.float())
 
     # -------------------------------------------------------------- train
     def backward(self, loss, retain_graph=False):
@@ -362,23 +387,29 @@ class ZeroStage3Optimizer:
             found = bool(t.item())
         return found
 
-    def _unscale_and_clip(self):
+    def _combined_scale(self):
+        """loss_scale x clip coefficient; folded into the Adam kernel as
+        grad_scale = 1/combined (zero extra passes over the 32 GB shards)."""
         scale = self.loss_scaler.loss_scale
         combined = scale
         if self.clip_grad > 0.0:
-            total_sq = torch.zeros(1, dtype=torch.float64, device=self.device)
-            for sg in self.sub_groups:
-                total_sq += sg.grad32.double().pow(2).sum()
+            grads = [sg.grad32 for sg in self.sub_groups]
+            if grads and grads[0].is_cuda:
+                from ...ops.loader import get_ext
+                total_sq = get_ext(required=True).l2norm_sq(grads).double()
+            else:
+                total_sq = torch.zeros(1, dtype=torch.float64,
+                                       device=self.device)
+                for g in grads:
+                    total_sq += g.double().pow(2).sum()
             if dist.is_initialized():
                 dist.all_reduce(total_sq, group=self.dp_group)
-            norm = (total_sq.sqrt().item()**1.0) / scale
+            norm = total_sq.sqrt().item() / scale
             self._global_grad_norm = norm
             clip = norm / self.clip_grad
             if clip > 1.0:
                 combined = scale * clip
-        if combined != 1.0:
-            for sg in self.sub_groups:
-                sg.grad32.mul_(1.0 / combined)
+        return combined
 
     def step(self, closure=None):
         assert closure is None
@@ -393,14 +424,27 @@ class ZeroStage3Optimizer:
                 self._clear_grads()
                 return
 
-        self._unscale_and_clip()
+        combined = self._combined_scale()
+
+        fused = hasattr(self.optimizer, "set_grad_scale")
+        if fused:
+            self.optimizer.set_grad_scale(1.0 / combined)
+            self.optimizer.set_fused_out16(
+                {sg.master32: sg.flat16 for sg in self.sub_groups})
+        elif combined != 1.0:
+            for sg in self.sub_groups:
+                sg.grad32.mul_(1.0 / combined)
 
         for sg in self.sub_groups:
             sg.master32.grad = sg.grad32
         self.optimizer.step()
         for sg in self.sub_groups:
             sg.master32.grad = None
-            sg.copy_master_to_shards()
+            if not fused:
+                sg.copy_master_to_shards()
+        if fused:
+            self.optimizer.set_grad_scale(1.0)
+            self.optimizer.set_fused_out16({})
         self._clear_grads()
         self._refresh_persistent_params()
 

@@ -35,6 +35,23 @@ def _pad_to(numel, multiple):
     return (numel + multiple - 1) // multiple * multiple
 
 
+def _avg_op(world, tensor):
+    """RCCL PreMulSum(1/world) — averages on the wire, no pre-divide pass.
+
+    Falls back to an explicit in-place pre-divide where unsupported (gloo).
+    """
+    if world <= 1:
+        return dist.ReduceOp.SUM
+    if tensor.is_cuda:
+        try:
+            import torch.distributed as td
+            return td._make_nccl_premul_sum(1.0 / world)
+        except Exception:
+            pass
+    tensor.div_(world)
+    return dist.ReduceOp.SUM
+
+
 class Bucket:
     """A flat slab of parameters partitioned across the DP group."""
 
@@ -225,12 +242,16 @@ class ZeroStage12Optimizer:
         else:
             ctx = _nullctx()
         with ctx:
-            # pre-divide for 16-bit range safety, then SUM-reduce-scatter
-            b.grad16.div_(self.world)
             shard = torch.empty(b.shard_numel, dtype=self.dtype,
                                 device=b.grad16.device)
-            dist.reduce_scatter_tensor(shard, b.grad16, group=self.dp_group)
-            b.grad32.add_(shard.float())
+            op = _avg_op(self.world, b.grad16)
+            dist.reduce_scatter_tensor(shard, b.grad16, op=op,
+                                       group=self.dp_group)
+            if shard.is_cuda and shard.dtype == torch.bfloat16:
+                from ...ops.loader import get_ext
+                get_ext(required=True).accum_bf16_to_f32(b.grad32, shard, 1.0)
+            else:
+                b.grad32.add_(shard.float())
             if stream is not None:
                 shard.record_stream(stream)
 
@@ -240,14 +261,19 @@ class ZeroStage12Optimizer:
 
     # -- step ---------------------------------------------------------------
 
-    def _unscale_and_clip(self):
+    def _combined_scale(self):
         scale = self.loss_scaler.loss_scale
-        # global grad norm over shards
         combined_scale = scale
         if self.clip_grad > 0.0:
-            total_sq = torch.zeros(1, dtype=torch.float64, device=self.device)
-            for b in self.buckets:
-                total_sq += b.grad32.double().pow(2).sum()
+            grads = [b.grad32 for b in self.buckets if b.grad32 is not None]
+            if grads and grads[0].is_cuda:
+                from ...ops.loader import get_ext
+                total_sq = get_ext(required=True).l2norm_sq(grads).double()
+            else:
+                total_sq = torch.zeros(1, dtype=torch.float64,
+                                       device=self.device)
+                for g in grads:
+                    total_sq += g.double().pow(2).sum()
             if dist.is_initialized():
                 dist.all_reduce(total_sq, group=self.dp_group)
             norm = (total_sq.sqrt() / scale).item()
@@ -255,9 +281,7 @@ class ZeroStage12Optimizer:
             clip = norm / self.clip_grad
             if clip > 1.0:
                 combined_scale = scale * clip
-        if combined_scale != 1.0:
-            for b in self.buckets:
-                b.grad32.mul_(1.0 / combined_scale)
+        return combined_scale
 
     def has_overflow(self):
         grads = [b.grad32 for b in self.buckets if b.grad32 is not None]
@@ -276,7 +300,15 @@ class ZeroStage12Optimizer:
                 self._clear_grads()
                 return
 
-        self._unscale_and_clip()
+        combined = self._combined_scale()
+        fused = hasattr(self.optimizer, "set_grad_scale")
+        if fused:
+            self.optimizer.set_grad_scale(1.0 / combined)
+            self.optimizer.set_fused_out16(
+                {b.master32: b.shard16 for b in self.buckets})
+        elif combined != 1.0:
+            for b in self.buckets:
+                b.grad32.mul_(1.0 / combined)
 
         for b in self.buckets:
             b.master32.grad = b.grad32
@@ -285,7 +317,11 @@ class ZeroStage12Optimizer:
 
         for b in self.buckets:
             b.master32.grad = None
-            b.shard16.copy_(b.master32.detach())
+            if not fused:
+                b.shard16.copy_(b.master32.detach())
+        if fused:
+            self.optimizer.set_grad_scale(1.0)
+            self.optimizer.set_fused_out16({})
 
         # all-gather updated 16-bit params, one call per bucket
         for b in self.buckets:
