@@ -11,6 +11,9 @@ void multi_tensor_adam(std::vector<at::Tensor> params,
                        double grad_scale);
 void accum_bf16_to_f32(at::Tensor dst, at::Tensor src, double scale);
 at::Tensor l2norm_sq(std::vector<at::Tensor> tensors);
+std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
+                                       at::Tensor v, bool causal,
+                                       double scale);
 void multi_tensor_lion(std::vector<at::Tensor> params,
                        std::vector<at::Tensor> grads,
                        std::vector<at::Tensor> exp_avgs, double lr,
@@ -46,6 +49,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("accum_bf16_to_f32", &accum_bf16_to_f32, py::arg("dst"),
         py::arg("src"), py::arg("scale") = 1.0);
   m.def("l2norm_sq", &l2norm_sq);
+  m.def("flash_attn_fwd", &flash_attn_fwd, py::arg("q"), py::arg("k"),
+        py::arg("v"), py::arg("causal") = true, py::arg("scale") = 0.0);
   m.def("multi_tensor_lion", &multi_tensor_lion, "fused Lion (gfx950)",
         py::arg("params"), py::arg("grads"), py::arg("exp_avgs"),
         py::arg("lr"), py::arg("beta1"), py::arg("beta2"),
