@@ -71,6 +71,8 @@ class LlamaAttention(nn.Module):
         self.k_proj = nn.Linear(h, cfg.num_key_value_heads * d, bias=False)
         self.v_proj = nn.Linear(h, cfg.num_key_value_heads * d, bias=False)
         self.o_proj = nn.Linear(cfg.num_attention_heads * d, h, bias=False)
+        self.sp_group = None  # set by enable_ulysses()
+        self._dist_attn = None
 
     def forward(self, x, cos, sin):
         B, S, _ = x.shape
@@ -78,10 +80,24 @@ class LlamaAttention(nn.Module):
         q = self.q_proj(x).view(B, S, -1, d)
         k = self.k_proj(x).view(B, S, -1, d)
         v = self.v_proj(x).view(B, S, -1, d)
+        # RoPE uses the caller's cos/sin slice (seq-offset aware under SP)
         q = apply_rope(q, cos, sin)
         k = apply_rope(k, cos, sin)
-        o = flash_attention(q, k, v, causal=True)
+        if self._dist_attn is not None:
+            o = self._dist_attn(q, k, v, causal=True)
+        else:
+            o = flash_attention(q, k, v, causal=True)
         return self.o_proj(o.reshape(B, S, -1))
+
+
+def enable_ulysses(model, sp_group=None):
+    """Route every attention through Ulysses head-scatter all-to-all."""
+    from ..sequence.layer import DistributedAttention
+    for mod in model.modules():
+        if isinstance(mod, LlamaAttention):
+            mod.sp_group = sp_group
+            mod._dist_attn = DistributedAttention(flash_attention, sp_group)
+    return model
 
 
 class LlamaMLP(nn.Module):
@@ -126,11 +142,11 @@ class LlamaModel(nn.Module):
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
 
-    def forward(self, input_ids):
+    def forward(self, input_ids, seq_offset=0):
         x = self.embed_tokens(input_ids)
         S = input_ids.shape[1]
-        cos = self.rope_cos[:S]
-        sin = self.rope_sin[:S]
+        cos = self.rope_cos[seq_offset:seq_offset + S]
+        sin = self.rope_sin[seq_offset:seq_offset + S]
         for layer in self.layers:
             if self.cfg.activation_checkpointing and self.training:
                 x = torch.utils.checkpoint.checkpoint(
@@ -159,8 +175,8 @@ class LlamaForCausalLM(nn.Module):
         elif isinstance(module, nn.Embedding):
             module.weight.data.normal_(0.0, std)
 
-    def forward(self, input_ids, labels=None):
-        h = self.model(input_ids)
+    def forward(self, input_ids, labels=None, seq_offset=0):
+        h = self.model(input_ids, seq_offset=seq_offset)
         logits = self.lm_head(h)
         if labels is None:
             return logits

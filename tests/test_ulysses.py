@@ -1,0 +1,108 @@
+"""Ulysses sequence parallelism: 2-rank sharded attention == full attention."""
+import torch
+
+from tests.common import run_distributed
+
+
+def _ulysses_forward():
+    import torch.distributed as tdist
+    from deepspeed_amd.comm import groups
+    from deepspeed_amd.sequence.layer import DistributedAttention
+    import torch.nn.functional as F
+
+    rank = tdist.get_rank()
+    world = tdist.get_world_size()
+    groups.reset_groups()
+    spg = groups.initialize_sequence_parallel(world)
+
+    torch.manual_seed(5)
+    B, S, H, D = 2, 16, 4, 8
+    q = torch.randn(B, S, H, D)
+    k = torch.randn(B, S, H, D)
+    v = torch.randn(B, S, H, D)
+
+    def local_attn(q_, k_, v_, causal=True):
+        qt, kt, vt = (t.transpose(1, 2) for t in (q_, k_, v_))
+        o = F.scaled_dot_product_attention(qt, kt, vt, is_causal=causal)
+        return o.transpose(1, 2)
+
+    # full-sequence reference (identical on both ranks)
+    ref = local_attn(q, k, v)
+
+    s = S // world
+    ql = q[:, rank * s:(rank + 1) * s].clone().requires_grad_(True)
+    kl = k[:, rank * s:(rank + 1) * s].clone().requires_grad_(True)
+    vl = v[:, rank * s:(rank + 1) * s].clone().requires_grad_(True)
+    dist_attn = DistributedAttention(local_attn, spg)
+    out = dist_attn(ql, kl, vl, causal=True)
+    assert out.shape == (B, s, H, D)
+    ref_local = ref[:, rank * s:(rank + 1) * s]
+    assert torch.allclose(out, ref_local, atol=1e-5), \
+        (out - ref_local).abs().max().item()
+
+    # backward: grads flow through both a2a
+    out.sum().backward()
+    qr = q.clone().requires_grad_(True)
+    kr = k.clone().requires_grad_(True)
+    vr = v.clone().requires_grad_(True)
+    local_attn(qr, kr, vr).sum().backward()
+    assert torch.allclose(ql.grad, qr.grad[:, rank * s:(rank + 1) * s],
+                          atol=1e-5)
+    assert torch.allclose(vl.grad, vr.grad[:, rank * s:(rank + 1) * s],
+                          atol=1e-5)
+    return True
+
+
+def test_ulysses_matches_full_attention():
+    results = run_distributed(_ulysses_forward, world_size=2)
+    assert all(results)
+
+
+def _ulysses_llama():
+    import torch.distributed as tdist
+    from deepspeed_amd.comm import groups
+    from deepspeed_amd.models.llama import (LLAMA_CONFIGS, LlamaForCausalLM,
+                                            enable_ulysses)
+
+    rank = tdist.get_rank()
+    world = tdist.get_world_size()
+    groups.reset_groups()
+    spg = groups.initialize_sequence_parallel(world)
+
+    cfg = LLAMA_CONFIGS["llama-tiny"]
+    torch.manual_seed(3)
+    model = LlamaForCausalLM(cfg).float()
+    S = 32
+    data = torch.randint(0, cfg.vocab_size, (1, S))
+    # full-model reference
+    ref_loss = model(data, labels=data)
+
+    s = S // world
+    sp_model = enable_ulysses(model, spg)
+    local = data[:, rank * s:(rank + 1) * s]
+    # NOTE: shifted-label CE differs at shard boundaries; compare logits
+    logits = sp_model(local, seq_offset=rank * s)
+    full_logits = model_full_logits(model, data)
+    ref_local = full_logits[:, rank * s:(rank + 1) * s]
+    assert torch.allclose(logits, ref_local, atol=1e-4), \
+        (logits - ref_local).abs().max().item()
+    return float(ref_loss)
+
+
+def model_full_logits(model, data):
+    from deepspeed_amd.models.llama import LlamaAttention
+    # temporarily disable SP
+    saved = []
+    for mod in model.modules():
+        if isinstance(mod, LlamaAttention):
+            saved.append((mod, mod._dist_attn))
+            mod._dist_attn = None
+    out = model(data)
+    for mod, da in saved:
+        mod._dist_attn = da
+    return out
+
+
+def test_ulysses_llama_logits_match():
+    results = run_distributed(_ulysses_llama, world_size=2)
+    assert abs(results[0] - results[1]) < 1e-5
