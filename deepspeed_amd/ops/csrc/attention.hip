@@ -152,7 +152,8 @@ __global__ void flash_fwd_kernel(
 
     // ---- online softmax -------------------------------------------------
     // lane holds S[row = l4*4 + r][col = l15 + 16*ct] (scaled below)
-    const bool btile = causal && (kvbase + KVTILE > qbase);
+    const bool btile = (causal && (kvbase + KVTILE > qbase)) ||
+                       (kvbase + KVTILE > S);
     float p_new[4][4];  // [ct][r]
     float corr[4];
 #pragma unroll
@@ -164,7 +165,7 @@ __global__ void flash_fwd_kernel(
         float sv = s_acc[ct][r] * scale;
         if (btile) {
           int gcol = kvbase + l15 + 16 * ct;
-          if (gcol > grow) sv = -INFINITY;
+          if ((causal && gcol > grow) || gcol >= S) sv = -INFINITY;
         }
         s_acc[ct][r] = sv;
         rowmax = fmaxf(rowmax, sv);

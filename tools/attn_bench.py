@@ -1,8 +1,11 @@
 #!/usr/bin/env python3
 """A/B attention microbenchmark: HIP flash fwd + chunked bwd vs torch SDPA."""
 import math
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
