@@ -57,14 +57,14 @@ def test_flash_fwd(B, S, Hq, Hk, causal):
 def test_flash_bwd(B, S, Hq, Hk):
     torch.manual_seed(1)
     D = 128
-    from deepspeed_amd.ops.attention import flash_attention
+    from deepspeed_amd.ops.attention import _FlashAttnFn
     q = torch.randn(B, S, Hq, D, device="cuda", dtype=torch.bfloat16,
                     requires_grad=True)
     k = torch.randn(B, S, Hk, D, device="cuda", dtype=torch.bfloat16,
                     requires_grad=True)
     v = torch.randn(B, S, Hk, D, device="cuda", dtype=torch.bfloat16,
                     requires_grad=True)
-    out = flash_attention(q, k, v, causal=True)
+    out = _FlashAttnFn.apply(q, k, v, True, 1.0 / math.sqrt(D))
     g = torch.randn_like(out)
     out.backward(g)
 
