@@ -23,7 +23,7 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--model", type=str, default="llama3-8b")
     p.add_argument("--seq-len", type=int, default=4096)
-    p.add_argument("--micro-batch", type=int, default=1)
+    p.add_argument("--micro-batch", type=int, default=4)
     p.add_argument("--grad-accum", type=int, default=1)
     p.add_argument("--zero-stage", type=int, default=3)
     p.add_argument("--act-ckpt", action="store_true",

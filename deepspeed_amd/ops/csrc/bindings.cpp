@@ -27,6 +27,8 @@ std::vector<at::Tensor> layernorm_fwd(at::Tensor x, at::Tensor w,
 std::vector<at::Tensor> layernorm_bwd(at::Tensor dy, at::Tensor x,
                                       at::Tensor w, at::Tensor mean,
                                       at::Tensor rstd);
+void rope(at::Tensor dst, at::Tensor src, at::Tensor cos, at::Tensor sin,
+          long pos0, bool backward);
 void rope_inplace(at::Tensor t, at::Tensor cos, at::Tensor sin, long pos0,
                   bool backward);
 at::Tensor swiglu_fwd(at::Tensor g, at::Tensor u);
@@ -59,6 +61,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_bwd", &rmsnorm_bwd);
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("rope", &rope);
   m.def("rope_inplace", &rope_inplace);
   m.def("swiglu_fwd", &swiglu_fwd);
   m.def("swiglu_bwd", &swiglu_bwd);

@@ -10,7 +10,8 @@
 
 // q/k layout: [B, S, H, D] bf16 contiguous; cos/sin: [S, D/2] fp32.
 // Llama "rotate_half" pairing: (d, d + D/2).
-__global__ void rope_kernel(short* __restrict__ t,
+__global__ void rope_kernel(short* __restrict__ dst,
+                            const short* __restrict__ src,
                             const float* __restrict__ cs,  // [S, D/2] cos
                             const float* __restrict__ sn,  // [S, D/2] sin
                             long long total_pairs, int S, int H, int D,
@@ -29,27 +30,33 @@ __global__ void rope_kernel(short* __restrict__ t,
     float c = cs[(long long)(s) * half + d];
     float sv = sn[(long long)(s) * half + d];
     if (backward) sv = -sv;
-    float x1 = bf2f(t[base + d]);
-    float x2 = bf2f(t[base + d + half]);
-    t[base + d] = f2bf(x1 * c - x2 * sv);
-    t[base + d + half] = f2bf(x2 * c + x1 * sv);
+    float x1 = bf2f(src[base + d]);
+    float x2 = bf2f(src[base + d + half]);
+    dst[base + d] = f2bf(x1 * c - x2 * sv);
+    dst[base + d + half] = f2bf(x2 * c + x1 * sv);
   }
 }
 
-void rope_inplace(at::Tensor t, at::Tensor cos, at::Tensor sin, long pos0,
-                  bool backward) {
-  TORCH_CHECK(t.scalar_type() == at::kBFloat16 && t.is_contiguous());
-  TORCH_CHECK(t.dim() == 4, "rope expects [B,S,H,D]");
-  int B = t.size(0), S = t.size(1), H = t.size(2), D = t.size(3);
+void rope(at::Tensor dst, at::Tensor src, at::Tensor cos, at::Tensor sin,
+          long pos0, bool backward) {
+  TORCH_CHECK(src.scalar_type() == at::kBFloat16 && src.is_contiguous());
+  TORCH_CHECK(src.dim() == 4, "rope expects [B,S,H,D]");
+  int B = src.size(0), S = src.size(1), H = src.size(2), D = src.size(3);
   long long total_pairs = (long long)B * S * H * (D / 2);
   auto stream = c10::hip::getCurrentHIPStream();
   int block = 256;
   int grid = grid_for(total_pairs, block);
   hipLaunchKernelGGL(rope_kernel, dim3(grid), dim3(block), 0, stream.stream(),
-                     reinterpret_cast<short*>(t.data_ptr()),
+                     reinterpret_cast<short*>(dst.data_ptr()),
+                     reinterpret_cast<const short*>(src.data_ptr()),
                      cos.data_ptr<float>(), sin.data_ptr<float>(), total_pairs,
                      S, H, D, (int)pos0, backward ? 1 : 0);
   HIP_CHECK_KERNEL();
+}
+
+void rope_inplace(at::Tensor t, at::Tensor cos, at::Tensor sin, long pos0,
+                  bool backward) {
+  rope(t, t, cos, sin, pos0, backward);
 }
 
 // -------- SwiGLU: y = silu(g) * u; packed gu = [..., 2*I] (g|u) ----------

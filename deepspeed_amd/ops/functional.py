@@ -111,8 +111,8 @@ class _RoPEFn(torch.autograd.Function):
         ctx.save_for_backward(cos, sin)
         if t.is_cuda:
             ext = get_ext(required=True)
-            out = t.contiguous().clone()
-            ext.rope_inplace(out, cos, sin, 0, False)
+            out = torch.empty_like(t)
+            ext.rope(out, t.contiguous(), cos, sin, 0, False)
         else:
             out = _rope_torch(t, cos, sin, sign=1.0)
         return out
@@ -122,8 +122,8 @@ class _RoPEFn(torch.autograd.Function):
         cos, sin = ctx.saved_tensors
         if dy.is_cuda:
             ext = get_ext(required=True)
-            dx = dy.contiguous().clone()
-            ext.rope_inplace(dx, cos, sin, 0, True)
+            dx = torch.empty_like(dy)
+            ext.rope(dx, dy.contiguous(), cos, sin, 0, True)
         else:
             dx = _rope_torch(dy, cos, sin, sign=-1.0)
         return dx, None, None

@@ -329,10 +329,13 @@ class ZeroStage3Optimizer:
         inputs = []
         for p in params:
             shard_numel = p.ds_tensor.numel()
-            padded = torch.empty(shard_numel * world, dtype=p.grad.dtype,
-                                 device=p.grad.device)
-            padded[:p.ds_numel].copy_(p.grad.reshape(-1))
-            if padded.numel() > p.ds_numel:
+            if shard_numel * world == p.ds_numel:
+                # aligned param: reduce-scatter the autograd grad in place
+                padded = p.grad.reshape(-1)
+            else:
+                padded = torch.empty(shard_numel * world, dtype=p.grad.dtype,
+                                     device=p.grad.device)
+                padded[:p.ds_numel].copy_(p.grad.reshape(-1))
                 padded[p.ds_numel:].zero_()
             inputs.append(padded)
             shards.append(torch.empty(shard_numel, dtype=p.grad.dtype,
