@@ -1,0 +1,59 @@
+"""End-to-end Llama-tiny training on 2 gloo ranks (mirrors bench.py path)."""
+import torch
+
+from tests.common import run_distributed
+
+
+def _train(stage):
+    import torch.distributed as tdist
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    from deepspeed_amd.models.llama import LLAMA_CONFIGS, LlamaForCausalLM
+    groups.reset_groups()
+    rank = tdist.get_rank()
+    cfg = LLAMA_CONFIGS["llama-tiny"]
+    torch.manual_seed(100 + rank)  # different init per rank: broadcast fixes
+    model = LlamaForCausalLM(cfg)
+    config = {
+        "train_micro_batch_size_per_gpu": 2,
+        "optimizer": {"type": "AdamW", "params": {"lr": 3e-4}},
+        "zero_optimization": {"stage": stage, "reduce_bucket_size": 100000,
+                              "sub_group_size": 300000},
+        "bf16": {"enabled": True},
+        "gradient_clipping": 1.0,
+    }
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config=config)
+    torch.manual_seed(7 + rank)
+    data = torch.randint(0, cfg.vocab_size, (2, 64))
+    losses = []
+    for _ in range(5):
+        loss = engine(data, labels=data)
+        engine.backward(loss)
+        engine.step()
+        losses.append(loss.item())
+    assert losses[-1] < losses[0], f"no progress: {losses}"
+    # params must agree across ranks (broadcast at init + synced updates)
+    if stage != 3:
+        sample = next(iter(engine.module.parameters())).detach().float()
+    else:
+        sample = next(iter(engine.module.parameters())).ds_tensor \
+            .detach().float()
+    s = sample.sum().item()
+    t = torch.tensor([s])
+    tdist.all_reduce(t, op=tdist.ReduceOp.MAX)
+    tmin = torch.tensor([s])
+    tdist.all_reduce(tmin, op=tdist.ReduceOp.MIN)
+    if stage != 3:  # stage-3 shards differ per rank by design
+        assert abs(t.item() - tmin.item()) < 1e-6
+    engine.destroy()
+    return losses
+
+
+def test_llama_zero2_2rank():
+    # per-rank losses differ (different data); progress + param sync asserted
+    # inside _train
+    run_distributed(_train, world_size=2, args=(2,))
+
+
+def test_llama_zero3_2rank():
+    run_distributed(_train, world_size=2, args=(3,))
