@@ -48,8 +48,19 @@ class DeepSpeedEngine(torch.nn.Module):
         self.training_data = training_data
         self.collate_fn = collate_fn
         self.mpu = mpu
+        # batch triple is defined over DATA-parallel ranks only
+        if mpu is not None and hasattr(mpu, "get_data_parallel_world_size"):
+            cfg_world = mpu.get_data_parallel_world_size()
+        else:
+            cfg_world = dist.get_world_size()
+            sp = 1
+            if isinstance(config, dict):
+                sp = config.get("sequence_parallel", {}).get(
+                    "sequence_parallel_size", 1)
+            if sp > 1:
+                cfg_world = max(1, cfg_world // sp)
         self._config = config_class or DeepSpeedConfig(
-            config, world_size=dist.get_world_size())
+            config, world_size=cfg_world)
 
         self.global_steps = 0
         self.global_samples = 0

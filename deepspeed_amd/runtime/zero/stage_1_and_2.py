@@ -197,6 +197,8 @@ class ZeroStage12Optimizer:
 
     # -- backward-time grad handling ---------------------------------------
 
+    _boundary_override = None
+
     def _attach_grad_views(self):
         """Point every p.grad at its slice of the bucket grad buffer so
         autograd accumulates in place across micro-steps."""
@@ -208,6 +210,10 @@ class ZeroStage12Optimizer:
                 if p.grad is None or p.grad.data_ptr() != b.grad_view(p).data_ptr():
                     p.grad = b.grad_view(p)
 
+    def _attach_grad_views_once(self):
+        # reset per-bucket flags only when not mid-boundary counting
+        self._attach_grad_views()
+
     def backward(self, loss, retain_graph=False):
         """Engine calls this. Scales loss (fp16) and runs autograd."""
         self.micro_step += 1
@@ -217,7 +223,17 @@ class ZeroStage12Optimizer:
             self.reduce_gradients()
 
     def is_gradient_accumulation_boundary(self):
+        if self._boundary_override is not None:
+            return self._boundary_override
         return self.micro_step % self.gradient_accumulation_steps == 0
+
+    def set_accumulation_boundary(self, is_boundary):
+        """Explicit boundary control (pipeline engine drives this)."""
+        self._boundary_override = is_boundary
+
+    def ensure_grad_views(self):
+        """Attach p.grad views without running loss scaling (PP path)."""
+        self._attach_grad_views_once()
 
     def _on_grad_ready(self, p, bucket):
         if not self.is_gradient_accumulation_boundary():
