@@ -1,0 +1,163 @@
+"""Universal checkpoint: topology-independent per-parameter format.
+
+Parity: reference `deepspeed/checkpoint/ds_to_universal.py`
+(`extract_zero_shards:124`, stage-3 variant @164) and
+`universal_checkpoint.py:149` (load_hp_checkpoint_state).
+
+Format: <out>/zero/<param_name>/{fp32,exp_avg,exp_avg_sq}.pt + meta.pt —
+convert once offline, then resume into ANY dp world size (elastic resume).
+"""
+import os
+from collections import OrderedDict
+
+import torch
+
+from ..utils.zero_to_fp32 import _load_zero_shards, _read_tag
+
+_STATE_KEYS = ("exp_avg", "exp_avg_sq")
+
+
+def _iter_params_with_states(shards):
+    """Yield (name, {'fp32': t, 'exp_avg': t, 'exp_avg_sq': t})."""
+    layout = shards[0]["shard_layout"]
+    base_states = [sd["base_optimizer_state"] for sd in shards]
+
+    def master_state(rank, master_idx, key):
+        st = base_states[rank]["state"].get(master_idx, {})
+        return st.get(key)
+
+    if layout["kind"] == "bucket":
+        flats_per_rank = [sd["single_partition_of_fp32_groups"]
+                          for sd in shards]
+        for bi, binfo in enumerate(layout["buckets"]):
+            full = {"fp32": torch.cat([f[bi].detach().float()
+                                       for f in flats_per_rank])}
+            for key in _STATE_KEYS:
+                parts = [master_state(r, bi, key)
+                         for r in range(len(shards))]
+                if all(p is not None for p in parts):
+                    full[key] = torch.cat([p.float() for p in parts])
+            for name, off, numel, shape in binfo["params"]:
+                yield name, {k: v[off:off + numel].reshape(shape).clone()
+                             for k, v in full.items()}
+    else:  # subgroup / stage 3
+        flats_per_rank = [sd["fp32_flat_groups"] for sd in shards]
+        for gi, ginfo in enumerate(layout["subgroups"]):
+            states = {"fp32": flats_per_rank}
+            extra = {}
+            for key in _STATE_KEYS:
+                parts = [master_state(r, gi, key)
+                         for r in range(len(shards))]
+                if all(p is not None for p in parts):
+                    extra[key] = parts
+            for name, off, shard_numel, full_numel, shape in ginfo["params"]:
+                out = {}
+                out["fp32"] = torch.cat(
+                    [f[gi].detach().float()[off:off + shard_numel]
+                     for f in flats_per_rank])[:full_numel] \
+                    .reshape(shape).clone()
+                for key, parts in extra.items():
+                    out[key] = torch.cat(
+                        [p.float()[off:off + shard_numel]
+                         for p in parts])[:full_numel].reshape(shape).clone()
+                yield name, out
+
+
+def ds_to_universal(checkpoint_dir, output_dir, tag=None):
+    """Convert a (dp-sharded) checkpoint into universal format."""
+    dirpath = _read_tag(checkpoint_dir, tag)
+    shards = _load_zero_shards(dirpath)
+    zero_dir = os.path.join(output_dir, "zero")
+    os.makedirs(zero_dir, exist_ok=True)
+    names = []
+    for name, tensors in _iter_params_with_states(shards):
+        pdir = os.path.join(zero_dir, name)
+        os.makedirs(pdir, exist_ok=True)
+        for key, t in tensors.items():
+            torch.save(t, os.path.join(pdir, f"{key}.pt"))
+        names.append(name)
+    # meta: step counts per group
+    steps = [g.get("step", 0)
+             for g in shards[0]["base_optimizer_state"]["param_groups"]]
+    torch.save({"param_names": names, "group_steps": steps,
+                "source_world": shards[0].get("partition_count", 1)},
+               os.path.join(output_dir, "meta.pt"))
+    return names
+
+
+def _load_param(universal_dir, name):
+    pdir = os.path.join(universal_dir, "zero", name)
+    out = {}
+    for key in ("fp32",) + _STATE_KEYS:
+        f = os.path.join(pdir, f"{key}.pt")
+        if os.path.exists(f):
+            out[key] = torch.load(f, map_location="cpu", weights_only=False)
+    return out
+
+
+def load_universal_into_optimizer(optimizer, universal_dir):
+    """Scatter a universal checkpoint into the CURRENT topology.
+
+    Works for ZeroStage12Optimizer (buckets) and ZeroStage3Optimizer
+    (subgroups) at any dp world size.
+    """
+    meta = torch.load(os.path.join(universal_dir, "meta.pt"),
+                      map_location="cpu", weights_only=False)
+    layout = optimizer.shard_layout()
+    world = layout["world"]
+    base = optimizer.optimizer
+
+    def set_state(master, key, flat_shard):
+        st = base.state.setdefault(master, {})
+        st[key] = flat_shard.to(master.device)
+
+    if layout["kind"] == "bucket":
+        rank = optimizer.rank
+        for bi, (b, binfo) in enumerate(zip(optimizer.buckets,
+                                            layout["buckets"])):
+            full = {"fp32": torch.zeros(binfo["numel_padded"])}
+            for key in _STATE_KEYS:
+                full[key] = torch.zeros(binfo["numel_padded"])
+            for name, off, numel, shape in binfo["params"]:
+                t = _load_param(universal_dir, name)
+                for key in full:
+                    if key in t:
+                        full[key][off:off + numel] = t[key].reshape(-1)
+            lo = rank * binfo["shard_numel"]
+            hi = lo + binfo["shard_numel"]
+            b.master32.data.copy_(full["fp32"][lo:hi])
+            master = b.master32
+            for key in _STATE_KEYS:
+                set_state(master, key, full[key][lo:hi])
+            b.shard16.copy_(b.master32.detach().to(b.shard16.dtype))
+        from .. import comm as dist
+        for b in optimizer.buckets:
+            dist.all_gather_into_tensor(b.flat16, b.shard16,
+                                        group=optimizer.dp_group)
+    else:  # stage 3 subgroups
+        rank = optimizer.rank
+        for gi, (sg, ginfo) in enumerate(zip(optimizer.sub_groups,
+                                             layout["subgroups"])):
+            fp32 = torch.zeros(sg.numel)
+            states = {key: torch.zeros(sg.numel) for key in _STATE_KEYS}
+            for name, off, shard_numel, full_numel, shape in ginfo["params"]:
+                t = _load_param(universal_dir, name)
+                flat = t["fp32"].reshape(-1)
+                lo = rank * shard_numel
+                hi = min(lo + shard_numel, full_numel)
+                if hi > lo:
+                    fp32[off:off + hi - lo] = flat[lo:hi]
+                for key in _STATE_KEYS:
+                    if key in t:
+                        states[key][off:off + hi - lo] = \
+                            t[key].reshape(-1)[lo:hi]
+            sg.master32.data.copy_(fp32.to(sg.master32.device))
+            for key in _STATE_KEYS:
+                set_state(sg.master32, key, states[key])
+            sg.copy_master_to_shards()
+        optimizer._refresh_persistent_params()
+
+    # restore per-group step counters
+    for g, step in zip(base.param_groups, meta.get("group_steps", [])):
+        if step:
+            g["step"] = step

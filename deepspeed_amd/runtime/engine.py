@@ -445,6 +445,11 @@ class DeepSpeedEngine(torch.nn.Module):
                      load_lr_scheduler_states=load_lr_scheduler_states,
                      load_module_only=load_module_only)
 
+    def load_universal_checkpoint(self, universal_dir):
+        """Resume from a universal checkpoint at ANY dp world size."""
+        from ..checkpoint.universal import load_universal_into_optimizer
+        load_universal_into_optimizer(self.optimizer, universal_dir)
+
     def module_state_dict(self, exclude_frozen_parameters=False):
         return self.module.state_dict()
 

@@ -477,6 +477,20 @@ class ZeroStage3Optimizer:
     def param_groups(self):
         return self.optimizer.param_groups
 
+    def shard_layout(self):
+        names = {id(p): n for n, p in self.module.named_parameters()}
+        subgroups = []
+        for i, sg in enumerate(self.sub_groups):
+            subgroups.append({
+                "group_idx": sg.group_idx,
+                "numel": sg.numel,
+                "params": [(names.get(id(p), f"param_{p.ds_id}"),
+                            sg.offsets[p], p.ds_tensor.numel(), p.ds_numel,
+                            list(p.ds_shape)) for p in sg.params],
+            })
+        return {"stage": 3, "world": self.world, "kind": "subgroup",
+                "subgroups": subgroups}
+
     def state_dict(self):
         return {
             "loss_scaler": self.loss_scaler,
@@ -485,6 +499,7 @@ class ZeroStage3Optimizer:
             "param_shapes": self._param_shapes(),
             "zero_stage": 3,
             "partition_count": self.world,
+            "shard_layout": self.shard_layout(),
         }
 
     def _param_shapes(self):

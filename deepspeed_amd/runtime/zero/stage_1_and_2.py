@@ -115,6 +115,7 @@ class ZeroStage12Optimizer:
                  ignore_unused_parameters=True,
                  mpu=None):
         self.optimizer = init_optimizer
+        self.engine = engine
         self.stage = stage
         self.dp_group = dp_process_group
         self.world = dist.get_world_size(self.dp_group)
@@ -369,6 +370,30 @@ class ZeroStage12Optimizer:
     def param_groups(self):
         return self.optimizer.param_groups
 
+    def _param_names(self):
+        names = {}
+        if self.engine is not None and getattr(self.engine, "module", None) \
+                is not None:
+            names = {id(p): n
+                     for n, p in self.engine.module.named_parameters()}
+        return names
+
+    def shard_layout(self):
+        """Reassembly metadata for zero_to_fp32/universal checkpoint."""
+        names = self._param_names()
+        buckets = []
+        for i, b in enumerate(self.buckets):
+            buckets.append({
+                "group_idx": b.group_idx,
+                "numel_padded": b.numel_padded,
+                "shard_numel": b.shard_numel,
+                "params": [(names.get(id(p), f"param_{i}_{j}"),
+                            b.offsets[p], p.numel(), list(p.shape))
+                           for j, p in enumerate(b.params)],
+            })
+        return {"stage": self.stage, "world": self.world, "kind": "bucket",
+                "buckets": buckets}
+
     def state_dict(self):
         sd = {}
         sd["loss_scaler"] = self.loss_scaler
@@ -376,6 +401,7 @@ class ZeroStage12Optimizer:
         sd["single_partition_of_fp32_groups"] = [b.master32 for b in self.buckets]
         sd["zero_stage"] = self.stage
         sd["partition_count"] = self.world
+        sd["shard_layout"] = self.shard_layout()
         return sd
 
     def load_state_dict(self, sd, load_optimizer_states=True):

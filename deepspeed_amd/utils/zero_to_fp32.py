@@ -1,0 +1,97 @@
+"""Reconstruct a full fp32 state_dict from ZeRO checkpoint shards.
+
+Parity: reference `deepspeed/utils/zero_to_fp32.py`
+(`get_fp32_state_dict_from_zero_checkpoint:563`,
+`convert_zero_checkpoint_to_fp32_state_dict:628`). Works offline on a
+checkpoint directory written by deepspeed_amd (stages 1/2/3).
+"""
+import argparse
+import glob
+import os
+from collections import OrderedDict
+
+import torch
+
+
+def _read_tag(ckpt_dir, tag):
+    if tag is None:
+        latest = os.path.join(ckpt_dir, "latest")
+        if os.path.exists(latest):
+            with open(latest) as f:
+                tag = f.read().strip()
+        else:
+            raise FileNotFoundError(f"no 'latest' file in {ckpt_dir}")
+    return os.path.join(ckpt_dir, str(tag))
+
+
+def _load_zero_shards(dirpath):
+    files = sorted(glob.glob(os.path.join(
+        dirpath, "zero_pp_rank_*_mp_rank_*_optim_states.pt")),
+        key=lambda f: int(os.path.basename(f).split("_")[3]))
+    if not files:
+        raise FileNotFoundError(f"no zero shard files in {dirpath}")
+    return [torch.load(f, map_location="cpu", weights_only=False)
+            ["optimizer_state_dict"] for f in files]
+
+
+def _reassemble(shards, key="flat"):
+    """Yield (name, fp32_tensor) from per-rank optimizer state dicts."""
+    layout = shards[0]["shard_layout"]
+    if layout["kind"] == "bucket":  # stages 1/2
+        flats_per_rank = [sd["single_partition_of_fp32_groups"]
+                          for sd in shards]
+        for bi, binfo in enumerate(layout["buckets"]):
+            full = torch.cat([flats[bi].detach().float()
+                              for flats in flats_per_rank])
+            for name, off, numel, shape in binfo["params"]:
+                yield name, full[off:off + numel].reshape(shape).clone()
+    elif layout["kind"] == "subgroup":  # stage 3
+        flats_per_rank = [sd["fp32_flat_groups"] for sd in shards]
+        for gi, ginfo in enumerate(layout["subgroups"]):
+            for name, off, shard_numel, full_numel, shape in ginfo["params"]:
+                pieces = [flats[gi].detach().float()[off:off + shard_numel]
+                          for flats in flats_per_rank]
+                yield name, torch.cat(pieces)[:full_numel] \
+                    .reshape(shape).clone()
+    else:
+        raise ValueError(f"unknown layout kind {layout['kind']}")
+
+
+def get_fp32_state_dict_from_zero_checkpoint(checkpoint_dir, tag=None):
+    dirpath = _read_tag(checkpoint_dir, tag)
+    shards = _load_zero_shards(dirpath)
+    state_dict = OrderedDict()
+    for name, tensor in _reassemble(shards):
+        state_dict[name] = tensor
+    # merge non-sharded buffers from the model states file
+    model_files = glob.glob(os.path.join(dirpath,
+                                         "mp_rank_*_model_states.pt"))
+    if model_files:
+        ms = torch.load(model_files[0], map_location="cpu",
+                        weights_only=False)
+        for bname in ms.get("buffer_names", []):
+            if bname in ms["module"]:
+                state_dict[bname] = ms["module"][bname]
+    return state_dict
+
+
+def convert_zero_checkpoint_to_fp32_state_dict(checkpoint_dir,
+                                               output_file, tag=None):
+    sd = get_fp32_state_dict_from_zero_checkpoint(checkpoint_dir, tag)
+    torch.save(sd, output_file)
+    print(f"saved fp32 state_dict ({len(sd)} entries) to {output_file}")
+    return sd
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("checkpoint_dir")
+    p.add_argument("output_file")
+    p.add_argument("-t", "--tag", default=None)
+    args = p.parse_args()
+    convert_zero_checkpoint_to_fp32_state_dict(args.checkpoint_dir,
+                                               args.output_file, args.tag)
+
+
+if __name__ == "__main__":
+    main()
