@@ -14,6 +14,10 @@ at::Tensor l2norm_sq(std::vector<at::Tensor> tensors);
 std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
                                        at::Tensor v, bool causal,
                                        double scale);
+void cpu_adam_step(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
+                   c10::optional<at::Tensor> out16, double lr, double beta1,
+                   double beta2, double eps, long step, long adamw,
+                   long bias_correction, double wd, double grad_scale);
 void multi_tensor_lion(std::vector<at::Tensor> params,
                        std::vector<at::Tensor> grads,
                        std::vector<at::Tensor> exp_avgs, double lr,
@@ -53,6 +57,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("l2norm_sq", &l2norm_sq);
   m.def("flash_attn_fwd", &flash_attn_fwd, py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("causal") = true, py::arg("scale") = 0.0);
+  m.def("cpu_adam_step", &cpu_adam_step, py::arg("p"), py::arg("g"),
+        py::arg("m"), py::arg("v"), py::arg("out16") = c10::nullopt,
+        py::arg("lr") = 1e-3, py::arg("beta1") = 0.9, py::arg("beta2") = 0.999,
+        py::arg("eps") = 1e-8, py::arg("step") = 1, py::arg("adamw") = 1,
+        py::arg("bias_correction") = 1, py::arg("wd") = 0.0,
+        py::arg("grad_scale") = 1.0);
   m.def("multi_tensor_lion", &multi_tensor_lion, "fused Lion (gfx950)",
         py::arg("params"), py::arg("grads"), py::arg("exp_avgs"),
         py::arg("lr"), py::arg("beta1"), py::arg("beta2"),

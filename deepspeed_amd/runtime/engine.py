@@ -189,6 +189,11 @@ class DeepSpeedEngine(torch.nn.Module):
         return MonitorMaster(self._config)
 
     # -------------------------------------------------------------- optimizer
+    def _offload_optimizer_enabled(self):
+        zc = self._config.zero_config
+        return (zc.stage >= 1 and zc.offload_optimizer is not None
+                and zc.offload_optimizer.device == "cpu")
+
     def _configure_basic_optimizer(self, model_parameters):
         cfg = self._config.optimizer
         if cfg is None:
@@ -197,6 +202,10 @@ class DeepSpeedEngine(torch.nn.Module):
         params = dict(cfg.params)
         params.pop("torch_adam", None)
         if name in ("adam", "adamw", "fusedadam"):
+            if self._offload_optimizer_enabled():
+                from ..ops.cpu_adam import DeepSpeedCPUAdam
+                params.setdefault("adamw_mode", name != "adam")
+                return DeepSpeedCPUAdam(model_parameters, **params)
             params.setdefault("adam_w_mode", name != "adam")
             return FusedAdam(model_parameters, **params)
         if name == "cpuadam" or name == "deepspeedcpuadam":

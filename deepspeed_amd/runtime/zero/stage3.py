@@ -40,16 +40,22 @@ class SubGroup:
     same pass as the fp32 update (no separate cast+copy over 16 GB)."""
 
     __slots__ = ("params", "offsets", "master32", "grad32", "flat16",
-                 "group_idx", "numel")
+                 "flat16_cpu", "grad_stage", "group_idx", "numel", "offload")
 
-    def __init__(self, params, offsets, numel, group_idx, device):
+    def __init__(self, params, offsets, numel, group_idx, device,
+                 offload=False):
         self.params = params
         self.offsets = offsets
         self.numel = numel
         self.group_idx = group_idx
+        self.offload = offload
         dtype16 = params[0].ds_tensor.dtype if params else torch.bfloat16
+        pin = offload and torch.cuda.is_available()
+        host = torch.device("cpu")
+        state_dev = host if offload else device
         self.flat16 = torch.empty(numel, dtype=dtype16, device=device)
-        self.master32 = torch.empty(numel, dtype=torch.float32, device=device)
+        self.master32 = torch.empty(numel, dtype=torch.float32,
+                                    device=state_dev, pin_memory=pin)
         for p in params:
             off = offsets[p]
             n = p.ds_tensor.numel()
@@ -57,14 +63,50 @@ class SubGroup:
             p.ds_tensor = self.flat16[off:off + n]
             self.master32[off:off + n].copy_(p.ds_tensor.float())
         self.master32 = self.master32.detach().requires_grad_(True)
-        self.grad32 = torch.zeros(numel, dtype=torch.float32, device=device)
+        self.grad32 = torch.zeros(numel, dtype=torch.float32,
+                                  device=state_dev, pin_memory=pin)
+        if offload:
+            # pinned staging: bf16 shard out (H2D) + grad shard in (D2H)
+            self.flat16_cpu = torch.empty(numel, dtype=dtype16, device=host,
+                                          pin_memory=pin)
+            self.grad_stage = torch.empty(numel, dtype=dtype16, device=host,
+                                          pin_memory=pin)
+        else:
+            self.flat16_cpu = None
+            self.grad_stage = None
 
     def grad_shard_view(self, p):
         off = self.offsets[p]
         return self.grad32[off:off + p.ds_tensor.numel()]
 
+    def accumulate_grad(self, p, shard16_gpu):
+        """shard16_gpu: this rank's reduced gradient shard (device)."""
+        dst = self.grad_shard_view(p)[:shard16_gpu.numel()]
+        if self.offload:
+            off = self.offsets[p]
+            stage = self.grad_stage[off:off + shard16_gpu.numel()]
+            stage.copy_(shard16_gpu, non_blocking=True)
+            if shard16_gpu.is_cuda:
+                torch.cuda.synchronize()
+            dst.add_(stage.float())
+        elif shard16_gpu.is_cuda and shard16_gpu.dtype == torch.bfloat16:
+            from ...ops.loader import get_ext
+            get_ext(required=True).accum_bf16_to_f32(dst, shard16_gpu, 1.0)
+        else:
+            dst.add_(shard16_gpu.float())
+
     def copy_master_to_shards(self):
-        self.flat16.copy_(self.master32.detach())
+        if self.offload:
+            self.flat16_cpu.copy_(self.master32.detach().to(torch.bfloat16)
+                                  if self.flat16_cpu.dtype == torch.bfloat16
+                                  else self.master32.detach())
+            self.flat16.copy_(self.flat16_cpu, non_blocking=True)
+        else:
+            self.flat16.copy_(self.master32.detach())
+
+    def publish_flat16(self):
+        """After a fused CPU-Adam step wrote flat16_cpu, push H2D."""
+        self.flat16.copy_(self.flat16_cpu, non_blocking=True)
 
 
 class ZeroStage3Optimizer:
@@ -168,14 +210,16 @@ class ZeroStage3Optimizer:
             for p in gparams:
                 if numel >= self.sub_group_size and cur:
                     self.sub_groups.append(
-                        SubGroup(cur, offsets, numel, gi, self.device))
+                        SubGroup(cur, offsets, numel, gi, self.device,
+                                 offload=self.offload_optimizer))
                     cur, offsets, numel = [], OrderedDict(), 0
                 offsets[p] = numel
                 cur.append(p)
                 numel += p.ds_tensor.numel()
             if cur:
                 self.sub_groups.append(
-                    SubGroup(cur, offsets, numel, gi, self.device))
+                    SubGroup(cur, offsets, numel, gi, self.device,
+                             offload=self.offload_optimizer))
         self.param_to_subgroup = {}
         for sg in self.sub_groups:
             for p in sg.params:
@@ -316,13 +360,7 @@ class ZeroStage3Optimizer:
         if world == 1:
             for p in params:
                 sg = self.param_to_subgroup[p]
-                dst = sg.grad_shard_view(p)[:p.ds_numel]
-                src = p.grad.reshape(-1)
-                if p.grad.is_cuda and src.dtype == torch.bfloat16:
-                    from ...ops.loader import get_ext
-                    get_ext(required=True).accum_bf16_to_f32(dst, src, 1.0)
-                else:
-                    dst.add_(src.float())
+                sg.accumulate_grad(p, p.grad.reshape(-1))
                 p.grad = None
             return
         from .stage_1_and_2 import _avg_op
@@ -361,12 +399,7 @@ class ZeroStage3Optimizer:
                                            group=self.dp_group)
         for p, shard in zip(params, shards):
             sg = self.param_to_subgroup[p]
-            dst = sg.grad_shard_view(p)
-            if shard.is_cuda and shard.dtype == torch.bfloat16:
-                from ...ops.loader import get_ext
-                get_ext(required=True).accum_bf16_to_f32(dst, shard, 1.0)
-            else:
-                dst.add_(shard.float())
+            sg.accumulate_grad(p, shard)
 
     # -------------------------------------------------------------- train
     def backward(self, loss, retain_graph=False):
@@ -401,11 +434,11 @@ class ZeroStage3Optimizer:
                 from ...ops.loader import get_ext
                 total_sq = get_ext(required=True).l2norm_sq(grads).double()
             else:
-                total_sq = torch.zeros(1, dtype=torch.float64,
-                                       device=self.device)
+                total_sq = torch.zeros(1, dtype=torch.float64)
                 for g in grads:
-                    total_sq += g.double().pow(2).sum()
+                    total_sq += float(torch.linalg.vector_norm(g))**2
             if dist.is_initialized():
+                total_sq = total_sq.to(self.device)
                 dist.all_reduce(total_sq, group=self.dp_group)
             norm = total_sq.sqrt().item() / scale
             self._global_grad_norm = norm
@@ -433,7 +466,8 @@ class ZeroStage3Optimizer:
         if fused:
             self.optimizer.set_grad_scale(1.0 / combined)
             self.optimizer.set_fused_out16(
-                {sg.master32: sg.flat16 for sg in self.sub_groups})
+                {sg.master32: (sg.flat16_cpu if sg.offload else sg.flat16)
+                 for sg in self.sub_groups})
         elif combined != 1.0:
             for sg in self.sub_groups:
                 sg.grad32.mul_(1.0 / combined)
@@ -445,6 +479,8 @@ class ZeroStage3Optimizer:
             sg.master32.grad = None
             if not fused:
                 sg.copy_master_to_shards()
+            elif sg.offload:
+                sg.publish_flat16()
         if fused:
             self.optimizer.set_grad_scale(1.0)
             self.optimizer.set_fused_out16({})
