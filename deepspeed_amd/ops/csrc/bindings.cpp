@@ -44,7 +44,10 @@ at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets,
                              at::Tensor lse, at::Tensor dloss,
                              long ignore_index);
 
+void bind_aio(py::module_& m);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  bind_aio(m);
   m.def("multi_tensor_adam", &multi_tensor_adam, "fused AdamW (gfx950)",
         py::arg("params"), py::arg("grads"), py::arg("exp_avgs"),
         py::arg("exp_avg_sqs"), py::arg("lr"), py::arg("beta1"),

@@ -145,8 +145,17 @@ class ZeroStage3Optimizer:
 
         self.device = (torch.device("cuda", torch.cuda.current_device())
                        if torch.cuda.is_available() else torch.device("cpu"))
-        self.offload_optimizer = (offload_optimizer is not None
-                                  and getattr(offload_optimizer, "device", "none") == "cpu")
+        self.offload_device = getattr(offload_optimizer, "device", "none") \
+            if offload_optimizer is not None else "none"
+        self.offload_optimizer = self.offload_device in ("cpu", "nvme")
+        self.nvme_swapper = None
+        if self.offload_device == "nvme":
+            from ..swap_tensor.optimizer_swapper import OptimizerStateSwapper
+            import os as _os
+            nvme_path = getattr(offload_optimizer, "nvme_path", None) \
+                or "/tmp/dsamd_nvme_swap"
+            nvme_path = _os.path.join(nvme_path, f"rank{self.rank}")
+            self.nvme_swapper = OptimizerStateSwapper(nvme_path)
 
         self.loss_scaler = CreateLossScaler(dtype, static_loss_scale,
                                             dynamic_loss_scale,
@@ -155,6 +164,8 @@ class ZeroStage3Optimizer:
 
         self._shard_module_params()
         self._build_sub_groups()
+        if self.nvme_swapper is not None:
+            self._init_nvme_state()
         self._gather_persistent_params()
         self._install_module_hooks()
         self._install_grad_hooks()
@@ -227,6 +238,53 @@ class ZeroStage3Optimizer:
         for gi, g in enumerate(self.optimizer.param_groups):
             g["params"] = [sg.master32 for sg in self.sub_groups
                            if sg.group_idx == gi]
+
+    def _init_nvme_state(self):
+        """Write initial fp32 masters to NVMe; drop resident copies."""
+        for i, sg in enumerate(self.sub_groups):
+            buf = self.nvme_swapper._buffer(i, "master", sg.numel)
+            buf.copy_(sg.master32.detach().cpu())
+            # materialize zero state files too
+            self.nvme_swapper._buffer(i, "exp_avg", sg.numel)
+            self.nvme_swapper._buffer(i, "exp_avg_sq", sg.numel)
+            self.nvme_swapper.swap_out(i)
+            self.nvme_swapper.release_buffers(i)
+            sg.master32 = None  # lives on NVMe now
+        for g in self.optimizer.param_groups:
+            g["params"] = []
+
+    def _nvme_step(self, combined_scale):
+        """Per-sub-group: swap in -> host AdamW -> swap out -> publish."""
+        from ...ops.loader import get_ext
+        ext = get_ext(required=False)
+        for g in self.optimizer.param_groups:
+            g["step"] = g.get("step", 0) + 1
+        for i, sg in enumerate(self.sub_groups):
+            group = self.optimizer.param_groups[sg.group_idx]
+            beta1, beta2 = group.get("betas", (0.9, 0.999))
+            master, ea, eas = self.nvme_swapper.swap_in(i, sg.numel)
+            grad = sg.grad32 if not sg.grad32.is_cuda else sg.grad32.cpu()
+            if ext is not None:
+                ext.cpu_adam_step(master, grad, ea, eas,
+                                  sg.flat16_cpu, group["lr"], beta1, beta2,
+                                  group.get("eps", 1e-8), group["step"], 1, 1,
+                                  group.get("weight_decay", 0.0),
+                                  1.0 / combined_scale)
+            else:
+                g32 = grad * (1.0 / combined_scale)
+                step = group["step"]
+                bc1 = 1 - beta1**step
+                bc2 = 1 - beta2**step
+                master.mul_(1.0 - group["lr"] *
+                            group.get("weight_decay", 0.0))
+                ea.mul_(beta1).add_(g32, alpha=1 - beta1)
+                eas.mul_(beta2).addcmul_(g32, g32, value=1 - beta2)
+                denom = (eas / bc2).sqrt().add_(group.get("eps", 1e-8))
+                master.addcdiv_(ea, denom, value=-group["lr"] / bc1)
+                sg.flat16_cpu.copy_(master.to(sg.flat16_cpu.dtype))
+            self.nvme_swapper.swap_out(i)
+            self.nvme_swapper.release_buffers(i)
+            sg.publish_flat16()
 
     def _gather_persistent_params(self):
         persist = [p for p in self._all_params if p.ds_persist]
@@ -461,6 +519,12 @@ class ZeroStage3Optimizer:
                 return
 
         combined = self._combined_scale()
+
+        if self.nvme_swapper is not None:
+            self._nvme_step(combined)
+            self._clear_grads()
+            self._refresh_persistent_params()
+            return
 
         fused = hasattr(self.optimizer, "set_grad_scale")
         if fused:
