@@ -172,14 +172,29 @@ class DeepSpeedEngine(torch.nn.Module):
         self.module.train()
 
     def _broadcast_model(self):
-        """Sync initial weights from DP rank 0 (ref engine.py:1715)."""
+        """Sync initial weights from DP rank 0 (ref engine.py:1715).
+
+        Expert (MoE) params broadcast over their expert-DP group — each EP
+        rank owns distinct experts.
+        """
         if self.dp_world_size <= 1:
             return
         src_rank = dist.get_global_rank(self.dp_group, 0) \
             if self.dp_group is not None else 0
         for p in self.module.parameters():
-            if torch.is_tensor(p):
-                dist.broadcast(p.data, src_rank, group=self.dp_group)
+            if not torch.is_tensor(p):
+                continue
+            gn = getattr(p, "group_name", None)
+            if gn is not None:
+                try:
+                    epg = groups.get_expert_data_parallel_group(gn)
+                except KeyError:
+                    continue
+                if dist.get_world_size(epg) > 1:
+                    dist.broadcast(p.data, dist.get_global_rank(epg, 0),
+                                   group=epg)
+                continue
+            dist.broadcast(p.data, src_rank, group=self.dp_group)
         for b in self.module.buffers():
             if torch.is_tensor(b) and b.numel() > 0:
                 dist.broadcast(b.data, src_rank, group=self.dp_group)

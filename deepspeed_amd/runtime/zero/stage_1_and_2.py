@@ -57,12 +57,16 @@ class Bucket:
 
     __slots__ = ("params", "flat16", "grad16", "shard16", "master32",
                  "grad32", "numel_padded", "shard_numel", "offsets",
-                 "group_idx", "ready_count", "reduced", "work")
+                 "group_idx", "ready_count", "reduced", "work", "pg",
+                 "world")
 
-    def __init__(self, params, offsets, group_idx, world, rank, device, dtype):
+    def __init__(self, params, offsets, group_idx, world, rank, device,
+                 dtype, pg=None):
         self.params = params
         self.offsets = offsets  # param -> start offset in flat
         self.group_idx = group_idx
+        self.pg = pg  # process group (None = default DP); expert-DP for MoE
+        self.world = world
         total = offsets[params[-1]] + params[-1].numel()
         self.numel_padded = _pad_to(total, world * ALIGN)
         self.shard_numel = self.numel_padded // world
@@ -137,13 +141,29 @@ class ZeroStage12Optimizer:
         self.overflow = False
         self._overflow_checker = CheckOverflow()
 
-        # Build buckets per param group
+        # Build buckets per param group; expert params (MoE, tagged with
+        # .group_name by Experts) get their own buckets partitioned/reduced
+        # over the expert-data-parallel group.
         self.buckets = []
         self.param_to_bucket = {}
         self._grad_acc_hooks = []
         for gi, group in enumerate(self.optimizer.param_groups):
             trainable = [p for p in group["params"] if p.requires_grad]
-            self._build_buckets(trainable, gi)
+            dense = [p for p in trainable
+                     if getattr(p, "group_name", None) is None]
+            self._build_buckets(dense, gi, pg=None)
+            expert_groups = {}
+            for p in trainable:
+                gn = getattr(p, "group_name", None)
+                if gn is not None:
+                    expert_groups.setdefault(gn, []).append(p)
+            for gn, eparams in expert_groups.items():
+                from ...comm import groups as grp
+                try:
+                    epg = grp.get_expert_data_parallel_group(gn)
+                except KeyError:
+                    epg = self.dp_group  # ep groups not built (ep_size==1)
+                self._build_buckets(eparams, gi, pg=epg)
             # swap group params for the fp32 masters of this group's buckets
             group["params"] = [b.master32 for b in self.buckets
                                if b.group_idx == gi]
@@ -162,24 +182,26 @@ class ZeroStage12Optimizer:
 
     # -- setup --------------------------------------------------------------
 
-    def _build_buckets(self, params, group_idx):
+    def _build_buckets(self, params, group_idx, pg=None):
         cur, offsets, cur_numel = [], OrderedDict(), 0
         for p in params:
             if p.dtype != self.dtype:
                 raise ValueError(
                     f"param dtype {p.dtype} != engine dtype {self.dtype}")
             if cur_numel >= self.reduce_bucket_size and cur:
-                self._make_bucket(cur, offsets, group_idx)
+                self._make_bucket(cur, offsets, group_idx, pg)
                 cur, offsets, cur_numel = [], OrderedDict(), 0
             offsets[p] = cur_numel
             cur.append(p)
             cur_numel += _pad_to(p.numel(), ALIGN)
         if cur:
-            self._make_bucket(cur, offsets, group_idx)
+            self._make_bucket(cur, offsets, group_idx, pg)
 
-    def _make_bucket(self, params, offsets, group_idx):
-        b = Bucket(params, offsets, group_idx, self.world, self.rank,
-                   self.device, self.dtype)
+    def _make_bucket(self, params, offsets, group_idx, pg=None):
+        world = dist.get_world_size(pg) if pg is not None else self.world
+        rank = dist.get_rank(pg) if pg is not None else self.rank
+        b = Bucket(params, offsets, group_idx, world, rank,
+                   self.device, self.dtype, pg=pg)
         self.buckets.append(b)
         for p in params:
             self.param_to_bucket[p] = b
@@ -261,9 +283,10 @@ class ZeroStage12Optimizer:
         with ctx:
             shard = torch.empty(b.shard_numel, dtype=self.dtype,
                                 device=b.grad16.device)
-            op = _avg_op(self.world, b.grad16)
+            pg = b.pg if b.pg is not None else self.dp_group
+            op = _avg_op(b.world, b.grad16)
             dist.reduce_scatter_tensor(shard, b.grad16, op=op,
-                                       group=self.dp_group)
+                                       group=pg)
             if shard.is_cuda and shard.dtype == torch.bfloat16:
                 from ...ops.loader import get_ext
                 get_ext(required=True).accum_bf16_to_f32(b.grad32, shard, 1.0)
@@ -342,8 +365,9 @@ class ZeroStage12Optimizer:
 
         # all-gather updated 16-bit params, one call per bucket
         for b in self.buckets:
-            dist.all_gather_into_tensor(b.flat16, b.shard16,
-                                        group=self.dp_group)
+            dist.all_gather_into_tensor(
+                b.flat16, b.shard16,
+                group=b.pg if b.pg is not None else self.dp_group)
         self._clear_grads()
 
     def _clear_grads(self):
