@@ -74,7 +74,7 @@ class LlamaAttention(nn.Module):
         self.sp_group = None  # set by enable_ulysses()
         self._dist_attn = None
 
-    def forward(self, x, cos, sin):
+    def forward(self, x, cos, sin, kv_cache=None):
         B, S, _ = x.shape
         d = self.cfg.head_dim
         q = self.q_proj(x).view(B, S, -1, d)
@@ -83,7 +83,11 @@ class LlamaAttention(nn.Module):
         # RoPE uses the caller's cos/sin slice (seq-offset aware under SP)
         q = apply_rope(q, cos, sin)
         k = apply_rope(k, cos, sin)
-        if self._dist_attn is not None:
+        if kv_cache is not None:
+            k, v = kv_cache.update(k, v)
+            causal = S > 1  # prefill chunk is causal; decode sees all past
+            o = flash_attention(q, k, v, causal=causal)
+        elif self._dist_attn is not None:
             o = self._dist_attn(q, k, v, causal=True)
         else:
             o = flash_attention(q, k, v, causal=True)
@@ -123,8 +127,9 @@ class LlamaDecoderLayer(nn.Module):
                                                      cfg.rms_norm_eps)
         self.mlp = LlamaMLP(cfg)
 
-    def forward(self, x, cos, sin):
-        x = x + self.self_attn(self.input_layernorm(x), cos, sin)
+    def forward(self, x, cos, sin, kv_cache=None):
+        x = x + self.self_attn(self.input_layernorm(x), cos, sin,
+                               kv_cache=kv_cache)
         x = x + self.mlp(self.post_attention_layernorm(x))
         return x
 
@@ -142,17 +147,18 @@ class LlamaModel(nn.Module):
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
 
-    def forward(self, input_ids, seq_offset=0):
+    def forward(self, input_ids, seq_offset=0, kv_caches=None):
         x = self.embed_tokens(input_ids)
         S = input_ids.shape[1]
         cos = self.rope_cos[seq_offset:seq_offset + S]
         sin = self.rope_sin[seq_offset:seq_offset + S]
-        for layer in self.layers:
+        for i, layer in enumerate(self.layers):
             if self.cfg.activation_checkpointing and self.training:
                 x = torch.utils.checkpoint.checkpoint(
                     layer, x, cos, sin, use_reentrant=False)
             else:
-                x = layer(x, cos, sin)
+                x = layer(x, cos, sin,
+                          kv_cache=kv_caches[i] if kv_caches else None)
         return self.norm(x)
 
 
@@ -175,8 +181,8 @@ class LlamaForCausalLM(nn.Module):
         elif isinstance(module, nn.Embedding):
             module.weight.data.normal_(0.0, std)
 
-    def forward(self, input_ids, labels=None, seq_offset=0):
-        h = self.model(input_ids, seq_offset=seq_offset)
+    def forward(self, input_ids, labels=None, seq_offset=0, kv_caches=None):
+        h = self.model(input_ids, seq_offset=seq_offset, kv_caches=kv_caches)
         logits = self.lm_head(h)
         if labels is None:
             return logits
