@@ -41,3 +41,70 @@ def test_llama_tiny_trains_gpu(stage):
         losses.append(loss.item())
     assert losses[-1] < losses[0] * 0.9, f"no training progress: {losses}"
     engine.destroy()
+
+
+def test_mixtral_tiny_gpu():
+    _init_env()
+    import deepspeed_amd
+    from deepspeed_amd.models.mixtral import MIXTRAL_CONFIGS, MixtralForCausalLM
+    cfg = MIXTRAL_CONFIGS["mixtral-tiny"]
+    torch.manual_seed(0)
+    with torch.device("cuda:0"):
+        model = MixtralForCausalLM(cfg)
+    config = {
+        "train_micro_batch_size_per_gpu": 2,
+        "optimizer": {"type": "AdamW", "params": {"lr": 3e-4}},
+        "zero_optimization": {"stage": 1},
+        "bf16": {"enabled": True},
+    }
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config=config)
+    data = torch.randint(0, cfg.vocab_size, (2, 64), device="cuda:0")
+    losses = []
+    for _ in range(6):
+        loss = engine(data, labels=data)
+        engine.backward(loss)
+        engine.step()
+        losses.append(loss.item())
+    assert losses[-1] < losses[0], losses
+    engine.destroy()
+
+
+def test_inference_generate_gpu():
+    _init_env()
+    import deepspeed_amd
+    from deepspeed_amd.models.llama import LLAMA_CONFIGS, LlamaForCausalLM
+    cfg = LLAMA_CONFIGS["llama-tiny"]
+    torch.manual_seed(0)
+    with torch.device("cuda:0"):
+        model = LlamaForCausalLM(cfg)
+    engine = deepspeed_amd.init_inference(model, config={})
+    ids = torch.randint(0, cfg.vocab_size, (2, 16), device="cuda:0")
+    out = engine.generate(ids, max_new_tokens=16)
+    assert out.shape == (2, 32)
+
+
+def test_zero3_cpu_offload_gpu():
+    _init_env()
+    import deepspeed_amd
+    from deepspeed_amd.models.llama import LLAMA_CONFIGS, LlamaForCausalLM
+    cfg = LLAMA_CONFIGS["llama-tiny"]
+    torch.manual_seed(0)
+    with torch.device("cuda:0"):
+        model = LlamaForCausalLM(cfg)
+    config = {
+        "train_micro_batch_size_per_gpu": 2,
+        "optimizer": {"type": "AdamW", "params": {"lr": 3e-4}},
+        "zero_optimization": {"stage": 3,
+                              "offload_optimizer": {"device": "cpu"}},
+        "bf16": {"enabled": True},
+    }
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config=config)
+    data = torch.randint(0, cfg.vocab_size, (2, 64), device="cuda:0")
+    losses = []
+    for _ in range(6):
+        loss = engine(data, labels=data)
+        engine.backward(loss)
+        engine.step()
+        losses.append(loss.item())
+    assert losses[-1] < losses[0], losses
+    engine.destroy()
