@@ -45,6 +45,10 @@ at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets,
                              long ignore_index);
 
 void bind_aio(py::module_& m);
+std::vector<at::Tensor> quantize_int8(at::Tensor x, long group_size);
+at::Tensor dequantize_int8(at::Tensor q, at::Tensor scales, long group_size);
+std::vector<at::Tensor> quantize_fp8(at::Tensor x, long group_size);
+at::Tensor dequantize_fp8(at::Tensor q, at::Tensor scales, long group_size);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   bind_aio(m);
@@ -58,6 +62,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("accum_bf16_to_f32", &accum_bf16_to_f32, py::arg("dst"),
         py::arg("src"), py::arg("scale") = 1.0);
   m.def("l2norm_sq", &l2norm_sq);
+  m.def("quantize_int8", &quantize_int8);
+  m.def("dequantize_int8", &dequantize_int8);
+  m.def("quantize_fp8", &quantize_fp8);
+  m.def("dequantize_fp8", &dequantize_fp8);
   m.def("flash_attn_fwd", &flash_attn_fwd, py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("causal") = true, py::arg("scale") = 0.0);
   m.def("cpu_adam_step", &cpu_adam_step, py::arg("p"), py::arg("g"),
