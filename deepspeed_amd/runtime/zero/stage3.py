@@ -438,19 +438,28 @@ class ZeroStage3Optimizer:
                                       device=p.grad.device))
             p.grad = None
         if use_coalescing:
-            from torch.distributed.distributed_c10d import _coalescing_manager
-            op = _avg_op(world, inputs[0]) if inputs[0].is_cuda else None
-            if op is None:
-                for inp in inputs:
-                    inp.div_(world)
-                op = dist.ReduceOp.SUM
-            with _coalescing_manager(self.dp_group, self.device,
-                                     async_ops=True) as cm:
-                for out, inp in zip(shards, inputs):
-                    dist.reduce_scatter_tensor(out, inp, op=op,
-                                               group=self.dp_group)
-            cm.wait()
-        else:
+            try:
+                from torch.distributed.distributed_c10d import \
+                    _coalescing_manager
+                op = _avg_op(world, inputs[0]) if inputs[0].is_cuda else None
+                if op is None:
+                    for inp in inputs:
+                        inp.div_(world)
+                    op = dist.ReduceOp.SUM
+                with _coalescing_manager(self.dp_group, self.device,
+                                         async_ops=True) as cm:
+                    for out, inp in zip(shards, inputs):
+                        dist.reduce_scatter_tensor(out, inp, op=op,
+                                                   group=self.dp_group)
+                cm.wait()
+            except Exception as e:
+                from .stage3_params import _COALESCE_OK  # noqa: F401
+                import deepspeed_amd.runtime.zero.stage3_params as s3p
+                s3p._COALESCE_OK = False
+                logger.warning(f"coalesced reduce-scatter failed ({e}); "
+                               "falling back to per-param calls")
+                use_coalescing = False
+        if not use_coalescing:
             for out, inp in zip(shards, inputs):
                 op = _avg_op(world, inp)
                 dist.reduce_scatter_tensor(out, inp, op=op,

@@ -43,8 +43,11 @@ def _pad_to(numel, multiple):
     return (numel + multiple - 1) // multiple * multiple
 
 
+_COALESCE_OK = True
+
+
 def _supports_coalescing(group):
-    if not torch.cuda.is_available():
+    if not _COALESCE_OK or not torch.cuda.is_available():
         return False
     try:
         import torch.distributed as td
@@ -136,16 +139,25 @@ def all_gather_params(params, dp_group, async_op=True):
         return h
     use_coalescing = _supports_coalescing(dp_group) and len(todo) > 1
     if use_coalescing:
-        from torch.distributed.distributed_c10d import _coalescing_manager
-        device = todo[0].ds_tensor.device
-        with _coalescing_manager(dp_group, device, async_ops=True) as cm:
-            for p in todo:
-                buf = torch.empty(p.ds_tensor.numel() * world,
-                                  dtype=p.ds_tensor.dtype, device=device)
-                dist.all_gather_into_tensor(buf, p.ds_tensor, group=dp_group)
-                buffers.append(buf)
-        works = [cm]
-    else:
+        try:
+            from torch.distributed.distributed_c10d import _coalescing_manager
+            device = todo[0].ds_tensor.device
+            with _coalescing_manager(dp_group, device, async_ops=True) as cm:
+                for p in todo:
+                    buf = torch.empty(p.ds_tensor.numel() * world,
+                                      dtype=p.ds_tensor.dtype, device=device)
+                    dist.all_gather_into_tensor(buf, p.ds_tensor,
+                                                group=dp_group)
+                    buffers.append(buf)
+            works = [cm]
+        except Exception as e:
+            log_dist(f"coalesced allgather unavailable ({e}); falling back",
+                     ranks=[0])
+            global _COALESCE_OK
+            _COALESCE_OK = False
+            buffers = []
+            use_coalescing = False
+    if not use_coalescing:
         for p in todo:
             buf = torch.empty(p.ds_tensor.numel() * world,
                               dtype=p.ds_tensor.dtype,
