@@ -1,0 +1,135 @@
+"""Compression: QAT weight/activation quantization + layer reduction.
+
+Parity: reference `compression/compress.py`, `basic_layer.py`
+(LinearLayer_Compress), `helper.py`. `init_compression` walks the module
+tree and swaps Linears for QAT-capable ones per config; `redundancy_clean`
+makes quantization permanent.
+"""
+import re
+
+import torch
+import torch.nn.functional as F
+
+from .utils.logging import log_dist
+
+
+class _FakeQuant(torch.autograd.Function):
+    """Straight-through symmetric fake quantization."""
+
+    @staticmethod
+    def forward(ctx, x, bits):
+        qmax = 2 ** (bits - 1) - 1
+        amax = x.abs().amax().clamp(min=1e-8)
+        scale = amax / qmax
+        return torch.clamp(torch.round(x / scale), -qmax, qmax) * scale
+
+    @staticmethod
+    def backward(ctx, g):
+        return g, None
+
+
+def fake_quantize(x, bits=8):
+    return _FakeQuant.apply(x, bits)
+
+
+class LinearLayerCompress(torch.nn.Linear):
+    """Linear with optional QAT weight/activation quantization and
+    row/head pruning masks."""
+
+    def __init__(self, in_features, out_features, bias=True):
+        super().__init__(in_features, out_features, bias=bias)
+        self.weight_quant_enabled = False
+        self.weight_bits = 8
+        self.act_quant_enabled = False
+        self.act_bits = 8
+        self.register_buffer("prune_mask", None)
+
+    @classmethod
+    def from_linear(cls, linear):
+        m = cls(linear.in_features, linear.out_features,
+                bias=linear.bias is not None)
+        m.weight.data.copy_(linear.weight.data)
+        if linear.bias is not None:
+            m.bias.data.copy_(linear.bias.data)
+        return m.to(linear.weight.dtype)
+
+    def enable_weight_quantization(self, bits=8):
+        self.weight_quant_enabled = True
+        self.weight_bits = bits
+
+    def enable_activation_quantization(self, bits=8):
+        self.act_quant_enabled = True
+        self.act_bits = bits
+
+    def enable_sparse_pruning(self, ratio, method="l1"):
+        w = self.weight.data.abs()
+        k = int(w.numel() * ratio)
+        if k > 0:
+            thresh = w.reshape(-1).kthvalue(k).values
+            self.prune_mask = (w > thresh).to(self.weight.dtype)
+
+    def forward(self, x):
+        w = self.weight
+        if self.prune_mask is not None:
+            w = w * self.prune_mask
+        if self.weight_quant_enabled:
+            w = fake_quantize(w.float(), self.weight_bits).to(x.dtype)
+        if self.act_quant_enabled:
+            x = fake_quantize(x.float(), self.act_bits).to(w.dtype)
+        return F.linear(x, w, self.bias)
+
+
+def init_compression(model, compression_config):
+    """Swap matching Linears for compressible ones; apply settings.
+
+    Config (subset of reference schema):
+      {"weight_quantization": {"shared_parameters": {"enabled": true},
+          "different_groups": {"wq1": {
+              "params": {"target_bits": 8},
+              "modules": ["attention.*", ".*mlp.*"]}}},
+       "sparse_pruning": {...}}
+    """
+    wq = compression_config.get("weight_quantization", {})
+    sp = compression_config.get("sparse_pruning", {})
+    n = 0
+    for name, module in list(model.named_modules()):
+        for child_name, child in list(module.named_children()):
+            if type(child) is torch.nn.Linear:
+                full = f"{name}.{child_name}" if name else child_name
+                new = None
+                for gname, g in wq.get("different_groups", {}).items():
+                    if any(re.search(pat, full)
+                           for pat in g.get("modules", [".*"])):
+                        new = LinearLayerCompress.from_linear(child)
+                        new.enable_weight_quantization(
+                            g.get("params", {}).get("target_bits", 8))
+                        break
+                for gname, g in sp.get("different_groups", {}).items():
+                    if any(re.search(pat, full)
+                           for pat in g.get("modules", [".*"])):
+                        if new is None:
+                            new = LinearLayerCompress.from_linear(child)
+                        new.enable_sparse_pruning(
+                            g.get("params", {}).get("dense_ratio", 0.5))
+                        break
+                if new is not None:
+                    setattr(module, child_name, new)
+                    n += 1
+    log_dist(f"init_compression: converted {n} linears", ranks=[0])
+    return model
+
+
+def redundancy_clean(model, compression_config=None):
+    """Make quantization/pruning permanent (bake into weights)."""
+    for module in model.modules():
+        if isinstance(module, LinearLayerCompress):
+            if module.prune_mask is not None:
+                module.weight.data.mul_(module.prune_mask)
+                module.prune_mask = None
+            if module.weight_quant_enabled:
+                module.weight.data.copy_(
+                    fake_quantize(module.weight.data.float(),
+                                  module.weight_bits)
+                    .to(module.weight.dtype))
+                module.weight_quant_enabled = False
+    return model

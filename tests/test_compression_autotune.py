@@ -1,0 +1,76 @@
+"""Compression (QAT/pruning), autotuner, quantized reduction collectives."""
+import torch
+
+from tests.common import run_distributed
+
+
+def test_fake_quant_straight_through():
+    from deepspeed_amd.compression import fake_quantize
+    x = torch.randn(100, requires_grad=True)
+    y = fake_quantize(x, 8)
+    assert (y - x).abs().max() < x.abs().max() / 100
+    y.sum().backward()
+    assert torch.allclose(x.grad, torch.ones_like(x))
+
+
+def test_init_compression_and_clean():
+    from deepspeed_amd.compression import (LinearLayerCompress,
+                                           init_compression,
+                                           redundancy_clean)
+    m = torch.nn.Sequential(torch.nn.Linear(16, 16), torch.nn.ReLU(),
+                            torch.nn.Linear(16, 8))
+    cfg = {"weight_quantization": {"different_groups": {
+        "wq": {"params": {"target_bits": 8}, "modules": [".*"]}}},
+        "sparse_pruning": {"different_groups": {
+            "sp": {"params": {"dense_ratio": 0.5}, "modules": ["0"]}}}}
+    init_compression(m, cfg)
+    assert isinstance(m[0], LinearLayerCompress)
+    assert m[0].prune_mask is not None
+    x = torch.randn(4, 16)
+    out = m(x)
+    assert out.shape == (4, 8)
+    redundancy_clean(m)
+    # ~half of m[0] weights pruned to zero
+    frac_zero = (m[0].weight == 0).float().mean().item()
+    assert 0.3 < frac_zero < 0.7
+
+
+def test_autotuner_cpu():
+    from deepspeed_amd.autotuning import Autotuner
+    from tests.simple_model import SimpleModel
+
+    def model_fn():
+        torch.manual_seed(0)
+        return SimpleModel(32)
+
+    def data_fn(cfg):
+        mb = cfg["train_micro_batch_size_per_gpu"]
+        return (torch.randn(mb, 32, dtype=torch.bfloat16),
+                torch.randn(mb, 32, dtype=torch.bfloat16))
+
+    base = {"optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+            "bf16": {"enabled": True}}
+    tuner = Autotuner(model_fn, data_fn, base, steps=2, warmup=1)
+    best, results = tuner.tune(micro_batches=(2, 4), stages=(1, 2))
+    assert best["train_micro_batch_size_per_gpu"] in (2, 4)
+    assert len([r for r in results if "error" not in r]) >= 1
+
+
+def _quant_reduce():
+    import torch.distributed as td
+    from deepspeed_amd.runtime.comm.coalesced_collectives import \
+        (all_to_all_quant_reduce, reduce_scatter_coalesced)
+    rank = td.get_rank()
+    world = td.get_world_size()
+    torch.manual_seed(17 + rank)
+    g = torch.randn(4096, dtype=torch.float32)
+    exact = reduce_scatter_coalesced([g.clone()])[0]
+    quant = all_to_all_quant_reduce([g.clone()])[0]
+    rel = (quant[:exact.numel()] - exact).abs().mean() / \
+        exact.abs().mean()
+    assert rel < 0.05, rel.item()
+    return True
+
+
+def test_quantized_grad_reduce_2rank():
+    assert all(run_distributed(_quant_reduce, world_size=2))
