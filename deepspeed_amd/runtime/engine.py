@@ -469,6 +469,12 @@ class DeepSpeedEngine(torch.nn.Module):
                      load_lr_scheduler_states=load_lr_scheduler_states,
                      load_module_only=load_module_only)
 
+    def generate(self, input_ids, **kwargs):
+        """RLHF-style generation with the training weights (hybrid engine:
+        gathers ZeRO-3 shards for the rollout, then releases)."""
+        from .hybrid_engine import generate
+        return generate(self, input_ids, **kwargs)
+
     def load_universal_checkpoint(self, universal_dir):
         """Resume from a universal checkpoint at ANY dp world size."""
         from ..checkpoint.universal import load_universal_into_optimizer

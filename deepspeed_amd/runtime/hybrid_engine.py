@@ -1,0 +1,83 @@
+"""Hybrid engine: RLHF train <-> generate switching on ZeRO-3 weights.
+
+Parity: reference `runtime/hybrid_engine.py:31` (DeepSpeedHybridEngine).
+`generate()` gathers the sharded parameters, runs KV-cache generation with
+the training weights, then releases them back to shards — no weight copies,
+no separate inference model.
+"""
+import contextlib
+
+import torch
+
+from ..inference.engine import KVCache
+from ..utils.logging import log_dist
+
+
+class HybridEngineMixin:
+    """Mixed into DeepSpeedEngine (see engine.generate)."""
+
+
+@contextlib.contextmanager
+def gathered_for_generation(engine):
+    """Gather all ZeRO-3 shards for the duration of generation."""
+    if engine.zero_optimization_stage() != 3:
+        yield
+        return
+    from .zero.stage3_params import (ZeroParamStatus, all_gather_params,
+                                     free_param, is_zero_param)
+    params = [p for p in engine.module.parameters() if is_zero_param(p)]
+    need = [p for p in params
+            if p.ds_status == ZeroParamStatus.NOT_AVAILABLE]
+    handle = all_gather_params(need, engine.optimizer.dp_group,
+                               async_op=False)
+    handle.wait()
+    try:
+        yield
+    finally:
+        for p in need:
+            if not p.ds_persist:
+                free_param(p)
+
+
+@torch.no_grad()
+def generate(engine, input_ids, max_new_tokens=32, temperature=0.0,
+             top_k=0, eos_token_id=None):
+    """Generate with the (possibly ZeRO-3-sharded) training weights."""
+    module = engine.module
+    cfg = getattr(module, "cfg", None)
+    assert cfg is not None, "model must expose .cfg"
+    was_training = module.training
+    was_ckpt = getattr(cfg, "activation_checkpointing", False)
+    cfg.activation_checkpointing = False
+    module.eval()
+    try:
+        with gathered_for_generation(engine):
+            B, S = input_ids.shape
+            max_seq = min(cfg.max_position_embeddings, S + max_new_tokens)
+            caches = [KVCache(B, max_seq, cfg.num_key_value_heads,
+                              cfg.head_dim, engine.config.dtype,
+                              input_ids.device)
+                      for _ in range(cfg.num_hidden_layers)]
+            logits = module(input_ids, kv_caches=caches)
+            out = input_ids
+            for _ in range(max_new_tokens):
+                nl = logits[:, -1, :].float()
+                if temperature > 0:
+                    nl = nl / temperature
+                    if top_k > 0:
+                        kth = torch.topk(nl, top_k, dim=-1).values[:, -1:]
+                        nl = nl.masked_fill(nl < kth, float("-inf"))
+                    nxt = torch.multinomial(torch.softmax(nl, -1), 1)
+                else:
+                    nxt = nl.argmax(-1, keepdim=True)
+                out = torch.cat([out, nxt], dim=1)
+                if eos_token_id is not None and (nxt == eos_token_id).all():
+                    break
+                if out.shape[1] >= max_seq:
+                    break
+                logits = module(nxt, seq_offset=out.shape[1] - 1,
+                                kv_caches=caches)
+            return out
+    finally:
+        cfg.activation_checkpointing = was_ckpt
+        module.train(was_training)
