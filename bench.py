@@ -21,8 +21,16 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=8)
     p.add_argument("--warmup", type=int, default=3)
-    p.add_argument("--model", type=str, default="llama3-8b")
+    p.add_argument("--model", type=str, default="llama3-8b",
+                   help="llama3-8b|llama3-70b|llama-small|mixtral-8x7b|...")
     p.add_argument("--seq-len", type=int, default=4096)
+    p.add_argument("--sp", type=int, default=1,
+                   help="Ulysses sequence-parallel degree (config #3: "
+                   "--sp 8 --seq-len 32768)")
+    p.add_argument("--ep", type=int, default=1,
+                   help="expert-parallel degree for MoE models")
+    p.add_argument("--offload", type=str, default="none",
+                   choices=["none", "cpu", "nvme"])
     p.add_argument("--micro-batch", type=int, default=4)
     p.add_argument("--grad-accum", type=int, default=1)
     p.add_argument("--zero-stage", type=int, default=3)
@@ -40,7 +48,10 @@ def main():
 
     import deepspeed_amd
     from deepspeed_amd import comm as dist
-    from deepspeed_amd.models.llama import LLAMA_CONFIGS, LlamaForCausalLM
+    from deepspeed_amd.models.llama import (LLAMA_CONFIGS, LlamaForCausalLM,
+                                            enable_ulysses)
+    from deepspeed_amd.models.mixtral import (MIXTRAL_CONFIGS,
+                                              MixtralForCausalLM)
 
     dist.init_distributed()
     rank = dist.get_rank()
@@ -50,13 +61,21 @@ def main():
     if torch.cuda.is_available():
         torch.cuda.set_device(device)
 
-    cfg = LLAMA_CONFIGS[args.model]
+    if args.model in MIXTRAL_CONFIGS:
+        cfg = MIXTRAL_CONFIGS[args.model]
+        cfg.ep_size = min(args.ep, world) if args.ep > 1 else 1
+        model_cls = MixtralForCausalLM
+    else:
+        cfg = LLAMA_CONFIGS[args.model]
+        model_cls = LlamaForCausalLM
     cfg.activation_checkpointing = args.act_ckpt
+    if args.seq_len > cfg.max_position_embeddings:
+        cfg.max_position_embeddings = args.seq_len
     torch.manual_seed(1234 + rank)
     t0 = time.time()
     # build directly on device: 8B bf16 = 16 GB, fits trivially in 288 GB
     with torch.device(device):
-        model = LlamaForCausalLM(cfg)
+        model = model_cls(cfg)
     if rank == 0:
         n_params = sum(p.numel() for p in model.parameters())
         print(f"# model {args.model}: {n_params/1e9:.2f}B params, "
@@ -73,13 +92,29 @@ def main():
         "gradient_clipping": 1.0,
         "steps_per_print": 1000000,
     }
+    if args.offload != "none":
+        ds_config["zero_optimization"]["offload_optimizer"] = {
+            "device": args.offload}
+    if args.sp > 1:
+        ds_config["sequence_parallel"] = {
+            "sequence_parallel_size": args.sp}
     engine, _, _, _ = deepspeed_amd.initialize(model=model, config=ds_config)
+
+    sp_rank, sp_world = 0, 1
+    if args.sp > 1:
+        from deepspeed_amd.comm import groups as _grp
+        spg = _grp.get_sequence_parallel_group()
+        enable_ulysses(engine.module, spg)
+        sp_rank = dist.get_rank(spg)
+        sp_world = dist.get_world_size(spg)
 
     S = args.seq_len
     B = args.micro_batch
     # rotating synthetic batches: fresh tokens each micro-step (no caching)
+    S_local = S // sp_world
+    seq_off = sp_rank * S_local
     n_bufs = 8
-    bufs = [torch.randint(0, cfg.vocab_size, (B, S), device=device)
+    bufs = [torch.randint(0, cfg.vocab_size, (B, S_local), device=device)
             for _ in range(n_bufs)]
     step_idx = [0]
 
@@ -87,7 +122,10 @@ def main():
         for _ in range(args.grad_accum):
             data = bufs[step_idx[0] % n_bufs]
             step_idx[0] += 1
-            loss = engine(data, labels=data)
+            if sp_world > 1:
+                loss = engine(data, labels=data, seq_offset=seq_off)
+            else:
+                loss = engine(data, labels=data)
             engine.backward(loss)
             engine.step()
         return loss
@@ -117,8 +155,9 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = t.item()
 
-    global_batch = B * args.grad_accum * world
-    tokens_per_step = global_batch * S
+    global_batch = B * args.grad_accum * (world // sp_world)
+    # with SP, ranks of one SP group share a sample (S_local tokens each)
+    tokens_per_step = B * args.grad_accum * world * S_local
     tokens_per_s = tokens_per_step * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
@@ -128,8 +167,11 @@ def main():
     tflops_per_gpu = 6 * n_params * tokens_per_s / world / 1e12
 
     if rank == 0:
+        metric = ("tokens/sec Llama-3-8B ZeRO-3"
+                  if args.model == "llama3-8b" and args.zero_stage == 3
+                  else f"tokens/sec {args.model} zero{args.zero_stage}")
         print(json.dumps({
-            "metric": "tokens/sec Llama-3-8B ZeRO-3",
+            "metric": metric,
             "value": round(tokens_per_s, 1),
             "unit": "tokens/s",
             "n_gpus": world,
@@ -142,7 +184,13 @@ def main():
             "dtype": "bf16",
             "data": "synthetic",
             "config": {"model": args.model, "global_batch": global_batch,
-                       "seq_len": S, "parallelism": f"zero{args.zero_stage}_dp{world}",
+                       "seq_len": S,
+                       "parallelism": (f"zero{args.zero_stage}_"
+                                       f"dp{world // sp_world}" +
+                                       (f"_sp{sp_world}" if sp_world > 1
+                                        else "") +
+                                       (f"_ep{args.ep}" if args.ep > 1
+                                        else "")),
                        "grad_accum": args.grad_accum,
                        "model_tflops_per_gpu_6PT": round(tflops_per_gpu, 1),
                        "final_loss": round(loss.item(), 4)},

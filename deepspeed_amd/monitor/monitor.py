@@ -72,3 +72,28 @@ class MonitorMaster(Monitor):
     def write_events(self, event_list):
         for m in self.monitors:
             m.write_events(event_list)
+
+
+class WandbMonitor(Monitor):
+    """Weights & Biases sink (ref deepspeed/monitor/wandb.py:12)."""
+
+    def __init__(self, config):
+        super().__init__(config)
+        self.enabled = getattr(config, "enabled", False) and \
+            dist.get_rank() == 0
+        self._wandb = None
+        if self.enabled:
+            try:
+                import wandb
+                self._wandb = wandb
+                wandb.init(project=getattr(config, "project", "dsamd"),
+                           group=getattr(config, "group", None),
+                           name=getattr(config, "job_name", None))
+            except ImportError:
+                self.enabled = False
+
+    def write_events(self, event_list):
+        if not self.enabled or self._wandb is None:
+            return
+        for name, value, step in event_list:
+            self._wandb.log({name: value}, step=step)
