@@ -34,6 +34,14 @@ DEV_INLINE int kswz(int row, int byte_off) {
   return row * (DHEAD * 2) + (byte_off ^ ((row & 7) << 4));
 }
 
+// Vt[d][kv] byte offset: pad-8 rows spread the d-read (2-way) and the
+// ((d>>5)&3)<<5 XOR spreads the 4 d-groups on the transpose write
+// (was an 8-way write conflict -> ~2-way). Alignment: both terms are
+// multiples of 16B for the b128 reads.
+DEV_INLINE int vtswz(int d, int kv_byte) {
+  return d * ((KVTILE + VT_PAD) * 2) + (kv_byte ^ (((d >> 5) & 3) << 5));
+}
+
 __launch_bounds__(256, 2)
 __global__ void flash_fwd_kernel(
     const short* __restrict__ q,  // [B,S,Hq,D]
@@ -97,27 +105,30 @@ __global__ void flash_fwd_kernel(
 
   for (int t = 0; t < n_tiles; ++t) {
     const int kvbase = t * KVTILE;
-    // ---- stage K (swizzled) and V (transposed) --------------------------
-    // 256 threads; K tile 64x128: each thread loads 2 rows' worth: thread i
-    // handles row = i/4, 32 cols starting (i%4)*32
+    // ---- stage K (async global_load_lds, pre-swizzled source) -----------
+    // wave w, pass p covers 1 KiB of linear K LDS: lane's dest is
+    // base + lane*16 (HW rule), so the SOURCE column carries the XOR
+    // (guide m173: swizzled layouts via pre-swizzled global address).
     {
+      for (int pass = 0; pass < 4; ++pass) {
+        int linear = (pass * 4 + wave) * 1024 + lane * 16;
+        int row = linear >> 8;             // /256 bytes per row
+        int colbyte = linear & 255;
+        int src_col = colbyte ^ ((row & 7) << 4);
+        int grow = kvbase + row;
+        int srow = grow < S ? grow : S - 1;  // clamped; masked via -inf
+        const short* src = kp + (long long)srow * kv_row_stride +
+                           (src_col >> 1);
+        __builtin_amdgcn_global_load_lds(
+            reinterpret_cast<const unsigned int*>(src),
+            reinterpret_cast<unsigned int*>(
+                reinterpret_cast<char*>(k_lds) + (pass * 4 + wave) * 1024),
+            16, 0, 0);
+      }
+      // V transposed: coalesced row reads, XOR-spread scatter writes
       int row = threadIdx.x >> 2;        // 0..63
       int c0 = (threadIdx.x & 3) * 32;   // 0,32,64,96
       int grow = kvbase + row;
-      const short* krow = kp + (long long)grow * kv_row_stride;
-#pragma unroll
-      for (int cc = 0; cc < 4; ++cc) {
-        int col = c0 + cc * 8;
-        bf16x8_t kv8;
-        if (grow < S) {
-          kv8 = *reinterpret_cast<const bf16x8_t*>(krow + col);
-        } else {
-          kv8 = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
-        }
-        *reinterpret_cast<bf16x8_t*>(
-            reinterpret_cast<char*>(k_lds) + kswz(row, col * 2)) = kv8;
-      }
-      // V transposed: read same shape, scatter to vt[d][kv]
       const short* vrow = vp + (long long)grow * kv_row_stride;
 #pragma unroll
       for (int cc = 0; cc < 4; ++cc) {
@@ -128,15 +139,19 @@ __global__ void flash_fwd_kernel(
         } else {
           vv8 = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
         }
+        char* vbase = reinterpret_cast<char*>(vt_lds);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          vt_lds[(col + j) * (KVTILE + VT_PAD) + row] = vv8[j];
+          *reinterpret_cast<short*>(vbase + vtswz(col + j, row * 2)) =
+              vv8[j];
       }
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     }
     __syncthreads();
 
     // ---- S = Q @ K^T  (4 col-tiles of 16) -------------------------------
     f32x4_t s_acc[4];
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ct = 0; ct < 4; ++ct) {
       f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
@@ -149,6 +164,7 @@ __global__ void flash_fwd_kernel(
       }
       s_acc[ct] = acc;
     }
+    __builtin_amdgcn_s_setprio(0);
 
     // ---- online softmax -------------------------------------------------
     // lane holds S[row = l4*4 + r][col = l15 + 16*ct] (scaled below)
@@ -211,6 +227,7 @@ __global__ void flash_fwd_kernel(
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
     // ---- O += P @ V  ----------------------------------------------------
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       bf16x8_t afrag = *reinterpret_cast<const bf16x8_t*>(
@@ -218,10 +235,12 @@ __global__ void flash_fwd_kernel(
 #pragma unroll
       for (int dt = 0; dt < 8; ++dt) {
         bf16x8_t bfrag = *reinterpret_cast<const bf16x8_t*>(
-            vt_lds + (l15 + 16 * dt) * (KVTILE + VT_PAD) + l4 * 8 + 32 * ks);
+            reinterpret_cast<char*>(vt_lds) +
+            vtswz(l15 + 16 * dt, (l4 * 8 + 32 * ks) * 2));
         o_acc[dt] = MFMA_BF16_16x16x32(afrag, bfrag, o_acc[dt], 0, 0, 0);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
     __syncthreads();  // before next tile's staging overwrites K/V
   }
 
