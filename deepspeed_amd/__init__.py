@@ -15,6 +15,9 @@ from . import ops  # noqa: F401
 from .config import DeepSpeedConfig  # noqa: F401
 from .runtime.engine import DeepSpeedEngine
 from .runtime import zero  # noqa: F401
+from .runtime.pipe import PipelineModule, LayerSpec  # noqa: F401
+from .moe.layer import MoE  # noqa: F401
+from .ops.adam import FusedAdam  # noqa: F401
 from .utils.logging import logger, log_dist  # noqa: F401
 
 
