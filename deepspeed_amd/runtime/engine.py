@@ -480,6 +480,31 @@ class DeepSpeedEngine(torch.nn.Module):
         from ..checkpoint.universal import load_universal_into_optimizer
         load_universal_into_optimizer(self.optimizer, universal_dir)
 
+    def save_16bit_model(self, save_dir, save_filename="pytorch_model.bin",
+                         exclude_frozen_parameters=False):
+        """Consolidated 16-bit weights (gathers ZeRO-3 shards on rank 0).
+
+        Ref engine.py:5567 / _zero3_consolidated_16bit_state_dict:5491.
+        """
+        sd = self._consolidated_16bit_state_dict()
+        if dist.get_rank() == 0 and sd is not None:
+            os.makedirs(save_dir, exist_ok=True)
+            torch.save(sd, os.path.join(save_dir, save_filename))
+        if dist.is_initialized():
+            dist.barrier()
+        return True
+
+    def _consolidated_16bit_state_dict(self):
+        if self.zero_optimization_stage() != 3:
+            return self.module.state_dict() if \
+                self.get_data_parallel_rank() == 0 else None
+        from .hybrid_engine import gathered_for_generation
+        with gathered_for_generation(self):
+            if self.get_data_parallel_rank() == 0:
+                return {k: v.detach().clone()
+                        for k, v in self.module.state_dict().items()}
+        return None
+
     def module_state_dict(self, exclude_frozen_parameters=False):
         return self.module.state_dict()
 

@@ -160,3 +160,25 @@ def test_universal_checkpoint_stage2_world2_to_1():
 
 def test_universal_checkpoint_stage3_world2_to_1():
     _check_universal(3)
+
+
+def _save16(stage, outdir):
+    import torch.distributed as tdist
+    from deepspeed_amd.comm import groups
+    groups.reset_groups()
+    e = _make_engine(stage)
+    batches = make_batches(4, 4, HIDDEN, dtype=torch.bfloat16)
+    _train_some(e, batches, 2)
+    e.save_16bit_model(outdir)
+    e.destroy()
+    return True
+
+
+def test_save_16bit_model_stage3():
+    with tempfile.TemporaryDirectory() as d:
+        assert all(run_distributed(_save16, world_size=2, args=(3, d)))
+        sd = torch.load(os.path.join(d, "pytorch_model.bin"),
+                        map_location="cpu", weights_only=False)
+        # full shapes restored
+        assert all(v.numel() > 0 for v in sd.values())
+        assert any("linears.0.weight" in k for k in sd)
