@@ -32,7 +32,7 @@ def parse_args():
     p.add_argument("--offload", type=str, default="none",
                    choices=["none", "cpu", "nvme"])
     p.add_argument("--micro-batch", type=int, default=4)
-    p.add_argument("--grad-accum", type=int, default=1)
+    p.add_argument("--grad-accum", type=int, default=2)
     p.add_argument("--zero-stage", type=int, default=3)
     p.add_argument("--act-ckpt", action="store_true",
                    help="enable activation checkpointing (default off: "

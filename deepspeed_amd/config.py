@@ -59,7 +59,7 @@ class ZeroConfig(DSConfigModel):
     contiguous_gradients: bool = True
     allgather_partitions: bool = True
     # stage-3
-    prefetch_bucket_size: int = Field(int(5e7), alias="stage3_prefetch_bucket_size")
+    prefetch_bucket_size: int = Field(int(2e8), alias="stage3_prefetch_bucket_size")
     param_persistence_threshold: int = Field(int(1e5), alias="stage3_param_persistence_threshold")
     model_persistence_threshold: int = Field(int(1e14), alias="stage3_model_persistence_threshold")
     max_live_parameters: int = Field(int(1e9), alias="stage3_max_live_parameters")

@@ -31,7 +31,7 @@ def test_zero_defaults():
     z = c.zero_config
     assert z.stage == 3
     assert z.reduce_bucket_size == int(5e8)
-    assert z.prefetch_bucket_size == int(5e7)
+    assert z.prefetch_bucket_size == int(2e8)
     assert z.param_persistence_threshold == int(1e5)
     assert z.overlap_comm is True
 
