@@ -37,10 +37,11 @@ DEV_INLINE int kswz(int row, int byte_off) {
   return row * (DHEAD * 2) + (byte_off ^ ((row & 7) << 4));
 }
 
-// Vt[d][kv]: pad-8 rows spread the d-read; ((d>>5)&3)<<5 XOR spreads the
-// 4 d-groups (d, d+32, d+64, d+96) hit by one transpose-write instruction.
+// Vt[d][kv]: pad-8 rows spread the d-read; ((d>>5)&3)<<4 XOR spreads the
+// d-groups hit by one transpose-write instruction. 16B granules keep the
+// XOR inside the 80B row and preserve b128 read alignment.
 DEV_INLINE int vtswz(int d, int kv_byte) {
-  return d * ((KVTILE + VT_PAD) * 2) + (kv_byte ^ (((d >> 5) & 3) << 5));
+  return d * ((KVTILE + VT_PAD) * 2) + (kv_byte ^ (((d >> 5) & 3) << 4));
 }
 
 __launch_bounds__(256, 2)
