@@ -26,7 +26,7 @@ using f32x4_t = __attribute__((ext_vector_type(4))) float;
 #define MFMA_BF16_16x16x32 __builtin_amdgcn_mfma_f32_16x16x32_bf16
 
 constexpr int QTILE = 64;   // q rows per block
-constexpr int KVTILE = 32;  // kv rows per tile
+constexpr int KVTILE = 64;  // kv rows per tile
 constexpr int DHEAD = 128;
 constexpr int VT_PAD = 8;   // Vt rows: [128][32+8]
 constexpr int P_PAD = 8;    // P rows:  [16][32+8]
@@ -104,15 +104,15 @@ __global__ void flash_fwd_kernel(
   const int kv_end = causal ? min(S, qbase + QTILE) : S;
   const int n_tiles = (kv_end + KVTILE - 1) / KVTILE;
 
-  // V staging state: thread owns row = tid>>3 (0..31), cols (tid&7)*16..+15
-  const int v_row = threadIdx.x >> 3;
-  const int v_c0 = (threadIdx.x & 7) * 16;
-  bf16x8_t vreg[2];
+  // V staging state: thread owns row = tid>>2 (0..63), cols (tid&3)*32..+31
+  const int v_row = threadIdx.x >> 2;
+  const int v_c0 = (threadIdx.x & 3) * 32;
+  bf16x8_t vreg[4];
 
   // ---- staging helpers -------------------------------------------------
   auto issue_k = [&](int buf, int kvbase) {
-    // wave w covers 2 KiB of linear K LDS (2 passes of 1 KiB)
-    for (int pass = 0; pass < 2; ++pass) {
+    // wave w covers 4 KiB of linear K LDS (4 passes of 1 KiB)
+    for (int pass = 0; pass < 4; ++pass) {
       int linear = (pass * 4 + wave) * 1024 + lane * 16;
       int row = linear >> 8;
       int colbyte = linear & 255;
@@ -133,13 +133,13 @@ __global__ void flash_fwd_kernel(
     int srow = grow < S ? grow : S - 1;
     const short* vrow_p = vp + (long long)srow * kv_row_stride;
 #pragma unroll
-    for (int cc = 0; cc < 2; ++cc)
+    for (int cc = 0; cc < 4; ++cc)
       vreg[cc] = *reinterpret_cast<const bf16x8_t*>(vrow_p + v_c0 + cc * 8);
   };
   auto write_v = [&](int buf) {
     char* vbase = reinterpret_cast<char*>(vt_lds[buf]);
 #pragma unroll
-    for (int cc = 0; cc < 2; ++cc)
+    for (int cc = 0; cc < 4; ++cc)
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         *reinterpret_cast<short*>(vbase + vtswz(v_c0 + cc * 8 + j,
@@ -164,10 +164,10 @@ __global__ void flash_fwd_kernel(
     }
 
     // ---- S = Q @ K^T  (2 col-tiles of 16) -------------------------------
-    f32x4_t s_acc[2];
+    f32x4_t s_acc[4];
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int ct = 0; ct < 2; ++ct) {
+    for (int ct = 0; ct < 4; ++ct) {
       f32x4_t acc = {0.f, 0.f, 0.f, 0.f};
       int krow = l15 + 16 * ct;
 #pragma unroll
@@ -184,14 +184,14 @@ __global__ void flash_fwd_kernel(
     // ---- online softmax -------------------------------------------------
     const bool btile = (causal && (kvbase + KVTILE > qbase)) ||
                        (kvbase + KVTILE > S);
-    float p_new[2][4];  // [ct][r]
+    float p_new[4][4];  // [ct][r]
     float corr[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       int grow = wave_q + l4 * 4 + r;
       float rowmax = -INFINITY;
 #pragma unroll
-      for (int ct = 0; ct < 2; ++ct) {
+      for (int ct = 0; ct < 4; ++ct) {
         float sv = s_acc[ct][r] * scale;
         if (btile) {
           int gcol = kvbase + l15 + 16 * ct;
@@ -208,7 +208,7 @@ __global__ void flash_fwd_kernel(
       corr[r] = c;
       float rowsum = 0.f;
 #pragma unroll
-      for (int ct = 0; ct < 2; ++ct) {
+      for (int ct = 0; ct < 4; ++ct) {
         float p = (s_acc[ct][r] == -INFINITY) ? 0.f
                                               : __expf(s_acc[ct][r] - m_new);
         p_new[ct][r] = p;
@@ -228,7 +228,7 @@ __global__ void flash_fwd_kernel(
     // ---- stage P (bf16) into per-wave LDS -------------------------------
     short* pw = p_lds[wave];
 #pragma unroll
-    for (int ct = 0; ct < 2; ++ct)
+    for (int ct = 0; ct < 4; ++ct)
 #pragma unroll
       for (int r = 0; r < 4; ++r)
         pw[(l4 * 4 + r) * (KVTILE + P_PAD) + l15 + 16 * ct] =
@@ -236,16 +236,19 @@ __global__ void flash_fwd_kernel(
     // wave-local P handoff: drain ds_writes before cross-lane ds_reads
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
-    // ---- O += P @ V  (single K=32 slice) --------------------------------
-    bf16x8_t afrag = *reinterpret_cast<const bf16x8_t*>(
-        pw + l15 * (KVTILE + P_PAD) + l4 * 8);
+    // ---- O += P @ V  (two K=32 slices) ----------------------------------
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int dt = 0; dt < 8; ++dt) {
-      bf16x8_t bfrag = *reinterpret_cast<const bf16x8_t*>(
-          reinterpret_cast<char*>(vt_lds[cur]) +
-          vtswz(l15 + 16 * dt, (l4 * 8) * 2));
-      o_acc[dt] = MFMA_BF16_16x16x32(afrag, bfrag, o_acc[dt], 0, 0, 0);
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8_t afrag = *reinterpret_cast<const bf16x8_t*>(
+          pw + l15 * (KVTILE + P_PAD) + l4 * 8 + 32 * ks);
+#pragma unroll
+      for (int dt = 0; dt < 8; ++dt) {
+        bf16x8_t bfrag = *reinterpret_cast<const bf16x8_t*>(
+            reinterpret_cast<char*>(vt_lds[cur]) +
+            vtswz(l15 + 16 * dt, (l4 * 8 + 32 * ks) * 2));
+        o_acc[dt] = MFMA_BF16_16x16x32(afrag, bfrag, o_acc[dt], 0, 0, 0);
+      }
     }
     __builtin_amdgcn_s_setprio(0);
 
