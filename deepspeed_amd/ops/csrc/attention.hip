@@ -2,19 +2,16 @@
 //
 // Role parity: reference uses CUTLASS/flash-attn kernels
 // (deepspeed/inference/v2/kernels, csrc/transformer/inference/csrc/softmax.cu).
-// MI355X-native design (guide §B "fused attention prefill"), v3:
-//   - 4 waves/block, 64 q-rows per block (16 per wave), KV tiles of 32,
-//     DOUBLE-BUFFERED K/V LDS: one barrier per tile, next tile's K staged
-//     via async global_load_lds and V via register loads issued BEFORE the
-//     current tile's MFMA work (T14 async-STAGE split)
+// MI355X-native design (guide §B "fused attention prefill"):
+//   - 4 waves/block, 64 q-rows per block (16 per wave), KV tiles of 64
 //   - mfma_f32_16x16x32_bf16; online softmax in fp32 registers
-//   - K LDS tile swizzled ((row&7)<<4) via PRE-SWIZZLED global source
-//     addresses (global_load_lds writes lane-linear; guide m173)
-//   - V staged TRANSPOSED with pad-8 rows + ((d>>5)&3)<<5 XOR: both the
-//     transpose write and the b128 read are ~2-way conflicts
+//   - K tile staged in LDS with the ((row&7)<<4) XOR byte-swizzle
+//     (row-major [64][128] bf16 is otherwise a 16-way bank conflict)
+//   - V staged TRANSPOSED in LDS so the PV B-operand is a ds_read_b128
 //   - P staged per-wave in LDS (pad 8) to convert C-layout -> A-layout
-//   - causal masking; GQA; per-row LSE output for the backward
-// Correctness verified against fp32 torch reference (tests/test_attention_gpu.py,
+//   - causal masking; GQA via head-group indexing; saves per-row LSE for the
+//     backward pass
+// Correctness verified against fp32 torch reference (tests/test_ops_gpu.py,
 // asymmetric random inputs per guide G9).
 #include <torch/extension.h>
 
@@ -28,20 +25,21 @@ using f32x4_t = __attribute__((ext_vector_type(4))) float;
 constexpr int QTILE = 64;   // q rows per block
 constexpr int KVTILE = 64;  // kv rows per tile
 constexpr int DHEAD = 128;
-constexpr int VT_PAD = 8;   // Vt rows: [128][32+8]
-constexpr int P_PAD = 8;    // P rows:  [16][32+8]
-constexpr int KBYTES = KVTILE * DHEAD * 2;  // 8 KiB per K buffer
+constexpr int KPAD = 0;            // K uses swizzle, no pad
+constexpr int VT_PAD = 8;          // Vt rows padded: [128][64+8]
+constexpr int P_PAD = 8;           // P rows padded: [16][64+8]
 
-// K row-major, ((row&7)<<4) XOR within the 256B row (16B granules)
+// swizzle byte offset within a K row (16B granules spread across banks)
 DEV_INLINE int kswz(int row, int byte_off) {
   return row * (DHEAD * 2) + (byte_off ^ ((row & 7) << 4));
 }
 
-// Vt[d][kv]: pad-8 rows spread the d-read; ((d>>5)&3)<<4 XOR spreads the
-// d-groups hit by one transpose-write instruction. 16B granules keep the
-// XOR inside the 80B row and preserve b128 read alignment.
+// Vt[d][kv] byte offset: pad-8 rows spread the d-read (2-way) and the
+// ((d>>5)&3)<<5 XOR spreads the 4 d-groups on the transpose write
+// (was an 8-way write conflict -> ~2-way). Alignment: both terms are
+// multiples of 16B for the b128 reads.
 DEV_INLINE int vtswz(int d, int kv_byte) {
-  return d * ((KVTILE + VT_PAD) * 2) + (kv_byte ^ (((d >> 5) & 3) << 4));
+  return d * ((KVTILE + VT_PAD) * 2) + (kv_byte ^ (((d >> 5) & 3) << 5));
 }
 
 __launch_bounds__(256, 2)
@@ -52,9 +50,9 @@ __global__ void flash_fwd_kernel(
     short* __restrict__ out,      // [B,S,Hq,D]
     float* __restrict__ lse_out,  // [B,Hq,S]
     int B, int S, int Hq, int Hk, float scale, int causal) {
-  __shared__ short k_lds[2][KVTILE * DHEAD];
-  __shared__ short vt_lds[2][DHEAD * (KVTILE + VT_PAD)];
-  __shared__ short p_lds[4][16 * (KVTILE + P_PAD)];
+  __shared__ short k_lds[KVTILE * DHEAD];            // swizzled
+  __shared__ short vt_lds[DHEAD * (KVTILE + VT_PAD)];  // transposed
+  __shared__ short p_lds[4][16 * (KVTILE + P_PAD)];    // per-wave P
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -91,6 +89,7 @@ __global__ void flash_fwd_kernel(
     }
   }
 
+  // online softmax state: rows wave_q + l4*4 + r  (r = 0..3)
   float m_run[4], l_run[4];
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -104,66 +103,53 @@ __global__ void flash_fwd_kernel(
   const int kv_end = causal ? min(S, qbase + QTILE) : S;
   const int n_tiles = (kv_end + KVTILE - 1) / KVTILE;
 
-  // V staging state: thread owns row = tid>>2 (0..63), cols (tid&3)*32..+31
-  const int v_row = threadIdx.x >> 2;
-  const int v_c0 = (threadIdx.x & 3) * 32;
-  bf16x8_t vreg[4];
-
-  // ---- staging helpers -------------------------------------------------
-  auto issue_k = [&](int buf, int kvbase) {
-    // wave w covers 4 KiB of linear K LDS (4 passes of 1 KiB)
-    for (int pass = 0; pass < 4; ++pass) {
-      int linear = (pass * 4 + wave) * 1024 + lane * 16;
-      int row = linear >> 8;
-      int colbyte = linear & 255;
-      int src_col = colbyte ^ ((row & 7) << 4);
-      int grow = kvbase + row;
-      int srow = grow < S ? grow : S - 1;  // clamped; masked via -inf later
-      const short* src = kp + (long long)srow * kv_row_stride +
-                         (src_col >> 1);
-      __builtin_amdgcn_global_load_lds(
-          reinterpret_cast<const unsigned int*>(src),
-          reinterpret_cast<unsigned int*>(
-              reinterpret_cast<char*>(k_lds[buf]) + (pass * 4 + wave) * 1024),
-          16, 0, 0);
-    }
-  };
-  auto issue_v = [&](int kvbase) {
-    int grow = kvbase + v_row;
-    int srow = grow < S ? grow : S - 1;
-    const short* vrow_p = vp + (long long)srow * kv_row_stride;
-#pragma unroll
-    for (int cc = 0; cc < 4; ++cc)
-      vreg[cc] = *reinterpret_cast<const bf16x8_t*>(vrow_p + v_c0 + cc * 8);
-  };
-  auto write_v = [&](int buf) {
-    char* vbase = reinterpret_cast<char*>(vt_lds[buf]);
-#pragma unroll
-    for (int cc = 0; cc < 4; ++cc)
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        *reinterpret_cast<short*>(vbase + vtswz(v_c0 + cc * 8 + j,
-                                                v_row * 2)) = vreg[cc][j];
-  };
-
-  // prologue: stage tile 0 into buffer 0
-  issue_k(0, 0);
-  issue_v(0);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  write_v(0);
-  __syncthreads();
-
   for (int t = 0; t < n_tiles; ++t) {
-    const int cur = t & 1;
     const int kvbase = t * KVTILE;
-    const bool have_next = t + 1 < n_tiles;
-    // T14 async-STAGE: issue next tile's loads before this tile's math
-    if (have_next) {
-      issue_k(cur ^ 1, kvbase + KVTILE);
-      issue_v(kvbase + KVTILE);
+    // ---- stage K (async global_load_lds, pre-swizzled source) -----------
+    // wave w, pass p covers 1 KiB of linear K LDS: lane's dest is
+    // base + lane*16 (HW rule), so the SOURCE column carries the XOR
+    // (guide m173: swizzled layouts via pre-swizzled global address).
+    {
+      for (int pass = 0; pass < 4; ++pass) {
+        int linear = (pass * 4 + wave) * 1024 + lane * 16;
+        int row = linear >> 8;             // /256 bytes per row
+        int colbyte = linear & 255;
+        int src_col = colbyte ^ ((row & 7) << 4);
+        int grow = kvbase + row;
+        int srow = grow < S ? grow : S - 1;  // clamped; masked via -inf
+        const short* src = kp + (long long)srow * kv_row_stride +
+                           (src_col >> 1);
+        __builtin_amdgcn_global_load_lds(
+            reinterpret_cast<const unsigned int*>(src),
+            reinterpret_cast<unsigned int*>(
+                reinterpret_cast<char*>(k_lds) + (pass * 4 + wave) * 1024),
+            16, 0, 0);
+      }
+      // V transposed: coalesced row reads, XOR-spread scatter writes
+      int row = threadIdx.x >> 2;        // 0..63
+      int c0 = (threadIdx.x & 3) * 32;   // 0,32,64,96
+      int grow = kvbase + row;
+      const short* vrow = vp + (long long)grow * kv_row_stride;
+#pragma unroll
+      for (int cc = 0; cc < 4; ++cc) {
+        int col = c0 + cc * 8;
+        bf16x8_t vv8;
+        if (grow < S) {
+          vv8 = *reinterpret_cast<const bf16x8_t*>(vrow + col);
+        } else {
+          vv8 = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};
+        }
+        char* vbase = reinterpret_cast<char*>(vt_lds);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          *reinterpret_cast<short*>(vbase + vtswz(col + j, row * 2)) =
+              vv8[j];
+      }
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     }
+    __syncthreads();
 
-    // ---- S = Q @ K^T  (2 col-tiles of 16) -------------------------------
+    // ---- S = Q @ K^T  (4 col-tiles of 16) -------------------------------
     f32x4_t s_acc[4];
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -173,8 +159,7 @@ __global__ void flash_fwd_kernel(
 #pragma unroll
       for (int ks = 0; ks < 4; ++ks) {
         bf16x8_t bfrag = *reinterpret_cast<const bf16x8_t*>(
-            reinterpret_cast<char*>(k_lds[cur]) +
-            kswz(krow, (l4 * 8 + 32 * ks) * 2));
+            reinterpret_cast<char*>(k_lds) + kswz(krow, (l4 * 8 + 32 * ks) * 2));
         acc = MFMA_BF16_16x16x32(qfrag[ks], bfrag, acc, 0, 0, 0);
       }
       s_acc[ct] = acc;
@@ -182,6 +167,7 @@ __global__ void flash_fwd_kernel(
     __builtin_amdgcn_s_setprio(0);
 
     // ---- online softmax -------------------------------------------------
+    // lane holds S[row = l4*4 + r][col = l15 + 16*ct] (scaled below)
     const bool btile = (causal && (kvbase + KVTILE > qbase)) ||
                        (kvbase + KVTILE > S);
     float p_new[4][4];  // [ct][r]
@@ -200,6 +186,7 @@ __global__ void flash_fwd_kernel(
         s_acc[ct][r] = sv;
         rowmax = fmaxf(rowmax, sv);
       }
+      // reduce across the 16 lanes sharing these rows
 #pragma unroll
       for (int off = 8; off > 0; off >>= 1)
         rowmax = fmaxf(rowmax, __shfl_xor(rowmax, off, 64));
@@ -220,6 +207,7 @@ __global__ void flash_fwd_kernel(
       l_run[r] = l_run[r] * c + rowsum;
       m_run[r] = m_new;
     }
+    // rescale O
 #pragma unroll
     for (int dt = 0; dt < 8; ++dt)
 #pragma unroll
@@ -233,10 +221,12 @@ __global__ void flash_fwd_kernel(
       for (int r = 0; r < 4; ++r)
         pw[(l4 * 4 + r) * (KVTILE + P_PAD) + l15 + 16 * ct] =
             f2bf(p_new[ct][r]);
-    // wave-local P handoff: drain ds_writes before cross-lane ds_reads
+    // wave-local P handoff: drain the ds_writes before cross-lane ds_reads
+    // (no cross-wave barrier needed — each wave reads only its own region;
+    // the "memory" clobber pins ordering, guide §5 rule 18)
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
-    // ---- O += P @ V  (two K=32 slices) ----------------------------------
+    // ---- O += P @ V  ----------------------------------------------------
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
@@ -245,19 +235,13 @@ __global__ void flash_fwd_kernel(
 #pragma unroll
       for (int dt = 0; dt < 8; ++dt) {
         bf16x8_t bfrag = *reinterpret_cast<const bf16x8_t*>(
-            reinterpret_cast<char*>(vt_lds[cur]) +
+            reinterpret_cast<char*>(vt_lds) +
             vtswz(l15 + 16 * dt, (l4 * 8 + 32 * ks) * 2));
         o_acc[dt] = MFMA_BF16_16x16x32(afrag, bfrag, o_acc[dt], 0, 0, 0);
       }
     }
     __builtin_amdgcn_s_setprio(0);
-
-    if (have_next) {
-      // K gload + V reg loads must have landed before writes/next reads
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      write_v(cur ^ 1);
-    }
-    __syncthreads();
+    __syncthreads();  // before next tile's staging overwrites K/V
   }
 
   // ---- epilogue: O /= l, write out + LSE --------------------------------
