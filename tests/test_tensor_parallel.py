@@ -56,3 +56,33 @@ def _tp_llama():
 
 def test_tp_llama_2rank_logits_match():
     assert all(run_distributed(_tp_llama, world_size=2))
+
+
+def _domino():
+    import torch.distributed as tdist
+    from deepspeed_amd.comm import groups
+    from deepspeed_amd.module_inject.layers import LinearAllreduce
+    from deepspeed_amd.runtime.domino import DominoLinearAllreduce
+    groups.reset_groups()
+    world = tdist.get_world_size()
+    rank = tdist.get_rank()
+    g = groups.initialize_tensor_parallel(world)
+    torch.manual_seed(5)
+    lin = torch.nn.Linear(16, 8)
+    x = torch.randn(6, 16, requires_grad=True)
+    ref_mod = LinearAllreduce.from_linear(lin, g, rank, world)
+    ref = ref_mod(x)
+    dom = DominoLinearAllreduce.from_linear(lin, g, rank, world, n_chunks=3)
+    dom.train()
+    out = dom(x)
+    assert torch.allclose(out, ref, atol=1e-5), (out - ref).abs().max()
+    out.sum().backward()
+    x2 = x.detach().clone().requires_grad_(True)
+    ref_mod(x2).sum().backward()
+    assert torch.allclose(x.grad, x2.grad, atol=1e-5)
+    assert torch.allclose(dom.weight.grad, ref_mod.weight.grad, atol=1e-5)
+    return True
+
+
+def test_domino_row_parallel_2rank():
+    assert all(run_distributed(_domino, world_size=2))
