@@ -69,9 +69,13 @@ def _domino():
     g = groups.initialize_tensor_parallel(world)
     torch.manual_seed(5)
     lin = torch.nn.Linear(16, 8)
-    x = torch.randn(6, 16, requires_grad=True)
+    xf = torch.randn(6, 16)
+    # row-parallel input is column-sharded (output of a column-parallel op)
+    sh = 16 // world
+    x = xf[:, rank * sh:(rank + 1) * sh].clone().requires_grad_(True)
     ref_mod = LinearAllreduce.from_linear(lin, g, rank, world)
     ref = ref_mod(x)
+    assert torch.allclose(ref, lin(xf), atol=1e-5)
     dom = DominoLinearAllreduce.from_linear(lin, g, rank, world, n_chunks=3)
     dom.train()
     out = dom(x)
