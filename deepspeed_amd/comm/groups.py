@@ -227,3 +227,28 @@ def get_tensor_parallel_world_size():
 def get_tensor_parallel_rank():
     g = get_tensor_parallel_group()
     return dist.get_rank(g) if g is not None else 0
+
+
+# -- MiCS: shard groups (adjacent, intra-node) x replica groups (strided) ---
+
+def initialize_mics(shard_size):
+    """Returns (shard_group, replica_group) for MiCS-style ZeRO-3:
+    params shard over `shard_size` adjacent ranks (intra-node xGMI
+    all-gathers); gradients additionally average across replica groups."""
+    _ensure_dist()
+    world = dist.get_world_size()
+    assert world % shard_size == 0, f"world {world} % shard {shard_size}"
+    rank = dist.get_rank()
+    shard_group = None
+    replica_group = None
+    for start in range(0, world, shard_size):
+        ranks = list(range(start, start + shard_size))
+        grp = dist.new_group(ranks)
+        if rank in ranks:
+            shard_group = grp
+    for offset in range(shard_size):
+        ranks = list(range(offset, world, shard_size))
+        grp = dist.new_group(ranks)
+        if rank in ranks:
+            replica_group = grp
+    return shard_group, replica_group

@@ -69,6 +69,7 @@ class ZeroConfig(DSConfigModel):
     offload_param: Optional[OffloadParamConfig] = None
     offload_optimizer: Optional[OffloadOptimizerConfig] = None
     zero_hpz_partition_size: int = 1
+    mics_shard_size: int = -1
     round_robin_gradients: bool = False
     ignore_unused_parameters: bool = True
 

@@ -251,11 +251,18 @@ class DeepSpeedEngine(torch.nn.Module):
 
         if stage == 3:
             from .zero.stage3 import ZeroStage3Optimizer
+            shard_group, replica_group = self.dp_group, None
+            if zc.mics_shard_size > 0:
+                shard_group, replica_group = groups.initialize_mics(
+                    zc.mics_shard_size)
+                log_dist(f"MiCS: shard_size={zc.mics_shard_size}, "
+                         "grads average across replica groups", ranks=[0])
             self.optimizer = ZeroStage3Optimizer(
                 basic,
                 module=self.module,
                 engine=self,
-                dp_process_group=self.dp_group,
+                dp_process_group=shard_group,
+                replica_group=replica_group,
                 reduce_bucket_size=zc.reduce_bucket_size,
                 prefetch_bucket_size=zc.prefetch_bucket_size,
                 param_persistence_threshold=zc.param_persistence_threshold,

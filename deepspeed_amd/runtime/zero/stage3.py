@@ -127,10 +127,16 @@ class ZeroStage3Optimizer:
                  dynamic_loss_scale=False,
                  dynamic_loss_args=None,
                  dtype=torch.bfloat16,
-                 gradient_accumulation_steps=1):
+                 gradient_accumulation_steps=1,
+                 replica_group=None):
         self.optimizer = init_optimizer
         self.module = module
         self.dp_group = dp_process_group
+        # MiCS: params shard over dp_group (sub-world); grads additionally
+        # average across replica_group (ref runtime/zero/mics.py)
+        self.replica_group = replica_group
+        self.replica_world = (dist.get_world_size(replica_group)
+                              if replica_group is not None else 1)
         self.world = dist.get_world_size(self.dp_group)
         self.rank = dist.get_rank(self.dp_group)
         self.dtype = dtype
@@ -191,13 +197,16 @@ class ZeroStage3Optimizer:
     def _shard_module_params(self):
         """Convert any unsharded params; move buffers to device."""
         params = list(self.module.parameters())
-        src = (dist.get_global_rank(self.dp_group, 0)
-               if self.dp_group is not None and self.world > 1 else 0)
+        bcast_group = None if self.replica_world > 1 else self.dp_group
+        bcast_world = dist.get_world_size(bcast_group) \
+            if dist.is_initialized() else 1
+        src = (dist.get_global_rank(bcast_group, 0)
+               if bcast_group is not None and bcast_world > 1 else 0)
         for p in params:
             if not is_zero_param(p):
                 p.data = p.data.to(self.device, self.dtype)
-                if self.world > 1:
-                    dist.broadcast(p.data, src, group=self.dp_group)
+                if bcast_world > 1:
+                    dist.broadcast(p.data, src, group=bcast_group)
                 convert_to_zero_param(p, self.dp_group, self.device,
                                       self.dtype, self.persist_threshold)
         for b in self.module.buffers():
@@ -418,7 +427,12 @@ class ZeroStage3Optimizer:
         if world == 1:
             for p in params:
                 sg = self.param_to_subgroup[p]
-                sg.accumulate_grad(p, p.grad.reshape(-1))
+                g = p.grad.reshape(-1)
+                if self.replica_world > 1:
+                    from .stage_1_and_2 import _avg_op
+                    op = _avg_op(self.replica_world, g)
+                    dist.all_reduce(g, op=op, group=self.replica_group)
+                sg.accumulate_grad(p, g)
                 p.grad = None
             return
         from .stage_1_and_2 import _avg_op
@@ -464,6 +478,11 @@ class ZeroStage3Optimizer:
                 op = _avg_op(world, inp)
                 dist.reduce_scatter_tensor(out, inp, op=op,
                                            group=self.dp_group)
+        if self.replica_world > 1:
+            from .stage_1_and_2 import _avg_op
+            for shard in shards:
+                op = _avg_op(self.replica_world, shard)
+                dist.all_reduce(shard, op=op, group=self.replica_group)
         for p, shard in zip(params, shards):
             sg = self.param_to_subgroup[p]
             sg.accumulate_grad(p, shard)
