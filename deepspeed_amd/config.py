@@ -145,6 +145,12 @@ class MonitorTensorBoardConfig(DSConfigModel):
     job_name: str = "DSAMDJob"
 
 
+class TorchAutocastConfig(DSConfigModel):
+    enabled: bool = False
+    dtype: str = "bfloat16"
+    lower_precision_safe_modules: List[str] = Field(default_factory=list)
+
+
 class UlyssesConfig(DSConfigModel):
     """Sequence-parallel (Ulysses) settings."""
     sequence_parallel_size: int = 1
@@ -207,6 +213,8 @@ class DeepSpeedConfig:
         self.csv_monitor = MonitorCSVConfig(**config.get("csv_monitor", {}))
         self.tensorboard = MonitorTensorBoardConfig(**config.get("tensorboard", {}))
         self.ulysses = UlyssesConfig(**config.get("sequence_parallel", {}))
+        self.torch_autocast = TorchAutocastConfig(
+            **config.get("torch_autocast", {}))
         self.pipeline = PipelineConfig(**config.get("pipeline", {}))
         self.moe = MoEConfig(**config.get("moe", {}))
         self.data_types = config.get("data_types", {})

@@ -348,13 +348,23 @@ class DeepSpeedEngine(torch.nn.Module):
     def set_gradient_accumulation_boundary(self, is_boundary):
         self._is_gradient_accumulation_boundary = is_boundary
 
+    def torch_autocast_enabled(self):
+        return self._config.torch_autocast.enabled
+
     def forward(self, *inputs, **kwargs):
         if self.flops_profiler is not None and \
                 self.global_steps == self._config.flops_profiler.profile_step:
             self.flops_profiler.start_profile(ignore_list=None)
         if self.wall_clock_breakdown():
             self.timers("forward").start()
-        loss = self.module(*inputs, **kwargs)
+        if self.torch_autocast_enabled():
+            ac_dtype = (torch.bfloat16 if "bf" in
+                        self._config.torch_autocast.dtype else torch.float16)
+            dev = "cuda" if torch.cuda.is_available() else "cpu"
+            with torch.autocast(device_type=dev, dtype=ac_dtype):
+                loss = self.module(*inputs, **kwargs)
+        else:
+            loss = self.module(*inputs, **kwargs)
         if self.wall_clock_breakdown():
             self.timers("forward").stop()
         return loss

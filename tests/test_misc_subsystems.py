@@ -99,3 +99,28 @@ def test_onebit_adam_smoke():
         losses.append(loss.item())
     assert losses[-1] < losses[0]
     assert opt.adam_freeze_key  # entered compressed stage
+
+
+def test_torch_autocast_engine():
+    import os
+    import deepspeed_amd
+    from tests.simple_model import SimpleModel
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    import torch.distributed as td
+    if not td.is_initialized():
+        os.environ.update(RANK="0", WORLD_SIZE="1", MASTER_PORT="29539")
+        td.init_process_group("gloo", rank=0, world_size=1)
+    torch.manual_seed(0)
+    model = SimpleModel(32)
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 4,
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+        "torch_autocast": {"enabled": True, "dtype": "bfloat16"},
+    })
+    x = torch.randn(4, 32)
+    y = torch.randn(4, 32)
+    loss = engine(x, y)
+    engine.backward(loss)
+    engine.step()
+    assert torch.isfinite(loss)
+    engine.destroy()
