@@ -176,6 +176,16 @@ class ZeroStage3Optimizer:
         self._install_module_hooks()
         self._install_grad_hooks()
 
+        # dedicated HIP streams: all-gathers (prefetch/fetch) and grad
+        # reduce-scatters overlap backward/forward compute on the default
+        # stream (ref stage3.py reduce_and_partition_stream /
+        # __allgather_stream)
+        use_streams = torch.cuda.is_available() and self.overlap_comm \
+            and self.world > 1
+        self.ag_stream = torch.cuda.Stream() if use_streams else None
+        self.rs_stream = torch.cuda.Stream() if use_streams else None
+        self._rs_refs = []  # tensors owned by in-flight reduce work
+
         # reduction state
         self._ipg_params = []
         self._ipg_numel = 0
@@ -359,7 +369,8 @@ class ZeroStage3Optimizer:
         need = [p for p in params
                 if p.ds_status == ZeroParamStatus.NOT_AVAILABLE]
         if need:
-            all_gather_params(need, self.dp_group).wait()
+            all_gather_params(need, self.dp_group,
+                              stream=self.ag_stream).wait()
         # publish data for params gathered by a shared handle
         for p in params:
             assert p.ds_status == ZeroParamStatus.AVAILABLE, \
@@ -384,7 +395,8 @@ class ZeroStage3Optimizer:
             if not need:
                 continue
             budget -= sum(p.ds_numel for p in need)
-            self._inflight[mod] = all_gather_params(need, self.dp_group)
+            self._inflight[mod] = all_gather_params(need, self.dp_group,
+                                                    stream=self.ag_stream)
             launched += 1
             if launched >= 8:
                 break
@@ -421,6 +433,18 @@ class ZeroStage3Optimizer:
         self._ipg_params = []
         self._ipg_numel = 0
         world = self.world
+        if self.rs_stream is not None:
+            # grads were produced on the default stream
+            self.rs_stream.wait_stream(torch.cuda.current_stream())
+            for p in params:
+                if p.grad is not None:
+                    p.grad.record_stream(self.rs_stream)
+            with torch.cuda.stream(self.rs_stream):
+                self._flush_ipg_body(params, world)
+            return
+        self._flush_ipg_body(params, world)
+
+    def _flush_ipg_body(self, params, world):
         use_coalescing = (torch.cuda.is_available() and world > 1
                           and len(params) > 1)
         shards = []
@@ -533,9 +557,17 @@ class ZeroStage3Optimizer:
                 combined = scale * clip
         return combined
 
+    def _sync_comm_streams(self):
+        if self.rs_stream is not None:
+            torch.cuda.current_stream().wait_stream(self.rs_stream)
+        if self.ag_stream is not None:
+            torch.cuda.current_stream().wait_stream(self.ag_stream)
+        self._rs_refs = []
+
     def step(self, closure=None):
         assert closure is None
         self._flush_ipg()
+        self._sync_comm_streams()
 
         if self.dtype == torch.float16:
             self.overflow = self.has_overflow()

@@ -97,11 +97,12 @@ def free_param(p):
 class AllGatherHandle:
     """Waits on an in-flight coalesced gather and publishes p.data."""
 
-    def __init__(self, params, works, buffers, group):
+    def __init__(self, params, works, buffers, group, stream=None):
         self.params = params
         self.works = works  # list of work objs or a coalescing-manager
         self.buffers = buffers
         self.group = group
+        self.stream = stream  # side comm stream the gather was enqueued on
         self.complete = False
 
     def wait(self):
@@ -110,6 +111,14 @@ class AllGatherHandle:
         for w in self.works:
             if w is not None:
                 w.wait()
+        if self.stream is not None:
+            # stream-level dependency: compute waits the allgather stream,
+            # host does not block; buffers were allocated on the side
+            # stream, so tell the allocator about the compute-stream use.
+            cur = torch.cuda.current_stream()
+            cur.wait_stream(self.stream)
+            for buf in self.buffers:
+                buf.record_stream(cur)
         for p, buf in zip(self.params, self.buffers):
             p.data = buf.narrow(0, 0, p.ds_numel).view(p.ds_shape)
             p.ds_full_buffer = buf
@@ -117,11 +126,14 @@ class AllGatherHandle:
         self.complete = True
 
 
-def all_gather_params(params, dp_group, async_op=True):
+def all_gather_params(params, dp_group, async_op=True, stream=None):
     """Launch coalesced all-gathers for NOT_AVAILABLE params.
 
     Returns an AllGatherHandle (already complete when nothing to do).
-    Must be called identically on all ranks of the group.
+    Must be called identically on all ranks of the group. When `stream` is
+    given, the collectives enqueue on that side stream (overlap with
+    compute on the default stream); handle.wait() installs the
+    stream-order dependency.
     """
     todo = [p for p in params if p.ds_status == ZeroParamStatus.NOT_AVAILABLE]
     for p in todo:
@@ -137,6 +149,17 @@ def all_gather_params(params, dp_group, async_op=True):
         h = AllGatherHandle(todo, [], buffers, dp_group)
         h.wait()
         return h
+    if stream is not None:
+        # side stream must see the up-to-date shards
+        stream.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(stream):
+            return _launch_gathers(todo, dp_group, world, async_op, stream)
+    return _launch_gathers(todo, dp_group, world, async_op, None)
+
+
+def _launch_gathers(todo, dp_group, world, async_op, stream):
+    buffers = []
+    works = []
     use_coalescing = _supports_coalescing(dp_group) and len(todo) > 1
     if use_coalescing:
         try:
@@ -166,7 +189,7 @@ def all_gather_params(params, dp_group, async_op=True):
                                             async_op=async_op)
             buffers.append(buf)
             works.append(w)
-    handle = AllGatherHandle(todo, works, buffers, dp_group)
+    handle = AllGatherHandle(todo, works, buffers, dp_group, stream=stream)
     if not async_op:
         handle.wait()
     return handle
