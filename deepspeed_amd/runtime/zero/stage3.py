@@ -180,8 +180,10 @@ class ZeroStage3Optimizer:
         # reduce-scatters overlap backward/forward compute on the default
         # stream (ref stage3.py reduce_and_partition_stream /
         # __allgather_stream)
+        import os as _os
         use_streams = torch.cuda.is_available() and self.overlap_comm \
-            and self.world > 1
+            and (self.world > 1
+                 or _os.environ.get("DSAMD_FORCE_STREAMS") == "1")
         self.ag_stream = torch.cuda.Stream() if use_streams else None
         self.rs_stream = torch.cuda.Stream() if use_streams else None
         self._rs_refs = []  # tensors owned by in-flight reduce work
