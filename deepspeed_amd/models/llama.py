@@ -94,6 +94,63 @@ class LlamaAttention(nn.Module):
         return self.o_proj(o.reshape(B, S, -1))
 
 
+class StaticKVCache:
+    """Graph-capturable cache: device-side position, static shapes.
+
+    All updates are device-tensor indexed (index_copy_/index_fill_), so a
+    decode step has no host-dependent control flow and can be captured in a
+    hipGraph (torch.cuda.CUDAGraph on ROCm).
+    """
+
+    def __init__(self, batch, max_seq, n_kv, head_dim, dtype, device):
+        self.k = torch.zeros(batch, max_seq, n_kv, head_dim, dtype=dtype,
+                             device=device)
+        self.v = torch.zeros_like(self.k)
+        self.max_seq = max_seq
+
+    def prefill(self, k, v):
+        S = k.shape[1]
+        self.k[:, :S] = k
+        self.v[:, :S] = v
+
+    def decode_update(self, k1, v1, pos_idx):
+        """k1/v1 [B,1,Hk,D]; pos_idx int64 device scalar-tensor [1]."""
+        self.k.index_copy_(1, pos_idx, k1)
+        self.v.index_copy_(1, pos_idx, v1)
+        return self.k, self.v
+
+
+def llama_decode_step(model, ids, caches, pos_idx, attn_mask, cos_t, sin_t):
+    """One static-shape decode step (graph-capturable).
+
+    ids [B,1] int64; pos_idx [1] int64 device; attn_mask [1,1,1,max_seq]
+    additive; cos_t/sin_t: full rope tables on device.
+    """
+    core = model.model
+    x = core.embed_tokens(ids)
+    cos = cos_t.index_select(0, pos_idx)
+    sin = sin_t.index_select(0, pos_idx)
+    for layer, cache in zip(core.layers, caches):
+        attn = layer.self_attn
+        h = layer.input_layernorm(x)
+        B = h.shape[0]
+        d = attn.cfg.head_dim
+        q = attn.q_proj(h).view(B, 1, -1, d)
+        k = attn.k_proj(h).view(B, 1, -1, d)
+        v = attn.v_proj(h).view(B, 1, -1, d)
+        q = apply_rope(q, cos, sin)
+        k = apply_rope(k, cos, sin)
+        kf, vf = cache.decode_update(k, v, pos_idx)
+        o = F.scaled_dot_product_attention(
+            q.transpose(1, 2), kf.transpose(1, 2), vf.transpose(1, 2),
+            attn_mask=attn_mask,
+            enable_gqa=(kf.shape[2] != q.shape[2]))
+        x = x + attn.o_proj(o.transpose(1, 2).reshape(B, 1, -1))
+        x = x + layer.mlp(layer.post_attention_layernorm(x))
+    x = core.norm(x)
+    return model.lm_head(x)
+
+
 def enable_ulysses(model, sp_group=None):
     """Route every attention through Ulysses head-scatter all-to-all."""
     from ..sequence.layer import DistributedAttention

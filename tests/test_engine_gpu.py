@@ -108,3 +108,20 @@ def test_zero3_cpu_offload_gpu():
         losses.append(loss.item())
     assert losses[-1] < losses[0], losses
     engine.destroy()
+
+
+def test_hipgraph_decode_matches_eager():
+    _init_env()
+    import deepspeed_amd
+    from deepspeed_amd.models.llama import LLAMA_CONFIGS, LlamaForCausalLM
+    cfg = LLAMA_CONFIGS["llama-tiny"]
+    torch.manual_seed(0)
+    with torch.device("cuda:0"):
+        model = LlamaForCausalLM(cfg)
+    engine = deepspeed_amd.init_inference(model, config={})
+    ids = torch.randint(0, cfg.vocab_size, (2, 12), device="cuda:0")
+    eager = engine.generate(ids, max_new_tokens=10)
+    graphed = engine.generate_hipgraph(ids, max_new_tokens=10)
+    assert graphed.shape == eager.shape
+    match = (graphed == eager).float().mean().item()
+    assert match > 0.95, f"token match only {match}: {graphed} vs {eager}"
