@@ -124,3 +124,36 @@ def test_torch_autocast_engine():
     engine.step()
     assert torch.isfinite(loss)
     engine.destroy()
+
+
+def test_fp16_overflow_skips_step_and_rescales():
+    import os
+    import deepspeed_amd
+    from tests.simple_model import SimpleModel
+    import torch.distributed as td
+    if not td.is_initialized():
+        os.environ.update(RANK="0", WORLD_SIZE="1",
+                          MASTER_ADDR="127.0.0.1", MASTER_PORT="29540")
+        td.init_process_group("gloo", rank=0, world_size=1)
+    torch.manual_seed(0)
+    model = SimpleModel(32)
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 4,
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+        "fp16": {"enabled": True, "initial_scale_power": 4},
+    })
+    opt = engine.optimizer
+    scale0 = opt.loss_scaler.cur_scale
+    x = torch.randn(4, 32, dtype=torch.float16)
+    y = torch.randn(4, 32, dtype=torch.float16)
+    loss = engine(x, y)
+    engine.backward(loss)
+    # poison a grad shard -> overflow detected, step skipped, scale halves
+    opt.buckets[0].grad32[0] = float("inf")
+    before = opt.buckets[0].master32.detach().clone()
+    engine.step()
+    assert opt.overflow
+    assert opt.loss_scaler.cur_scale < scale0 or \
+        opt.loss_scaler.cur_hysteresis < opt.loss_scaler.delayed_shift
+    assert torch.equal(before, opt.buckets[0].master32.detach())
+    engine.destroy()
