@@ -120,6 +120,18 @@ class StaticKVCache:
         return self.k, self.v
 
 
+def _linear_1t(lin, x):
+    """Decode-time linear: skinny-GEMV HIP kernel when applicable."""
+    if x.is_cuda and x.dtype == torch.bfloat16:
+        from ..ops.loader import get_ext
+        ext = get_ext(required=False)
+        if ext is not None:
+            B = x.numel() // x.shape[-1]
+            y = ext.gemv_bf16(lin.weight, x.reshape(B, -1).contiguous())
+            return y.reshape(*x.shape[:-1], lin.weight.shape[0])
+    return lin(x)
+
+
 def llama_decode_step(model, ids, caches, pos_idx, attn_mask, cos_t, sin_t):
     """One static-shape decode step (graph-capturable).
 
@@ -135,9 +147,9 @@ def llama_decode_step(model, ids, caches, pos_idx, attn_mask, cos_t, sin_t):
         h = layer.input_layernorm(x)
         B = h.shape[0]
         d = attn.cfg.head_dim
-        q = attn.q_proj(h).view(B, 1, -1, d)
-        k = attn.k_proj(h).view(B, 1, -1, d)
-        v = attn.v_proj(h).view(B, 1, -1, d)
+        q = _linear_1t(attn.q_proj, h).view(B, 1, -1, d)
+        k = _linear_1t(attn.k_proj, h).view(B, 1, -1, d)
+        v = _linear_1t(attn.v_proj, h).view(B, 1, -1, d)
         q = apply_rope(q, cos, sin)
         k = apply_rope(k, cos, sin)
         kf, vf = cache.decode_update(k, v, pos_idx)
@@ -145,10 +157,14 @@ def llama_decode_step(model, ids, caches, pos_idx, attn_mask, cos_t, sin_t):
             q.transpose(1, 2), kf.transpose(1, 2), vf.transpose(1, 2),
             attn_mask=attn_mask,
             enable_gqa=(kf.shape[2] != q.shape[2]))
-        x = x + attn.o_proj(o.transpose(1, 2).reshape(B, 1, -1))
-        x = x + layer.mlp(layer.post_attention_layernorm(x))
+        x = x + _linear_1t(attn.o_proj, o.transpose(1, 2).reshape(B, 1, -1))
+        h2 = layer.post_attention_layernorm(x)
+        mlp = layer.mlp
+        x = x + _linear_1t(mlp.down_proj,
+                           swiglu(_linear_1t(mlp.gate_proj, h2),
+                                  _linear_1t(mlp.up_proj, h2)))
     x = core.norm(x)
-    return model.lm_head(x)
+    return _linear_1t(model.lm_head, x)
 
 
 def enable_ulysses(model, sp_group=None):

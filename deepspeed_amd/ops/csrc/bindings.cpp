@@ -45,6 +45,7 @@ at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets,
                              long ignore_index);
 
 void bind_aio(py::module_& m);
+at::Tensor gemv_bf16(at::Tensor W, at::Tensor x);
 std::vector<at::Tensor> quantize_int8(at::Tensor x, long group_size);
 at::Tensor dequantize_int8(at::Tensor q, at::Tensor scales, long group_size);
 std::vector<at::Tensor> quantize_fp8(at::Tensor x, long group_size);
@@ -62,6 +63,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("accum_bf16_to_f32", &accum_bf16_to_f32, py::arg("dst"),
         py::arg("src"), py::arg("scale") = 1.0);
   m.def("l2norm_sq", &l2norm_sq);
+  m.def("gemv_bf16", &gemv_bf16);
   m.def("quantize_int8", &quantize_int8);
   m.def("dequantize_int8", &dequantize_int8);
   m.def("quantize_fp8", &quantize_fp8);
