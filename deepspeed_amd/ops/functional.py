@@ -12,7 +12,7 @@ from .loader import get_ext
 class _RMSNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, eps):
-        if x.is_cuda:
+        if x.is_cuda and x.dtype == torch.bfloat16:
             ext = get_ext(required=True)
             y, rstd = ext.rmsnorm_fwd(x.contiguous(), weight.contiguous(), eps)
         else:
@@ -26,7 +26,7 @@ class _RMSNormFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, weight, rstd = ctx.saved_tensors
-        if x.is_cuda:
+        if x.is_cuda and x.dtype == torch.bfloat16:
             ext = get_ext(required=True)
             dx, dw = ext.rmsnorm_bwd(dy.contiguous(), x, weight, rstd)
         else:
@@ -50,7 +50,7 @@ def rms_norm(x, weight, eps=1e-5):
 class _LayerNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, eps):
-        if x.is_cuda:
+        if x.is_cuda and x.dtype == torch.bfloat16:
             ext = get_ext(required=True)
             y, mean, rstd = ext.layernorm_fwd(x.contiguous(),
                                               weight.contiguous(),
@@ -74,7 +74,7 @@ class _LayerNormFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         x, weight, mean, rstd = ctx.saved_tensors
-        if x.is_cuda:
+        if x.is_cuda and x.dtype == torch.bfloat16:
             ext = get_ext(required=True)
             dx, dw, db = ext.layernorm_bwd(dy.contiguous(), x, weight, mean,
                                            rstd)
@@ -109,7 +109,7 @@ class _RoPEFn(torch.autograd.Function):
         cos = cos.float()
         sin = sin.float()
         ctx.save_for_backward(cos, sin)
-        if t.is_cuda:
+        if t.is_cuda and t.dtype == torch.bfloat16:
             ext = get_ext(required=True)
             out = torch.empty_like(t)
             ext.rope(out, t.contiguous(), cos, sin, 0, False)
@@ -120,7 +120,7 @@ class _RoPEFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         cos, sin = ctx.saved_tensors
-        if dy.is_cuda:
+        if dy.is_cuda and dy.dtype == torch.bfloat16:
             ext = get_ext(required=True)
             dx = torch.empty_like(dy)
             ext.rope(dx, dy.contiguous(), cos, sin, 0, True)
@@ -162,7 +162,7 @@ class _SwiGLUFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, g, u):
         ctx.save_for_backward(g, u)
-        if g.is_cuda:
+        if g.is_cuda and g.dtype == torch.bfloat16:
             ext = get_ext(required=True)
             return ext.swiglu_fwd(g.contiguous(), u.contiguous())
         g32 = g.float()
@@ -171,7 +171,7 @@ class _SwiGLUFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy):
         g, u = ctx.saved_tensors
-        if g.is_cuda:
+        if g.is_cuda and g.dtype == torch.bfloat16:
             ext = get_ext(required=True)
             dg, du = ext.swiglu_bwd(dy.contiguous(), g.contiguous(),
                                     u.contiguous())
@@ -194,7 +194,7 @@ class _FusedCrossEntropyFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, logits, targets, ignore_index):
-        if logits.is_cuda:
+        if logits.is_cuda and logits.dtype == torch.bfloat16:
             ext = get_ext(required=True)
             loss, lse = ext.cross_entropy_fwd(logits.contiguous(), targets,
                                               ignore_index)
@@ -217,7 +217,7 @@ class _FusedCrossEntropyFn(torch.autograd.Function):
         logits, targets, lse, n_valid = ctx.saved_tensors
         scale = (dloss.float() / n_valid.float())
         dloss_rows = scale.expand(logits.shape[0]).contiguous()
-        if logits.is_cuda:
+        if logits.is_cuda and logits.dtype == torch.bfloat16:
             ext = get_ext(required=True)
             dlogits = ext.cross_entropy_bwd(logits, targets, lse, dloss_rows,
                                             ctx.ignore_index)
