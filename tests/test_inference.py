@@ -18,7 +18,7 @@ def test_kv_cache_matches_full_forward():
     engine = deepspeed_amd.init_inference(model,
                                           config={"dtype": torch.float32})
     torch.manual_seed(1)
-    ids = torch.randint(0, cfg.vocab_size, (2, 16))
+    ids = torch.randint(0, cfg.vocab_size, (2, 16)).to(engine.device)
     # full forward logits
     full = engine.forward(ids)
     # prefill 12 then decode 4 with cache
