@@ -182,3 +182,23 @@ def test_save_16bit_model_stage3():
         # full shapes restored
         assert all(v.numel() > 0 for v in sd.values())
         assert any("linears.0.weight" in k for k in sd)
+
+
+def test_fast_checkpoint_engine_roundtrip():
+    from deepspeed_amd.runtime.checkpoint_engine import (
+        FastCheckpointEngine, TorchCheckpointEngine, make_checkpoint_engine)
+    eng = make_checkpoint_engine("fast")
+    assert isinstance(eng, FastCheckpointEngine)
+    sd = {"module": {"w": torch.randn(17, 3),
+                     "b16": torch.randn(8).to(torch.bfloat16)},
+          "step": 7, "nested": [torch.arange(5), "tag", (1, 2)]}
+    with tempfile.TemporaryDirectory() as d:
+        path = os.path.join(d, "ckpt.meta")
+        eng.save(sd, path)
+        back = eng.load(path)
+    assert torch.equal(back["module"]["w"], sd["module"]["w"])
+    assert torch.equal(back["module"]["b16"], sd["module"]["b16"])
+    assert back["step"] == 7
+    assert torch.equal(back["nested"][0], sd["nested"][0])
+    assert back["nested"][1] == "tag"
+    assert back["nested"][2] == (1, 2)
