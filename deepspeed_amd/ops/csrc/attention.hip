@@ -22,7 +22,7 @@ using f32x4_t = __attribute__((ext_vector_type(4))) float;
 
 #define MFMA_BF16_16x16x32 __builtin_amdgcn_mfma_f32_16x16x32_bf16
 
-constexpr int QTILE = 64;   // q rows per block
+constexpr int QTILE = 128;  // q rows per block (8 waves x 16)
 constexpr int KVTILE = 64;  // kv rows per tile
 constexpr int DHEAD = 128;
 constexpr int KPAD = 0;            // K uses swizzle, no pad
@@ -42,7 +42,7 @@ DEV_INLINE int vtswz(int d, int kv_byte) {
   return d * ((KVTILE + VT_PAD) * 2) + (kv_byte ^ (((d >> 5) & 3) << 5));
 }
 
-__launch_bounds__(256, 2)
+__launch_bounds__(512, 2)
 __global__ void flash_fwd_kernel(
     const short* __restrict__ q,  // [B,S,Hq,D]
     const short* __restrict__ k,  // [B,S,Hk,D]
@@ -52,7 +52,7 @@ __global__ void flash_fwd_kernel(
     int B, int S, int Hq, int Hk, float scale, int causal) {
   __shared__ short k_lds[KVTILE * DHEAD];            // swizzled
   __shared__ short vt_lds[DHEAD * (KVTILE + VT_PAD)];  // transposed
-  __shared__ short p_lds[4][16 * (KVTILE + P_PAD)];    // per-wave P
+  __shared__ short p_lds[8][16 * (KVTILE + P_PAD)];    // per-wave P
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -110,8 +110,8 @@ __global__ void flash_fwd_kernel(
     // base + lane*16 (HW rule), so the SOURCE column carries the XOR
     // (guide m173: swizzled layouts via pre-swizzled global address).
     {
-      for (int pass = 0; pass < 4; ++pass) {
-        int linear = (pass * 4 + wave) * 1024 + lane * 16;
+      for (int pass = 0; pass < 2; ++pass) {
+        int linear = (pass * 8 + wave) * 1024 + lane * 16;
         int row = linear >> 8;             // /256 bytes per row
         int colbyte = linear & 255;
         int src_col = colbyte ^ ((row & 7) << 4);
@@ -122,16 +122,16 @@ __global__ void flash_fwd_kernel(
         __builtin_amdgcn_global_load_lds(
             reinterpret_cast<const unsigned int*>(src),
             reinterpret_cast<unsigned int*>(
-                reinterpret_cast<char*>(k_lds) + (pass * 4 + wave) * 1024),
+                reinterpret_cast<char*>(k_lds) + (pass * 8 + wave) * 1024),
             16, 0, 0);
       }
       // V transposed: coalesced row reads, XOR-spread scatter writes
-      int row = threadIdx.x >> 2;        // 0..63
-      int c0 = (threadIdx.x & 3) * 32;   // 0,32,64,96
+      int row = threadIdx.x >> 3;        // 0..63
+      int c0 = (threadIdx.x & 7) * 16;   // 0..112 step 16
       int grow = kvbase + row;
       const short* vrow = vp + (long long)grow * kv_row_stride;
 #pragma unroll
-      for (int cc = 0; cc < 4; ++cc) {
+      for (int cc = 0; cc < 2; ++cc) {
         int col = c0 + cc * 8;
         bf16x8_t vv8;
         if (grow < S) {
@@ -273,7 +273,7 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
   auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStream();
   dim3 grid((S + QTILE - 1) / QTILE, B * Hq);
-  hipLaunchKernelGGL(flash_fwd_kernel, grid, dim3(256), 0, stream.stream(),
+  hipLaunchKernelGGL(flash_fwd_kernel, grid, dim3(512), 0, stream.stream(),
                      reinterpret_cast<const short*>(q.data_ptr()),
                      reinterpret_cast<const short*>(k.data_ptr()),
                      reinterpret_cast<const short*>(v.data_ptr()),
