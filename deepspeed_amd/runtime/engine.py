@@ -522,6 +522,17 @@ class DeepSpeedEngine(torch.nn.Module):
                         for k, v in self.module.state_dict().items()}
         return None
 
+    def save_fp16_model(self, save_dir, save_filename="pytorch_model.bin"):
+        """Reference-compat alias of save_16bit_model."""
+        return self.save_16bit_model(save_dir, save_filename)
+
+    def set_lr(self, lr):
+        for g in self.optimizer.param_groups:
+            g["lr"] = lr
+
+    def get_loss_scale(self):
+        return getattr(self.optimizer, "loss_scale", 1.0)
+
     def module_state_dict(self, exclude_frozen_parameters=False):
         return self.module.state_dict()
 
