@@ -46,6 +46,8 @@ at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor targets,
 
 void bind_aio(py::module_& m);
 at::Tensor gemv_bf16(at::Tensor W, at::Tensor x);
+at::Tensor spatial_bias_add(at::Tensor a, at::Tensor bias,
+                            c10::optional<at::Tensor> other);
 std::vector<at::Tensor> quantize_int8(at::Tensor x, long group_size);
 at::Tensor dequantize_int8(at::Tensor q, at::Tensor scales, long group_size);
 std::vector<at::Tensor> quantize_fp8(at::Tensor x, long group_size);
@@ -64,6 +66,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("src"), py::arg("scale") = 1.0);
   m.def("l2norm_sq", &l2norm_sq);
   m.def("gemv_bf16", &gemv_bf16);
+  m.def("spatial_bias_add", &spatial_bias_add, py::arg("a"), py::arg("bias"),
+        py::arg("other") = c10::nullopt);
   m.def("quantize_int8", &quantize_int8);
   m.def("dequantize_int8", &dequantize_int8);
   m.def("quantize_fp8", &quantize_fp8);
