@@ -168,6 +168,17 @@ class ZeroStage3Optimizer:
                                             dynamic_loss_args)
         self.overflow = False
 
+        from ...comm import groups as grp
+        for p in module.parameters():
+            gn = getattr(p, "group_name", None)
+            if gn is not None and gn in grp.get_expert_parallel_group_dict():
+                if dist.get_world_size(
+                        grp.get_expert_parallel_group(gn)) > 1:
+                    raise NotImplementedError(
+                        "MoE expert parameters with ZeRO stage 3 + "
+                        "ep_size>1 are not supported yet — use stage 1/2 "
+                        "(expert-DP bucket groups) for expert-parallel "
+                        "models")
         self._shard_module_params()
         self._build_sub_groups()
         if self.nvme_swapper is not None:
@@ -373,7 +384,14 @@ class ZeroStage3Optimizer:
         if need:
             all_gather_params(need, self.dp_group,
                               stream=self.ag_stream).wait()
-        # publish data for params gathered by a shared handle
+        # tied params may be INFLIGHT under another module's handle
+        pending = [p for p in params
+                   if p.ds_status == ZeroParamStatus.INFLIGHT]
+        if pending:
+            for key, handle in list(self._inflight.items()):
+                if any(p in handle.params for p in pending):
+                    handle.wait()
+                    self._inflight.pop(key, None)
         for p in params:
             assert p.ds_status == ZeroParamStatus.AVAILABLE, \
                 f"param {p.ds_id} not available after fetch"
