@@ -341,9 +341,17 @@ class DeepSpeedEngine(torch.nn.Module):
 
     # ------------------------------------------------------------- train
     def is_gradient_accumulation_boundary(self):
+        """True when the just-completed micro-step closes a GAS window.
+
+        self.micro_steps counts COMPLETED backward passes (incremented at
+        the end of engine.backward), so the boundary is micro_steps % GAS
+        == 0 — evaluated between backward() and step().
+        """
         if self._is_gradient_accumulation_boundary is not None:
             return self._is_gradient_accumulation_boundary
-        return (self.micro_steps + 1) % self.gradient_accumulation_steps() == 0
+        if self.micro_steps == 0:
+            return False
+        return self.micro_steps % self.gradient_accumulation_steps() == 0
 
     def set_gradient_accumulation_boundary(self, is_boundary):
         self._is_gradient_accumulation_boundary = is_boundary

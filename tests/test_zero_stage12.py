@@ -82,3 +82,51 @@ def test_zero2_bf16_grad_accum():
 
 def test_zero1_fp16():
     _check(1, "fp16")
+
+
+def _gas_equivalence():
+    """gas=2/mb=2 must equal gas=1/mb=4 on the same sample stream."""
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+
+    def train(mb, gas):
+        groups.reset_groups()
+        torch.manual_seed(11)
+        model = SimpleModel(HIDDEN)
+        config = {
+            "train_micro_batch_size_per_gpu": mb,
+            "gradient_accumulation_steps": gas,
+            "optimizer": {"type": "AdamW", "params": {"lr": LR}},
+            "zero_optimization": {"stage": 2, "reduce_bucket_size": 2000},
+            "bf16": {"enabled": True},
+        }
+        engine, _, _, _ = deepspeed_amd.initialize(model=model,
+                                                   config=config)
+        data = make_batches(12, 4, HIDDEN, dtype=torch.bfloat16)
+        flat = [(x.reshape(-1, HIDDEN), y.reshape(-1, HIDDEN))
+                for x, y in data]
+        xs = torch.cat([x for x, _ in flat])
+        ys = torch.cat([y for _, y in flat])
+        i = 0
+        for s in range(3):
+            for g in range(gas):
+                x = xs[i:i + mb]
+                y = ys[i:i + mb]
+                i += mb
+                loss = engine(x, y)
+                engine.backward(loss)
+                engine.step()
+        out = [p.detach().float().clone() for p in model.parameters()]
+        engine.destroy()
+        return out
+
+    a = train(4, 1)
+    b = train(2, 2)
+    for pa, pb in zip(a, b):
+        assert torch.allclose(pa, pb, atol=3e-3), (pa - pb).abs().max()
+    return True
+
+
+def test_gas_equivalence_single_rank():
+    results = run_distributed(_gas_equivalence, world_size=1)
+    assert all(results)
