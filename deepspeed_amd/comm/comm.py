@@ -249,6 +249,12 @@ allgather_fn = all_gather_into_tensor
 reduce_scatter_fn = reduce_scatter_tensor
 
 
+@timed_op
+def inference_all_reduce(tensor, op=ReduceOp.SUM, group=None):
+    """TP all-reduce used by injected inference modules (ref comm.py:662)."""
+    return torch_dist.all_reduce(tensor, op=op, group=group)
+
+
 def log_summary(show_straggler=False):
     comms_logger.log_all()
 

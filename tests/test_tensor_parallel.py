@@ -90,3 +90,28 @@ def _domino():
 
 def test_domino_row_parallel_2rank():
     assert all(run_distributed(_domino, world_size=2))
+
+
+def _tp_mixtral():
+    import torch.distributed as tdist
+    from deepspeed_amd.comm import groups
+    from deepspeed_amd.models.mixtral import (MIXTRAL_CONFIGS,
+                                              MixtralForCausalLM)
+    from deepspeed_amd.module_inject.auto_tp import apply_tensor_parallel
+    groups.reset_groups()
+    world = tdist.get_world_size()
+    g = groups.initialize_tensor_parallel(world)
+    cfg = MIXTRAL_CONFIGS["mixtral-tiny"]
+    torch.manual_seed(3)
+    model = MixtralForCausalLM(cfg).float().eval()
+    torch.manual_seed(9)
+    ids = torch.randint(0, cfg.vocab_size, (1, 16))
+    ref = model(ids)
+    apply_tensor_parallel(model, g)  # experts' MLPs + attention sharded
+    out = model(ids)
+    assert torch.allclose(out, ref, atol=1e-4), (out - ref).abs().max()
+    return True
+
+
+def test_tp_mixtral_2rank_logits_match():
+    assert all(run_distributed(_tp_mixtral, world_size=2))
