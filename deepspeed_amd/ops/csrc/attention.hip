@@ -42,7 +42,7 @@ DEV_INLINE int vtswz(int d, int kv_byte) {
   return d * ((KVTILE + VT_PAD) * 2) + (kv_byte ^ (((d >> 5) & 3) << 5));
 }
 
-__launch_bounds__(512, 2)
+__launch_bounds__(512, 3)
 __global__ void flash_fwd_kernel(
     const short* __restrict__ q,  // [B,S,Hq,D]
     const short* __restrict__ k,  // [B,S,Hk,D]
