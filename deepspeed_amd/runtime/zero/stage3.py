@@ -395,19 +395,31 @@ class ZeroStage3Optimizer:
         for p in params:
             assert p.ds_status == ZeroParamStatus.AVAILABLE, \
                 f"param {p.ds_id} not available after fetch"
-        if forward and self._trace_complete:
-            self._prefetch(forward=True)
+        if self._trace_complete:
+            self._prefetch(forward=forward, current_mod=mod)
 
-    def _prefetch(self, forward=True):
-        """Launch lookahead all-gathers along the recorded trace."""
+    def _prefetch(self, forward=True, current_mod=None):
+        """Launch lookahead all-gathers along the recorded trace.
+
+        Forward walks the trace ahead of _trace_pos; backward walks it in
+        reverse starting just before the current module (backward re-fetch
+        order is the reverse of the forward trace).
+        """
         budget = self.prefetch_bucket_size
-        # advance trace_pos to current position lazily
-        pos = self._trace_pos
         n = len(self._trace)
         launched = 0
-        while pos < n and budget > 0:
+        if forward:
+            pos = self._trace_pos
+            step = 1
+        else:
+            if not hasattr(self, "_trace_index"):
+                self._trace_index = {id(m): i
+                                     for i, m in enumerate(self._trace)}
+            pos = self._trace_index.get(id(current_mod), 0) - 1
+            step = -1
+        while 0 <= pos < n and budget > 0:
             mod = self._trace[pos]
-            pos += 1
+            pos += step
             if mod in self._inflight:
                 continue
             need = [p for p in mod._ds_direct_params
@@ -420,7 +432,8 @@ class ZeroStage3Optimizer:
             launched += 1
             if launched >= 8:
                 break
-        self._trace_pos = min(pos, n)
+        if forward:
+            self._trace_pos = min(max(pos, 0), n)
 
     def release_sub_module(self, mod):
         for p in mod._ds_direct_params:
@@ -584,10 +597,21 @@ class ZeroStage3Optimizer:
             torch.cuda.current_stream().wait_stream(self.ag_stream)
         self._rs_refs = []
 
+    def _drain_inflight(self):
+        """Consume leftover prefetch handles (e.g. frozen-param modules)
+        so no pre-step gathered copy survives the optimizer update."""
+        for mod, h in list(self._inflight.items()):
+            h.wait()
+            for p in h.params:
+                if not p.ds_persist:
+                    free_param(p)
+        self._inflight.clear()
+
     def step(self, closure=None):
         assert closure is None
         self._flush_ipg()
         self._sync_comm_streams()
+        self._drain_inflight()
 
         if self.dtype == torch.float16:
             self.overflow = self.has_overflow()
