@@ -157,3 +157,18 @@ def test_fp16_overflow_skips_step_and_rescales():
         opt.loss_scaler.cur_hysteresis < opt.loss_scaler.delayed_shift
     assert torch.equal(before, opt.buckets[0].master32.detach())
     engine.destroy()
+
+
+def test_launcher_hostfile_parsing():
+    import tempfile
+    from deepspeed_amd.launcher.runner import (fetch_hostfile,
+                                               _filter_resources)
+    with tempfile.NamedTemporaryFile("w", suffix=".txt", delete=False) as f:
+        f.write("nodeA slots=8\nnodeB slots=8  # comment\n# full comment\n")
+        path = f.name
+    res = fetch_hostfile(path)
+    assert res == {"nodeA": 8, "nodeB": 8}
+    keep = _filter_resources(res, include="nodeA:0,1,2,3", exclude="")
+    assert keep == {"nodeA": [0, 1, 2, 3]}
+    keep = _filter_resources(res, include="", exclude="nodeB")
+    assert keep == {"nodeA": list(range(8))}
