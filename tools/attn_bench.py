@@ -9,7 +9,10 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
-B, S, Hq, Hk, D = 1, 4096, 32, 8, 128
+import os as _os
+B = int(_os.environ.get("AB_B", 1))
+S = int(_os.environ.get("AB_S", 4096))
+Hq, Hk, D = 32, 8, 128
 it = 10
 
 
