@@ -14,6 +14,10 @@ at::Tensor l2norm_sq(std::vector<at::Tensor> tensors);
 std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
                                        at::Tensor v, bool causal,
                                        double scale);
+std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
+                                       at::Tensor v, at::Tensor dout,
+                                       at::Tensor lse, at::Tensor drow,
+                                       bool causal, double scale);
 void cpu_adam_step(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
                    c10::optional<at::Tensor> out16, double lr, double beta1,
                    double beta2, double eps, long step, long adamw,
@@ -74,6 +78,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dequantize_fp8", &dequantize_fp8);
   m.def("flash_attn_fwd", &flash_attn_fwd, py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("causal") = true, py::arg("scale") = 0.0);
+  m.def("flash_attn_bwd", &flash_attn_bwd);
   m.def("cpu_adam_step", &cpu_adam_step, py::arg("p"), py::arg("g"),
         py::arg("m"), py::arg("v"), py::arg("out16") = c10::nullopt,
         py::arg("lr") = 1e-3, py::arg("beta1") = 0.9, py::arg("beta2") = 0.999,
