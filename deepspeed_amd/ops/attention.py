@@ -39,16 +39,8 @@ class _FlashAttnFn(torch.autograd.Function):
 
         if os.environ.get("DSAMD_FLASH_BWD", "1") == "1":
             ext = get_ext(required=True)
-            dout_c = dout.contiguous()
-            drow = (dout_c.float() * out.float()).sum(-1) \
-                .permute(0, 2, 1).contiguous()  # [B,Hq,S]
-            dq, dk_ph, dv_ph = ext.flash_attn_bwd(
-                q, k, v, dout_c, lse, drow, causal, scale)
-            if G > 1:
-                dk = dk_ph.view(B, S, Hk, G, D).sum(3).to(k.dtype)
-                dv = dv_ph.view(B, S, Hk, G, D).sum(3).to(v.dtype)
-            else:
-                dk, dv = dk_ph, dv_ph
+            dq, dk, dv = ext.flash_attn_bwd(
+                q, k, v, dout.contiguous(), out, lse, causal, scale)
             return dq, dk, dv, None, None
 
         # head-major views [B,H,S,D]

@@ -10,7 +10,9 @@
 // row=(lane>>4)*4+reg, col=lane&15; A-operand lane holds A[l15][l4*8+j];
 // row-major LDS tiles use the ((row&7)<<4) XOR swizzle with pre-swizzled
 // staging; transposed tiles use pad-8 rows + a 16B-granule XOR.
-// GQA: grads computed per q-head; the wrapper group-sums dK/dV.
+// GQA: pass 1 accumulates over the q-head group in-kernel (K/V fragments
+// stay in registers across the group), writing dK/dV at [B,S,Hk,D]
+// directly — no per-q-head buffers, no wrapper reduction.
 #include <torch/extension.h>
 
 #include "common.h"
@@ -23,6 +25,34 @@ using f32x4_t = __attribute__((ext_vector_type(4))) float;
 namespace bwd {
 
 constexpr int DHEAD = 128;
+
+// drow[b,h,s] = sum_d dout[b,s,h,d] * out[b,s,h,d]  (one wave per row)
+__global__ void drow_kernel(const short* __restrict__ dout,
+                            const short* __restrict__ out,
+                            float* __restrict__ drow, int S, int Hq,
+                            long long n_rows) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  long long row = (long long)blockIdx.x * 4 + wave;  // over B*S*Hq
+  if (row >= n_rows) return;
+  const long long nbsh = (long long)S * Hq;
+  const short* a = dout + row * DHEAD;
+  const short* b = out + row * DHEAD;
+  // 128 elems over 64 lanes: one 2-elem (4 B) load per lane
+  unsigned int au = reinterpret_cast<const unsigned int*>(a)[lane];
+  unsigned int bu = reinterpret_cast<const unsigned int*>(b)[lane];
+  float acc = bf2f((short)(au & 0xffff)) * bf2f((short)(bu & 0xffff)) +
+              bf2f((short)(au >> 16)) * bf2f((short)(bu >> 16));
+  acc = wave_reduce_sum(acc);
+  if (lane == 0) {
+    // row index [b, s, h] -> drow[b, h, s]
+    long long bb = row / nbsh;
+    long long rem = row % nbsh;
+    long long s = rem / Hq;
+    long long h = rem % Hq;
+    drow[(bb * Hq + h) * S + s] = acc;
+  }
+}
 
 // row-major [R][128] bf16 with ((row&7)<<4) XOR
 DEV_INLINE int rmswz(int row, int byte_off) {
@@ -49,8 +79,8 @@ __global__ void flash_bwd_dkv_kernel(
     const short* __restrict__ dout,  // [B,S,Hq,D]
     const float* __restrict__ lse,   // [B,Hq,S]
     const float* __restrict__ drow,  // [B,Hq,S]
-    short* __restrict__ dk_ph,       // [B,S,Hq,D] per-q-head
-    short* __restrict__ dv_ph,       // [B,S,Hq,D]
+    short* __restrict__ dk,          // [B,S,Hk,D]
+    short* __restrict__ dv,          // [B,S,Hk,D]
     int B, int S, int Hq, int Hk, float scale, int causal) {
   constexpr int QIT = 32;
   __shared__ short q_lds[QIT * DHEAD];               // row-major swz
@@ -66,21 +96,17 @@ __global__ void flash_bwd_dkv_kernel(
   const int l4 = lane >> 4;
 
   const int kv_block = blockIdx.x;      // 64 kv rows per block
-  const int bh = blockIdx.y;
-  const int b = bh / Hq;
-  const int h = bh % Hq;
-  const int hk = h / (Hq / Hk);
+  const int bh = blockIdx.y;            // over B * Hk (kv heads)
+  const int b = bh / Hk;
+  const int hk = bh % Hk;
+  const int G = Hq / Hk;
   const int kv_base = kv_block * 64;
   const int wave_kv = kv_base + wave * 16;
 
   const long long qrs = (long long)Hq * DHEAD;
   const long long kvrs = (long long)Hk * DHEAD;
-  const short* qp = q + ((long long)b * S) * qrs + h * DHEAD;
-  const short* dop = dout + ((long long)b * S) * qrs + h * DHEAD;
   const short* kp = k + ((long long)b * S) * kvrs + hk * DHEAD;
   const short* vp = v + ((long long)b * S) * kvrs + hk * DHEAD;
-  const float* lsep = lse + ((long long)b * Hq + h) * S;
-  const float* drp = drow + ((long long)b * Hq + h) * S;
 
   // K/V fragments in registers (A-operand layout, 4 d-slices)
   bf16x8_t kfrag[4], vfrag[4];
@@ -104,6 +130,13 @@ __global__ void flash_bwd_dkv_kernel(
   }
 
   const int q_start = causal ? (kv_base / QIT) * QIT : 0;
+  // accumulate over the GQA group: all q-heads sharing this kv head
+  for (int g = 0; g < G; ++g) {
+  const int h = hk * G + g;
+  const short* qp = q + ((long long)b * S) * qrs + h * DHEAD;
+  const short* dop = dout + ((long long)b * S) * qrs + h * DHEAD;
+  const float* lsep = lse + ((long long)b * Hq + h) * S;
+  const float* drp = drow + ((long long)b * Hq + h) * S;
   for (int qb = q_start; qb < S; qb += QIT) {
     // ---- stage Q/dO (row-major swizzled + transposed) ------------------
     {
@@ -199,14 +232,15 @@ __global__ void flash_bwd_dkv_kernel(
     __builtin_amdgcn_s_setprio(0);
     __syncthreads();
   }
+  }  // group loop
 
-  // ---- epilogue: write dK/dV (per-q-head buffers) -----------------------
+  // ---- epilogue: write dK/dV [B,S,Hk,D] ---------------------------------
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     int gkv = wave_kv + l4 * 4 + r;
     if (gkv >= S) continue;
-    short* dkr = dk_ph + ((long long)b * S + gkv) * qrs + h * DHEAD;
-    short* dvr = dv_ph + ((long long)b * S + gkv) * qrs + h * DHEAD;
+    short* dkr = dk + ((long long)b * S + gkv) * kvrs + hk * DHEAD;
+    short* dvr = dv + ((long long)b * S + gkv) * kvrs + hk * DHEAD;
 #pragma unroll
     for (int dt = 0; dt < 8; ++dt) {
       dkr[l15 + 16 * dt] = f2bf_(dk_acc[dt][r]);
@@ -383,19 +417,27 @@ __global__ void flash_bwd_dq_kernel(
 
 std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
                                        at::Tensor v, at::Tensor dout,
-                                       at::Tensor lse, at::Tensor drow,
+                                       at::Tensor out, at::Tensor lse,
                                        bool causal, double scale) {
   TORCH_CHECK(q.scalar_type() == at::kBFloat16);
   TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous() &&
-              dout.is_contiguous());
+              dout.is_contiguous() && out.is_contiguous());
   int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
   int Hk = k.size(2);
   TORCH_CHECK(D == 128, "flash_attn_bwd: head_dim must be 128");
   auto dq = at::empty_like(q);
-  auto dk_ph = at::empty_like(q);  // per-q-head; wrapper group-sums
-  auto dv_ph = at::empty_like(q);
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  auto drow = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStream();
-  dim3 grid1((S + 63) / 64, B * Hq);
+  long long rows = (long long)B * S * Hq;
+  dim3 gridd((rows + 3) / 4);
+  hipLaunchKernelGGL(bwd::drow_kernel, gridd, dim3(256), 0, stream.stream(),
+                     reinterpret_cast<const short*>(dout.data_ptr()),
+                     reinterpret_cast<const short*>(out.data_ptr()),
+                     drow.data_ptr<float>(), S, Hq, rows);
+  HIP_CHECK_KERNEL();
+  dim3 grid1((S + 63) / 64, B * Hk);
   hipLaunchKernelGGL(bwd::flash_bwd_dkv_kernel, grid1, dim3(256), 0,
                      stream.stream(),
                      reinterpret_cast<const short*>(q.data_ptr()),
@@ -403,8 +445,8 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
                      reinterpret_cast<const short*>(v.data_ptr()),
                      reinterpret_cast<const short*>(dout.data_ptr()),
                      lse.data_ptr<float>(), drow.data_ptr<float>(),
-                     reinterpret_cast<short*>(dk_ph.data_ptr()),
-                     reinterpret_cast<short*>(dv_ph.data_ptr()), B, S, Hq,
+                     reinterpret_cast<short*>(dk.data_ptr()),
+                     reinterpret_cast<short*>(dv.data_ptr()), B, S, Hq,
                      Hk, (float)scale, causal ? 1 : 0);
   HIP_CHECK_KERNEL();
   dim3 grid2((S + 63) / 64, B * Hq);
@@ -418,5 +460,5 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
                      reinterpret_cast<short*>(dq.data_ptr()), B, S, Hq, Hk,
                      (float)scale, causal ? 1 : 0);
   HIP_CHECK_KERNEL();
-  return {dq, dk_ph, dv_ph};
+  return {dq, dk, dv};
 }

@@ -16,7 +16,7 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
                                        double scale);
 std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
                                        at::Tensor v, at::Tensor dout,
-                                       at::Tensor lse, at::Tensor drow,
+                                       at::Tensor out, at::Tensor lse,
                                        bool causal, double scale);
 void cpu_adam_step(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
                    c10::optional<at::Tensor> out16, double lr, double beta1,
