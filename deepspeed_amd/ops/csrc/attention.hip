@@ -186,10 +186,8 @@ __global__ void flash_fwd_kernel(
         s_acc[ct][r] = sv;
         rowmax = fmaxf(rowmax, sv);
       }
-      // reduce across the 16 lanes sharing these rows
-#pragma unroll
-      for (int off = 8; off > 0; off >>= 1)
-        rowmax = fmaxf(rowmax, __shfl_xor(rowmax, off, 64));
+      // reduce across the 16 lanes sharing these rows (DPP, VALU-only)
+      rowmax = row16_reduce_max(rowmax);
       float m_new = fmaxf(m_run[r], rowmax);
       float c = (m_run[r] == -INFINITY) ? 0.f : __expf(m_run[r] - m_new);
       corr[r] = c;
@@ -201,9 +199,7 @@ __global__ void flash_fwd_kernel(
         p_new[ct][r] = p;
         rowsum += p;
       }
-#pragma unroll
-      for (int off = 8; off > 0; off >>= 1)
-        rowsum += __shfl_xor(rowsum, off, 64);
+      rowsum = row16_reduce_sum(rowsum);
       l_run[r] = l_run[r] * c + rowsum;
       m_run[r] = m_new;
     }
@@ -260,6 +256,293 @@ __global__ void flash_fwd_kernel(
   }
 }
 
+
+// ===========================================================================
+// v5: 8-wave 32x32 "swapped QK^T" forward (guide §B 8-warp ladder).
+//   - each wave owns 32 q rows (block = 256 q rows), KV tiles of 64
+//   - S^T = K·Q^T via mfma_f32_32x32x16_bf16: C col = q = lane&31, so the
+//     softmax row (over kv) is LANE-LOCAL: in-lane reduce + one half-swap
+//   - P stays in registers: cvt_pk to bf16 pairs + permlane32_swap build
+//     the PV A/B fragments directly — no P LDS staging, no cross-wave sync
+//   - PV computed as O^T = V^T·P^T so O's q is also lane-local and the
+//     online-softmax rescale is a scalar multiply
+//   - K staged via global_load_lds (pre-swizzled source), V transposed
+// C/D layout (32x32x16): col=lane&31, row=(reg&3)+8*(reg>>2)+4*(lane>>5).
+// A/B operand: row(col)=lane&31, k=(lane>>5)*8+j.
+// ===========================================================================
+
+using f32x16_t = __attribute__((ext_vector_type(16))) float;
+#define MFMA_BF16_32x32x16 __builtin_amdgcn_mfma_f32_32x32x16_bf16
+
+constexpr int QB5 = 32;    // q rows per wave
+constexpr int QT5 = 256;   // q rows per block (8 waves)
+constexpr int KT5 = 64;    // kv rows per tile
+
+// pack two f32 into a u32 of 2 bf16 (compiler emits v_cvt_pk_bf16_f32)
+DEV_INLINE unsigned int cvtpk2(float lo, float hi) {
+  unsigned short a = __builtin_bit_cast(unsigned short, (__bf16)lo);
+  unsigned short b = __builtin_bit_cast(unsigned short, (__bf16)hi);
+  return (unsigned int)a | ((unsigned int)b << 16);
+}
+
+__launch_bounds__(512, 2)
+__global__ void flash_fwd_v5_kernel(
+    const short* __restrict__ q,  // [B,S,Hq,D]
+    const short* __restrict__ k,  // [B,S,Hk,D]
+    const short* __restrict__ v,  // [B,S,Hk,D]
+    short* __restrict__ out,      // [B,S,Hq,D]
+    float* __restrict__ lse_out,  // [B,Hq,S]
+    int B, int S, int Hq, int Hk, float scale, int causal) {
+  // K double-buffered: tile t+1's global_load_lds issues during tile t's
+  // compute (async-STAGE, guide T14); V global loads land in registers one
+  // tile early and scatter to LDS after the barrier.
+  __shared__ short k_lds[2][KT5 * DHEAD];              // swizzled row-major
+  __shared__ short vt_lds[DHEAD * (KT5 + VT_PAD)];     // transposed
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int l31 = lane & 31;
+  const int hi = lane >> 5;  // half-wave: 0 or 1
+
+  const int qtile = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / Hq;
+  const int h = bh % Hq;
+  const int hk = h / (Hq / Hk);
+  const int qbase = qtile * QT5;
+  const int wave_q = qbase + wave * QB5;
+  const int my_q = wave_q + l31;  // this lane's q row (lane-local softmax)
+
+  const long long qrs = (long long)Hq * DHEAD;
+  const long long kvrs = (long long)Hk * DHEAD;
+  const short* qp = q + ((long long)b * S) * qrs + h * DHEAD;
+  const short* kp = k + ((long long)b * S) * kvrs + hk * DHEAD;
+  const short* vp = v + ((long long)b * S) * kvrs + hk * DHEAD;
+
+  // ---- Q fragments: qfrag[ks] = Q[my_q][16*ks + hi*8 .. +7] -------------
+  bf16x8_t qfrag[8];
+  {
+    int srow = my_q < S ? my_q : S - 1;
+    const short* src = qp + (long long)srow * qrs + hi * 8;
+#pragma unroll
+    for (int ks = 0; ks < 8; ++ks)
+      qfrag[ks] = *reinterpret_cast<const bf16x8_t*>(src + 16 * ks);
+  }
+
+  // softmax runs in the exp2 domain (v_exp_f32 is natively 2^x): S values
+  // are scaled by scale*log2(e) once, so every exp is a bare v_exp with no
+  // hidden *log2e multiply; LSE converts back with ln2 at the epilogue.
+  const float sc2 = scale * 1.44269504f;
+  float m_run = -INFINITY, l_run = 0.f;
+  f32x16_t ot[4];  // O^T accum: rows d = 32*mt + (reg&3)+8*(reg>>2)+4*hi
+#pragma unroll
+  for (int mt = 0; mt < 4; ++mt)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) ot[mt][r] = 0.f;
+
+  const int kv_end = causal ? min(S, qbase + QT5) : S;
+  const int n_tiles = (kv_end + KT5 - 1) / KT5;
+
+  // async-STAGE helpers -----------------------------------------------------
+  const int st_row = threadIdx.x >> 3;        // V staging: row this thread
+  const int st_c0 = (threadIdx.x & 7) * 16;   //            first col
+
+#define ISSUE_K(t_)                                                          \
+  {                                                                          \
+    int kvb_ = (t_)*KT5;                                                     \
+    short* kbuf_ = k_lds[(t_)&1];                                            \
+    for (int pass = 0; pass < 2; ++pass) {                                   \
+      int linear = (pass * 8 + wave) * 1024 + lane * 16;                     \
+      int row = linear >> 8;                                                 \
+      int colbyte = linear & 255;                                            \
+      int src_col = colbyte ^ ((row & 7) << 4);                              \
+      int grow = kvb_ + row;                                                 \
+      int srow = grow < S ? grow : S - 1;                                    \
+      const short* src = kp + (long long)srow * kvrs + (src_col >> 1);       \
+      __builtin_amdgcn_global_load_lds(                                      \
+          reinterpret_cast<const unsigned int*>(src),                        \
+          reinterpret_cast<unsigned int*>(reinterpret_cast<char*>(kbuf_) +   \
+                                          (pass * 8 + wave) * 1024),         \
+          16, 0, 0);                                                         \
+    }                                                                        \
+  }
+
+#define LOAD_V(t_, dst_)                                                     \
+  {                                                                          \
+    int grow = (t_)*KT5 + st_row;                                            \
+    const short* vrow = vp + (long long)grow * kvrs;                         \
+    if (grow < S) {                                                          \
+      dst_[0] = *reinterpret_cast<const bf16x8_t*>(vrow + st_c0);            \
+      dst_[1] = *reinterpret_cast<const bf16x8_t*>(vrow + st_c0 + 8);        \
+    } else {                                                                 \
+      dst_[0] = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};                            \
+      dst_[1] = bf16x8_t{0, 0, 0, 0, 0, 0, 0, 0};                            \
+    }                                                                        \
+  }
+
+  bf16x8_t vreg[2], vnext[2];
+  ISSUE_K(0);
+  LOAD_V(0, vreg);
+
+  for (int t = 0; t < n_tiles; ++t) {
+    const int kvbase = t * KT5;
+    const bool has_next = (t + 1) < n_tiles;
+    if (has_next) ISSUE_K(t + 1);  // lands in the other K buffer
+    // scatter this tile's V registers into LDS (the compiler's wait for
+    // vreg also retires this tile's older global_load_lds -> k_lds)
+    {
+      char* vbase = reinterpret_cast<char*>(vt_lds);
+#pragma unroll
+      for (int cc = 0; cc < 2; ++cc) {
+        int col = st_c0 + cc * 8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          *reinterpret_cast<short*>(vbase + vtswz(col + j, st_row * 2)) =
+              vreg[cc][j];
+      }
+    }
+    if (has_next) LOAD_V(t + 1, vnext);  // in flight under this tile's MFMAs
+    __syncthreads();
+
+    // ---- S^T = K Q^T: st[mt] rows kv = kvbase+32mt+(r&3)+8*(r>>2)+4hi ---
+    f32x16_t st[2];
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt) {
+      f32x16_t acc;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) acc[r] = 0.f;
+      int krow = 32 * mt + l31;
+#pragma unroll
+      for (int ks = 0; ks < 8; ++ks) {
+        bf16x8_t kf = *reinterpret_cast<const bf16x8_t*>(
+            reinterpret_cast<char*>(k_lds[t & 1]) +
+            kswz(krow, (16 * ks + hi * 8) * 2));
+        acc = MFMA_BF16_32x32x16(kf, qfrag[ks], acc, 0, 0, 0);
+      }
+      st[mt] = acc;
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- online softmax (lane-local q row) ------------------------------
+    // btile is wave-uniform: interior tiles take the cmp-free path
+    const bool btile = (causal && (kvbase + KT5 > qbase)) ||
+                       (kvbase + KT5 > S);
+    float tmax = -INFINITY;
+    if (btile) {
+#pragma unroll
+      for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          float sv = st[mt][r] * sc2;
+          int gkv = kvbase + 32 * mt + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          if ((causal && gkv > my_q) || gkv >= S) sv = -INFINITY;
+          st[mt][r] = sv;
+          tmax = fmaxf(tmax, sv);
+        }
+    } else {
+#pragma unroll
+      for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          float sv = st[mt][r] * sc2;
+          st[mt][r] = sv;
+          tmax = fmaxf(tmax, sv);
+        }
+    }
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+    // defer-max (guide T13): only rescale when the running max grew by more
+    // than 8/ln2 (P is then bounded by e^8, which f32 accum tolerates; LSE
+    // stays exact). Masked values rely on exp2(-inf)=0 — no per-value select.
+    if (!__all(tmax - m_run <= 11.5416f)) {
+      float m_new = fmaxf(m_run, tmax);
+      float c = (m_run == -INFINITY) ? 0.f : __builtin_amdgcn_exp2f(m_run - m_new);
+      l_run *= c;
+#pragma unroll
+      for (int mt = 0; mt < 4; ++mt)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) ot[mt][r] *= c;
+      m_run = m_new;
+    }
+    float psum = 0.f;
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        float p = __builtin_amdgcn_exp2f(st[mt][r] - m_run);
+        st[mt][r] = p;  // P overwrites S in-register
+        psum += p;
+      }
+    psum += __shfl_xor(psum, 32, 64);
+    l_run += psum;
+
+    // ---- build P fragments in-register ----------------------------------
+    // pfrag[ks] = P[q=l31][kv = kvbase + 16ks + hi*8 + j]; own C regs hold
+    // kv (r&3)+8*(r>>2)+4hi — the other half-wave holds the 4-offset rows.
+    // permlane32_swap(wA, wB): x = {lo: wA_lo, hi: wB_lo},
+    //                          y = {lo: wA_hi, hi: wB_hi}
+    // with wA = own rows +0..3 packed, wB = own rows +4..7 packed gives
+    // each half the partner words it needs (guide T12).
+    bf16x8_t pfrag[4];
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      int mt = ks >> 1, s = ks & 1;  // kv 16s..16s+15 of m-tile mt
+      // own regs 8s..8s+3 = rows 16s+4hi+0..3 ; 8s+4..8s+7 = +8
+      unsigned int wA0 = cvtpk2(st[mt][8 * s + 0], st[mt][8 * s + 1]);
+      unsigned int wA1 = cvtpk2(st[mt][8 * s + 2], st[mt][8 * s + 3]);
+      unsigned int wB0 = cvtpk2(st[mt][8 * s + 4], st[mt][8 * s + 5]);
+      unsigned int wB1 = cvtpk2(st[mt][8 * s + 6], st[mt][8 * s + 7]);
+      auto r0 = __builtin_amdgcn_permlane32_swap((int)wA0, (int)wB0,
+                                                 false, false);
+      auto r1 = __builtin_amdgcn_permlane32_swap((int)wA1, (int)wB1,
+                                                 false, false);
+      unsigned int w[4] = {(unsigned int)r0[0], (unsigned int)r1[0],
+                           (unsigned int)r0[1], (unsigned int)r1[1]};
+      pfrag[ks] = *reinterpret_cast<bf16x8_t*>(w);
+    }
+
+    // ---- O^T += V^T P^T  (4 d m-tiles x 4 kv k-slices) ------------------
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int mt = 0; mt < 4; ++mt) {
+      int drow = 32 * mt + l31;
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        bf16x8_t vf = *reinterpret_cast<const bf16x8_t*>(
+            reinterpret_cast<char*>(vt_lds) +
+            vtswz(drow, (16 * ks + hi * 8) * 2));
+        ot[mt] = MFMA_BF16_32x32x16(vf, pfrag[ks], ot[mt], 0, 0, 0);
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+    __syncthreads();
+    vreg[0] = vnext[0];
+    vreg[1] = vnext[1];
+  }
+#undef ISSUE_K
+#undef LOAD_V
+
+  // ---- epilogue: O^T lane holds d rows for its q col --------------------
+  if (my_q < S) {
+    float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
+    short* orow = out + ((long long)b * S + my_q) * qrs + h * DHEAD;
+#pragma unroll
+    for (int mt = 0; mt < 4; ++mt)
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        int d0 = 32 * mt + 8 * g + 4 * hi;  // 4 consecutive d
+        unsigned int w[2];
+        w[0] = cvtpk2(ot[mt][4 * g + 0] * inv_l, ot[mt][4 * g + 1] * inv_l);
+        w[1] = cvtpk2(ot[mt][4 * g + 2] * inv_l, ot[mt][4 * g + 3] * inv_l);
+        *reinterpret_cast<unsigned long long*>(orow + d0) =
+            *reinterpret_cast<unsigned long long*>(w);
+      }
+    if (hi == 0 && lse_out)  // back to base-e: lse = ln2*(m2 + log2 l)
+      lse_out[((long long)b * Hq + h) * S + my_q] =
+          0.693147181f * (m_run + __log2f(fmaxf(l_run, 1e-30f)));
+  }
+}
+
 std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
                                        at::Tensor v, bool causal,
                                        double scale) {
@@ -272,14 +555,30 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
   auto out = at::empty_like(q);
   auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStream();
-  dim3 grid((S + QTILE - 1) / QTILE, B * Hq);
-  hipLaunchKernelGGL(flash_fwd_kernel, grid, dim3(512), 0, stream.stream(),
-                     reinterpret_cast<const short*>(q.data_ptr()),
-                     reinterpret_cast<const short*>(k.data_ptr()),
-                     reinterpret_cast<const short*>(v.data_ptr()),
-                     reinterpret_cast<short*>(out.data_ptr()),
-                     lse.data_ptr<float>(), B, S, Hq, Hk, (float)scale,
-                     causal ? 1 : 0);
+  static const bool use_v4 = []() {
+    const char* e = getenv("DSAMD_ATTN_FWD_V4");
+    return e && e[0] == '1';
+  }();
+  if (use_v4) {
+    dim3 grid((S + QTILE - 1) / QTILE, B * Hq);
+    hipLaunchKernelGGL(flash_fwd_kernel, grid, dim3(512), 0, stream.stream(),
+                       reinterpret_cast<const short*>(q.data_ptr()),
+                       reinterpret_cast<const short*>(k.data_ptr()),
+                       reinterpret_cast<const short*>(v.data_ptr()),
+                       reinterpret_cast<short*>(out.data_ptr()),
+                       lse.data_ptr<float>(), B, S, Hq, Hk, (float)scale,
+                       causal ? 1 : 0);
+  } else {
+    dim3 grid((S + QT5 - 1) / QT5, B * Hq);
+    hipLaunchKernelGGL(flash_fwd_v5_kernel, grid, dim3(512), 0,
+                       stream.stream(),
+                       reinterpret_cast<const short*>(q.data_ptr()),
+                       reinterpret_cast<const short*>(k.data_ptr()),
+                       reinterpret_cast<const short*>(v.data_ptr()),
+                       reinterpret_cast<short*>(out.data_ptr()),
+                       lse.data_ptr<float>(), B, S, Hq, Hk, (float)scale,
+                       causal ? 1 : 0);
+  }
   HIP_CHECK_KERNEL();
   return {out, lse};
 }

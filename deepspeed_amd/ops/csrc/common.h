@@ -38,6 +38,11 @@ DEV_INLINE float bf2f(short b) {
 }
 
 DEV_INLINE short f2bf(float f) {
+#if defined(__gfx950__)
+  // native RNE convert: v_cvt_pk_bf16_f32 (one VALU op vs 4 integer ops)
+  __bf16 b = (__bf16)f;
+  return __builtin_bit_cast(short, b);
+#else
   union {
     float f;
     unsigned int u;
@@ -47,9 +52,45 @@ DEV_INLINE short f2bf(float f) {
   unsigned int lsb = (cvt.u >> 16) & 1;
   cvt.u += 0x7fff + lsb;
   return (short)(cvt.u >> 16);
+#endif
 }
 
 // wave-level reduction over 64 lanes
+// 16-lane (DPP row) reductions: row_ror rotates within each group of 16
+// lanes on the VALU — no LDS-pipe traffic (vs __shfl_xor -> ds_swizzle).
+// After the 4 steps every lane of the group holds the full reduction.
+DEV_INLINE float dpp_ror16(float x, const int ctrl) {
+  int i = __builtin_bit_cast(int, x);
+  int r;
+  switch (ctrl) {  // update_dpp needs literal ctrl
+    case 0x128: r = __builtin_amdgcn_update_dpp(0, i, 0x128, 0xf, 0xf, true);
+                break;
+    case 0x124: r = __builtin_amdgcn_update_dpp(0, i, 0x124, 0xf, 0xf, true);
+                break;
+    case 0x122: r = __builtin_amdgcn_update_dpp(0, i, 0x122, 0xf, 0xf, true);
+                break;
+    default:    r = __builtin_amdgcn_update_dpp(0, i, 0x121, 0xf, 0xf, true);
+                break;
+  }
+  return __builtin_bit_cast(float, r);
+}
+
+DEV_INLINE float row16_reduce_sum(float x) {
+  x += dpp_ror16(x, 0x128);
+  x += dpp_ror16(x, 0x124);
+  x += dpp_ror16(x, 0x122);
+  x += dpp_ror16(x, 0x121);
+  return x;
+}
+
+DEV_INLINE float row16_reduce_max(float x) {
+  x = fmaxf(x, dpp_ror16(x, 0x128));
+  x = fmaxf(x, dpp_ror16(x, 0x124));
+  x = fmaxf(x, dpp_ror16(x, 0x122));
+  x = fmaxf(x, dpp_ror16(x, 0x121));
+  return x;
+}
+
 DEV_INLINE float wave_reduce_sum(float x) {
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) x += __shfl_down(x, off, WAVE);
