@@ -278,6 +278,13 @@ constexpr int QB5 = 32;    // q rows per wave
 constexpr int QT5 = 256;   // q rows per block (8 waves)
 constexpr int KT5 = 64;    // kv rows per tile
 
+// exchange a scalar with the other half-wave (VALU permlane, no LDS pipe)
+DEV_INLINE float half_swap(float x, int hi) {
+  int i = __builtin_bit_cast(int, x);
+  auto r = __builtin_amdgcn_permlane32_swap(i, i, false, false);
+  return __builtin_bit_cast(float, hi ? r[0] : r[1]);
+}
+
 // pack two f32 into a u32 of 2 bf16 (compiler emits v_cvt_pk_bf16_f32)
 DEV_INLINE unsigned int cvtpk2(float lo, float hi) {
   unsigned short a = __builtin_bit_cast(unsigned short, (__bf16)lo);
@@ -450,7 +457,7 @@ __global__ void flash_fwd_v5_kernel(
           tmax = fmaxf(tmax, sv);
         }
     }
-    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+    tmax = fmaxf(tmax, half_swap(tmax, hi));
     // defer-max (guide T13): only rescale when the running max grew by more
     // than 8/ln2 (P is then bounded by e^8, which f32 accum tolerates; LSE
     // stays exact). Masked values rely on exp2(-inf)=0 — no per-value select.
@@ -473,7 +480,7 @@ __global__ void flash_fwd_v5_kernel(
         st[mt][r] = p;  // P overwrites S in-register
         psum += p;
       }
-    psum += __shfl_xor(psum, 32, 64);
+    psum += half_swap(psum, hi);
     l_run += psum;
 
     // ---- build P fragments in-register ----------------------------------
