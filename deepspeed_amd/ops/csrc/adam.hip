@@ -154,3 +154,151 @@ void multi_tensor_lion(std::vector<at::Tensor> params,
     HIP_CHECK_KERNEL();
   }
 }
+
+// Adagrad (ref csrc/adagrad/cpu_adagrad.cpp GPU analogue): fp32 master.
+__global__ void adagrad_kernel_f32(float* __restrict__ p,
+                                   const float* __restrict__ g,
+                                   float* __restrict__ h,
+                                   short* __restrict__ out16, long long n,
+                                   float lr, float eps, float wd) {
+  long long i0 = (long long)(blockIdx.x) * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = i0; i < n; i += stride) {
+    float gk = g[i], pk = p[i];
+    if (wd != 0.f) gk += wd * pk;
+    float hk = h[i] + gk * gk;
+    pk -= lr * gk / (sqrtf(hk) + eps);
+    h[i] = hk;
+    p[i] = pk;
+    if (out16) out16[i] = f2bf(pk);
+  }
+}
+
+void multi_tensor_adagrad(std::vector<at::Tensor> params,
+                          std::vector<at::Tensor> grads,
+                          std::vector<at::Tensor> sq_accums, double lr,
+                          double eps, double weight_decay,
+                          std::vector<at::Tensor> out16) {
+  auto stream = c10::hip::getCurrentHIPStream();
+  for (size_t t = 0; t < params.size(); ++t) {
+    long long n = params[t].numel();
+    short* o16 = out16.empty()
+                     ? nullptr
+                     : reinterpret_cast<short*>(out16[t].data_ptr());
+    int block = 256;
+    int grid = grid_for(n, block);
+    hipLaunchKernelGGL(adagrad_kernel_f32, dim3(grid), dim3(block), 0,
+                       stream.stream(), params[t].data_ptr<float>(),
+                       grads[t].data_ptr<float>(),
+                       sq_accums[t].data_ptr<float>(), o16, n, (float)lr,
+                       (float)eps, (float)weight_decay);
+    HIP_CHECK_KERNEL();
+  }
+}
+
+// LAMB (ref csrc/lamb/fused_lamb_cuda_kernel.cu). Two passes, trust ratio
+// computed DEVICE-side so there is no host sync per parameter:
+//   pass 1: adam-moment update, u = mhat/(sqrt(vhat)+eps) + wd*p written
+//           into the grad buffer (scratch), block-reduced ||p||^2,||u||^2
+//           atomically accumulated into a 2-float workspace
+//   pass 2: p -= lr * (||p||/||u||) * u
+__global__ void lamb_phase1_f32(float* __restrict__ p,
+                                float* __restrict__ g,  // in: grad, out: u
+                                float* __restrict__ m, float* __restrict__ v,
+                                float* __restrict__ ws, long long n,
+                                float beta1, float beta2, float eps,
+                                float bc1, float bc2, float wd,
+                                float gscale) {
+  __shared__ float red[2][4];
+  long long i0 = (long long)(blockIdx.x) * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  float psq = 0.f, usq = 0.f;
+  for (long long i = i0; i < n; i += stride) {
+    float gk = g[i] * gscale, pk = p[i];
+    float mk = m[i] * beta1 + gk * (1.f - beta1);
+    float vk = v[i] * beta2 + gk * gk * (1.f - beta2);
+    float u = (mk / bc1) / (sqrtf(vk / bc2) + eps) + wd * pk;
+    m[i] = mk;
+    v[i] = vk;
+    g[i] = u;
+    psq += pk * pk;
+    usq += u * u;
+  }
+  psq = wave_reduce_sum(psq);
+  usq = wave_reduce_sum(usq);
+  int wid = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) {
+    red[0][wid] = psq;
+    red[1][wid] = usq;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float ps = 0.f, us = 0.f;
+    for (int w = 0; w < (int)(blockDim.x >> 6); ++w) {
+      ps += red[0][w];
+      us += red[1][w];
+    }
+    atomicAdd(&ws[0], ps);
+    atomicAdd(&ws[1], us);
+  }
+}
+
+__global__ void lamb_phase2_f32(float* __restrict__ p,
+                                const float* __restrict__ u,
+                                const float* __restrict__ ws,
+                                short* __restrict__ out16, long long n,
+                                float lr) {
+  float r1 = sqrtf(ws[0]), r2 = sqrtf(ws[1]);
+  float ratio = (r1 > 0.f && r2 > 0.f) ? r1 / r2 : 1.f;
+  long long i0 = (long long)(blockIdx.x) * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = i0; i < n; i += stride) {
+    float pk = p[i] - lr * ratio * u[i];
+    p[i] = pk;
+    if (out16) out16[i] = f2bf(pk);
+  }
+}
+
+void multi_tensor_lamb(std::vector<at::Tensor> params,
+                       std::vector<at::Tensor> grads,
+                       std::vector<at::Tensor> exp_avgs,
+                       std::vector<at::Tensor> exp_avg_sqs,
+                       at::Tensor workspace,  // fp32 [2*nparams], zeroed
+                       double lr, double beta1, double beta2, double eps,
+                       long step, long bias_correction, double weight_decay,
+                       std::vector<at::Tensor> out16, double grad_scale) {
+  float bc1 = 1.f, bc2 = 1.f;
+  if (bias_correction) {
+    bc1 = 1.f - powf((float)beta1, (float)step);
+    bc2 = 1.f - powf((float)beta2, (float)step);
+  }
+  TORCH_CHECK(workspace.numel() >= (long)(2 * params.size()));
+  auto stream = c10::hip::getCurrentHIPStream();
+  float* ws = workspace.data_ptr<float>();
+  for (size_t t = 0; t < params.size(); ++t) {
+    long long n = params[t].numel();
+    int block = 256;
+    int grid = grid_for(n, block);
+    hipLaunchKernelGGL(lamb_phase1_f32, dim3(grid), dim3(block), 0,
+                       stream.stream(), params[t].data_ptr<float>(),
+                       grads[t].data_ptr<float>(),
+                       exp_avgs[t].data_ptr<float>(),
+                       exp_avg_sqs[t].data_ptr<float>(), ws + 2 * t, n,
+                       (float)beta1, (float)beta2, (float)eps, bc1, bc2,
+                       (float)weight_decay, (float)grad_scale);
+    HIP_CHECK_KERNEL();
+  }
+  for (size_t t = 0; t < params.size(); ++t) {
+    long long n = params[t].numel();
+    short* o16 = out16.empty()
+                     ? nullptr
+                     : reinterpret_cast<short*>(out16[t].data_ptr());
+    int block = 256;
+    int grid = grid_for(n, block);
+    hipLaunchKernelGGL(lamb_phase2_f32, dim3(grid), dim3(block), 0,
+                       stream.stream(), params[t].data_ptr<float>(),
+                       grads[t].data_ptr<float>(), ws + 2 * t, o16, n,
+                       (float)lr);
+    HIP_CHECK_KERNEL();
+  }
+}

@@ -18,6 +18,25 @@ std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
                                        at::Tensor v, at::Tensor dout,
                                        at::Tensor out, at::Tensor lse,
                                        bool causal, double scale);
+void multi_tensor_adagrad(std::vector<at::Tensor> params,
+                          std::vector<at::Tensor> grads,
+                          std::vector<at::Tensor> sq_accums, double lr,
+                          double eps, double weight_decay,
+                          std::vector<at::Tensor> out16);
+void multi_tensor_lamb(std::vector<at::Tensor> params,
+                       std::vector<at::Tensor> grads,
+                       std::vector<at::Tensor> exp_avgs,
+                       std::vector<at::Tensor> exp_avg_sqs,
+                       at::Tensor workspace, double lr, double beta1,
+                       double beta2, double eps, long step,
+                       long bias_correction, double weight_decay,
+                       std::vector<at::Tensor> out16, double grad_scale);
+void cpu_lion_step(at::Tensor p, at::Tensor g, at::Tensor m,
+                   c10::optional<at::Tensor> out16, double lr, double beta1,
+                   double beta2, double wd);
+void cpu_adagrad_step(at::Tensor p, at::Tensor g, at::Tensor h,
+                      c10::optional<at::Tensor> out16, double lr, double eps,
+                      double wd);
 void cpu_adam_step(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
                    c10::optional<at::Tensor> out16, double lr, double beta1,
                    double beta2, double eps, long step, long adamw,
@@ -79,6 +98,25 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_attn_fwd", &flash_attn_fwd, py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("causal") = true, py::arg("scale") = 0.0);
   m.def("flash_attn_bwd", &flash_attn_bwd);
+  m.def("multi_tensor_adagrad", &multi_tensor_adagrad, py::arg("params"),
+        py::arg("grads"), py::arg("sq_accums"), py::arg("lr"),
+        py::arg("eps") = 1e-8, py::arg("weight_decay") = 0.0,
+        py::arg("out16") = std::vector<at::Tensor>{});
+  m.def("multi_tensor_lamb", &multi_tensor_lamb, py::arg("params"),
+        py::arg("grads"), py::arg("exp_avgs"), py::arg("exp_avg_sqs"),
+        py::arg("workspace"), py::arg("lr"), py::arg("beta1") = 0.9,
+        py::arg("beta2") = 0.999, py::arg("eps") = 1e-6, py::arg("step") = 1,
+        py::arg("bias_correction") = 1, py::arg("weight_decay") = 0.0,
+        py::arg("out16") = std::vector<at::Tensor>{},
+        py::arg("grad_scale") = 1.0);
+  m.def("cpu_lion_step", &cpu_lion_step, py::arg("p"), py::arg("g"),
+        py::arg("m"), py::arg("out16") = c10::nullopt, py::arg("lr") = 1e-4,
+        py::arg("beta1") = 0.9, py::arg("beta2") = 0.99, py::arg("wd") = 0.0,
+        py::call_guard<py::gil_scoped_release>());
+  m.def("cpu_adagrad_step", &cpu_adagrad_step, py::arg("p"), py::arg("g"),
+        py::arg("h"), py::arg("out16") = c10::nullopt, py::arg("lr") = 1e-2,
+        py::arg("eps") = 1e-8, py::arg("wd") = 0.0,
+        py::call_guard<py::gil_scoped_release>());
   m.def("cpu_adam_step", &cpu_adam_step, py::arg("p"), py::arg("g"),
         py::arg("m"), py::arg("v"), py::arg("out16") = c10::nullopt,
         py::arg("lr") = 1e-3, py::arg("beta1") = 0.9, py::arg("beta2") = 0.999,

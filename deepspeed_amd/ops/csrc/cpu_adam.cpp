@@ -67,3 +67,53 @@ void cpu_adam_step(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v,
     if (op) op[i] = f32_to_bf16(pk);
   }
 }
+
+// Host Lion for ZeRO-Offload (ref csrc/lion/cpu_lion_impl.cpp).
+void cpu_lion_step(at::Tensor p, at::Tensor g, at::Tensor m,
+                   c10::optional<at::Tensor> out16, double lr_, double beta1_,
+                   double beta2_, double wd_) {
+  TORCH_CHECK(p.device().is_cpu() && p.scalar_type() == at::kFloat);
+  const float lr = (float)lr_, beta1 = (float)beta1_, beta2 = (float)beta2_;
+  const float wd = (float)wd_;
+  const int64_t n = p.numel();
+  float* pp = p.data_ptr<float>();
+  const float* gp = g.data_ptr<float>();
+  float* mp = m.data_ptr<float>();
+  uint16_t* op = nullptr;
+  if (out16.has_value())
+    op = reinterpret_cast<uint16_t*>(out16->data_ptr());
+#pragma omp parallel for schedule(static)
+  for (int64_t i = 0; i < n; ++i) {
+    float gk = gp[i], pk = pp[i] * (1.f - lr * wd), mk = mp[i];
+    float u = mk * beta1 + gk * (1.f - beta1);
+    pk -= lr * (u > 0.f ? 1.f : (u < 0.f ? -1.f : 0.f));
+    mp[i] = mk * beta2 + gk * (1.f - beta2);
+    pp[i] = pk;
+    if (op) op[i] = f32_to_bf16(pk);
+  }
+}
+
+// Host Adagrad for ZeRO-Offload (ref csrc/adagrad/cpu_adagrad.cpp).
+void cpu_adagrad_step(at::Tensor p, at::Tensor g, at::Tensor h,
+                      c10::optional<at::Tensor> out16, double lr_,
+                      double eps_, double wd_) {
+  TORCH_CHECK(p.device().is_cpu() && p.scalar_type() == at::kFloat);
+  const float lr = (float)lr_, eps = (float)eps_, wd = (float)wd_;
+  const int64_t n = p.numel();
+  float* pp = p.data_ptr<float>();
+  const float* gp = g.data_ptr<float>();
+  float* hp = h.data_ptr<float>();
+  uint16_t* op = nullptr;
+  if (out16.has_value())
+    op = reinterpret_cast<uint16_t*>(out16->data_ptr());
+#pragma omp parallel for schedule(static)
+  for (int64_t i = 0; i < n; ++i) {
+    float gk = gp[i], pk = pp[i];
+    if (wd != 0.f) gk += wd * pk;
+    float hk = hp[i] + gk * gk;
+    pk -= lr * gk / (sqrtf(hk) + eps);
+    hp[i] = hk;
+    pp[i] = pk;
+    if (op) op[i] = f32_to_bf16(pk);
+  }
+}

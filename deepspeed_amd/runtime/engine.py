@@ -227,8 +227,32 @@ class DeepSpeedEngine(torch.nn.Module):
             from ..ops.cpu_adam import DeepSpeedCPUAdam
             return DeepSpeedCPUAdam(model_parameters, **params)
         if name == "lion":
+            if self._offload_optimizer_enabled():
+                from ..ops.lion import DeepSpeedCPULion
+                return DeepSpeedCPULion(model_parameters, **params)
             from ..ops.lion import FusedLion
             return FusedLion(model_parameters, **params)
+        if name == "adagrad":
+            if self._offload_optimizer_enabled():
+                from ..ops.adagrad import DeepSpeedCPUAdagrad
+                return DeepSpeedCPUAdagrad(model_parameters, **params)
+            from ..ops.adagrad import FusedAdagrad
+            return FusedAdagrad(model_parameters, **params)
+        if name == "lamb" or name == "fusedlamb":
+            from ..ops.lamb import FusedLamb
+            return FusedLamb(model_parameters, **params)
+        if name == "onebitadam":
+            from ..ops.onebit_adam import OnebitAdam
+            return OnebitAdam(model_parameters, deepspeed=self, **params)
+        if name == "zerooneadam":
+            from ..ops.onebit_adam import ZeroOneAdam
+            return ZeroOneAdam(model_parameters, deepspeed=self, **params)
+        if name == "onebitlamb":
+            from ..ops.onebit_adam import OnebitLamb
+            return OnebitLamb(model_parameters, deepspeed=self, **params)
+        if name == "muon":
+            from ..ops.muon import Muon
+            return Muon(model_parameters, **params)
         if name == "sgd":
             return torch.optim.SGD(model_parameters, **params)
         raise ValueError(f"unknown optimizer type {cfg.type}")
