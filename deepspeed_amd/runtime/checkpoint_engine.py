@@ -124,3 +124,32 @@ def make_checkpoint_engine(name="torch", config_params=None):
         return FastCheckpointEngine(config_params)
     raise ValueError(f"unknown checkpoint engine {name} "
                      "(nebula/datastates are external services)")
+
+
+class NebulaCheckpointEngine(CheckpointEngine):
+    """Parity stub for reference `runtime/checkpoint_engine/nebula_*`.
+
+    Nebula is an Azure-proprietary asynchronous checkpoint service; its
+    client SDK ("torch_nebula") is not available on this stack. The
+    MI355X-native equivalent of its value proposition — async, tiered
+    checkpoint writes that overlap training — is FastCheckpointEngine
+    (O_DIRECT aio writer thread). Instantiating this class raises with
+    that pointer rather than silently degrading.
+    """
+
+    def __init__(self, config_params=None):
+        raise RuntimeError(
+            "Nebula requires the proprietary torch_nebula SDK (Azure). "
+            "Use checkpoint.engine=fast (FastCheckpointEngine) for async "
+            "checkpointing on MI355X nodes.")
+
+
+class DataStatesCheckpointEngine(CheckpointEngine):
+    """Parity stub for reference DataStates-LLM engine (ANL host-memory
+    async checkpointing). Same guidance as Nebula: FastCheckpointEngine
+    provides the in-tree async path."""
+
+    def __init__(self, config_params=None):
+        raise RuntimeError(
+            "datastates-llm is not installed; use checkpoint.engine=fast "
+            "(FastCheckpointEngine) for async checkpointing.")

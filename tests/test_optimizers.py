@@ -159,3 +159,38 @@ def test_fused_gpu_matches_torch(kind):
                 step_ref(q, st, it)
     for p, q in zip(ps, qs):
         assert torch.allclose(p, q, atol=2e-4), (p - q).abs().max()
+
+
+def _bf16_opt_run():
+    from deepspeed_amd.ops import FusedAdam
+    from deepspeed_amd.runtime.bf16_optimizer import BF16_Optimizer
+    torch.manual_seed(0)
+    m = torch.nn.Linear(32, 32).to(torch.bfloat16)
+    base = FusedAdam(m.parameters(), lr=1e-2)
+    opt = BF16_Optimizer(base)
+    assert len(opt.fp32_groups_flat_partition) >= 1
+    ref = torch.nn.Linear(32, 32)
+    with torch.no_grad():
+        ref.weight.copy_(m.weight.float())
+        ref.bias.copy_(m.bias.float())
+    ropt = torch.optim.AdamW(ref.parameters(), lr=1e-2, betas=(0.9, 0.999),
+                             eps=1e-8, weight_decay=0.0)
+    x = torch.randn(8, 32)
+    for _ in range(3):
+        loss = m(x.bfloat16()).float().pow(2).mean()
+        opt.backward(loss)
+        opt.step()
+        opt.zero_grad()
+        rl = ref(x).pow(2).mean()
+        rl.backward()
+        ropt.step()
+        ropt.zero_grad()
+    opt.update_lp_params()
+    err = (m.weight.float() - ref.weight).abs().max().item()
+    assert err < 0.05, err
+
+
+def test_bf16_optimizer_wrapper():
+    """BF16_Optimizer = named stage-1 fp32-master path (single process)."""
+    from tests.common import run_distributed
+    run_distributed(_bf16_opt_run, world_size=1)
