@@ -75,6 +75,9 @@ std::vector<at::Tensor> quantize_int8(at::Tensor x, long group_size);
 at::Tensor dequantize_int8(at::Tensor q, at::Tensor scales, long group_size);
 std::vector<at::Tensor> quantize_fp8(at::Tensor x, long group_size);
 at::Tensor dequantize_fp8(at::Tensor q, at::Tensor scales, long group_size);
+std::vector<at::Tensor> quantize_fp_em(at::Tensor x, long qbits, long group);
+at::Tensor dequantize_fp_em(at::Tensor q, at::Tensor scales, long qbits,
+                            long group, std::vector<long> shape);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   bind_aio(m);
@@ -95,6 +98,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dequantize_int8", &dequantize_int8);
   m.def("quantize_fp8", &quantize_fp8);
   m.def("dequantize_fp8", &dequantize_fp8);
+  m.def("quantize_fp_em", &quantize_fp_em, py::arg("x"), py::arg("qbits"),
+        py::arg("group") = 512);
+  m.def("dequantize_fp_em", &dequantize_fp_em, py::arg("q"),
+        py::arg("scales"), py::arg("qbits"), py::arg("group"),
+        py::arg("shape"));
   m.def("flash_attn_fwd", &flash_attn_fwd, py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("causal") = true, py::arg("scale") = 0.0);
   m.def("flash_attn_bwd", &flash_attn_bwd);

@@ -66,3 +66,58 @@ def test_quantizers_gpu():
         rel = ((back.float() - x.float()).abs() /
                (x.float().abs() + 1e-3)).median().item()
         assert rel < tol, (quant.__name__, rel)
+
+
+# ------------------------------------------------------- FP_Quantize family
+def _fp_roundtrip_cpu(q_bits, tol):
+    import torch
+    from deepspeed_amd.ops.fp_quantizer import FP_Quantize
+    torch.manual_seed(0)
+    x = torch.randn(1000) * 3
+    fpq = FP_Quantize(group_size=128, q_bits=q_bits)
+    q = fpq.quantize(x)
+    y = fpq.dequantize(q).float()
+    rel = (x - y).abs().max() / x.abs().max()
+    assert rel < tol, f"q{q_bits}: rel err {rel}"
+
+
+def test_fp4_roundtrip_cpu():
+    _fp_roundtrip_cpu(4, 0.20)
+
+
+def test_fp6_roundtrip_cpu():
+    _fp_roundtrip_cpu(6, 0.08)
+
+
+def test_fp12_roundtrip_cpu():
+    _fp_roundtrip_cpu(12, 0.005)
+
+
+def test_fp_quantize_exact_values_cpu():
+    """Codec spot checks: e2m1 grid is {0,.5,1,1.5,2,3,4,6} x sign."""
+    import torch
+    from deepspeed_amd.ops.fp_quantizer import _decode_torch, _encode_torch
+    vals = torch.tensor([0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0,
+                         -1.5, -6.0])
+    bits = _encode_torch(vals, 2, 1)
+    back = _decode_torch(bits, 2, 1)
+    assert torch.equal(back, vals)
+    # rounding: 2.4 -> 2, 2.6 -> 3, 7 -> saturate 6
+    r = _decode_torch(_encode_torch(torch.tensor([2.4, 2.6, 7.0]), 2, 1),
+                      2, 1)
+    assert torch.equal(r, torch.tensor([2.0, 3.0, 6.0]))
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("q_bits", [4, 6, 12])
+def test_fp_quantize_gpu_matches_cpu_codec(q_bits):
+    import torch
+    from deepspeed_amd.ops.fp_quantizer import FP_Quantize
+    torch.manual_seed(0)
+    x = torch.randn(4096, dtype=torch.bfloat16) * 2
+    cpu = FP_Quantize(group_size=256, q_bits=q_bits)
+    yc = cpu.dequantize(cpu.quantize(x.float()))
+    gpu = FP_Quantize(group_size=256, q_bits=q_bits)
+    yg = gpu.dequantize(gpu.quantize(x.cuda())).cpu()
+    assert torch.allclose(yc.float(), yg.float(), atol=1e-4), \
+        (yc.float() - yg.float()).abs().max()
