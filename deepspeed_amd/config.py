@@ -68,6 +68,7 @@ class ZeroConfig(DSConfigModel):
     sub_group_size: int = int(1e9)
     offload_param: Optional[OffloadParamConfig] = None
     offload_optimizer: Optional[OffloadOptimizerConfig] = None
+    zenflow: Optional[dict] = None  # ZenFlow selective-offload (stage 1/2)
     zero_hpz_partition_size: int = 1
     mics_shard_size: int = -1
     round_robin_gradients: bool = False

@@ -306,8 +306,15 @@ class DeepSpeedEngine(torch.nn.Module):
                 log_dist("dtype is 16-bit with ZeRO disabled: using "
                          "stage-1 partitioned fp32-master optimizer "
                          "(identical numerics, lower memory)", ranks=[0])
-            self.optimizer = ZeroStage12Optimizer(
+            opt_cls = ZeroStage12Optimizer
+            extra = {}
+            if zc.zenflow is not None:
+                from .zenflow import ZenFlowZeroOptimizer
+                opt_cls = ZenFlowZeroOptimizer
+                extra["zenflow_config"] = zc.zenflow
+            self.optimizer = opt_cls(
                 basic,
+                **extra,
                 engine=self,
                 stage=max(stage, 1),
                 dp_process_group=self.dp_group,

@@ -1,0 +1,185 @@
+"""ZenFlow — importance-aware selective optimizer offload for ZeRO-1/2.
+
+Parity: reference `deepspeed/runtime/zenflow/zenflow_stage_1_and_2.py`
+(+ `zenflow_config.py`, `ops/adam/zenflow_cpu_adam.py`): the fp32
+optimizer state lives on the host; every step the top-k most important
+gradient channels are updated IMMEDIATELY on the GPU (hot set), while
+the remaining gradients accumulate on the host and are applied by a
+batched CPU-Adam pass every `update_interval` steps. This keeps the
+update path off the critical step for ~99% of the state while the loss
+trajectory tracks dense Adam closely.
+
+MI355X-native design: the unit of selection is the flat fp32 grad shard
+of a stage-1/2 bucket (already reduce-scattered over RCCL, so importance
+and hot sets are rank-local — no extra comm). The hot set's master/m/v
+vectors are cached on the GPU between re-selections; the lazy pass is
+the OpenMP `cpu_adam_step` on the pinned host slabs, and the bf16 shard
+is refreshed with one H2D per bucket only on lazy boundaries.
+"""
+from typing import Optional
+
+import torch
+from pydantic import BaseModel
+
+from .. import comm as dist
+from ..utils.logging import log_dist
+from .zero.stage_1_and_2 import ZeroStage12Optimizer
+
+
+class ZenFlowConfig(BaseModel):
+    topk_ratio: float = 0.01
+    update_interval: int = 4
+    select_interval: int = 16    # re-pick the hot set every N steps
+    select_strategy: str = "auto"
+    full_warm_up_rounds: int = 0
+    overlap_step: bool = False   # reserved: async CPU step thread
+
+
+class ZenFlowZeroOptimizer(ZeroStage12Optimizer):
+    def __init__(self, init_optimizer, zenflow_config: Optional[dict] = None,
+                 **kw):
+        super().__init__(init_optimizer, **kw)
+        zf = zenflow_config or {}
+        self.zf = zf if isinstance(zf, ZenFlowConfig) else ZenFlowConfig(**zf)
+        self._zf_step = 0
+        # host-resident optimizer state (pinned when CUDA is present);
+        # Bucket is __slots__, so ZenFlow state lives in a side table
+        pin = torch.cuda.is_available()
+        self._zf = []
+        for b in self.buckets:
+            cpu = b.master32.detach().to("cpu")
+            st = dict(master_cpu=cpu.pin_memory() if pin else cpu,
+                      acc_steps=0, hot_idx=None, hot_master=None,
+                      hot_m=None, hot_v=None)
+            st["m_cpu"] = torch.zeros_like(st["master_cpu"])
+            st["v_cpu"] = torch.zeros_like(st["master_cpu"])
+            st["acc_cpu"] = torch.zeros_like(st["master_cpu"])
+            self._zf.append(st)
+            # the GPU master slab is not used by ZenFlow's dense pass
+            b.master32 = None
+        # drop the wrapped optimizer's references so the GPU slabs free
+        for group in self.optimizer.param_groups:
+            group["params"] = []
+        log_dist(f"ZenFlow: topk={self.zf.topk_ratio}, "
+                 f"update_interval={self.zf.update_interval}", ranks=[0])
+
+    # -- selection --------------------------------------------------------
+    def _writeback_hot(self, st):
+        if st["hot_idx"] is None:
+            return
+        idx = st["hot_idx"].cpu()
+        st["master_cpu"][idx] = st["hot_master"].cpu()
+        st["m_cpu"][idx] = st["hot_m"].cpu()
+        st["v_cpu"][idx] = st["hot_v"].cpu()
+
+    def _select_hot(self, st, g):
+        k = max(1, int(self.zf.topk_ratio * g.numel()))
+        self._writeback_hot(st)
+        st["hot_idx"] = g.abs().topk(k).indices
+        idx_cpu = st["hot_idx"].cpu()
+        dev = g.device
+        st["hot_master"] = st["master_cpu"][idx_cpu].to(dev,
+                                                        non_blocking=True)
+        st["hot_m"] = st["m_cpu"][idx_cpu].to(dev, non_blocking=True)
+        st["hot_v"] = st["v_cpu"][idx_cpu].to(dev, non_blocking=True)
+
+    # -- step -------------------------------------------------------------
+    @torch.no_grad()
+    def step(self, closure=None):
+        assert closure is None, "closure not supported"
+        self._sync_comm()
+        self._zf_step += 1
+        combined = self._combined_scale()
+        group = self.optimizer.param_groups[0]
+        lr = group["lr"]
+        beta1, beta2 = group.get("betas", (0.9, 0.999))
+        eps = group.get("eps", 1e-8)
+        wd = group.get("weight_decay", 0.0)
+        t = self._zf_step
+        bc1 = 1 - beta1 ** t
+        bc2 = 1 - beta2 ** t
+
+        lazy = (self._zf_step % self.zf.update_interval) == 0
+        for b, st in zip(self.buckets, self._zf):
+            g = b.grad32
+            if combined != 1.0:
+                g.mul_(1.0 / combined)
+            if st["hot_idx"] is None or \
+                    (self._zf_step % self.zf.select_interval) == 1:
+                self._select_hot(st, g)
+            idx = st["hot_idx"]
+            # ---- hot channels: immediate GPU Adam -----------------------
+            gh = g[idx]
+            st["hot_m"].mul_(beta1).add_(gh, alpha=1 - beta1)
+            st["hot_v"].mul_(beta2).addcmul_(gh, gh, value=1 - beta2)
+            if wd:
+                st["hot_master"].mul_(1 - lr * wd)
+            denom = (st["hot_v"] / bc2).sqrt().add_(eps)
+            st["hot_master"].addcdiv_(st["hot_m"] / bc1, denom, value=-lr)
+            flat_shard = b.shard16.view(-1)
+            flat_shard[idx] = st["hot_master"].to(b.shard16.dtype)
+            # ---- cold channels: accumulate on host ----------------------
+            g.index_fill_(0, idx, 0.0)
+            st["acc_cpu"].add_(g.to("cpu"))
+            st["acc_steps"] += 1
+            if lazy:
+                self._writeback_hot(st)
+                st["acc_cpu"].div_(max(st["acc_steps"], 1))
+                from ..ops.loader import has_ext, get_ext
+                if has_ext():
+                    get_ext().cpu_adam_step(
+                        st["master_cpu"], st["acc_cpu"], st["m_cpu"],
+                        st["v_cpu"], None, lr, beta1, beta2, eps, t, 1, 1,
+                        wd, 1.0)
+                else:
+                    m, v = st["m_cpu"], st["v_cpu"]
+                    m.mul_(beta1).add_(st["acc_cpu"], alpha=1 - beta1)
+                    v.mul_(beta2).addcmul_(st["acc_cpu"], st["acc_cpu"],
+                                           value=1 - beta2)
+                    st["master_cpu"].mul_(1 - lr * wd)
+                    st["master_cpu"].addcdiv_(
+                        m / bc1, (v / bc2).sqrt().add_(eps), value=-lr)
+                st["acc_cpu"].zero_()
+                st["acc_steps"] = 0
+                b.shard16.copy_(
+                    st["master_cpu"].to(b.shard16.device, non_blocking=True)
+                    .to(b.shard16.dtype))
+                # refresh the hot cache from the new masters
+                idx_cpu = idx.cpu()
+                st["hot_master"] = st["master_cpu"][idx_cpu].to(g.device)
+                st["hot_m"] = st["m_cpu"][idx_cpu].to(g.device)
+                st["hot_v"] = st["v_cpu"][idx_cpu].to(g.device)
+
+        for b in self.buckets:
+            dist.all_gather_into_tensor(
+                b.flat16, b.shard16,
+                group=b.pg if b.pg is not None else self.dp_group)
+        self._clear_grads()
+
+    # checkpointing: host masters replace the GPU ones
+    def state_dict(self):
+        return {
+            "zf_step": self._zf_step,
+            "single_partition_of_fp32_groups":
+                [st["master_cpu"] for st in self._zf],
+            "m": [st["m_cpu"] for st in self._zf],
+            "v": [st["v_cpu"] for st in self._zf],
+            "loss_scaler": self.loss_scaler.state_dict()
+            if hasattr(self.loss_scaler, "state_dict") else None,
+        }
+
+    def load_state_dict(self, sd):
+        self._zf_step = sd.get("zf_step", 0)
+        for b, st, mc, m, v in zip(self.buckets, self._zf,
+                                   sd["single_partition_of_fp32_groups"],
+                                   sd["m"], sd["v"]):
+            st["master_cpu"].copy_(mc)
+            st["m_cpu"].copy_(m)
+            st["v_cpu"].copy_(v)
+            st["hot_idx"] = None
+            b.shard16.copy_(st["master_cpu"].to(b.shard16.device)
+                            .to(b.shard16.dtype))
+        for b in self.buckets:
+            dist.all_gather_into_tensor(
+                b.flat16, b.shard16,
+                group=b.pg if b.pg is not None else self.dp_group)
