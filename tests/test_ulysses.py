@@ -106,3 +106,36 @@ def model_full_logits(model, data):
 def test_ulysses_llama_logits_match():
     results = run_distributed(_ulysses_llama, world_size=2)
     assert abs(results[0] - results[1]) < 1e-5
+
+
+# ----------------------------------------------------------------- AutoSP
+def test_autosp_pick_degree():
+    from deepspeed_amd.sequence.auto_sp import pick_sp_degree
+    # short sequences: no SP
+    assert pick_sp_degree(2048, 32, 8, world_size=8) == 1
+    # long sequence: largest divisor of world/heads/kv under the need
+    assert pick_sp_degree(32768, 32, 8, world_size=8) == 4
+    assert pick_sp_degree(65536, 32, 8, world_size=8) == 8
+    # kv head limit caps the degree
+    assert pick_sp_degree(65536, 32, 2, world_size=8) == 2
+    # degree must divide world
+    assert pick_sp_degree(65536, 32, 8, world_size=6) == 2
+
+
+def _autosp_apply():
+    from deepspeed_amd.models.llama import LlamaModel, LLAMA_CONFIGS
+    from deepspeed_amd.sequence.auto_sp import configure_auto_sp
+    from deepspeed_amd.sequence.layer import DistributedAttention
+    cfg = LLAMA_CONFIGS["llama-tiny"]
+    m = LlamaModel(cfg)
+    deg = configure_auto_sp(m, seq_len=65536, seq_threshold=4096)
+    assert deg == 2, deg
+    wrapped = [mod for mod in m.modules()
+               if getattr(mod, "_dist_attn", None) is not None]
+    assert wrapped, "no attention was wrapped"
+    assert isinstance(wrapped[0]._dist_attn, DistributedAttention)
+
+
+def test_autosp_applies_to_llama_world2():
+    from tests.common import run_distributed
+    run_distributed(_autosp_apply, world_size=2)
