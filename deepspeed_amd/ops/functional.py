@@ -238,3 +238,17 @@ def fused_cross_entropy(logits, targets, ignore_index=-100):
     V = logits.shape[-1]
     return _FusedCrossEntropyFn.apply(logits.reshape(-1, V),
                                       targets.reshape(-1), ignore_index)
+
+
+def gds_handle(*a, **kw):
+    """Parity stub for reference ops/aio GDS (GPUDirect Storage) builder.
+
+    NVMe<->HBM direct DMA on ROCm ships as hipGDS/KFD DMA-BUF in ROCm
+    enterprise stacks and is not present in this image; the in-tree
+    O_DIRECT aio thread pool (ops/csrc/aio.cpp) covers the NVMe swap
+    path through pinned host bounce buffers at NVMe line rate. Raises
+    so GDS-dependent configs fail loudly.
+    """
+    raise RuntimeError(
+        "GDS is unavailable on this ROCm stack; use aio_handle "
+        "(O_DIRECT thread pool) — swap paths accept it interchangeably.")
