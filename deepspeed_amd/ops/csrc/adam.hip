@@ -203,7 +203,8 @@ void multi_tensor_adagrad(std::vector<at::Tensor> params,
 //           atomically accumulated into a 2-float workspace
 //   pass 2: p -= lr * (||p||/||u||) * u
 __global__ void lamb_phase1_f32(float* __restrict__ p,
-                                float* __restrict__ g,  // in: grad, out: u
+                                const float* __restrict__ g,
+                                float* __restrict__ u_out,
                                 float* __restrict__ m, float* __restrict__ v,
                                 float* __restrict__ ws, long long n,
                                 float beta1, float beta2, float eps,
@@ -220,7 +221,7 @@ __global__ void lamb_phase1_f32(float* __restrict__ p,
     float u = (mk / bc1) / (sqrtf(vk / bc2) + eps) + wd * pk;
     m[i] = mk;
     v[i] = vk;
-    g[i] = u;
+    u_out[i] = u;
     psq += pk * pk;
     usq += u * u;
   }
@@ -261,6 +262,7 @@ __global__ void lamb_phase2_f32(float* __restrict__ p,
 
 void multi_tensor_lamb(std::vector<at::Tensor> params,
                        std::vector<at::Tensor> grads,
+                       std::vector<at::Tensor> updates,  // scratch, grad-sized
                        std::vector<at::Tensor> exp_avgs,
                        std::vector<at::Tensor> exp_avg_sqs,
                        at::Tensor workspace,  // fp32 [2*nparams], zeroed
@@ -282,6 +284,7 @@ void multi_tensor_lamb(std::vector<at::Tensor> params,
     hipLaunchKernelGGL(lamb_phase1_f32, dim3(grid), dim3(block), 0,
                        stream.stream(), params[t].data_ptr<float>(),
                        grads[t].data_ptr<float>(),
+                       updates[t].data_ptr<float>(),
                        exp_avgs[t].data_ptr<float>(),
                        exp_avg_sqs[t].data_ptr<float>(), ws + 2 * t, n,
                        (float)beta1, (float)beta2, (float)eps, bc1, bc2,
@@ -297,7 +300,7 @@ void multi_tensor_lamb(std::vector<at::Tensor> params,
     int grid = grid_for(n, block);
     hipLaunchKernelGGL(lamb_phase2_f32, dim3(grid), dim3(block), 0,
                        stream.stream(), params[t].data_ptr<float>(),
-                       grads[t].data_ptr<float>(), ws + 2 * t, o16, n,
+                       updates[t].data_ptr<float>(), ws + 2 * t, o16, n,
                        (float)lr);
     HIP_CHECK_KERNEL();
   }

@@ -25,6 +25,7 @@ void multi_tensor_adagrad(std::vector<at::Tensor> params,
                           std::vector<at::Tensor> out16);
 void multi_tensor_lamb(std::vector<at::Tensor> params,
                        std::vector<at::Tensor> grads,
+                       std::vector<at::Tensor> updates,
                        std::vector<at::Tensor> exp_avgs,
                        std::vector<at::Tensor> exp_avg_sqs,
                        at::Tensor workspace, double lr, double beta1,
@@ -111,7 +112,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("eps") = 1e-8, py::arg("weight_decay") = 0.0,
         py::arg("out16") = std::vector<at::Tensor>{});
   m.def("multi_tensor_lamb", &multi_tensor_lamb, py::arg("params"),
-        py::arg("grads"), py::arg("exp_avgs"), py::arg("exp_avg_sqs"),
+        py::arg("grads"), py::arg("updates"), py::arg("exp_avgs"),
+        py::arg("exp_avg_sqs"),
         py::arg("workspace"), py::arg("lr"), py::arg("beta1") = 0.9,
         py::arg("beta2") = 0.999, py::arg("eps") = 1e-6, py::arg("step") = 1,
         py::arg("bias_correction") = 1, py::arg("weight_decay") = 0.0,

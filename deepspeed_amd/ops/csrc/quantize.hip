@@ -230,9 +230,9 @@ __global__ void quantize_fp_em_kernel(const short* __restrict__ x,
     scales[g] = sc;
   }
   __syncthreads();
-  float inv = 1.f / s_scale;
+  // divide (not reciprocal-multiply): keeps bit-parity with the host codec
   for (int i = threadIdx.x; i < len; i += blockDim.x)
-    qg[i] = (unsigned short)enc_fp_em(bf2f(xg[i]) * inv, E, M);
+    qg[i] = (unsigned short)enc_fp_em(bf2f(xg[i]) / s_scale, E, M);
 }
 
 template <int E, int M>

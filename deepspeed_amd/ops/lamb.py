@@ -44,7 +44,7 @@ class FusedLamb(torch.optim.Optimizer):
                 group["step"] = 0
             group["step"] += 1
             beta1, beta2 = group["betas"]
-            ps, gs, ms, vs = [], [], [], []
+            ps, gs, us, ms, vs = [], [], [], [], []
             for p in group["params"]:
                 if p.grad is None:
                     continue
@@ -54,8 +54,12 @@ class FusedLamb(torch.optim.Optimizer):
                         p, dtype=torch.float32)
                     state["exp_avg_sq"] = torch.zeros_like(
                         p, dtype=torch.float32)
+                    if p.is_cuda:
+                        state["update"] = torch.empty_like(
+                            p, dtype=torch.float32)
                 ps.append(p)
                 gs.append(p.grad)
+                us.append(state.get("update"))
                 ms.append(state["exp_avg"])
                 vs.append(state["exp_avg_sq"])
             if not ps:
@@ -67,8 +71,8 @@ class FusedLamb(torch.optim.Optimizer):
                 else:
                     self._ws.zero_()
                 get_ext().multi_tensor_lamb(
-                    ps, gs, ms, vs, self._ws, group["lr"], beta1, beta2,
-                    group["eps"], group["step"],
+                    ps, gs, us, ms, vs, self._ws, group["lr"], beta1,
+                    beta2, group["eps"], group["step"],
                     1 if group["bias_correction"] else 0,
                     group["weight_decay"], [], 1.0)
             else:
