@@ -119,5 +119,7 @@ def test_fp_quantize_gpu_matches_cpu_codec(q_bits):
     yc = cpu.dequantize(cpu.quantize(x.float()))
     gpu = FP_Quantize(group_size=256, q_bits=q_bits)
     yg = gpu.dequantize(gpu.quantize(x.cuda())).cpu()
-    assert torch.allclose(yc.float(), yg.float(), atol=1e-4), \
-        (yc.float() - yg.float()).abs().max()
+    # GPU path returns bf16; compare after identical final rounding
+    assert torch.allclose(yc.to(torch.bfloat16).float(), yg.float(),
+                          atol=1e-5), \
+        (yc.to(torch.bfloat16).float() - yg.float()).abs().max()
