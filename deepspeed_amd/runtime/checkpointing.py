@@ -27,6 +27,29 @@ def _zero_ckpt_name(dp_rank, mp_rank=0):
     return f"zero_pp_rank_{dp_rank}_mp_rank_{mp_rank:02d}_optim_states.pt"
 
 
+def _expert_ckpt_name(ep_rank, mp_rank=0):
+    return f"expert_ep_rank_{ep_rank}_mp_rank_{mp_rank:02d}_model_states.pt"
+
+
+def _ensure_expert_groups(gn):
+    """Expert process groups are created lazily on first forward; a
+    checkpoint before any step must create them (collective)."""
+    from ..comm import groups
+    if gn not in groups.get_expert_parallel_group_dict():
+        groups.create_expert_and_data_parallel(int(gn.rsplit("_", 1)[1]))
+
+
+def _expert_param_names(module):
+    """Params tagged by the MoE Experts bank (p.allreduce=False,
+    p.group_name=<ep group>). Returns {name: group_name}."""
+    out = {}
+    for n, p in module.named_parameters():
+        gn = getattr(p, "group_name", None)
+        if gn is not None and not getattr(p, "allreduce", True):
+            out[n] = gn
+    return out
+
+
 def save_checkpoint(engine, save_dir, tag=None, client_state=None,
                     save_latest=True, exclude_frozen_parameters=False):
     tag = _tag(engine, tag)
@@ -35,6 +58,22 @@ def save_checkpoint(engine, save_dir, tag=None, client_state=None,
     dp_rank = engine.get_data_parallel_rank()
     is_zero = hasattr(engine.optimizer, "state_dict") and \
         not isinstance(engine.optimizer, torch.optim.Optimizer)
+
+    # MoE: expert params differ per expert-parallel rank — they go into
+    # separate expert_ep_rank files (ref _save_moe_checkpoint,
+    # runtime/engine.py:4921) and are excluded from the dense states.
+    expert_names = _expert_param_names(engine.module)
+    if expert_names:
+        from ..comm import groups
+        full_sd = engine.module.state_dict()
+        gn = next(iter(expert_names.values()))
+        _ensure_expert_groups(gn)
+        ep_rank = groups.get_expert_parallel_rank(gn)
+        edp_rank = dist.get_rank(groups.get_expert_data_parallel_group(gn))
+        if edp_rank == 0:
+            esd = {k: v for k, v in full_sd.items() if k in expert_names}
+            torch.save({"module": esd, "ds_version": VERSION},
+                       os.path.join(ckpt_dir, _expert_ckpt_name(ep_rank)))
 
     # model states: rank 0 of each DP group (mp_rank 0 — no TP yet)
     if dp_rank == 0:
@@ -56,6 +95,9 @@ def save_checkpoint(engine, save_dir, tag=None, client_state=None,
         }
         if client_state:
             state.update(client_state)
+        if expert_names and state.get("module"):
+            state["module"] = {k: v for k, v in state["module"].items()
+                               if k not in expert_names}
         torch.save(state, os.path.join(ckpt_dir, _model_states_name()))
 
     # zero shards: every dp rank
@@ -90,9 +132,20 @@ def load_checkpoint(engine, load_dir, tag=None, load_module_strict=True,
     is_zero = hasattr(engine.optimizer, "load_state_dict") and \
         not isinstance(engine.optimizer, torch.optim.Optimizer)
 
+    expert_names = _expert_param_names(engine.module)
     if engine.zero_optimization_stage() != 3:
         engine.load_module_state_dict(state["module"],
-                                      strict=load_module_strict)
+                                      strict=load_module_strict
+                                      and not expert_names)
+    if expert_names:
+        from ..comm import groups
+        gn = next(iter(expert_names.values()))
+        _ensure_expert_groups(gn)
+        ep_rank = groups.get_expert_parallel_rank(gn)
+        efile = os.path.join(ckpt_dir, _expert_ckpt_name(ep_rank))
+        if os.path.exists(efile):
+            esd = torch.load(efile, map_location="cpu", weights_only=False)
+            engine.module.load_state_dict(esd["module"], strict=False)
 
     if not load_module_only:
         if is_zero:

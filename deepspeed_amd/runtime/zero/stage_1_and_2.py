@@ -162,7 +162,15 @@ class ZeroStage12Optimizer:
                 try:
                     epg = grp.get_expert_data_parallel_group(gn)
                 except KeyError:
-                    epg = self.dp_group  # ep groups not built (ep_size==1)
+                    # groups are normally built lazily on first forward;
+                    # the optimizer partitions BEFORE that, so build them
+                    # here (collective — every rank reaches this point)
+                    try:
+                        ep_size = int(gn.rsplit("_", 1)[1])
+                        grp.create_expert_and_data_parallel(ep_size)
+                        epg = grp.get_expert_data_parallel_group(gn)
+                    except (ValueError, KeyError, AssertionError):
+                        epg = self.dp_group
                 self._build_buckets(eparams, gi, pg=epg)
             # swap group params for the fp32 masters of this group's buckets
             group["params"] = [b.master32 for b in self.buckets
@@ -440,8 +448,9 @@ class ZeroStage12Optimizer:
             b.master32.data.copy_(s.data)
             b.shard16.copy_(b.master32.detach())
         for b in self.buckets:
-            dist.all_gather_into_tensor(b.flat16, b.shard16,
-                                        group=self.dp_group)
+            dist.all_gather_into_tensor(
+                b.flat16, b.shard16,
+                group=b.pg if b.pg is not None else self.dp_group)
 
     def refresh_fp32_params(self):
         """Re-copy 16-bit params into fp32 masters (after external load)."""

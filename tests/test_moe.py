@@ -98,3 +98,61 @@ def _moe_ep_equivalence():
 def test_moe_ep2_equivalence():
     results = run_distributed(_moe_ep_equivalence, world_size=2)
     assert all(results)
+
+
+def _moe_expert_ckpt():
+    import os
+    import tempfile
+    import torch.distributed as dist
+    import deepspeed_amd as ds
+    from deepspeed_amd.moe.layer import MoE
+
+    torch.manual_seed(dist.get_rank())
+    M = 16
+
+    class Net(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.inp = torch.nn.Linear(M, M)
+            self.moe = MoE(M, torch.nn.Linear(M, M), num_experts=4,
+                           ep_size=2, k=1, capacity_factor=8.0)
+            self.out = torch.nn.Linear(M, 4)
+
+        def forward(self, x):
+            h = self.inp(x)
+            h, _, _ = self.moe(h)
+            return self.out(h)
+
+    cfg = {"train_micro_batch_size_per_gpu": 2,
+           "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+           "bf16": {"enabled": True},
+           "zero_optimization": {"stage": 1}}
+    engine, _, _, _ = ds.initialize(model=Net(), config=cfg)
+    # make expert weights rank-distinct
+    before = {n: p.detach().clone()
+              for n, p in engine.module.named_parameters()
+              if getattr(p, "group_name", None) is not None}
+    assert before, "no expert params found"
+    tmp = tempfile.mkdtemp(prefix=f"moeckpt_shared")
+    # all ranks must agree on the dir: broadcast rank0's
+    obj = [tmp]
+    dist.broadcast_object_list(obj, src=0)
+    tmp = obj[0]
+    engine.save_checkpoint(tmp, tag="t0")
+    files = sorted(os.listdir(os.path.join(tmp, "t0")))
+    assert any(f.startswith("expert_ep_rank_0") for f in files), files
+    assert any(f.startswith("expert_ep_rank_1") for f in files), files
+    # perturb experts, then reload and verify restoration
+    with torch.no_grad():
+        for n, p in engine.module.named_parameters():
+            if n in before:
+                p.add_(1.0)
+    engine.load_checkpoint(tmp, tag="t0")
+    for n, p in engine.module.named_parameters():
+        if n in before:
+            assert torch.allclose(p.detach().float(),
+                                  before[n].float(), atol=1e-6), n
+
+
+def test_moe_expert_checkpoint_ep2():
+    run_distributed(_moe_expert_ckpt, world_size=2)
