@@ -87,3 +87,59 @@ def compute_elastic_config(ds_config, target_deepspeed_version=None,
     if return_microbatch:
         return final_batch, valid_gpus, None
     return final_batch, valid_gpus
+
+
+def _try_import_elastic_agent():
+    try:
+        from torch.distributed.elastic.agent.server.local_elastic_agent \
+            import LocalElasticAgent
+        return LocalElasticAgent
+    except ImportError:
+        return None
+
+
+_Base = _try_import_elastic_agent()
+
+if _Base is not None:
+
+    class DSElasticAgent(_Base):
+        """Elastic agent with DeepSpeed worker environment.
+
+        Parity: reference `elasticity/elastic_agent.py:32` (DSElasticAgent
+        overriding `_set_master_addr_port` / worker env so restarted
+        workers re-enter `deepspeed.initialize` cleanly at the new scale).
+        MI355X note: rendezvous over 127.0.0.1 on single-node pools (the
+        container hostname may not resolve); on restart, workers should
+        reload the latest checkpoint and call `compute_elastic_config`
+        with the surviving GPU count to re-derive the batch triple.
+        """
+
+        def __init__(self, spec, logs_specs=None, start_method="spawn",
+                     exit_barrier_timeout=300, log_line_prefix_template=None,
+                     ds_env=None, **kw):
+            self._ds_env = dict(ds_env or {})
+            try:
+                super().__init__(
+                    spec, logs_specs=logs_specs, start_method=start_method,
+                    exit_barrier_timeout=exit_barrier_timeout,
+                    log_line_prefix_template=log_line_prefix_template, **kw)
+            except TypeError:  # older torch signature
+                super().__init__(spec, start_method=start_method,
+                                 exit_barrier_timeout=exit_barrier_timeout)
+
+        def _start_workers(self, worker_group):
+            import os
+            spec = worker_group.spec
+            for k, v in self._ds_env.items():
+                os.environ.setdefault(k, str(v))
+            os.environ.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+            # one process per GPU: LOCAL_RANK -> HIP device is implicit
+            os.environ.setdefault("NCCL_MIN_NCHANNELS", "32")
+            worker_ids = super()._start_workers(worker_group)
+            from .utils.logging import logger
+            logger.info(
+                f"DSElasticAgent: (re)started {spec.local_world_size} "
+                f"workers for group role={spec.role}")
+            return worker_ids
+else:  # pragma: no cover - torch without elastic
+    DSElasticAgent = None

@@ -201,3 +201,32 @@ def call_to_str(base, *args, **kwargs):
         name += ", ".join(f"{k}={v!r}" for k, v in kwargs.items())
     name += ")"
     return name
+
+
+def assert_ints_same_as_other_ranks(ints, group=None, tag=""):
+    """Debug-mode cross-rank agreement check (ref stage3.py:1594).
+
+    Collectives deadlock or silently corrupt when ranks disagree on the
+    order/identity of the tensors they reduce; this asserts the given
+    list of ints (e.g. param ids in a reduction bucket) is identical on
+    every rank of the group. Enable via DSAMD_SANITY=1 — it costs one
+    small all_gather per call, cheap insurance on a 7-link async fabric.
+    """
+    from .. import comm as dist
+    if not dist.is_initialized():
+        return
+    world = dist.get_world_size(group)
+    if world <= 1:
+        return
+    t = torch.tensor(list(ints), dtype=torch.long)
+    if torch.cuda.is_available():
+        t = t.cuda()
+    sizes = [torch.zeros(1, dtype=torch.long, device=t.device)
+             for _ in range(world)]
+    dist.all_gather(sizes, torch.tensor([t.numel()], dtype=torch.long,
+                                        device=t.device), group=group)
+    assert all(s.item() == t.numel() for s in sizes),         f"[sanity:{tag}] rank list LENGTHS differ: "         f"{[s.item() for s in sizes]}"
+    gathered = [torch.zeros_like(t) for _ in range(world)]
+    dist.all_gather(gathered, t, group=group)
+    for r, g in enumerate(gathered):
+        assert torch.equal(g, gathered[0]),             f"[sanity:{tag}] rank {r} disagrees with rank 0 on ids"

@@ -211,6 +211,9 @@ class ZeroStage3Optimizer:
         self._inflight = {}         # module -> AllGatherHandle
 
         self._global_grad_norm = 0.0
+        # opt-in cross-rank sanity asserts (DSAMD_SANITY=1): verify every
+        # rank reduces the same params in the same order
+        self._sanity = _os.environ.get("DSAMD_SANITY") == "1"
         log_dist(
             f"ZeRO-3: {sum(p.ds_numel for p in self._all_params)/1e9:.2f}B "
             f"params, {len(self.sub_groups)} sub-groups, world {self.world}",
@@ -465,6 +468,11 @@ class ZeroStage3Optimizer:
         params = self._ipg_params
         self._ipg_params = []
         self._ipg_numel = 0
+        if self._sanity:
+            from ..utils import assert_ints_same_as_other_ranks
+            assert_ints_same_as_other_ranks(
+                [p.ds_id for p in params], group=self.dp_group,
+                tag="ipg_flush")
         world = self.world
         if self.rs_stream is not None:
             # grads were produced on the default stream

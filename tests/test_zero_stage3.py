@@ -142,3 +142,18 @@ def test_zero3_mics_shard1_matches_reference():
                 (g - e).abs().max()
     for g0, g1 in zip(results[0], results[1]):
         assert torch.equal(g0, g1)  # replicas identical
+
+
+def _zero3_sanity_mode(steps=2, grad_accum=1, persist_threshold=10):
+    import os
+    os.environ["DSAMD_SANITY"] = "1"
+    try:
+        return _zero3_train(steps=steps, grad_accum=grad_accum,
+                            persist_threshold=persist_threshold)
+    finally:
+        os.environ.pop("DSAMD_SANITY", None)
+
+
+def test_zero3_sanity_asserts_world2():
+    """DSAMD_SANITY cross-rank id checks pass on an honest run."""
+    run_distributed(_zero3_sanity_mode, world_size=2)
