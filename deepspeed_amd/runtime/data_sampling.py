@@ -100,3 +100,107 @@ class DeepSpeedDataSampler:
     def load_state_dict(self, sd):
         self.global_step = sd["global_step"]
         self.scheduler.load_state_dict(sd["scheduler"])
+
+
+class DataAnalyzer:
+    """Offline map-reduce over a dataset computing per-sample difficulty
+    metrics for curriculum learning.
+
+    Parity: reference `runtime/data_pipeline/data_analyzer.py`
+    (DataAnalyzer/DistributedDataAnalyzer): each worker scans its shard
+    of sample indices, applies `metric_function` per batch, and the
+    results merge into (a) `<metric>_sample_to_metric` — metric value per
+    sample index — and (b) `<metric>_metric_to_sample` — sample indices
+    bucketed by metric value — both stored as IndexedDatasets so the
+    curriculum sampler can mmap them.
+
+    When torch.distributed is initialized the sample range splits across
+    ranks and rank 0 merges (gloo-friendly; the scan is CPU work).
+    """
+
+    def __init__(self, dataset, metric_names, metric_functions,
+                 output_path, batch_size=64, metric_dtype=np.int64):
+        assert len(metric_names) == len(metric_functions)
+        self.dataset = dataset
+        self.metric_names = metric_names
+        self.metric_functions = metric_functions
+        self.output_path = output_path
+        self.batch_size = batch_size
+        self.metric_dtype = metric_dtype
+
+    def _my_range(self):
+        import torch.distributed as dist
+        n = len(self.dataset)
+        if dist.is_initialized() and dist.get_world_size() > 1:
+            world, rank = dist.get_world_size(), dist.get_rank()
+            per = (n + world - 1) // world
+            return range(rank * per, min(n, (rank + 1) * per))
+        return range(n)
+
+    def run_map_reduce(self):
+        import torch.distributed as dist
+        os.makedirs(self.output_path, exist_ok=True)
+        local = {m: [] for m in self.metric_names}
+        idxs = list(self._my_range())
+        for b0 in range(0, len(idxs), self.batch_size):
+            batch_idx = idxs[b0:b0 + self.batch_size]
+            batch = [self.dataset[i] for i in batch_idx]
+            for name, fn in zip(self.metric_names, self.metric_functions):
+                vals = fn(batch)
+                if torch.is_tensor(vals):
+                    vals = vals.tolist()
+                local[name].extend(int(v) for v in vals)
+        world = dist.get_world_size() if dist.is_initialized() else 1
+        if world > 1:
+            gathered = [None] * world
+            dist.all_gather_object(gathered, (list(self._my_range()),
+                                              local))
+            if dist.get_rank() != 0:
+                dist.barrier()
+                return self.output_path
+            merged = {m: {} for m in self.metric_names}
+            for rng, loc in gathered:
+                for m in self.metric_names:
+                    for i, v in zip(rng, loc[m]):
+                        merged[m][i] = v
+            final = {m: [merged[m][i] for i in range(len(self.dataset))]
+                     for m in self.metric_names}
+        else:
+            final = local
+        for m in self.metric_names:
+            vals = final[m]
+            b = IndexedDatasetBuilder(
+                os.path.join(self.output_path, f"{m}_sample_to_metric"),
+                dtype=np.int64)
+            for v in vals:
+                b.add_item([v])
+            b.finalize()
+            buckets = {}
+            for i, v in enumerate(vals):
+                buckets.setdefault(v, []).append(i)
+            b = IndexedDatasetBuilder(
+                os.path.join(self.output_path, f"{m}_metric_to_sample"),
+                dtype=np.int64)
+            self._bucket_keys = sorted(buckets)
+            for k in sorted(buckets):
+                b.add_item(buckets[k])
+            b.finalize()
+            with open(os.path.join(self.output_path,
+                                   f"{m}_metric_values.txt"), "w") as f:
+                f.write("\n".join(str(k) for k in sorted(buckets)))
+        if world > 1:
+            dist.barrier()
+        return self.output_path
+
+
+def load_metric_index(output_path, metric_name):
+    """(sample_to_metric IndexedDataset, sorted metric values,
+    metric_to_sample IndexedDataset) for a DataAnalyzer output."""
+    s2m = IndexedDataset(
+        os.path.join(output_path, f"{metric_name}_sample_to_metric"))
+    m2s = IndexedDataset(
+        os.path.join(output_path, f"{metric_name}_metric_to_sample"))
+    with open(os.path.join(output_path,
+                           f"{metric_name}_metric_values.txt")) as f:
+        values = [int(x) for x in f.read().split()]
+    return s2m, values, m2s

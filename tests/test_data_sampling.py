@@ -37,3 +37,25 @@ def test_curriculum_sampler_truncates():
     assert lens[0] == 4            # min difficulty at step 0
     assert lens[-1] == 16          # full length at the end
     assert all(a <= b for a, b in zip(lens, lens[1:]))  # monotone
+
+
+def test_data_analyzer_map_reduce(tmp_path):
+    import numpy as np
+    from deepspeed_amd.runtime.data_sampling import (DataAnalyzer,
+                                                     load_metric_index)
+    data = [list(range(3 + (i % 5))) for i in range(23)]  # var-length docs
+
+    def seqlen_metric(batch):
+        return [len(x) for x in batch]
+
+    an = DataAnalyzer(data, ["seqlen"], [seqlen_metric], str(tmp_path),
+                      batch_size=4)
+    an.run_map_reduce()
+    s2m, values, m2s = load_metric_index(str(tmp_path), "seqlen")
+    assert len(s2m) == 23
+    assert [int(s2m[i][0]) for i in range(23)] == [3 + (i % 5)
+                                                   for i in range(23)]
+    assert values == [3, 4, 5, 6, 7]
+    # bucket for value 3 holds exactly the samples with len 3
+    b3 = [int(x) for x in m2s[0]]
+    assert b3 == [i for i in range(23) if (i % 5) == 0]
