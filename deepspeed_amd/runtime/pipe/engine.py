@@ -168,6 +168,8 @@ class PipelineEngine(DeepSpeedEngine):
         self._bufs["grads"][b] = buf
 
     def _exec_reduce_grads(self, cmd):
+        if getattr(self.module, "tied_comms", None):
+            self.module.allreduce_tied_weight_gradients()
         if hasattr(self.optimizer, "set_accumulation_boundary"):
             self.optimizer.set_accumulation_boundary(True)
         if hasattr(self.optimizer, "reduce_gradients"):

@@ -104,9 +104,23 @@ class PipelineModule(torch.nn.Module):
         self._local_stop = stop
         self.forward_funcs = []
         mods = torch.nn.ModuleDict()
+        tied_local = {}   # key -> module instance (same-stage weight share)
         for i, spec in enumerate(self._layer_specs[start:stop]):
             idx = start + i
-            if isinstance(spec, LayerSpec):
+            if isinstance(spec, TiedLayerSpec):
+                if spec.key in tied_local:
+                    mod = tied_local[spec.key]  # share the instance
+                else:
+                    mod = spec.build()
+                    tied_local[spec.key] = mod
+                    mods[str(idx)] = mod
+                if spec.forward_fn is not None:
+                    fn = spec.forward_fn
+                    self.forward_funcs.append(
+                        (lambda m, f: (lambda x: f(m, x)))(mod, fn))
+                else:
+                    self.forward_funcs.append(mod)
+            elif isinstance(spec, LayerSpec):
                 mod = spec.build()
                 mods[str(idx)] = mod
                 self.forward_funcs.append(mod)
@@ -116,6 +130,52 @@ class PipelineModule(torch.nn.Module):
             else:  # plain callable (e.g. lambda reshaping)
                 self.forward_funcs.append(spec)
         self.stage_modules = mods
+        self._tied_local = tied_local
+        self._setup_tied_comms()
+
+    def _stage_of_layer(self, idx):
+        for st in range(self.num_stages):
+            if self.parts[st] <= idx < self.parts[st + 1]:
+                return st
+        return self.num_stages - 1
+
+    def _setup_tied_comms(self):
+        """Cross-stage weight tying (ref module.py tied_comms +
+        engine tied-grad allreduce): stages holding the same TiedLayerSpec
+        key form a process group; weights broadcast from the lowest
+        owner at init and grads all-reduce (SUM) after each backward so
+        every replica takes identical optimizer steps."""
+        self.tied_comms = {}
+        keys = {}
+        for idx, spec in enumerate(self._layer_specs):
+            if isinstance(spec, TiedLayerSpec):
+                keys.setdefault(spec.key, []).append(idx)
+        for key in sorted(keys):
+            stages = sorted({self._stage_of_layer(i) for i in keys[key]})
+            if len(stages) <= 1:
+                continue
+            # every rank must take part in new_group creation
+            for dp in range(self._grid.data_parallel_size):
+                ranks = [st * self._grid.data_parallel_size + dp
+                         for st in stages]
+                grp = dist.new_group(ranks)
+                if dist.get_rank() in ranks:
+                    my_group = grp
+            if self.stage_id in stages:
+                spec = self._layer_specs[keys[key][0]]
+                mod = self._tied_local[key]
+                weight = getattr(mod, spec.tied_weight_attr)
+                self.tied_comms[key] = (my_group, weight,
+                                        min(stages), stages)
+                # initial sync: lowest owner stage broadcasts
+                src = (min(stages) * self._grid.data_parallel_size
+                       + self._grid.data_parallel_id)
+                dist.broadcast(weight.data, src=src, group=my_group)
+
+    def allreduce_tied_weight_gradients(self):
+        for key, (grp, weight, _, _) in self.tied_comms.items():
+            if weight.grad is not None:
+                dist.all_reduce(weight.grad, group=grp)
 
     def forward(self, x):
         def run(funcs, inp):

@@ -58,3 +58,12 @@ def test_zenflow_stage2_cpu():
 
 def test_zenflow_world2():
     run_distributed(_zenflow_train, world_size=2)
+
+
+import pytest  # noqa: E402
+
+
+@pytest.mark.gpu
+def test_zenflow_gpu_single():
+    """Pinned-host masters + hot-set GPU updates on a real device."""
+    run_distributed(_zenflow_train, world_size=1, backend="nccl")
