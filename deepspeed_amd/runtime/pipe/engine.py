@@ -29,11 +29,6 @@ class PipelineEngine(DeepSpeedEngine):
         self.next_rank = (self.grid.stage_to_global(self.stage_id + 1)
                           if not self.is_last_stage else None)
 
-        if self.fp16_enabled():
-            raise NotImplementedError(
-                "pipeline engine supports bf16/fp32 (fp16 loss scaling with "
-                "PP lands with the fp16 PP optimizer)")
-
         self._act_meta_sent = False
         self._recv_act_template = None
         self._recv_grad_template = None
@@ -98,6 +93,10 @@ class PipelineEngine(DeepSpeedEngine):
             inputs, labels = batch, None
         if self.is_first_stage:
             x = inputs.to(self.device)
+            dt = self._config.dtype
+            if x.is_floating_point() and \
+                    dt in (torch.float16, torch.bfloat16):
+                x = x.to(dt)
             self._bufs["inputs"][cmd.buffer_id] = x
         if self.is_last_stage and labels is not None:
             self._bufs["labels"][cmd.buffer_id] = labels.to(self.device)
@@ -126,7 +125,14 @@ class PipelineEngine(DeepSpeedEngine):
             self.optimizer.ensure_grad_views()
         if self.is_last_stage:
             loss = self._bufs["losses"][b] / self.micro_batches
-            loss.backward()
+            # fp16: scale the loss so grads stay in half range; the
+            # partitioned optimizer folds 1/scale into the fused step
+            scaler = getattr(self.optimizer, "loss_scaler", None)
+            ls = getattr(scaler, "loss_scale", 1.0) if scaler else 1.0
+            if ls != 1.0:
+                (loss.float() * ls).backward()
+            else:
+                loss.backward()
         else:
             out = self._bufs["outputs"][b]
             grad = self._bufs["grads"][b]

@@ -162,3 +162,35 @@ def test_pipeline_tied_weights_sync():
     """Tied layer on stage 0 and 1: broadcast at init, grads all-reduced,
     replicas stay bit-identical through optimizer steps."""
     run_distributed(_pipe_tied_train, world_size=2)
+
+
+def _pipe_train_fp16(steps=3):
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    from deepspeed_amd.runtime.pipe.module import PipelineModule
+    groups.reset_groups()
+    model = PipelineModule(layers=_make_specs(), num_stages=2,
+                           loss_fn=_loss_fn, partition_method="uniform")
+    config = {
+        "train_micro_batch_size_per_gpu": MICRO,
+        "gradient_accumulation_steps": GAS,
+        "optimizer": {"type": "AdamW",
+                      "params": {"lr": LR, "weight_decay": 0.0}},
+        "fp16": {"enabled": True, "loss_scale": 128.0},
+    }
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config=config)
+    it = iter(_data(steps * GAS))
+    losses = [engine.train_batch(data_iter=it) for _ in range(steps)]
+    return losses
+
+
+def test_pipeline_fp16_loss_scaled():
+    """fp16 PP with static loss scale tracks the fp32 run."""
+    res = run_distributed(_pipe_train_fp16, world_size=2)
+    ref = _reference()
+    # rank 1 (last stage) reports the loss
+    fp16_losses = [l for l in res[1][0] if l is not None] \
+        if isinstance(res[1], tuple) else res[1]
+    ref_losses = ref[0] if isinstance(ref, tuple) else ref
+    for a, b in zip(fp16_losses, ref_losses):
+        assert abs(float(a) - float(b)) < 0.05, (a, b)
