@@ -1,0 +1,173 @@
+"""Continuous batching serving engine (inference v2).
+
+Parity: reference `inference/v2/ragged/*` + `mii` continuous batching:
+requests enter and leave the running batch at token granularity — new
+prompts prefill into free KV slots while in-flight sequences keep
+decoding; every decode step runs ONE batched forward over all active
+slots at their individual sequence lengths.
+
+MI355X-native design: the KV pool is one tensor per layer
+`[max_batch, max_seq, Hk, D]` (288 GB of HBM3E holds thousands of 8B
+slots — paged/blocked KV is not needed to avoid fragmentation at this
+scale, so slot granularity keeps the layout hipGraph-friendly). Ragged
+lengths are handled with per-row RoPE positions (`positions` arg) and
+an additive length mask through SDPA; the dedicated ragged-decode HIP
+kernel (skinny GEMV + per-row KV scan) is tracked in ROADMAP.
+"""
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+import torch
+
+
+@dataclass
+class Request:
+    prompt: torch.Tensor                 # 1-D token ids
+    max_new_tokens: int = 32
+    eos_token_id: Optional[int] = None
+    rid: int = -1
+    slot: int = -1
+    generated: List[int] = field(default_factory=list)
+    done: bool = False
+
+    @property
+    def tokens(self):
+        return torch.cat([self.prompt,
+                          torch.tensor(self.generated,
+                                       dtype=self.prompt.dtype)])
+
+
+class RaggedKVCache:
+    """One layer's slot-granular KV pool with per-slot lengths."""
+
+    def __init__(self, max_batch, max_seq, n_kv, d, dtype, device):
+        self.k = torch.zeros(max_batch, max_seq, n_kv, d, dtype=dtype,
+                             device=device)
+        self.v = torch.zeros_like(self.k)
+        self.lens = torch.zeros(max_batch, dtype=torch.long, device=device)
+        self.rows = None        # active slot ids for this forward
+        self.attn_mask = None
+        self._prefill_slot = None
+
+    # -- model-facing update ------------------------------------------------
+    def update(self, k, v):
+        if self._prefill_slot is not None:   # [1, S, Hk, D] prompt chunk
+            s = self._prefill_slot
+            S = k.shape[1]
+            self.k[s, :S] = k[0]
+            self.v[s, :S] = v[0]
+            self.lens[s] = S
+            self.attn_mask = None
+            return k, v                      # causal prefill over itself
+        rows = self.rows                     # decode: [n, 1, Hk, D]
+        pos = self.lens[rows]
+        self.k[rows, pos] = k[:, 0]
+        self.v[rows, pos] = v[:, 0]
+        self.lens[rows] = pos + 1
+        lens = self.lens[rows]
+        maxlen = int(lens.max())
+        ar = torch.arange(maxlen, device=k.device)
+        valid = ar.unsqueeze(0) < lens.unsqueeze(1)          # [n, maxlen]
+        self.attn_mask = torch.where(
+            valid, 0.0, float("-inf")).view(-1, 1, 1, maxlen).float()
+        return self.k[rows, :maxlen], self.v[rows, :maxlen]
+
+
+class ContinuousBatchingEngine:
+    """Token-level continuous batching over a native model."""
+
+    def __init__(self, model, max_batch=8, max_seq=None, device=None):
+        self.model = model
+        cfg = model.cfg if hasattr(model, "cfg") else model.config
+        self.cfg = cfg
+        self.device = device or next(model.parameters()).device
+        self.max_seq = max_seq or cfg.max_position_embeddings
+        self.max_batch = max_batch
+        dtype = next(model.parameters()).dtype
+        self.caches = [RaggedKVCache(max_batch, self.max_seq,
+                                     cfg.num_key_value_heads, cfg.head_dim,
+                                     dtype, self.device)
+                       for _ in range(cfg.num_hidden_layers)]
+        self.free_slots = list(range(max_batch))
+        self.pending: List[Request] = []
+        self.running: List[Request] = []
+        self._next_rid = 0
+        self._was_ckpt = getattr(cfg, "activation_checkpointing", False)
+        cfg.activation_checkpointing = False
+
+    def add_request(self, prompt, max_new_tokens=32, eos_token_id=None):
+        req = Request(prompt=prompt.to("cpu").long().view(-1),
+                      max_new_tokens=max_new_tokens,
+                      eos_token_id=eos_token_id, rid=self._next_rid)
+        self._next_rid += 1
+        self.pending.append(req)
+        return req.rid
+
+    # -- internals ----------------------------------------------------------
+    @torch.no_grad()
+    def _prefill(self, req):
+        req.slot = self.free_slots.pop()
+        for c in self.caches:
+            c._prefill_slot = req.slot
+        ids = req.prompt.view(1, -1).to(self.device)
+        logits = self.model(ids, kv_caches=self.caches)
+        for c in self.caches:
+            c._prefill_slot = None
+        nxt = int(logits[0, -1].float().argmax())
+        req.generated.append(nxt)
+        self._check_done(req)
+
+    @torch.no_grad()
+    def _decode(self, active):
+        rows = torch.tensor([r.slot for r in active], dtype=torch.long,
+                            device=self.device)
+        for c in self.caches:
+            c.rows = rows
+        last = torch.tensor([[r.generated[-1]] for r in active],
+                            dtype=torch.long, device=self.device)
+        positions = self.caches[0].lens[rows].view(-1, 1)
+        logits = self.model(last, kv_caches=self.caches,
+                            positions=positions)
+        for c in self.caches:
+            c.rows = None
+        nxt = logits[:, -1].float().argmax(-1)
+        for r, t in zip(active, nxt.tolist()):
+            r.generated.append(int(t))
+            self._check_done(r)
+
+    def _check_done(self, req):
+        if len(req.generated) >= req.max_new_tokens or \
+                (req.eos_token_id is not None
+                 and req.generated[-1] == req.eos_token_id) or \
+                len(req.prompt) + len(req.generated) >= self.max_seq:
+            req.done = True
+
+    # -- public loop --------------------------------------------------------
+    @torch.no_grad()
+    def step(self):
+        """Admit + one decode step. Returns requests finished this step."""
+        while self.pending and self.free_slots:
+            req = self.pending.pop(0)
+            self._prefill(req)
+            self.running.append(req)
+        active = [r for r in self.running if not r.done]
+        if active:
+            self._decode(active)
+        finished = [r for r in self.running if r.done]
+        for r in finished:
+            self.free_slots.append(r.slot)
+            self.running.remove(r)
+        return finished
+
+    def run(self, max_steps=10000):
+        """Drain all pending/running requests; returns rid -> tokens."""
+        out = {}
+        steps = 0
+        while (self.pending or self.running) and steps < max_steps:
+            for r in self.step():
+                out[r.rid] = r.tokens
+            steps += 1
+        return out
+
+    def has_capacity(self):
+        return bool(self.free_slots)

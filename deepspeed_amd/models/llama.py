@@ -85,8 +85,10 @@ class LlamaAttention(nn.Module):
         k = apply_rope(k, cos, sin)
         if kv_cache is not None:
             k, v = kv_cache.update(k, v)
+            mask = getattr(kv_cache, "attn_mask", None)
             causal = S > 1  # prefill chunk is causal; decode sees all past
-            o = flash_attention(q, k, v, causal=causal)
+            o = flash_attention(q, k, v, causal=causal and mask is None,
+                                attn_mask=mask)
         elif self._dist_attn is not None:
             o = self._dist_attn(q, k, v, causal=True)
         else:
@@ -220,11 +222,16 @@ class LlamaModel(nn.Module):
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
 
-    def forward(self, input_ids, seq_offset=0, kv_caches=None):
+    def forward(self, input_ids, seq_offset=0, kv_caches=None,
+                positions=None):
         x = self.embed_tokens(input_ids)
         S = input_ids.shape[1]
-        cos = self.rope_cos[seq_offset:seq_offset + S]
-        sin = self.rope_sin[seq_offset:seq_offset + S]
+        if positions is not None:  # per-row positions (ragged decode)
+            cos = self.rope_cos[positions]
+            sin = self.rope_sin[positions]
+        else:
+            cos = self.rope_cos[seq_offset:seq_offset + S]
+            sin = self.rope_sin[seq_offset:seq_offset + S]
         for i, layer in enumerate(self.layers):
             if self.cfg.activation_checkpointing and self.training:
                 x = torch.utils.checkpoint.checkpoint(
@@ -254,8 +261,10 @@ class LlamaForCausalLM(nn.Module):
         elif isinstance(module, nn.Embedding):
             module.weight.data.normal_(0.0, std)
 
-    def forward(self, input_ids, labels=None, seq_offset=0, kv_caches=None):
-        h = self.model(input_ids, seq_offset=seq_offset, kv_caches=kv_caches)
+    def forward(self, input_ids, labels=None, seq_offset=0, kv_caches=None,
+                positions=None):
+        h = self.model(input_ids, seq_offset=seq_offset, kv_caches=kv_caches,
+                       positions=positions)
         logits = self.lm_head(h)
         if labels is None:
             return logits

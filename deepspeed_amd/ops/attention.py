@@ -104,9 +104,18 @@ class _FlashAttnFn(torch.autograd.Function):
 _USE_HIP_FLASH = os.environ.get("DSAMD_FLASH", "1") == "1"
 
 
-def flash_attention(q, k, v, causal=True):
-    """q [B,S,Hq,D], k/v [B,S,Hkv,D] -> [B,S,Hq,D]. GQA-aware."""
+def flash_attention(q, k, v, causal=True, attn_mask=None):
+    """q [B,S,Hq,D], k/v [B,S,Hkv,D] -> [B,S,Hq,D]. GQA-aware.
+
+    attn_mask: optional additive float mask broadcastable to
+    [B, Hq, Sq, Sk] (ragged-batch decode); routes through SDPA."""
     scale = 1.0 / math.sqrt(q.shape[-1])
+    if attn_mask is not None:
+        out = F.scaled_dot_product_attention(
+            q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2),
+            attn_mask=attn_mask, is_causal=False,
+            enable_gqa=(k.shape[2] != q.shape[2]))
+        return out.transpose(1, 2)
     if _USE_HIP_FLASH and q.is_cuda and q.dtype == torch.bfloat16 \
             and q.shape[-1] == 128:
         return _FlashAttnFn.apply(q.contiguous(), k.contiguous(),

@@ -109,18 +109,18 @@ class _RoPEFn(torch.autograd.Function):
         cos = cos.float()
         sin = sin.float()
         ctx.save_for_backward(cos, sin)
-        if t.is_cuda and t.dtype == torch.bfloat16:
+        if t.is_cuda and t.dtype == torch.bfloat16 and cos.dim() == 2:
             ext = get_ext(required=True)
             out = torch.empty_like(t)
             ext.rope(out, t.contiguous(), cos, sin, 0, False)
-        else:
+        else:  # eager path also serves per-row positions (cos [B,S,half])
             out = _rope_torch(t, cos, sin, sign=1.0)
         return out
 
     @staticmethod
     def backward(ctx, dy):
         cos, sin = ctx.saved_tensors
-        if dy.is_cuda and dy.dtype == torch.bfloat16:
+        if dy.is_cuda and dy.dtype == torch.bfloat16 and cos.dim() == 2:
             ext = get_ext(required=True)
             dx = torch.empty_like(dy)
             ext.rope(dx, dy.contiguous(), cos, sin, 0, True)
@@ -135,8 +135,12 @@ def _rope_torch(t, cos, sin, sign):
     t32 = t.float()
     x1 = t32[..., :half]
     x2 = t32[..., half:]
-    c = cos[:S].reshape(1, S, 1, half)
-    s = sin[:S].reshape(1, S, 1, half) * sign
+    if cos.dim() == 3:  # per-row positions: [B, S, half]
+        c = cos.reshape(B, S, 1, half)
+        s = sin.reshape(B, S, 1, half) * sign
+    else:
+        c = cos[:S].reshape(1, S, 1, half)
+        s = sin[:S].reshape(1, S, 1, half) * sign
     o1 = x1 * c - x2 * s
     o2 = x2 * c + x1 * s
     return torch.cat([o1, o2], dim=-1).to(t.dtype)
