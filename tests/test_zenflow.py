@@ -22,8 +22,9 @@ def _zenflow_train():
     engine, opt, _, _ = ds.initialize(model=model, config=cfg)
     from deepspeed_amd.runtime.zenflow import ZenFlowZeroOptimizer
     assert isinstance(engine.optimizer, ZenFlowZeroOptimizer)
-    x = torch.randn(4, 64).bfloat16()
-    y = torch.randn(4, 8).bfloat16()
+    dev = "cuda" if torch.cuda.is_available() else "cpu"
+    x = torch.randn(4, 64, device=dev).bfloat16()
+    y = torch.randn(4, 8, device=dev).bfloat16()
     losses = []
     shard_before_lazy = None
     zf = engine.optimizer
