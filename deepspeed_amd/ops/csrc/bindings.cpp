@@ -76,6 +76,9 @@ std::vector<at::Tensor> quantize_int8(at::Tensor x, long group_size);
 at::Tensor dequantize_int8(at::Tensor q, at::Tensor scales, long group_size);
 std::vector<at::Tensor> quantize_fp8(at::Tensor x, long group_size);
 at::Tensor dequantize_fp8(at::Tensor q, at::Tensor scales, long group_size);
+std::vector<at::Tensor> quantize_int4(at::Tensor x, long group);
+at::Tensor dequantize_int4(at::Tensor q, at::Tensor scales, long group,
+                           long numel);
 std::vector<at::Tensor> quantize_fp_em(at::Tensor x, long qbits, long group);
 at::Tensor dequantize_fp_em(at::Tensor q, at::Tensor scales, long qbits,
                             long group, std::vector<long> shape);
@@ -99,6 +102,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dequantize_int8", &dequantize_int8);
   m.def("quantize_fp8", &quantize_fp8);
   m.def("dequantize_fp8", &dequantize_fp8);
+  m.def("quantize_int4", &quantize_int4, py::arg("x"),
+        py::arg("group") = 2048);
+  m.def("dequantize_int4", &dequantize_int4, py::arg("q"),
+        py::arg("scales"), py::arg("group"), py::arg("numel"));
   m.def("quantize_fp_em", &quantize_fp_em, py::arg("x"), py::arg("qbits"),
         py::arg("group") = 512);
   m.def("dequantize_fp_em", &dequantize_fp_em, py::arg("q"),

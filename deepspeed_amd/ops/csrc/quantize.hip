@@ -305,3 +305,88 @@ at::Tensor dequantize_fp_em(at::Tensor q, at::Tensor scales, long qbits,
   std::vector<int64_t> sh(shape.begin(), shape.end());
   return x.reshape(sh);
 }
+
+// symmetric int4, packed two nibbles per byte, per-group fp scale
+__global__ void quantize_int4_kernel(const short* __restrict__ x,
+                                     unsigned char* __restrict__ q,
+                                     float* __restrict__ scales,
+                                     long long n, int group) {
+  long long g = blockIdx.x;
+  const short* xg = x + g * group;
+  unsigned char* qg = q + g * (group / 2);
+  long long rem = n - g * group;
+  int len = rem < group ? (int)rem : group;
+  __shared__ float red[4];
+  float amax = 0.f;
+  for (int i = threadIdx.x; i < len; i += blockDim.x)
+    amax = fmaxf(amax, fabsf(bf2f(xg[i])));
+  amax = block_reduce_max<256>(amax, red);
+  __shared__ float s_scale;
+  if (threadIdx.x == 0) {
+    float sc = amax > 0.f ? amax / 7.f : 1.f;
+    s_scale = sc;
+    scales[g] = sc;
+  }
+  __syncthreads();
+  float inv = 1.f / s_scale;
+  for (int i = threadIdx.x; i * 2 < len; i += blockDim.x) {
+    int a = (int)rintf(bf2f(xg[2 * i]) * inv);
+    a = a < -7 ? -7 : (a > 7 ? 7 : a);
+    int b = 0;
+    if (2 * i + 1 < len) {
+      b = (int)rintf(bf2f(xg[2 * i + 1]) * inv);
+      b = b < -7 ? -7 : (b > 7 ? 7 : b);
+    }
+    qg[i] = (unsigned char)((a & 0xf) | ((b & 0xf) << 4));
+  }
+}
+
+__global__ void dequantize_int4_kernel(const unsigned char* __restrict__ q,
+                                       const float* __restrict__ scales,
+                                       short* __restrict__ x, long long n,
+                                       int group) {
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (2 * i >= n) return;
+  float sc = scales[(2 * i) / group];
+  unsigned char byte = q[i];
+  int a = (int)(byte & 0xf);
+  if (a > 7) a -= 16;
+  int b = (int)(byte >> 4);
+  if (b > 7) b -= 16;
+  x[2 * i] = f2bf(a * sc);
+  if (2 * i + 1 < n) x[2 * i + 1] = f2bf(b * sc);
+}
+
+std::vector<at::Tensor> quantize_int4(at::Tensor x, long group) {
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 && x.is_contiguous());
+  TORCH_CHECK(group % 2 == 0);
+  long long n = x.numel();
+  long long ngroups = (n + group - 1) / group;
+  auto q = at::empty({ngroups * (group / 2)}, x.options().dtype(at::kByte));
+  auto scales = at::empty({ngroups}, x.options().dtype(at::kFloat));
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(quantize_int4_kernel, dim3(ngroups), dim3(256), 0,
+                     stream.stream(),
+                     reinterpret_cast<const short*>(x.data_ptr()),
+                     reinterpret_cast<unsigned char*>(q.data_ptr()),
+                     scales.data_ptr<float>(), n, (int)group);
+  HIP_CHECK_KERNEL();
+  return {q, scales};
+}
+
+at::Tensor dequantize_int4(at::Tensor q, at::Tensor scales, long group,
+                           long numel) {
+  auto x = at::empty({numel}, q.options().dtype(at::kBFloat16));
+  auto stream = c10::hip::getCurrentHIPStream();
+  long long pairs = (numel + 1) / 2;
+  int block = 256;
+  long long grid = (pairs + block - 1) / block;
+  hipLaunchKernelGGL(dequantize_int4_kernel, dim3(grid), dim3(block), 0,
+                     stream.stream(),
+                     reinterpret_cast<const unsigned char*>(q.data_ptr()),
+                     scales.data_ptr<float>(),
+                     reinterpret_cast<short*>(x.data_ptr()), numel,
+                     (int)group);
+  HIP_CHECK_KERNEL();
+  return x;
+}

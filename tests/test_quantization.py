@@ -123,3 +123,27 @@ def test_fp_quantize_gpu_matches_cpu_codec(q_bits):
     assert torch.allclose(yc.to(torch.bfloat16).float(), yg.float(),
                           atol=1e-5), \
         (yc.to(torch.bfloat16).float() - yg.float()).abs().max()
+
+
+def test_int4_roundtrip_cpu():
+    from deepspeed_amd.ops.quantizer import dequantize_int4, quantize_int4
+    torch.manual_seed(0)
+    x = (torch.randn(33, 64) * 2).bfloat16()
+    q, scales, shape = quantize_int4(x.float(), group_size=128)
+    y = dequantize_int4(q, scales, shape, group_size=128)
+    rel = (x.float() - y.float()).abs().max() / x.float().abs().max()
+    assert y.shape == x.shape
+    assert rel < 0.12, rel
+
+
+@pytest.mark.gpu
+def test_int4_gpu_matches_cpu():
+    from deepspeed_amd.ops.quantizer import dequantize_int4, quantize_int4
+    torch.manual_seed(0)
+    x = (torch.randn(4096) * 3).bfloat16()
+    qc, sc, shp = quantize_int4(x.float(), group_size=256)
+    yc = dequantize_int4(qc, sc, shp, group_size=256)
+    qg, sg, _ = quantize_int4(x.cuda(), group_size=256)
+    yg = dequantize_int4(qg, sg, shp, group_size=256).cpu()
+    assert torch.allclose(yc.float(), yg.float(), atol=1e-5), \
+        (yc.float() - yg.float()).abs().max()
