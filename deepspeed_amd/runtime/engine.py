@@ -525,6 +525,22 @@ class DeepSpeedEngine(torch.nn.Module):
                      load_lr_scheduler_states=load_lr_scheduler_states,
                      load_module_only=load_module_only)
 
+    def compile(self, sample_input=None, backend="hipgraph",
+                num_warmup_iters=3):
+        """Compile the training step (ref engine.compile:5706).
+
+        backend "hipgraph": capture the module's fwd+bwd as hipGraphs
+        (see deepspeed_amd/compile.py — ZeRO comm stays eager/overlapped).
+        """
+        if backend != "hipgraph":
+            raise ValueError(
+                "only the 'hipgraph' backend exists on this stack "
+                "(no Triton/inductor codegen in ROCm torch here)")
+        if sample_input is None:
+            raise ValueError("compile() needs a static-shape sample_input")
+        from ..compile import engine_compile
+        return engine_compile(self, sample_input, num_warmup_iters)
+
     def generate(self, input_ids, **kwargs):
         """RLHF-style generation with the training weights (hybrid engine:
         gathers ZeRO-3 shards for the rollout, then releases)."""

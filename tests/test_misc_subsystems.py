@@ -172,3 +172,13 @@ def test_launcher_hostfile_parsing():
     assert keep == {"nodeA": [0, 1, 2, 3]}
     keep = _filter_resources(res, include="", exclude="nodeB")
     assert keep == {"nodeA": list(range(8))}
+
+
+def test_graph_compile_requires_gpu_and_stage():
+    import pytest
+    import torch
+    from deepspeed_amd.compile import graph_compile
+    m = torch.nn.Linear(4, 4)
+    if not torch.cuda.is_available():
+        with pytest.raises(RuntimeError, match="requires a GPU"):
+            graph_compile(m, torch.randn(2, 4))
