@@ -28,23 +28,59 @@ import torch
 from .utils.logging import log_dist
 
 
-def graph_compile(module, sample_input, num_warmup_iters=3):
+class _PositionalLM(torch.nn.Module):
+    """Graph capture takes positional args only; adapt (ids, labels)."""
+
+    def __init__(self, mod):
+        super().__init__()
+        self.mod = mod
+
+    def forward(self, input_ids, labels):
+        return self.mod(input_ids, labels=labels)
+
+
+class _GraphedLM(torch.nn.Module):
+    """Training calls replay the captured graphs; label-free calls
+    (generation) fall back to the eager module."""
+
+    def __init__(self, graphed, eager):
+        super().__init__()
+        self.graphed = graphed
+        self.eager = eager
+
+    def forward(self, input_ids, labels=None, **kw):
+        if labels is not None and not kw:
+            return self.graphed(input_ids, labels)
+        return self.eager(input_ids, labels=labels, **kw)
+
+
+def graph_compile(module, sample_input, sample_labels=None,
+                  num_warmup_iters=3):
     """Return `module` with fwd+bwd captured as hipGraphs.
 
-    sample_input: a representative input tensor (static shape). The
-    module must be on the GPU already, in its training dtype.
+    sample_input (and sample_labels for loss-computing LMs) must carry
+    the static training shapes; the module must already be on the GPU in
+    its training dtype.
     """
     if not torch.cuda.is_available():
         raise RuntimeError("graph_compile requires a GPU (hipGraph capture)")
-    sample = sample_input.to(next(module.parameters()).device)
-    graphed = torch.cuda.make_graphed_callables(
-        module, (sample,), num_warmup_iters=num_warmup_iters)
+    dev = next(module.parameters()).device
+    sample = sample_input.to(dev)
+    if sample_labels is not None:
+        graphed = torch.cuda.make_graphed_callables(
+            _PositionalLM(module), (sample, sample_labels.to(dev)),
+            num_warmup_iters=num_warmup_iters)
+        out = _GraphedLM(graphed, module)
+    else:
+        out = torch.cuda.make_graphed_callables(
+            module, (sample,), num_warmup_iters=num_warmup_iters)
     log_dist("hipGraph-compiled module forward+backward "
              f"(input {tuple(sample.shape)})", ranks=[0])
-    return graphed
+    return out
 
 
-def engine_compile(engine, sample_input, num_warmup_iters=3):
+def engine_compile(engine, sample_input, sample_labels=None,
+                   num_warmup_iters=3):
     """In-place `engine.compile()`: swap the engine's module for the
     graphed one. ZeRO stage must be < 3."""
     stage = engine.zero_optimization_stage()
@@ -54,5 +90,5 @@ def engine_compile(engine, sample_input, num_warmup_iters=3):
             "parameter gathering; use stage <= 2 (or rely on ZeRO-3's "
             "stream overlap, which already hides launch gaps)")
     engine.module = graph_compile(engine.module, sample_input,
-                                  num_warmup_iters)
+                                  sample_labels, num_warmup_iters)
     return engine

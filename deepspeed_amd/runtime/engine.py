@@ -525,8 +525,8 @@ class DeepSpeedEngine(torch.nn.Module):
                      load_lr_scheduler_states=load_lr_scheduler_states,
                      load_module_only=load_module_only)
 
-    def compile(self, sample_input=None, backend="hipgraph",
-                num_warmup_iters=3):
+    def compile(self, sample_input=None, sample_labels=None,
+                backend="hipgraph", num_warmup_iters=3):
         """Compile the training step (ref engine.compile:5706).
 
         backend "hipgraph": capture the module's fwd+bwd as hipGraphs
@@ -539,7 +539,8 @@ class DeepSpeedEngine(torch.nn.Module):
         if sample_input is None:
             raise ValueError("compile() needs a static-shape sample_input")
         from ..compile import engine_compile
-        return engine_compile(self, sample_input, num_warmup_iters)
+        return engine_compile(self, sample_input, sample_labels,
+                              num_warmup_iters)
 
     def generate(self, input_ids, **kwargs):
         """RLHF-style generation with the training weights (hybrid engine:

@@ -150,7 +150,7 @@ def test_engine_compile_hipgraph():
     def run(compiled):
         e, _, _, _ = ds.initialize(model=build(), config=cfg)
         if compiled:
-            e.compile(sample_input=ids)
+            e.compile(sample_input=ids, sample_labels=ids)
         losses = []
         for _ in range(4):
             loss = e(ids, labels=ids)
