@@ -18,6 +18,8 @@ is refreshed with one H2D per bucket only on lazy boundaries.
 """
 from typing import Optional
 
+import threading
+
 import torch
 from pydantic import BaseModel
 
@@ -32,7 +34,7 @@ class ZenFlowConfig(BaseModel):
     select_interval: int = 16    # re-pick the hot set every N steps
     select_strategy: str = "auto"
     full_warm_up_rounds: int = 0
-    overlap_step: bool = False   # reserved: async CPU step thread
+    overlap_step: bool = False   # run the lazy CPU Adam in a worker thread
 
 
 class ZenFlowZeroOptimizer(ZeroStage12Optimizer):
@@ -54,6 +56,9 @@ class ZenFlowZeroOptimizer(ZeroStage12Optimizer):
             st["m_cpu"] = torch.zeros_like(st["master_cpu"])
             st["v_cpu"] = torch.zeros_like(st["master_cpu"])
             st["acc_cpu"] = torch.zeros_like(st["master_cpu"])
+            st["acc_snap"] = torch.zeros_like(st["master_cpu"]) \
+                if self.zf.overlap_step else None
+            st["thread"] = None
             self._zf.append(st)
             # the GPU master slab is not used by ZenFlow's dense pass
             b.master32 = None
@@ -101,6 +106,9 @@ class ZenFlowZeroOptimizer(ZeroStage12Optimizer):
 
         lazy = (self._zf_step % self.zf.update_interval) == 0
         for b, st in zip(self.buckets, self._zf):
+            # publish a finished overlapped CPU step before touching state
+            if st["thread"] is not None and not st["thread"].is_alive():
+                self._publish_cpu_step(b, st)
             g = b.grad32
             if combined != 1.0:
                 g.mul_(1.0 / combined)
@@ -123,32 +131,30 @@ class ZenFlowZeroOptimizer(ZeroStage12Optimizer):
             st["acc_cpu"].add_(g.to("cpu"))
             st["acc_steps"] += 1
             if lazy:
+                if st["thread"] is not None:  # previous overlap still busy
+                    st["thread"].join()
+                    self._publish_cpu_step(b, st)
                 self._writeback_hot(st)
                 st["acc_cpu"].div_(max(st["acc_steps"], 1))
-                from ..ops.loader import has_ext, get_ext
-                if has_ext():
-                    get_ext().cpu_adam_step(
-                        st["master_cpu"], st["acc_cpu"], st["m_cpu"],
-                        st["v_cpu"], None, lr, beta1, beta2, eps, t, 1, 1,
-                        wd, 1.0)
+                if self.zf.overlap_step:
+                    # hand the batch to a worker; accumulation continues
+                    # into the other buffer, publication happens when the
+                    # thread finishes (ZenFlow's bounded staleness)
+                    st["acc_cpu"], st["acc_snap"] = (st["acc_snap"],
+                                                     st["acc_cpu"])
+                    args = (st["master_cpu"], st["acc_snap"], st["m_cpu"],
+                            st["v_cpu"], lr, beta1, beta2, eps, t, wd,
+                            bc1, bc2)
+                    st["thread"] = threading.Thread(
+                        target=self._host_adam, args=args, daemon=True)
+                    st["thread"].start()
                 else:
-                    m, v = st["m_cpu"], st["v_cpu"]
-                    m.mul_(beta1).add_(st["acc_cpu"], alpha=1 - beta1)
-                    v.mul_(beta2).addcmul_(st["acc_cpu"], st["acc_cpu"],
-                                           value=1 - beta2)
-                    st["master_cpu"].mul_(1 - lr * wd)
-                    st["master_cpu"].addcdiv_(
-                        m / bc1, (v / bc2).sqrt().add_(eps), value=-lr)
-                st["acc_cpu"].zero_()
+                    self._host_adam(st["master_cpu"], st["acc_cpu"],
+                                    st["m_cpu"], st["v_cpu"], lr, beta1,
+                                    beta2, eps, t, wd, bc1, bc2)
+                    st["acc_cpu"].zero_()
+                    self._publish_cpu_step(b, st)
                 st["acc_steps"] = 0
-                b.shard16.copy_(
-                    st["master_cpu"].to(b.shard16.device, non_blocking=True)
-                    .to(b.shard16.dtype))
-                # refresh the hot cache from the new masters
-                idx_cpu = idx.cpu()
-                st["hot_master"] = st["master_cpu"][idx_cpu].to(g.device)
-                st["hot_m"] = st["m_cpu"][idx_cpu].to(g.device)
-                st["hot_v"] = st["v_cpu"][idx_cpu].to(g.device)
 
         for b in self.buckets:
             dist.all_gather_into_tensor(
@@ -156,8 +162,43 @@ class ZenFlowZeroOptimizer(ZeroStage12Optimizer):
                 group=b.pg if b.pg is not None else self.dp_group)
         self._clear_grads()
 
+    @staticmethod
+    def _host_adam(master, grad, m, v, lr, beta1, beta2, eps, t, wd,
+                   bc1, bc2):
+        from ..ops.loader import has_ext, get_ext
+        if has_ext():
+            get_ext().cpu_adam_step(master, grad, m, v, None, lr, beta1,
+                                    beta2, eps, t, 1, 1, wd, 1.0)
+        else:
+            m.mul_(beta1).add_(grad, alpha=1 - beta1)
+            v.mul_(beta2).addcmul_(grad, grad, value=1 - beta2)
+            master.mul_(1 - lr * wd)
+            master.addcdiv_(m / bc1, (v / bc2).sqrt().add_(eps), value=-lr)
+
+    def _publish_cpu_step(self, b, st):
+        """H2D the freshly updated masters + refresh the hot cache."""
+        st["thread"] = None
+        if st["acc_snap"] is not None:
+            st["acc_snap"].zero_()
+        b.shard16.copy_(
+            st["master_cpu"].to(b.shard16.device, non_blocking=True)
+            .to(b.shard16.dtype))
+        if st["hot_idx"] is not None:
+            idx_cpu = st["hot_idx"].cpu()
+            dev = b.shard16.device
+            st["hot_master"] = st["master_cpu"][idx_cpu].to(dev)
+            st["hot_m"] = st["m_cpu"][idx_cpu].to(dev)
+            st["hot_v"] = st["v_cpu"][idx_cpu].to(dev)
+
+    def _drain_threads(self):
+        for b, st in zip(self.buckets, self._zf):
+            if st["thread"] is not None:
+                st["thread"].join()
+                self._publish_cpu_step(b, st)
+
     # checkpointing: host masters replace the GPU ones
     def state_dict(self):
+        self._drain_threads()
         return {
             "zf_step": self._zf_step,
             "single_partition_of_fp32_groups":

@@ -4,7 +4,7 @@ import torch
 from tests.common import run_distributed
 
 
-def _zenflow_train():
+def _zenflow_train(overlap=False):
     import deepspeed_amd as ds
     torch.manual_seed(0)
     model = torch.nn.Sequential(torch.nn.Linear(64, 64), torch.nn.Tanh(),
@@ -16,7 +16,7 @@ def _zenflow_train():
         "zero_optimization": {
             "stage": 2,
             "zenflow": {"topk_ratio": 0.05, "update_interval": 4,
-                        "select_interval": 8},
+                        "select_interval": 8, "overlap_step": overlap},
         },
     }
     engine, opt, _, _ = ds.initialize(model=model, config=cfg)
@@ -59,6 +59,11 @@ def test_zenflow_stage2_cpu():
 
 def test_zenflow_world2():
     run_distributed(_zenflow_train, world_size=2)
+
+
+def test_zenflow_overlap_step():
+    """Async host-Adam worker: converges, threads drained at ckpt."""
+    run_distributed(_zenflow_train, world_size=1, kwargs={"overlap": True})
 
 
 import pytest  # noqa: E402
