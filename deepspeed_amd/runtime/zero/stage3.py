@@ -704,6 +704,28 @@ class ZeroStage3Optimizer:
                 "subgroups": subgroups}
 
     def state_dict(self):
+        if self.nvme_swapper is not None:
+            # masters + Adam state live on NVMe: read them back so the
+            # checkpoint is self-contained (resume works without the
+            # original swap files)
+            masters, eas, eass = [], [], []
+            for i, sg in enumerate(self.sub_groups):
+                m, ea, eas_ = self.nvme_swapper.swap_in(i, sg.numel)
+                masters.append(m.clone())
+                eas.append(ea.clone())
+                eass.append(eas_.clone())
+                self.nvme_swapper.release_buffers(i)
+            return {
+                "loss_scaler": self.loss_scaler,
+                "base_optimizer_state": self.optimizer.state_dict(),
+                "fp32_flat_groups": masters,
+                "nvme_exp_avg": eas,
+                "nvme_exp_avg_sq": eass,
+                "param_shapes": self._param_shapes(),
+                "zero_stage": 3,
+                "partition_count": self.world,
+                "shard_layout": self.shard_layout(),
+            }
         return {
             "loss_scaler": self.loss_scaler,
             "base_optimizer_state": self.optimizer.state_dict(),
@@ -733,6 +755,21 @@ class ZeroStage3Optimizer:
             self.optimizer.load_state_dict(sd["base_optimizer_state"])
         saved = sd.get("fp32_flat_groups", [])
         assert len(saved) == len(self.sub_groups)
+        if self.nvme_swapper is not None:
+            eas = sd.get("nvme_exp_avg")
+            eass = sd.get("nvme_exp_avg_sq")
+            for i, (sg, m) in enumerate(zip(self.sub_groups, saved)):
+                mb, eab, easb = self.nvme_swapper.swap_in(i, sg.numel)
+                mb.copy_(m.cpu())
+                if eas is not None:
+                    eab.copy_(eas[i].cpu())
+                    easb.copy_(eass[i].cpu())
+                sg.flat16_cpu.copy_(mb.to(sg.flat16_cpu.dtype))
+                self.nvme_swapper.swap_out(i)
+                self.nvme_swapper.release_buffers(i)
+                sg.publish_flat16()
+            self._refresh_persistent_params()
+            return
         for sg, s in zip(self.sub_groups, saved):
             sg.master32.data.copy_(s.data.to(sg.master32.device))
             sg.copy_master_to_shards()
