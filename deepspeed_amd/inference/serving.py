@@ -25,6 +25,8 @@ class Request:
     prompt: torch.Tensor                 # 1-D token ids
     max_new_tokens: int = 32
     eos_token_id: Optional[int] = None
+    temperature: float = 0.0             # 0 = greedy
+    top_k: int = 0
     rid: int = -1
     slot: int = -1
     generated: List[int] = field(default_factory=list)
@@ -95,10 +97,12 @@ class ContinuousBatchingEngine:
         self._was_ckpt = getattr(cfg, "activation_checkpointing", False)
         cfg.activation_checkpointing = False
 
-    def add_request(self, prompt, max_new_tokens=32, eos_token_id=None):
+    def add_request(self, prompt, max_new_tokens=32, eos_token_id=None,
+                    temperature=0.0, top_k=0):
         req = Request(prompt=prompt.to("cpu").long().view(-1),
                       max_new_tokens=max_new_tokens,
-                      eos_token_id=eos_token_id, rid=self._next_rid)
+                      eos_token_id=eos_token_id, temperature=temperature,
+                      top_k=top_k, rid=self._next_rid)
         self._next_rid += 1
         self.pending.append(req)
         return req.rid
@@ -113,8 +117,7 @@ class ContinuousBatchingEngine:
         logits = self.model(ids, kv_caches=self.caches)
         for c in self.caches:
             c._prefill_slot = None
-        nxt = int(logits[0, -1].float().argmax())
-        req.generated.append(nxt)
+        req.generated.append(self._sample(logits[0, -1], req))
         self._check_done(req)
 
     @torch.no_grad()
@@ -130,10 +133,21 @@ class ContinuousBatchingEngine:
                             positions=positions)
         for c in self.caches:
             c.rows = None
-        nxt = logits[:, -1].float().argmax(-1)
-        for r, t in zip(active, nxt.tolist()):
-            r.generated.append(int(t))
+        for r, row in zip(active, logits[:, -1]):
+            r.generated.append(self._sample(row, r))
             self._check_done(r)
+
+    @staticmethod
+    def _sample(logits, req):
+        logits = logits.float()
+        if req.temperature <= 0:
+            return int(logits.argmax())
+        logits = logits / req.temperature
+        if req.top_k > 0:
+            kth = torch.topk(logits, req.top_k).values[-1]
+            logits = logits.masked_fill(logits < kth, float("-inf"))
+        probs = torch.softmax(logits, dim=-1)
+        return int(torch.multinomial(probs, 1))
 
     def _check_done(self, req):
         if len(req.generated) >= req.max_new_tokens or \

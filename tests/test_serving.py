@@ -82,3 +82,17 @@ def test_continuous_batching_gpu():
         # kernels; require the vast majority of tokens to agree
         agree = (out[rid].cuda() == ref).float().mean().item()
         assert agree >= 0.75, (rid, agree)
+
+
+def test_serving_sampling_params():
+    model = _model()
+    cb = ContinuousBatchingEngine(model, max_batch=2)
+    torch.manual_seed(3)
+    p = torch.randint(0, 2000, (6,))
+    r1 = cb.add_request(p, max_new_tokens=5)                  # greedy
+    r2 = cb.add_request(p, max_new_tokens=5, temperature=0.8,
+                        top_k=10)                             # sampled
+    out = cb.run()
+    assert len(out[r1]) == 11 and len(out[r2]) == 11
+    # sampled tokens stay within the vocab
+    assert out[r2][6:].max() < 2048
