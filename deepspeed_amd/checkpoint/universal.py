@@ -14,13 +14,26 @@ import torch
 
 from ..utils.zero_to_fp32 import _load_zero_shards, _read_tag
 
-_STATE_KEYS = ("exp_avg", "exp_avg_sq")
+# Adam's moments by default; other fused optimizers (Lion: exp_avg only,
+# Adagrad: sq_accum, Muon: momentum) are discovered dynamically from the
+# base optimizer state at convert/load time.
+_DEFAULT_STATE_KEYS = ("exp_avg", "exp_avg_sq")
+
+
+def _discover_state_keys(base_states):
+    keys = set()
+    for bs in base_states:
+        for st in bs.get("state", {}).values():
+            keys.update(k for k, v in st.items()
+                        if torch.is_tensor(v) and v.dim() >= 1)
+    return tuple(sorted(keys)) or _DEFAULT_STATE_KEYS
 
 
 def _iter_params_with_states(shards):
     """Yield (name, {'fp32': t, 'exp_avg': t, 'exp_avg_sq': t})."""
     layout = shards[0]["shard_layout"]
     base_states = [sd["base_optimizer_state"] for sd in shards]
+    state_keys = _discover_state_keys(base_states)
 
     def master_state(rank, master_idx, key):
         st = base_states[rank]["state"].get(master_idx, {})
@@ -32,7 +45,7 @@ def _iter_params_with_states(shards):
         for bi, binfo in enumerate(layout["buckets"]):
             full = {"fp32": torch.cat([f[bi].detach().float()
                                        for f in flats_per_rank])}
-            for key in _STATE_KEYS:
+            for key in state_keys:
                 parts = [master_state(r, bi, key)
                          for r in range(len(shards))]
                 if all(p is not None for p in parts):
@@ -45,7 +58,7 @@ def _iter_params_with_states(shards):
         for gi, ginfo in enumerate(layout["subgroups"]):
             states = {"fp32": flats_per_rank}
             extra = {}
-            for key in _STATE_KEYS:
+            for key in state_keys:
                 parts = [master_state(r, gi, key)
                          for r in range(len(shards))]
                 if all(p is not None for p in parts):
@@ -79,16 +92,19 @@ def ds_to_universal(checkpoint_dir, output_dir, tag=None):
     # meta: step counts per group
     steps = [g.get("step", 0)
              for g in shards[0]["base_optimizer_state"]["param_groups"]]
+    keys = _discover_state_keys(
+        [sd["base_optimizer_state"] for sd in shards])
     torch.save({"param_names": names, "group_steps": steps,
+                "state_keys": list(keys),
                 "source_world": shards[0].get("partition_count", 1)},
                os.path.join(output_dir, "meta.pt"))
     return names
 
 
-def _load_param(universal_dir, name):
+def _load_param(universal_dir, name, state_keys=_DEFAULT_STATE_KEYS):
     pdir = os.path.join(universal_dir, "zero", name)
     out = {}
-    for key in ("fp32",) + _STATE_KEYS:
+    for key in ("fp32",) + tuple(state_keys):
         f = os.path.join(pdir, f"{key}.pt")
         if os.path.exists(f):
             out[key] = torch.load(f, map_location="cpu", weights_only=False)
@@ -103,6 +119,7 @@ def load_universal_into_optimizer(optimizer, universal_dir):
     """
     meta = torch.load(os.path.join(universal_dir, "meta.pt"),
                       map_location="cpu", weights_only=False)
+    state_keys = tuple(meta.get("state_keys", _DEFAULT_STATE_KEYS))
     layout = optimizer.shard_layout()
     world = layout["world"]
     base = optimizer.optimizer
@@ -116,10 +133,10 @@ def load_universal_into_optimizer(optimizer, universal_dir):
         for bi, (b, binfo) in enumerate(zip(optimizer.buckets,
                                             layout["buckets"])):
             full = {"fp32": torch.zeros(binfo["numel_padded"])}
-            for key in _STATE_KEYS:
+            for key in state_keys:
                 full[key] = torch.zeros(binfo["numel_padded"])
             for name, off, numel, shape in binfo["params"]:
-                t = _load_param(universal_dir, name)
+                t = _load_param(universal_dir, name, state_keys)
                 for key in full:
                     if key in t:
                         full[key][off:off + numel] = t[key].reshape(-1)
@@ -127,7 +144,7 @@ def load_universal_into_optimizer(optimizer, universal_dir):
             hi = lo + binfo["shard_numel"]
             b.master32.data.copy_(full["fp32"][lo:hi])
             master = b.master32
-            for key in _STATE_KEYS:
+            for key in state_keys:
                 set_state(master, key, full[key][lo:hi])
             b.shard16.copy_(b.master32.detach().to(b.shard16.dtype))
         from .. import comm as dist
@@ -139,20 +156,20 @@ def load_universal_into_optimizer(optimizer, universal_dir):
         for gi, (sg, ginfo) in enumerate(zip(optimizer.sub_groups,
                                              layout["subgroups"])):
             fp32 = torch.zeros(sg.numel)
-            states = {key: torch.zeros(sg.numel) for key in _STATE_KEYS}
+            states = {key: torch.zeros(sg.numel) for key in state_keys}
             for name, off, shard_numel, full_numel, shape in ginfo["params"]:
-                t = _load_param(universal_dir, name)
+                t = _load_param(universal_dir, name, state_keys)
                 flat = t["fp32"].reshape(-1)
                 lo = rank * shard_numel
                 hi = min(lo + shard_numel, full_numel)
                 if hi > lo:
                     fp32[off:off + hi - lo] = flat[lo:hi]
-                for key in _STATE_KEYS:
+                for key in state_keys:
                     if key in t:
                         states[key][off:off + hi - lo] = \
                             t[key].reshape(-1)[lo:hi]
             sg.master32.data.copy_(fp32.to(sg.master32.device))
-            for key in _STATE_KEYS:
+            for key in state_keys:
                 set_state(sg.master32, key, states[key])
             sg.copy_master_to_shards()
         optimizer._refresh_persistent_params()
