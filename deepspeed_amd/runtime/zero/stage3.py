@@ -168,17 +168,28 @@ class ZeroStage3Optimizer:
                                             dynamic_loss_args)
         self.overflow = False
 
+        # MoE expert params partition/reduce over their expert-DP group
+        # (the replicas of the same expert), like the stage-1/2 buckets.
         from ...comm import groups as grp
+        self._expert_pg = {}
         for p in module.parameters():
             gn = getattr(p, "group_name", None)
-            if gn is not None and gn in grp.get_expert_parallel_group_dict():
-                if dist.get_world_size(
-                        grp.get_expert_parallel_group(gn)) > 1:
-                    raise NotImplementedError(
-                        "MoE expert parameters with ZeRO stage 3 + "
-                        "ep_size>1 are not supported yet — use stage 1/2 "
-                        "(expert-DP bucket groups) for expert-parallel "
-                        "models")
+            if gn is None:
+                continue
+            if gn not in self._expert_pg:
+                if gn not in grp.get_expert_parallel_group_dict():
+                    try:  # groups are lazily built on first forward
+                        grp.create_expert_and_data_parallel(
+                            int(gn.rsplit("_", 1)[1]))
+                    except (ValueError, AssertionError):
+                        pass
+                try:
+                    self._expert_pg[gn] = \
+                        grp.get_expert_data_parallel_group(gn)
+                except KeyError:
+                    self._expert_pg[gn] = None
+        if self._expert_pg and self.replica_world > 1:
+            raise NotImplementedError("MiCS + expert parallelism")
         self._shard_module_params()
         self._build_sub_groups()
         if self.nvme_swapper is not None:
@@ -231,6 +242,18 @@ class ZeroStage3Optimizer:
         for p in params:
             if not is_zero_param(p):
                 p.data = p.data.to(self.device, self.dtype)
+                gn = getattr(p, "group_name", None)
+                pg = self._expert_pg.get(gn) if gn is not None else None
+                if pg is not None:
+                    # replicate only across expert-DP (distinct experts
+                    # stay rank-distinct), then shard over that group
+                    if dist.get_world_size(pg) > 1:
+                        dist.broadcast(p.data,
+                                       dist.get_global_rank(pg, 0),
+                                       group=pg)
+                    convert_to_zero_param(p, pg, self.device, self.dtype,
+                                          self.persist_threshold)
+                    continue
                 if bcast_world > 1:
                     dist.broadcast(p.data, src, group=bcast_group)
                 convert_to_zero_param(p, self.dp_group, self.device,
@@ -321,10 +344,35 @@ class ZeroStage3Optimizer:
             self.nvme_swapper.release_buffers(i)
             sg.publish_flat16()
 
+    def _param_pg(self, p):
+        return getattr(p, "ds_group", None) or self.dp_group
+
+    def _gather_grouped(self, params, async_op=True, stream=None):
+        """Launch one gather per distinct partitioning group (dense
+        params share dp_group; expert params use their expert-DP group).
+        Returns a handle with .wait()."""
+        by_pg = {}
+        for p in params:
+            by_pg.setdefault(id(self._param_pg(p)), []).append(p)
+        handles = [all_gather_params(ps, self._param_pg(ps[0]),
+                                     async_op=async_op, stream=stream)
+                   for ps in by_pg.values()]
+        if len(handles) == 1:
+            return handles[0]
+
+        class _Multi:
+            def __init__(self, hs):
+                self._hs = hs
+
+            def wait(self):
+                for h in self._hs:
+                    h.wait()
+        return _Multi(handles)
+
     def _gather_persistent_params(self):
         persist = [p for p in self._all_params if p.ds_persist]
         if persist:
-            all_gather_params(persist, self.dp_group, async_op=False).wait()
+            self._gather_grouped(persist, async_op=False).wait()
 
     # ------------------------------------------------------------- hooks
     def _install_module_hooks(self):
@@ -385,7 +433,7 @@ class ZeroStage3Optimizer:
         need = [p for p in params
                 if p.ds_status == ZeroParamStatus.NOT_AVAILABLE]
         if need:
-            all_gather_params(need, self.dp_group,
+            self._gather_grouped(need,
                               stream=self.ag_stream).wait()
         # tied params may be INFLIGHT under another module's handle
         pending = [p for p in params
@@ -430,7 +478,7 @@ class ZeroStage3Optimizer:
             if not need:
                 continue
             budget -= sum(p.ds_numel for p in need)
-            self._inflight[mod] = all_gather_params(need, self.dp_group,
+            self._inflight[mod] = self._gather_grouped(need,
                                                     stream=self.ag_stream)
             launched += 1
             if launched >= 8:
@@ -473,7 +521,9 @@ class ZeroStage3Optimizer:
             assert_ints_same_as_other_ranks(
                 [p.ds_id for p in params], group=self.dp_group,
                 tag="ipg_flush")
-        world = self.world
+        by_pg = {}
+        for p in params:
+            by_pg.setdefault(id(self._param_pg(p)), []).append(p)
         if self.rs_stream is not None:
             # grads were produced on the default stream
             self.rs_stream.wait_stream(torch.cuda.current_stream())
@@ -481,11 +531,17 @@ class ZeroStage3Optimizer:
                 if p.grad is not None:
                     p.grad.record_stream(self.rs_stream)
             with torch.cuda.stream(self.rs_stream):
-                self._flush_ipg_body(params, world)
+                for ps in by_pg.values():
+                    pg = self._param_pg(ps[0])
+                    self._flush_ipg_body(ps, dist.get_world_size(pg), pg)
             return
-        self._flush_ipg_body(params, world)
+        for ps in by_pg.values():
+            pg = self._param_pg(ps[0])
+            self._flush_ipg_body(ps, dist.get_world_size(pg), pg)
 
-    def _flush_ipg_body(self, params, world):
+    def _flush_ipg_body(self, params, world, pg=None):
+        if pg is None:
+            pg = self.dp_group
         use_coalescing = (torch.cuda.is_available() and world > 1
                           and len(params) > 1)
         shards = []
@@ -525,11 +581,11 @@ class ZeroStage3Optimizer:
                     for inp in inputs:
                         inp.div_(world)
                     op = dist.ReduceOp.SUM
-                with _coalescing_manager(self.dp_group, self.device,
+                with _coalescing_manager(pg, self.device,
                                          async_ops=True) as cm:
                     for out, inp in zip(shards, inputs):
                         dist.reduce_scatter_tensor(out, inp, op=op,
-                                                   group=self.dp_group)
+                                                   group=pg)
                 cm.wait()
             except Exception as e:
                 from .stage3_params import _COALESCE_OK  # noqa: F401
@@ -542,7 +598,7 @@ class ZeroStage3Optimizer:
             for out, inp in zip(shards, inputs):
                 op = _avg_op(world, inp)
                 dist.reduce_scatter_tensor(out, inp, op=op,
-                                           group=self.dp_group)
+                                           group=pg)
         if self.replica_world > 1:
             from .stage_1_and_2 import _avg_op
             for shard in shards:
@@ -668,7 +724,7 @@ class ZeroStage3Optimizer:
         for p in persist:
             p.ds_status = ZeroParamStatus.NOT_AVAILABLE
         if persist:
-            all_gather_params(persist, self.dp_group, async_op=False).wait()
+            self._gather_grouped(persist, async_op=False).wait()
 
     def _clear_grads(self):
         for sg in self.sub_groups:

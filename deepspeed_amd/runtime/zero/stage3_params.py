@@ -77,6 +77,7 @@ def convert_to_zero_param(p, dp_group, device, dtype,
     if end > start:
         ds_tensor[:end - start].copy_(flat[start:end])
     p.ds_tensor = ds_tensor
+    p.ds_group = dp_group  # partitioning group (expert-DP for MoE experts)
     p.ds_persist = p.ds_numel <= persist_threshold
     p.ds_status = ZeroParamStatus.NOT_AVAILABLE
     p.ds_active_sub_modules = set()

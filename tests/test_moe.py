@@ -156,3 +156,64 @@ def _moe_expert_ckpt():
 
 def test_moe_expert_checkpoint_ep2():
     run_distributed(_moe_expert_ckpt, world_size=2)
+
+
+def _moe_zero3_train():
+    import torch.distributed as dist
+    import deepspeed_amd as ds
+    from deepspeed_amd.moe.layer import MoE
+
+    torch.manual_seed(dist.get_rank())
+    M = 16
+
+    class Net(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.inp = torch.nn.Linear(M, M)
+            self.moe = MoE(M, torch.nn.Linear(M, M), num_experts=4,
+                           ep_size=2, k=1, capacity_factor=8.0)
+            self.out = torch.nn.Linear(M, 4)
+
+        def forward(self, x):
+            h = self.inp(x)
+            h, _, _ = self.moe(h)
+            return self.out(h)
+
+    cfg = {"train_micro_batch_size_per_gpu": 4,
+           "optimizer": {"type": "AdamW", "params": {"lr": 5e-3}},
+           "bf16": {"enabled": True},
+           "zero_optimization": {"stage": 3, "sub_group_size": 500,
+                                 "param_persistence_threshold": 4}}
+    engine, _, _, _ = ds.initialize(model=Net(), config=cfg)
+    # expert params shard over the (singleton) expert-DP group: full-size
+    # local shards, rank-distinct values; dense params shard over DP
+    from deepspeed_amd.comm import groups
+    epg = groups.get_expert_data_parallel_group("ep_size_2")
+    for p in engine.module.parameters():
+        if getattr(p, "group_name", None) is not None:
+            assert p.ds_group is epg
+            assert p.ds_tensor.numel() >= p.ds_numel  # world-1 shard
+    torch.manual_seed(7)  # same data on both ranks
+    x = torch.randn(4, M).bfloat16()
+    y = torch.randn(4, 4).bfloat16()
+    losses = []
+    for _ in range(6):
+        out = engine(x)
+        loss = (out - y).float().pow(2).mean()
+        engine.backward(loss)
+        engine.step()
+        losses.append(loss.item())
+    assert losses[-1] < losses[0], losses
+    # experts remain rank-distinct (no cross-expert mixing)
+    w = [p for n, p in engine.module.named_parameters()
+         if "deepspeed_experts.0.weight" in n][0]
+    from deepspeed_amd.runtime.zero.stage3_params import all_gather_params
+    s = w.ds_tensor.float().sum()
+    buf = [torch.zeros_like(s) for _ in range(2)]
+    dist.all_gather(buf, s)
+    assert not torch.allclose(buf[0], buf[1]), "experts were averaged!"
+    return losses[-1]
+
+
+def test_moe_zero3_ep2():
+    run_distributed(_moe_zero3_train, world_size=2)
