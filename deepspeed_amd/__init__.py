@@ -93,3 +93,5 @@ def add_config_arguments(parser):
     group.add_argument("--deepscale", default=False, action="store_true")
     group.add_argument("--local_rank", type=int, default=-1)
     return parser
+
+from .accelerator import get_accelerator  # noqa: F401,E402

@@ -182,3 +182,14 @@ def test_graph_compile_requires_gpu_and_stage():
     if not torch.cuda.is_available():
         with pytest.raises(RuntimeError, match="requires a GPU"):
             graph_compile(m, torch.randn(2, 4))
+
+
+def test_accelerator_facade():
+    import torch
+    from deepspeed_amd import get_accelerator
+    acc = get_accelerator()
+    assert acc.device_name(0) == "cuda:0"
+    assert acc.communication_backend_name() == "nccl"
+    assert torch.bfloat16 in acc.supported_dtypes()
+    t = torch.randn(4)
+    assert not acc.on_accelerator(t)
