@@ -52,6 +52,18 @@ class InferenceEngine(torch.nn.Module):
         self.device = (torch.device("cuda", torch.cuda.current_device())
                        if torch.cuda.is_available()
                        else torch.device("cpu"))
+        tp = self._config.tensor_parallel or {}
+        tp_size = tp.get("tp_size", 1) if isinstance(tp, dict) \
+            else getattr(tp, "tp_size", 1)
+        if tp_size > 1:
+            # shard attention/MLP linears across the TP group
+            # (ref init_inference replace_with_kernel_inject + AutoTP)
+            import torch.distributed as tdist
+            from ..module_inject.auto_tp import apply_tensor_parallel
+            assert tdist.is_initialized(), \
+                "tensor_parallel.tp_size > 1 requires torch.distributed"
+            group = tp.get("tp_group") if isinstance(tp, dict) else None
+            apply_tensor_parallel(self.module, group)
         self.module.to(self._config.dtype).to(self.device)
         self.module.eval()
         self._caches = None

@@ -57,3 +57,26 @@ def test_generate_sampling_runs():
     ids = torch.randint(0, cfg.vocab_size, (2, 4))
     out = engine.generate(ids, max_new_tokens=4, temperature=0.8, top_k=10)
     assert out.shape == (2, 8)
+
+
+def _init_inference_tp2():
+    import torch.distributed as dist
+    import deepspeed_amd as ds
+    from deepspeed_amd.models.llama import LLAMA_CONFIGS, LlamaForCausalLM
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(LLAMA_CONFIGS["llama-tiny"]).eval()
+    torch.manual_seed(5)
+    ids = torch.randint(0, 2000, (1, 12))
+    with torch.no_grad():
+        ref = model(ids).float()
+    eng = ds.init_inference(model, dtype=torch.float32,
+                            tensor_parallel={"tp_size": 2})
+    with torch.no_grad():
+        out = eng(ids).float()
+    err = (out - ref).abs().max().item()
+    assert err < 1e-4, err
+
+
+def test_init_inference_tensor_parallel_world2():
+    from tests.common import run_distributed
+    run_distributed(_init_inference_tp2, world_size=2)
