@@ -217,3 +217,24 @@ def _moe_zero3_train():
 
 def test_moe_zero3_ep2():
     run_distributed(_moe_zero3_train, world_size=2)
+
+
+def test_split_params_into_moe_groups():
+    from deepspeed_amd.moe.utils import (
+        is_moe_param, split_params_into_different_moe_groups_for_optimizer)
+    dense = torch.nn.Parameter(torch.randn(4))
+    e1 = torch.nn.Parameter(torch.randn(4))
+    e1.allreduce = False
+    e1.group_name = "ep_size_2"
+    e2 = torch.nn.Parameter(torch.randn(4))
+    e2.allreduce = False
+    e2.group_name = "ep_size_2"
+    assert not is_moe_param(dense) and is_moe_param(e1)
+    groups = split_params_into_different_moe_groups_for_optimizer(
+        {"params": [dense, e1, e2], "lr": 1e-3})
+    assert len(groups) == 2
+    assert groups[0]["params"] == [dense]
+    assert groups[1]["moe"] and groups[1]["name"] == "ep_size_2"
+    assert len(groups[1]["params"]) == 2
+    # torch optimizer accepts the result
+    torch.optim.AdamW(groups)
