@@ -219,6 +219,14 @@ class DeepSpeedConfig:
         self.pipeline = PipelineConfig(**config.get("pipeline", {}))
         self.moe = MoEConfig(**config.get("moe", {}))
         self.data_types = config.get("data_types", {})
+        # curriculum learning: legacy top-level key or
+        # data_efficiency.data_sampling.curriculum_learning (reference
+        # runtime/data_pipeline/config.py)
+        cl = config.get("curriculum_learning")
+        if cl is None:
+            cl = config.get("data_efficiency", {}) \
+                .get("data_sampling", {}).get("curriculum_learning")
+        self.curriculum_learning = cl if (cl or {}).get("enabled") else None
 
         if self.fp16.enabled and self.bf16.enabled:
             raise ValueError("fp16 and bf16 cannot both be enabled")

@@ -14,7 +14,12 @@ class CurriculumScheduler:
 
     def __init__(self, config):
         self.state = {}
-        self.schedule_type = config["curriculum_type"]
+        # reference schema: curriculum_type names the metric ("seqlen"),
+        # schedule_type the curve; legacy configs put the curve in
+        # curriculum_type — accept both
+        self.metric = config.get("curriculum_type", "seqlen")
+        self.schedule_type = config.get("schedule_type",
+                                        config.get("curriculum_type"))
         self.min_difficulty = config["min_difficulty"]
         self.max_difficulty = config["max_difficulty"]
         cfg = config.get("schedule_config", config)

@@ -65,6 +65,12 @@ class DeepSpeedEngine(torch.nn.Module):
         self.global_steps = 0
         self.global_samples = 0
         self.micro_steps = 0
+        cl_cfg = getattr(self._config, "curriculum_learning", None)
+        if cl_cfg:
+            from .data_pipeline import CurriculumScheduler
+            self.curriculum_scheduler = CurriculumScheduler(cl_cfg)
+        else:
+            self.curriculum_scheduler = None
         self.skipped_steps = 0
         self._is_gradient_accumulation_boundary = None
 
@@ -453,6 +459,8 @@ class DeepSpeedEngine(torch.nn.Module):
             buf.copy_(synced)
 
     def step(self, lr_kwargs=None):
+        if self.curriculum_scheduler is not None:
+            self.curriculum_scheduler.update_difficulty(self.global_steps + 1)
         if self.wall_clock_breakdown():
             self.timers("step").start()
         if self.is_gradient_accumulation_boundary():

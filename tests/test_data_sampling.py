@@ -59,3 +59,39 @@ def test_data_analyzer_map_reduce(tmp_path):
     # bucket for value 3 holds exactly the samples with len 3
     b3 = [int(x) for x in m2s[0]]
     assert b3 == [i for i in range(23) if (i % 5) == 0]
+
+
+def test_engine_curriculum_wiring():
+    import torch
+    from tests.common import run_distributed
+
+    run_distributed(_curriculum_engine, world_size=1)
+
+
+def _curriculum_engine():
+    import torch
+    import deepspeed_amd as ds
+    from tests.simple_model import SimpleModel
+    cfg = {
+        "train_micro_batch_size_per_gpu": 2,
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+        "curriculum_learning": {
+            "enabled": True, "curriculum_type": "seqlen",
+            "min_difficulty": 8, "max_difficulty": 32,
+            "schedule_type": "fixed_linear",
+            "schedule_config": {"total_curriculum_step": 4,
+                                "difficulty_step": 8},
+        },
+    }
+    engine, _, _, _ = ds.initialize(model=SimpleModel(32), config=cfg)
+    assert engine.curriculum_scheduler is not None
+    d0 = engine.curriculum_scheduler.get_current_difficulty()
+    x = torch.randn(2, 32)
+    y = torch.randn(2, 32)
+    for _ in range(5):
+        loss = engine(x, y)
+        engine.backward(loss)
+        engine.step()
+    d1 = engine.curriculum_scheduler.get_current_difficulty()
+    assert d1 > d0, (d0, d1)
+    assert d1 == 32
