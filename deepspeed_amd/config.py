@@ -227,6 +227,15 @@ class DeepSpeedConfig:
             cl = config.get("data_efficiency", {}) \
                 .get("data_sampling", {}).get("curriculum_learning")
         self.curriculum_learning = cl if (cl or {}).get("enabled") else None
+        pld = config.get("progressive_layer_drop", {})
+        self.progressive_layer_drop = pld if pld.get("enabled") else None
+        rltd = config.get("data_efficiency", {}) \
+            .get("data_routing", {}).get("random_ltd", {})
+        self.random_ltd = rltd if rltd.get("enabled") else None
+        ev = config.get("eigenvalue", {})
+        self.eigenvalue = ev if ev.get("enabled") else None
+        self.wandb = config.get("wandb", {})
+        self.compression_training = config.get("compression_training")
 
         if self.fp16.enabled and self.bf16.enabled:
             raise ValueError("fp16 and bf16 cannot both be enabled")

@@ -71,6 +71,14 @@ class DeepSpeedEngine(torch.nn.Module):
             self.curriculum_scheduler = CurriculumScheduler(cl_cfg)
         else:
             self.curriculum_scheduler = None
+        pld_cfg = getattr(self._config, "progressive_layer_drop", None)
+        if pld_cfg:
+            from .data_pipeline import ProgressiveLayerDrop
+            self.progressive_layer_drop = ProgressiveLayerDrop(
+                theta=pld_cfg.get("theta", 0.5),
+                gamma=pld_cfg.get("gamma", 0.001))
+        else:
+            self.progressive_layer_drop = None
         self.skipped_steps = 0
         self._is_gradient_accumulation_boundary = None
 
@@ -461,6 +469,8 @@ class DeepSpeedEngine(torch.nn.Module):
     def step(self, lr_kwargs=None):
         if self.curriculum_scheduler is not None:
             self.curriculum_scheduler.update_difficulty(self.global_steps + 1)
+        if self.progressive_layer_drop is not None:
+            self.progressive_layer_drop.update_state(self.global_steps + 1)
         if self.wall_clock_breakdown():
             self.timers("step").start()
         if self.is_gradient_accumulation_boundary():
