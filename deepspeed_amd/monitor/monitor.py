@@ -67,6 +67,12 @@ class MonitorMaster(Monitor):
             self.monitors.append(CSVMonitor(ds_config.csv_monitor))
         if getattr(ds_config, "tensorboard", None) and ds_config.tensorboard.enabled:
             self.monitors.append(TensorBoardMonitor(ds_config.tensorboard))
+        wb = getattr(ds_config, "wandb", None)
+        if wb and (wb.get("enabled") if isinstance(wb, dict)
+                   else getattr(wb, "enabled", False)):
+            from types import SimpleNamespace
+            cfg = SimpleNamespace(**wb) if isinstance(wb, dict) else wb
+            self.monitors.append(WandbMonitor(cfg))
         self.enabled = len(self.monitors) > 0
 
     def write_events(self, event_list):
