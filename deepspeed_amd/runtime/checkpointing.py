@@ -62,7 +62,12 @@ def save_checkpoint(engine, save_dir, tag=None, client_state=None,
     # MoE: expert params differ per expert-parallel rank — they go into
     # separate expert_ep_rank files (ref _save_moe_checkpoint,
     # runtime/engine.py:4921) and are excluded from the dense states.
-    expert_names = _expert_param_names(engine.module)
+    # Under ZeRO-3 the per-dp-rank optimizer shards already carry each
+    # rank's expert state (experts partition over expert-DP), so the
+    # separate expert files are only needed for stages <= 2 where the
+    # dense model_states would otherwise lose rank-distinct experts.
+    expert_names = _expert_param_names(engine.module) \
+        if engine.zero_optimization_stage() != 3 else {}
     if expert_names:
         from ..comm import groups
         full_sd = engine.module.state_dict()
@@ -132,7 +137,8 @@ def load_checkpoint(engine, load_dir, tag=None, load_module_strict=True,
     is_zero = hasattr(engine.optimizer, "load_state_dict") and \
         not isinstance(engine.optimizer, torch.optim.Optimizer)
 
-    expert_names = _expert_param_names(engine.module)
+    expert_names = _expert_param_names(engine.module) \
+        if engine.zero_optimization_stage() != 3 else {}
     if engine.zero_optimization_stage() != 3:
         engine.load_module_state_dict(state["module"],
                                       strict=load_module_strict
