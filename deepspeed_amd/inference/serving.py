@@ -185,3 +185,15 @@ class ContinuousBatchingEngine:
 
     def has_capacity(self):
         return bool(self.free_slots)
+
+    def close(self):
+        """Free the KV pool and restore the model's training config."""
+        self.caches = []
+        self.cfg.activation_checkpointing = self._was_ckpt
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *exc):
+        self.close()
+        return False

@@ -93,6 +93,13 @@ class ZenFlowZeroOptimizer(ZeroStage12Optimizer):
     def step(self, closure=None):
         assert closure is None, "closure not supported"
         self._sync_comm()
+        if self.dtype == torch.float16:
+            self.overflow = self.has_overflow()
+            self.loss_scaler.update_scale(self.overflow)
+            if self.overflow:
+                log_dist("ZenFlow: OVERFLOW, skipping step", ranks=[0])
+                self._clear_grads()
+                return
         self._zf_step += 1
         combined = self._combined_scale()
         group = self.optimizer.param_groups[0]
