@@ -194,3 +194,40 @@ def test_pipeline_fp16_loss_scaled():
     ref_losses = ref[0] if isinstance(ref, tuple) else ref
     for a, b in zip(fp16_losses, ref_losses):
         assert abs(float(a) - float(b)) < 0.05, (a, b)
+
+
+def _pipe_train_fp16_dynamic(steps=4):
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    from deepspeed_amd.runtime.pipe.module import PipelineModule
+    groups.reset_groups()
+    model = PipelineModule(layers=_make_specs(), num_stages=2,
+                           loss_fn=_loss_fn, partition_method="uniform")
+    config = {
+        "train_micro_batch_size_per_gpu": MICRO,
+        "gradient_accumulation_steps": GAS,
+        "optimizer": {"type": "AdamW",
+                      "params": {"lr": LR, "weight_decay": 0.0}},
+        # huge initial scale forces an overflow -> skip -> backoff
+        "fp16": {"enabled": True, "loss_scale": 0,
+                 "initial_scale_power": 32, "loss_scale_window": 2},
+    }
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config=config)
+    it = iter(_data(steps * GAS))
+    losses = [engine.train_batch(data_iter=it) for _ in range(steps)]
+    scale = engine.optimizer.loss_scaler.loss_scale
+    skipped = getattr(engine.optimizer, "overflow_count", None)
+    return losses, float(scale)
+
+
+def test_pipeline_fp16_dynamic_scale():
+    """fp16 PP with DYNAMIC loss scale: overflow at 2^32 backs off, then
+    training proceeds and the final loss tracks the fp32 reference."""
+    res = run_distributed(_pipe_train_fp16_dynamic, world_size=2)
+    losses, scale = res[1] if res[1] else res[0]
+    assert scale < 2.0 ** 32, f"scale never backed off: {scale}"
+    ref = _reference(steps=4)
+    ref_losses = ref[0] if isinstance(ref, tuple) else ref
+    # later steps (after backoff) track the reference
+    final = [l for l in losses if l is not None][-1]
+    assert abs(float(final) - float(ref_losses[-1])) < 0.1
