@@ -363,6 +363,7 @@ class ZeroStage3Optimizer:
         class _Multi:
             def __init__(self, hs):
                 self._hs = hs
+                self.params = [p for h in hs for p in h.params]
 
             def wait(self):
                 for h in self._hs:
