@@ -139,3 +139,41 @@ def _autosp_apply():
 def test_autosp_applies_to_llama_world2():
     from tests.common import run_distributed
     run_distributed(_autosp_apply, world_size=2)
+
+
+def _ulysses_uneven_heads():
+    import torch.distributed as dist
+    import torch.nn.functional as F
+    from deepspeed_amd.sequence.layer import DistributedAttention
+
+    torch.manual_seed(0)
+    B, S, H, D = 2, 8, 3, 4  # 3 heads over 2 ranks -> [2, 1]
+    world = dist.get_world_size()
+    rank = dist.get_rank()
+    q = torch.randn(B, S, H, D)
+    k = torch.randn(B, S, H, D)
+    v = torch.randn(B, S, H, D)
+    # reference: full attention on the whole sequence
+    ref = F.scaled_dot_product_attention(
+        q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2),
+        is_causal=True).transpose(1, 2)
+
+    def local_attn(q_, k_, v_, **kw):
+        return F.scaled_dot_product_attention(
+            q_.transpose(1, 2), k_.transpose(1, 2), v_.transpose(1, 2),
+            is_causal=True).transpose(1, 2)
+
+    attn = DistributedAttention(local_attn, dist.group.WORLD)
+    s = S // world
+    sl = slice(rank * s, (rank + 1) * s)
+    ql = q[:, sl].clone().requires_grad_(True)
+    out = attn(ql, k[:, sl], v[:, sl])
+    assert out.shape == (B, s, H, D)
+    err = (out - ref[:, sl]).abs().max().item()
+    assert err < 1e-5, err
+    out.sum().backward()  # uneven backward path runs
+    assert ql.grad is not None and torch.isfinite(ql.grad).all()
+
+
+def test_ulysses_uneven_heads_world2():
+    run_distributed(_ulysses_uneven_heads, world_size=2)
