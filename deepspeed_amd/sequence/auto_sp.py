@@ -61,11 +61,14 @@ def pick_sp_degree(seq_len, num_heads, num_kv_heads=None, world_size=None,
         return 1
     # how many ways we'd LIKE to split to bring local seq under threshold
     want = max(1, (seq_len + seq_threshold - 1) // seq_threshold)
+    mha = num_kv_heads is None or num_kv_heads == num_heads
     best = 1
     for d in range(1, world_size + 1):
-        if world_size % d or num_heads % d:
+        if world_size % d:
             continue
-        if num_kv_heads is not None and num_kv_heads % d:
+        if num_heads % d and not (mha and d <= num_heads):
+            continue  # uneven distribution only supported for MHA
+        if num_kv_heads is not None and num_kv_heads % d and not mha:
             continue
         if d <= want:
             best = d
