@@ -193,3 +193,24 @@ def test_accelerator_facade():
     assert torch.bfloat16 in acc.supported_dtypes()
     t = torch.randn(4)
     assert not acc.on_accelerator(t)
+
+
+def test_llama70b_meta_build_and_memory_estimate(capsys):
+    """70B config builds on meta and the ZeRO-3 estimate fits 288 GB."""
+    import torch
+    from deepspeed_amd.models.llama import LLAMA_CONFIGS, LlamaForCausalLM
+    from deepspeed_amd.utils.memory_estimators import \
+        estimate_zero3_model_states_mem_needs
+    cfg = LLAMA_CONFIGS["llama3-70b"]
+    with torch.device("meta"):
+        model = LlamaForCausalLM(cfg)
+    total = sum(p.numel() for p in model.parameters())
+    assert 68e9 < total < 73e9, total
+    gpu, cpu = estimate_zero3_model_states_mem_needs(
+        total, largest_layer_params=int(1.5e9), num_gpus_per_node=8)
+    # 70B zero-3 on 8 GPUs: ~140 GB model states per GPU << 288 GB
+    assert gpu < 288 * (1 << 30), gpu / (1 << 30)
+    gpu_off, cpu_off = estimate_zero3_model_states_mem_needs(
+        total, largest_layer_params=int(1.5e9), num_gpus_per_node=8,
+        cpu_offload=True)
+    assert gpu_off < gpu
