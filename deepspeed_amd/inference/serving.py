@@ -27,6 +27,7 @@ class Request:
     eos_token_id: Optional[int] = None
     temperature: float = 0.0             # 0 = greedy
     top_k: int = 0
+    top_p: float = 1.0
     rid: int = -1
     slot: int = -1
     generated: List[int] = field(default_factory=list)
@@ -98,11 +99,11 @@ class ContinuousBatchingEngine:
         cfg.activation_checkpointing = False
 
     def add_request(self, prompt, max_new_tokens=32, eos_token_id=None,
-                    temperature=0.0, top_k=0):
+                    temperature=0.0, top_k=0, top_p=1.0):
         req = Request(prompt=prompt.to("cpu").long().view(-1),
                       max_new_tokens=max_new_tokens,
                       eos_token_id=eos_token_id, temperature=temperature,
-                      top_k=top_k, rid=self._next_rid)
+                      top_k=top_k, top_p=top_p, rid=self._next_rid)
         self._next_rid += 1
         self.pending.append(req)
         return req.rid
@@ -147,6 +148,13 @@ class ContinuousBatchingEngine:
             kth = torch.topk(logits, req.top_k).values[-1]
             logits = logits.masked_fill(logits < kth, float("-inf"))
         probs = torch.softmax(logits, dim=-1)
+        if req.top_p < 1.0:
+            sp, idx = probs.sort(descending=True)
+            cum = sp.cumsum(-1)
+            keep = cum - sp < req.top_p  # keep until mass reaches top_p
+            sp = sp * keep
+            sp = sp / sp.sum()
+            return int(idx[torch.multinomial(sp, 1)])
         return int(torch.multinomial(probs, 1))
 
     def _check_done(self, req):

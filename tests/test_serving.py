@@ -96,3 +96,13 @@ def test_serving_sampling_params():
     assert len(out[r1]) == 11 and len(out[r2]) == 11
     # sampled tokens stay within the vocab
     assert out[r2][6:].max() < 2048
+
+
+def test_serving_top_p():
+    model = _model()
+    cb = ContinuousBatchingEngine(model, max_batch=1)
+    torch.manual_seed(4)
+    p = torch.randint(0, 2000, (6,))
+    rid = cb.add_request(p, max_new_tokens=5, temperature=1.0, top_p=0.9)
+    out = cb.run()
+    assert len(out[rid]) == 11
