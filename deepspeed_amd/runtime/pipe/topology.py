@@ -77,3 +77,75 @@ class PipelineParallelGrid:
 
     def get_model_parallel_world_size(self):
         return 1
+
+
+class ProcessTopology:
+    """Generic cartesian process topology (ref runtime/pipe/topology.py:12).
+
+    Maps ranks <-> named-axis coordinates, row-major in axis order:
+    ProcessTopology(axes=['pipe','data'], dims=[2,4]) puts data innermost.
+    """
+
+    def __init__(self, axes, dims):
+        assert len(axes) == len(dims)
+        self.axes = list(axes)
+        self.dims = list(dims)
+
+    def world_size(self):
+        n = 1
+        for d in self.dims:
+            n *= d
+        return n
+
+    def get_dim(self, axis):
+        return self.dims[self.axes.index(axis)]
+
+    def get_coord(self, rank):
+        coords = {}
+        for axis, dim in zip(reversed(self.axes), reversed(self.dims)):
+            coords[axis] = rank % dim
+            rank //= dim
+        from collections import namedtuple
+        Coord = namedtuple("Coord", self.axes)
+        return Coord(**coords)
+
+    def get_rank(self, **coords):
+        rank = 0
+        for axis, dim in zip(self.axes, self.dims):
+            c = coords[axis]
+            assert 0 <= c < dim, f"{axis}={c} out of range {dim}"
+            rank = rank * dim + c
+        return rank
+
+    def get_axis_list(self, axis, idx):
+        """All ranks whose `axis` coordinate equals idx."""
+        return [r for r in range(self.world_size())
+                if getattr(self.get_coord(r), axis) == idx]
+
+    def get_axis_comm_lists(self, axis):
+        """Rank lists that differ only along `axis` (comm groups)."""
+        lists = {}
+        for r in range(self.world_size()):
+            c = self.get_coord(r)
+            key = tuple(v for a, v in zip(self.axes, c) if a != axis)
+            lists.setdefault(key, []).append(r)
+        return list(lists.values())
+
+    def filter_match(self, **filters):
+        out = []
+        for r in range(self.world_size()):
+            c = self.get_coord(r)
+            if all(getattr(c, a) == v for a, v in filters.items()):
+                out.append(r)
+        return out
+
+
+class PipeDataParallelTopology(ProcessTopology):
+    def __init__(self, num_pp, num_dp):
+        super().__init__(axes=["pipe", "data"], dims=[num_pp, num_dp])
+
+
+class PipeModelDataParallelTopology(ProcessTopology):
+    def __init__(self, num_pp, num_mp, num_dp):
+        super().__init__(axes=["pipe", "data", "model"],
+                         dims=[num_pp, num_dp, num_mp])

@@ -231,3 +231,20 @@ def test_pipeline_fp16_dynamic_scale():
     # later steps (after backoff) track the reference
     final = [l for l in losses if l is not None][-1]
     assert abs(float(final) - float(ref_losses[-1])) < 0.1
+
+
+def test_process_topology_coords():
+    from deepspeed_amd.runtime.pipe import (PipeDataParallelTopology,
+                                            ProcessTopology)
+    topo = ProcessTopology(axes=["pipe", "data"], dims=[2, 4])
+    assert topo.world_size() == 8
+    assert topo.get_dim("data") == 4
+    c = topo.get_coord(5)
+    assert c.pipe == 1 and c.data == 1
+    assert topo.get_rank(pipe=1, data=1) == 5
+    assert topo.get_axis_list("pipe", 0) == [0, 1, 2, 3]
+    groups = topo.get_axis_comm_lists("data")
+    assert [0, 1, 2, 3] in groups and [4, 5, 6, 7] in groups
+    assert topo.filter_match(pipe=1) == [4, 5, 6, 7]
+    pd = PipeDataParallelTopology(2, 2)
+    assert pd.world_size() == 4
