@@ -8,8 +8,9 @@ MX-scaled MFMA consumes — fp8 = e4m3fn (hardware convert), fp12 = e5m6.
 
 GPU path: HIP kernels (csrc/quantize.hip, enc_fp_em/dec_fp_em). CPU path:
 the identical codec written against torch.frexp, used for tests and as a
-host fallback. Containers are byte-aligned (u16 per element for 4/6/12;
-dense bit-packing is tracked in ROADMAP).
+host fallback. Compute containers are byte-aligned u16; pack_codes /
+unpack_codes provide the dense bitstream wire format for storage and
+comms (4 bits -> 2/byte, 6 bits -> 4/3B, 12 bits -> 2/3B).
 """
 import torch
 
@@ -127,3 +128,58 @@ class FP_Quantize:
             fp_out.copy_(out.to(fp_out.dtype))
             return fp_out
         return out
+
+
+# ---------------------------------------------------------------- packing
+# Dense bit-packed wire format for the 4/6/12-bit codes (u16 working
+# containers stay for compute; packing is for storage/comms). Little-
+# endian bitstream, vectorized with tensor shifts — runs on CPU or GPU.
+
+def pack_codes(codes, bits):
+    """uint16 codes [n] -> uint8 bitstream [ceil(n*bits/8)]."""
+    assert bits in (4, 6, 12)
+    c = codes.to(torch.int32)
+    n = c.numel()
+    if bits == 4:
+        pad = (-n) % 2
+        if pad:
+            c = torch.nn.functional.pad(c, (0, pad))
+        c = c.view(-1, 2)
+        return (c[:, 0] | (c[:, 1] << 4)).to(torch.uint8)
+    if bits == 6:  # 4 codes -> 3 bytes
+        pad = (-n) % 4
+        if pad:
+            c = torch.nn.functional.pad(c, (0, pad))
+        c = c.view(-1, 4)
+        w = c[:, 0] | (c[:, 1] << 6) | (c[:, 2] << 12) | (c[:, 3] << 18)
+        out = torch.stack([w & 0xFF, (w >> 8) & 0xFF, (w >> 16) & 0xFF],
+                          dim=1)
+        return out.reshape(-1).to(torch.uint8)
+    # 12: 2 codes -> 3 bytes
+    pad = (-n) % 2
+    if pad:
+        c = torch.nn.functional.pad(c, (0, pad))
+    c = c.view(-1, 2)
+    w = c[:, 0] | (c[:, 1] << 12)
+    out = torch.stack([w & 0xFF, (w >> 8) & 0xFF, (w >> 16) & 0xFF], dim=1)
+    return out.reshape(-1).to(torch.uint8)
+
+
+def unpack_codes(packed, bits, n):
+    """uint8 bitstream -> uint16 codes [n]."""
+    assert bits in (4, 6, 12)
+    b = packed.to(torch.int32)
+    if bits == 4:
+        lo = b & 0xF
+        hi = (b >> 4) & 0xF
+        return torch.stack([lo, hi], 1).reshape(-1)[:n].to(torch.uint16)
+    if bits == 6:
+        b = b.view(-1, 3)
+        w = b[:, 0] | (b[:, 1] << 8) | (b[:, 2] << 16)
+        c = torch.stack([w & 0x3F, (w >> 6) & 0x3F, (w >> 12) & 0x3F,
+                         (w >> 18) & 0x3F], dim=1)
+        return c.reshape(-1)[:n].to(torch.uint16)
+    b = b.view(-1, 3)
+    w = b[:, 0] | (b[:, 1] << 8) | (b[:, 2] << 16)
+    c = torch.stack([w & 0xFFF, (w >> 12) & 0xFFF], dim=1)
+    return c.reshape(-1)[:n].to(torch.uint16)

@@ -147,3 +147,33 @@ def test_int4_gpu_matches_cpu():
     yg = dequantize_int4(qg, sg, shp, group_size=256).cpu()
     assert torch.allclose(yc.float(), yg.float(), atol=1e-5), \
         (yc.float() - yg.float()).abs().max()
+
+
+@pytest.mark.parametrize("bits", [4, 6, 12])
+def test_fp_code_bit_packing(bits):
+    import torch
+    from deepspeed_amd.ops.fp_quantizer import pack_codes, unpack_codes
+    torch.manual_seed(0)
+    n = 1001
+    codes = torch.randint(0, 1 << bits, (n,), dtype=torch.uint16)
+    packed = pack_codes(codes, bits)
+    # dense: ceil(n*bits/8) bytes (up to one group of padding)
+    assert packed.numel() <= (n * bits + 7) // 8 + 3
+    back = unpack_codes(packed, bits, n)
+    assert torch.equal(back, codes)
+
+
+def test_fp6_roundtrip_packed():
+    import torch
+    from deepspeed_amd.ops.fp_quantizer import (FP_Quantize, pack_codes,
+                                                unpack_codes)
+    torch.manual_seed(0)
+    x = torch.randn(777) * 2
+    fpq = FP_Quantize(group_size=128, q_bits=6)
+    q = fpq.quantize(x)
+    packed = pack_codes(q, 6)
+    assert packed.numel() < q.numel()  # denser than u16 container
+    q2 = unpack_codes(packed, 6, q.numel())
+    y = fpq.dequantize(q2)
+    y_ref = fpq.dequantize(q)
+    assert torch.equal(y, y_ref)
