@@ -80,3 +80,24 @@ def _init_inference_tp2():
 def test_init_inference_tensor_parallel_world2():
     from tests.common import run_distributed
     run_distributed(_init_inference_tp2, world_size=2)
+
+
+def test_hf_llama_import_logits_match():
+    """Native model loaded from an HF checkpoint reproduces HF logits."""
+    from transformers import LlamaConfig as HFConfig
+    from transformers import LlamaForCausalLM as HFModel
+    from deepspeed_amd.models.hf import load_hf_llama
+    torch.manual_seed(0)
+    hf_cfg = HFConfig(hidden_size=128, intermediate_size=256,
+                      num_hidden_layers=2, num_attention_heads=8,
+                      num_key_value_heads=4, vocab_size=512,
+                      max_position_embeddings=128, rms_norm_eps=1e-5,
+                      rope_theta=10000.0)
+    hf = HFModel(hf_cfg).eval()
+    native = load_hf_llama(hf).eval()
+    ids = torch.randint(0, 512, (2, 16))
+    with torch.no_grad():
+        ref = hf(ids).logits
+        out = native(ids)
+    err = (out - ref).abs().max().item()
+    assert err < 2e-4, err
