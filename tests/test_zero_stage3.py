@@ -157,3 +157,33 @@ def _zero3_sanity_mode(steps=2, grad_accum=1, persist_threshold=10):
 def test_zero3_sanity_asserts_world2():
     """DSAMD_SANITY cross-rank id checks pass on an honest run."""
     run_distributed(_zero3_sanity_mode, world_size=2)
+
+
+def _zero3_fp16_dynamic(steps=6):
+    import deepspeed_amd as ds
+    from tests.simple_model import SimpleModel
+    torch.manual_seed(0)
+    cfg = {
+        "train_micro_batch_size_per_gpu": 4,
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+        "fp16": {"enabled": True, "loss_scale": 0,
+                 "initial_scale_power": 24, "loss_scale_window": 2},
+        "zero_optimization": {"stage": 3, "sub_group_size": 200},
+    }
+    engine, _, _, _ = ds.initialize(model=SimpleModel(32), config=cfg)
+    x = torch.randn(4, 32).half() if not torch.cuda.is_available() \
+        else torch.randn(4, 32, device="cuda").half()
+    y = torch.randn_like(x)
+    losses = []
+    for _ in range(steps):
+        loss = engine(x, y)
+        engine.backward(loss)
+        engine.step()
+        losses.append(float(loss))
+    scale = engine.optimizer.loss_scaler.loss_scale
+    assert scale < 2.0 ** 24, f"overflow never backed the scale off: {scale}"
+    assert losses[-1] < losses[0] * 1.5  # training proceeds post-backoff
+
+
+def test_zero3_fp16_dynamic_scale_world2():
+    run_distributed(_zero3_fp16_dynamic, world_size=2)
