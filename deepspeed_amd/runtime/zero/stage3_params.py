@@ -271,8 +271,14 @@ class GatheredParameters:
     def __enter__(self):
         if not self.enabled:
             return
-        h = all_gather_params(self.params, self.dp_group)
-        h.wait()
+        # group by each param's partitioning group (expert params carry
+        # their expert-DP group in ds_group)
+        by_pg = {}
+        for p in self.params:
+            pg = getattr(p, "ds_group", None) or self.dp_group
+            by_pg.setdefault(id(pg), (pg, []))[1].append(p)
+        for pg, ps in by_pg.values():
+            all_gather_params(ps, pg).wait()
 
     def __exit__(self, *exc):
         if not self.enabled:

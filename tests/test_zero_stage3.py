@@ -187,3 +187,30 @@ def _zero3_fp16_dynamic(steps=6):
 
 def test_zero3_fp16_dynamic_scale_world2():
     run_distributed(_zero3_fp16_dynamic, world_size=2)
+
+
+def _zero_init_and_gather():
+    import torch.distributed as dist
+    from deepspeed_amd.runtime.zero.stage3_params import (GatheredParameters,
+                                                          Init, free_param)
+    with Init(param_persistence_threshold=0):
+        m = torch.nn.Linear(64, 64, bias=False)
+    p = m.weight
+    assert hasattr(p, "ds_tensor")
+    assert p.data.numel() == 0  # sharded away
+    world = dist.get_world_size()
+    assert p.ds_tensor.numel() * world >= p.ds_numel
+    with GatheredParameters(p):
+        assert p.data.numel() == 64 * 64
+        full = p.data.clone()
+    # modifier_rank: rank 0's edit propagates to all shards
+    with GatheredParameters(p, modifier_rank=0):
+        if dist.get_rank() == 0:
+            p.data.fill_(3.5)
+    with GatheredParameters(p):
+        assert torch.all(p.data == 3.5)
+    return float(full.sum())
+
+
+def test_zero_init_gathered_parameters_world2():
+    run_distributed(_zero_init_and_gather, world_size=2)
