@@ -61,6 +61,32 @@ class LinearLayerCompress(torch.nn.Linear):
         self.act_quant_enabled = True
         self.act_bits = bits
 
+    def enable_row_pruning(self, ratio, method="l1"):
+        """Zero whole output rows by L1 importance (structured)."""
+        imp = self.weight.data.abs().sum(dim=1)
+        k = int(imp.numel() * ratio)
+        if k > 0:
+            thresh = imp.kthvalue(k).values
+            mask = (imp > thresh).to(self.weight.dtype)
+            self.prune_mask = mask.unsqueeze(1).expand_as(self.weight) \
+                .contiguous()
+
+    def enable_head_pruning(self, ratio, num_heads):
+        """Zero whole attention heads (blocks of output rows) by L2
+        importance — for the concatenated q/k/v/o projections."""
+        O = self.weight.shape[0]
+        assert O % num_heads == 0
+        hs = O // num_heads
+        imp = self.weight.data.reshape(num_heads, hs, -1) \
+            .float().pow(2).sum(dim=(1, 2)).sqrt()
+        k = int(num_heads * ratio)
+        if k > 0:
+            thresh = imp.kthvalue(k).values
+            mask = (imp > thresh).to(self.weight.dtype)
+            self.prune_mask = mask.view(num_heads, 1, 1).expand(
+                num_heads, hs, self.weight.shape[1]) \
+                .reshape_as(self.weight).contiguous()
+
     def enable_sparse_pruning(self, ratio, method="l1"):
         w = self.weight.data.abs()
         k = int(w.numel() * ratio)
@@ -91,6 +117,9 @@ def init_compression(model, compression_config):
     """
     wq = compression_config.get("weight_quantization", {})
     sp = compression_config.get("sparse_pruning", {})
+    aq = compression_config.get("activation_quantization", {})
+    rp = compression_config.get("row_pruning", {})
+    hp = compression_config.get("head_pruning", {})
     n = 0
     for name, module in list(model.named_modules()):
         for child_name, child in list(module.named_children()):
@@ -112,6 +141,33 @@ def init_compression(model, compression_config):
                         new.enable_sparse_pruning(
                             g.get("params", {}).get("dense_ratio", 0.5))
                         break
+                for gname, g in aq.get("different_groups", {}).items():
+                    if any(re.search(pat, full)
+                           for pat in g.get("modules", [".*"])):
+                        if new is None:
+                            new = LinearLayerCompress.from_linear(child)
+                        new.enable_activation_quantization(
+                            g.get("params", {}).get("bits", 8))
+                        break
+                for gname, g in rp.get("different_groups", {}).items():
+                    if any(re.search(pat, full)
+                           for pat in g.get("modules", [".*"])):
+                        if new is None:
+                            new = LinearLayerCompress.from_linear(child)
+                        new.enable_row_pruning(
+                            1.0 - g.get("params", {}).get("dense_ratio",
+                                                          0.5))
+                        break
+                for gname, g in hp.get("different_groups", {}).items():
+                    if any(re.search(pat, full)
+                           for pat in g.get("modules", [".*"])):
+                        if new is None:
+                            new = LinearLayerCompress.from_linear(child)
+                        new.enable_head_pruning(
+                            1.0 - g.get("params", {}).get("dense_ratio",
+                                                          0.5),
+                            g.get("params", {}).get("num_heads", 1))
+                        break
                 if new is not None:
                     setattr(module, child_name, new)
                     n += 1
@@ -132,4 +188,27 @@ def redundancy_clean(model, compression_config=None):
                                   module.weight_bits)
                     .to(module.weight.dtype))
                 module.weight_quant_enabled = False
+    return model
+
+
+def apply_layer_reduction(model, layer_reduction_config):
+    """Student-model layer reduction (ref compression layer_reduction):
+    keep `keep_number` layers of the ModuleList at `module_name_prefix`,
+    selecting the listed `teacher_layer` indices."""
+    cfg = layer_reduction_config
+    prefix = cfg["module_name_prefix"]
+    keep = cfg.get("teacher_layer",
+                   list(range(cfg["keep_number"])))
+    holder = model
+    for part in prefix.split("."):
+        holder = getattr(holder, part)
+    assert isinstance(holder, torch.nn.ModuleList), prefix
+    new_layers = torch.nn.ModuleList([holder[i] for i in keep])
+    parent = model
+    parts = prefix.split(".")
+    for part in parts[:-1]:
+        parent = getattr(parent, part)
+    setattr(parent, parts[-1], new_layers)
+    log_dist(f"layer_reduction: kept {len(new_layers)} layers {keep}",
+             ranks=[0])
     return model

@@ -74,3 +74,51 @@ def _quant_reduce():
 
 def test_quantized_grad_reduce_2rank():
     assert all(run_distributed(_quant_reduce, world_size=2))
+
+
+def test_structured_pruning_and_layer_reduction():
+    from deepspeed_amd.compression import (LinearLayerCompress,
+                                           apply_layer_reduction,
+                                           init_compression)
+    torch.manual_seed(0)
+
+    class Tiny(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.layers = torch.nn.ModuleList(
+                [torch.nn.Linear(16, 16) for _ in range(6)])
+
+        def forward(self, x):
+            for l in self.layers:
+                x = torch.tanh(l(x))
+            return x
+
+    m = Tiny()
+    cfg = {"row_pruning": {"different_groups": {
+               "rp": {"params": {"dense_ratio": 0.5},
+                      "modules": ["layers.0"]}}},
+           "head_pruning": {"different_groups": {
+               "hp": {"params": {"dense_ratio": 0.5, "num_heads": 4},
+                      "modules": ["layers.1"]}}},
+           "activation_quantization": {"different_groups": {
+               "aq": {"params": {"bits": 8}, "modules": ["layers.2"]}}}}
+    init_compression(m, cfg)
+    assert isinstance(m.layers[0], LinearLayerCompress)
+    # row pruning: ~half the rows fully zero after clean
+    w0 = m.layers[0]
+    rows_zero = (w0.prune_mask.sum(1) == 0).float().mean().item()
+    assert 0.3 < rows_zero < 0.7
+    # head pruning: mask zeroes whole 4-row head blocks
+    w1 = m.layers[1]
+    head_mask = w1.prune_mask.reshape(4, 4, 16)
+    per_head = head_mask.amax(dim=(1, 2))
+    assert set(per_head.tolist()) <= {0.0, 1.0}
+    assert 0 < per_head.sum() < 4
+    out = m(torch.randn(2, 16))
+    assert out.shape == (2, 16)
+    # layer reduction: 6 -> 3 layers picking teacher indices
+    apply_layer_reduction(m, {"module_name_prefix": "layers",
+                              "keep_number": 3,
+                              "teacher_layer": [0, 2, 4]})
+    assert len(m.layers) == 3
+    assert m(torch.randn(2, 16)).shape == (2, 16)
