@@ -1,0 +1,2 @@
+from .auto_tp import apply_tensor_parallel  # noqa: F401
+from .layers import LinearAllreduce, LinearLayer  # noqa: F401

@@ -1,0 +1,2 @@
+from .monitor import (CSVMonitor, Monitor, MonitorMaster,  # noqa: F401
+                      TensorBoardMonitor, WandbMonitor)
