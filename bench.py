@@ -132,7 +132,7 @@ def main():
 
     for i in range(args.warmup):
         loss = one_step()
-    if rank == 0:
+    if rank == 0 and args.warmup > 0:
         print(f"# warmup done, loss={loss.item():.4f}", flush=True)
 
     if dist.is_initialized() and world > 1:
