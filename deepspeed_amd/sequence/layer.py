@@ -146,8 +146,17 @@ class DistributedAttention(torch.nn.Module):
             else groups.get_sequence_parallel_group()
         world = dist.get_world_size(spg)
         hq = query.shape[2]
-        uneven_q = hq if hq % world != 0 else None
         hk = key.shape[2]
+        if hk != hq and hk % world != 0 and world % hk == 0 \
+                and hq % world == 0:
+            # SP degree beyond the KV-head count (GQA/MQA): replicate KV
+            # heads so every rank holds whole groups; autograd sums the
+            # replica grads back (ref layer.py uneven/kv-replication path)
+            rep = world // hk
+            key = key.repeat_interleave(rep, dim=2)
+            value = value.repeat_interleave(rep, dim=2)
+            hk = key.shape[2]
+        uneven_q = hq if hq % world != 0 else None
         uneven_k = hk if hk % world != 0 else None
         if uneven_q or uneven_k:
             assert hq == hk, \

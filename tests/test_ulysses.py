@@ -177,3 +177,40 @@ def _ulysses_uneven_heads():
 
 def test_ulysses_uneven_heads_world2():
     run_distributed(_ulysses_uneven_heads, world_size=2)
+
+
+def _ulysses_kv_replication():
+    import torch.distributed as dist
+    import torch.nn.functional as F
+    from deepspeed_amd.sequence.layer import DistributedAttention
+
+    torch.manual_seed(0)
+    B, S, Hq, Hk, D = 2, 8, 4, 1, 4  # MQA, sp=2 > kv heads
+    world = dist.get_world_size()
+    rank = dist.get_rank()
+    q = torch.randn(B, S, Hq, D)
+    k = torch.randn(B, S, Hk, D)
+    v = torch.randn(B, S, Hk, D)
+    ref = F.scaled_dot_product_attention(
+        q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2),
+        is_causal=True, enable_gqa=True).transpose(1, 2)
+
+    def local_attn(q_, k_, v_, **kw):
+        return F.scaled_dot_product_attention(
+            q_.transpose(1, 2), k_.transpose(1, 2), v_.transpose(1, 2),
+            is_causal=True,
+            enable_gqa=(k_.shape[2] != q_.shape[2])).transpose(1, 2)
+
+    attn = DistributedAttention(local_attn, dist.group.WORLD)
+    s = S // world
+    sl = slice(rank * s, (rank + 1) * s)
+    kl = k[:, sl].clone().requires_grad_(True)
+    out = attn(q[:, sl], kl, v[:, sl])
+    err = (out - ref[:, sl]).abs().max().item()
+    assert err < 1e-5, err
+    out.sum().backward()
+    assert kl.grad is not None and torch.isfinite(kl.grad).all()
+
+
+def test_ulysses_kv_replication_world2():
+    run_distributed(_ulysses_kv_replication, world_size=2)
