@@ -68,8 +68,8 @@ def pick_sp_degree(seq_len, num_heads, num_kv_heads=None, world_size=None,
             continue
         if num_heads % d and not (mha and d <= num_heads):
             continue  # uneven distribution only supported for MHA
-        if num_kv_heads is not None and num_kv_heads % d and not mha:
-            continue
+        if not mha and num_kv_heads % d and d % num_kv_heads:
+            continue  # GQA: degree must divide kv heads OR replicate them
         if d <= want:
             best = d
     return best

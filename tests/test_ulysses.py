@@ -116,8 +116,10 @@ def test_autosp_pick_degree():
     # long sequence: largest divisor of world/heads/kv under the need
     assert pick_sp_degree(32768, 32, 8, world_size=8) == 4
     assert pick_sp_degree(65536, 32, 8, world_size=8) == 8
-    # kv head limit caps the degree
-    assert pick_sp_degree(65536, 32, 2, world_size=8) == 2
+    # kv heads below the degree: replication lifts the cap
+    assert pick_sp_degree(65536, 32, 2, world_size=8) == 8
+    # ...but a degree that splits a kv group unevenly is rejected
+    assert pick_sp_degree(65536, 32, 3, world_size=8) == 1
     # degree must divide world
     assert pick_sp_degree(65536, 32, 8, world_size=6) == 2
 
