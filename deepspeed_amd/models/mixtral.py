@@ -52,8 +52,9 @@ class MixtralDecoderLayer(nn.Module):
                                     ep_size=cfg.ep_size, k=cfg.top_k,
                                     capacity_factor=cfg.capacity_factor)
 
-    def forward(self, x, cos, sin):
-        x = x + self.self_attn(self.input_layernorm(x), cos, sin)
+    def forward(self, x, cos, sin, kv_cache=None):
+        x = x + self.self_attn(self.input_layernorm(x), cos, sin,
+                               kv_cache=kv_cache)
         moe_out, l_aux, _ = self.block_sparse_moe(
             self.post_attention_layernorm(x))
         return x + moe_out, l_aux
@@ -83,18 +84,25 @@ class MixtralForCausalLM(nn.Module):
         elif isinstance(module, nn.Embedding):
             module.weight.data.normal_(0.0, std)
 
-    def forward(self, input_ids, labels=None):
+    def forward(self, input_ids, labels=None, seq_offset=0, kv_caches=None,
+                positions=None):
         x = self.embed_tokens(input_ids)
         S = input_ids.shape[1]
-        cos = self.rope_cos[:S]
-        sin = self.rope_sin[:S]
+        if positions is not None:  # per-row positions (ragged decode)
+            cos = self.rope_cos[positions]
+            sin = self.rope_sin[positions]
+        else:
+            cos = self.rope_cos[seq_offset:seq_offset + S]
+            sin = self.rope_sin[seq_offset:seq_offset + S]
         aux_total = 0.0
-        for layer in self.layers:
-            if self.cfg.activation_checkpointing and self.training:
+        for i, layer in enumerate(self.layers):
+            kv = kv_caches[i] if kv_caches is not None else None
+            if self.cfg.activation_checkpointing and self.training \
+                    and kv is None:
                 x, l_aux = torch.utils.checkpoint.checkpoint(
                     layer, x, cos, sin, use_reentrant=False)
             else:
-                x, l_aux = layer(x, cos, sin)
+                x, l_aux = layer(x, cos, sin, kv_cache=kv)
             aux_total = aux_total + l_aux
         x = self.norm(x)
         logits = self.lm_head(x)

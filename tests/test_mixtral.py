@@ -57,3 +57,25 @@ def _mixtral_ep2(steps=4):
 
 def test_mixtral_ep2_zero1():
     run_distributed(_mixtral_ep2, world_size=2)
+
+
+def test_mixtral_generate_and_serving_cpu():
+    """MoE decode path: KV-cache generate + continuous batching."""
+    import torch
+    from deepspeed_amd.inference.engine import InferenceEngine
+    from deepspeed_amd.inference.serving import ContinuousBatchingEngine
+    from deepspeed_amd.models.mixtral import (MIXTRAL_CONFIGS,
+                                              MixtralForCausalLM)
+    torch.manual_seed(0)
+    model = MixtralForCausalLM(MIXTRAL_CONFIGS["mixtral-tiny"]).eval()
+    eng = InferenceEngine(model)
+    ids = torch.randint(0, model.cfg.vocab_size, (1, 7))
+    out = eng.generate(ids, max_new_tokens=6)
+    assert out.shape == (1, 13)
+    # NOTE: capacity-based gating sees different token counts in decode
+    # vs full recompute, so exact logits equality is not expected for
+    # MoE; determinism between the two CACHED paths is.
+    cb = ContinuousBatchingEngine(model, max_batch=2)
+    rid = cb.add_request(ids[0], max_new_tokens=6)
+    res = cb.run()
+    assert torch.equal(res[rid], out[0])
