@@ -167,9 +167,23 @@ class InferenceEngine(torch.nn.Module):
 
     @torch.no_grad()
     def generate(self, input_ids, max_new_tokens=32, temperature=0.0,
-                 top_k=0, eos_token_id=None):
-        """Greedy (temperature=0) or sampled generation with KV cache."""
+                 top_k=0, eos_token_id=None, **hf_kwargs):
+        """Greedy (temperature=0) or sampled generation with KV cache.
+
+        Native models decode through the in-tree cache path; HF models
+        (no .cfg) delegate to their own generate() with cache enabled.
+        """
         input_ids = input_ids.to(self.device)
+        if self._model_cfg is None and hasattr(self.module, "generate"):
+            kw = dict(max_new_tokens=max_new_tokens,
+                      do_sample=temperature > 0, **hf_kwargs)
+            if temperature > 0:
+                kw.update(temperature=temperature)
+                if top_k:
+                    kw.update(top_k=top_k)
+            if eos_token_id is not None:
+                kw.update(eos_token_id=eos_token_id)
+            return self.module.generate(input_ids, **kw)
         B, S = input_ids.shape
         cfg = self._model_cfg
         max_seq = min(cfg.max_position_embeddings,

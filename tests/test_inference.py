@@ -101,3 +101,18 @@ def test_hf_llama_import_logits_match():
         out = native(ids)
     err = (out - ref).abs().max().item()
     assert err < 2e-4, err
+
+
+def test_init_inference_hf_model_delegates_generate():
+    from transformers import LlamaConfig as HFConfig
+    from transformers import LlamaForCausalLM as HFModel
+    import deepspeed_amd as ds
+    torch.manual_seed(0)
+    hf = HFModel(HFConfig(hidden_size=64, intermediate_size=128,
+                          num_hidden_layers=2, num_attention_heads=4,
+                          num_key_value_heads=2, vocab_size=128,
+                          max_position_embeddings=64)).eval()
+    eng = ds.init_inference(hf, dtype=torch.float32)
+    ids = torch.randint(0, 128, (1, 8))
+    out = eng.generate(ids, max_new_tokens=4)
+    assert out.shape[1] == 12
