@@ -377,14 +377,17 @@ class ZeroStage3Optimizer:
 
     # ------------------------------------------------------------- hooks
     def _install_module_hooks(self):
-        """fetch/release hooks on every module owning direct params."""
+        """fetch/release hooks on every module owning direct params (or
+        registered external ones, see register_external_parameter)."""
         self._module_hooks = []
         for mod in self.module.modules():
             direct = [p for p in mod.parameters(recurse=False)
                       if is_zero_param(p)]
-            if not direct:
+            external = [p for p in getattr(mod, "_ds_external_params", [])
+                        if is_zero_param(p)]
+            if not direct and not external:
                 continue
-            mod._ds_direct_params = direct
+            mod._ds_direct_params = direct + external
             self._module_hooks.append(mod.register_forward_pre_hook(
                 self._pre_forward_hook))
             self._module_hooks.append(mod.register_forward_hook(

@@ -307,3 +307,22 @@ def _repartition(p):
     p.ds_status = ZeroParamStatus.AVAILABLE
     if not p.ds_persist:
         free_param(p)
+
+
+def register_external_parameter(module, parameter):
+    """Declare that `module`'s forward uses a parameter OWNED by another
+    module (ref runtime/zero/partition_parameters.py:
+    register_external_parameter) — e.g. a tied embedding weight read by
+    the output head. ZeRO-3 then gathers/releases it around this
+    module's forward/backward like a direct parameter. Call before
+    deepspeed.initialize().
+    """
+    if not hasattr(module, "_ds_external_params"):
+        module._ds_external_params = []
+    if parameter not in module._ds_external_params:
+        module._ds_external_params.append(parameter)
+
+
+def unregister_external_parameter(module, parameter):
+    if parameter in getattr(module, "_ds_external_params", []):
+        module._ds_external_params.remove(parameter)

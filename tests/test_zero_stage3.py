@@ -214,3 +214,54 @@ def _zero_init_and_gather():
 
 def test_zero_init_gathered_parameters_world2():
     run_distributed(_zero_init_and_gather, world_size=2)
+
+
+def _zero3_external_param(steps=3):
+    import deepspeed_amd as ds
+    from deepspeed_amd.runtime.zero import register_external_parameter
+
+    torch.manual_seed(0)
+
+    class TiedHead(torch.nn.Module):
+        """Output head that reads the embedding's weight (tied)."""
+
+        def __init__(self, embed):
+            super().__init__()
+            self._tied_weight = [embed.weight]  # not a submodule/param
+
+        def forward(self, x):
+            return x @ self._tied_weight[0].t()
+
+    class Net(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.embed = torch.nn.Embedding(64, 32)
+            self.mid = torch.nn.Linear(32, 32)
+            self.head = TiedHead(self.embed)
+            register_external_parameter(self.head, self.embed.weight)
+
+        def forward(self, ids):
+            h = torch.tanh(self.mid(self.embed(ids)))
+            return self.head(h)
+
+    cfg = {"train_micro_batch_size_per_gpu": 2,
+           "optimizer": {"type": "AdamW", "params": {"lr": 1e-2}},
+           "bf16": {"enabled": True},
+           "zero_optimization": {"stage": 3, "sub_group_size": 100,
+                                 "param_persistence_threshold": 0}}
+    engine, _, _, _ = ds.initialize(model=Net(), config=cfg)
+    ids = torch.randint(0, 64, (2, 6))
+    losses = []
+    for _ in range(steps):
+        logits = engine(ids)
+        loss = torch.nn.functional.cross_entropy(
+            logits.float().flatten(0, 1), ids.flatten())
+        engine.backward(loss)
+        engine.step()
+        losses.append(loss.item())
+    assert losses[-1] < losses[0], losses
+
+
+def test_zero3_register_external_parameter_world2():
+    """Tied weight used across modules: gathered for the head's forward."""
+    run_distributed(_zero3_external_param, world_size=2)
