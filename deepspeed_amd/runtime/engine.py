@@ -607,6 +607,10 @@ class DeepSpeedEngine(torch.nn.Module):
     def get_loss_scale(self):
         return getattr(self.optimizer, "loss_scale", 1.0)
 
+    def empty_partition_cache(self):
+        if hasattr(self.optimizer, "empty_partition_cache"):
+            self.optimizer.empty_partition_cache()
+
     def module_state_dict(self, exclude_frozen_parameters=False):
         return self.module.state_dict()
 

@@ -835,6 +835,16 @@ class ZeroStage3Optimizer:
             sg.copy_master_to_shards()
         self._refresh_persistent_params()
 
+    def empty_partition_cache(self):
+        """Release every gathered (non-persistent) param and return the
+        HBM to the allocator (ref engine.empty_partition_cache)."""
+        self._drain_inflight()
+        for p in self._all_params:
+            if not p.ds_persist:
+                free_param(p)
+        if torch.cuda.is_available():
+            torch.cuda.empty_cache()
+
     def destroy(self):
         for h in self._grad_hooks + self._module_hooks:
             h.remove()
