@@ -236,6 +236,25 @@ def scatter(tensor, scatter_list=None, src=0, group=None, async_op=False):
                               group=group, async_op=async_op)
 
 
+@timed_op
+def reduce_scatter(output, input_list, op=ReduceOp.SUM, group=None,
+                   async_op=False):
+    return torch_dist.reduce_scatter(output, input_list, op=op,
+                                     group=group, async_op=async_op)
+
+
+def all_gather_object(object_list, obj, group=None):
+    return torch_dist.all_gather_object(object_list, obj, group=group)
+
+
+def monitored_barrier(group=None, timeout=None, wait_all_ranks=False):
+    if torch_dist.get_backend(group) == "nccl":
+        # RCCL has no monitored barrier; plain barrier is the equivalent
+        return torch_dist.barrier(group=group)
+    return torch_dist.monitored_barrier(group=group, timeout=timeout,
+                                        wait_all_ranks=wait_all_ranks)
+
+
 def get_all_gather_function():
     return all_gather_into_tensor
 

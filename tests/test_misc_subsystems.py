@@ -214,3 +214,22 @@ def test_llama70b_meta_build_and_memory_estimate(capsys):
         total, largest_layer_params=int(1.5e9), num_gpus_per_node=8,
         cpu_offload=True)
     assert gpu_off < gpu
+
+
+def _comm_extra_ops():
+    import torch
+    import deepspeed_amd.comm as dist
+    world = dist.get_world_size()
+    out = torch.zeros(2)
+    ins = [torch.ones(2) * (dist.get_rank() + 1) for _ in range(world)]
+    dist.reduce_scatter(out, ins)
+    objs = [None] * world
+    dist.all_gather_object(objs, {"rank": dist.get_rank()})
+    assert [o["rank"] for o in objs] == list(range(world))
+    dist.monitored_barrier()
+    return float(out.sum())
+
+
+def test_comm_extra_ops_world2():
+    from tests.common import run_distributed
+    run_distributed(_comm_extra_ops, world_size=2)
