@@ -233,3 +233,12 @@ def _comm_extra_ops():
 def test_comm_extra_ops_world2():
     from tests.common import run_distributed
     run_distributed(_comm_extra_ops, world_size=2)
+
+
+def test_op_builder_shim():
+    from deepspeed_amd.ops.op_builder import (CPUAdamBuilder,
+                                              FusedAdamBuilder)
+    ext = FusedAdamBuilder().load()
+    assert hasattr(ext, "multi_tensor_adam")
+    assert CPUAdamBuilder().is_compatible()
+    assert CPUAdamBuilder().jit_load() is ext
