@@ -153,6 +153,21 @@ def load_checkpoint(engine, load_dir, tag=None, load_module_strict=True,
             esd = torch.load(efile, map_location="cpu", weights_only=False)
             engine.module.load_state_dict(esd["module"], strict=False)
 
+    if load_module_only and is_zero:
+        # Module-only load under ZeRO still has to reconcile the optimizer's
+        # fp32 masters with the freshly loaded 16-bit weights, or the stale
+        # masters overwrite them at the first step (ref engine.py:4525).
+        if engine.zero_optimization_stage() == 3:
+            # Stage 3: module weights live only in the zero shards — restore
+            # them from the per-dp-rank zero file without optimizer state.
+            dp_rank = engine.get_data_parallel_rank()
+            zfile = os.path.join(ckpt_dir, _zero_ckpt_name(dp_rank))
+            zstate = torch.load(zfile, map_location="cpu", weights_only=False)
+            engine.optimizer.load_state_dict(zstate["optimizer_state_dict"],
+                                             load_optimizer_states=False)
+        elif hasattr(engine.optimizer, "refresh_fp32_params"):
+            engine.optimizer.refresh_fp32_params()
+
     if not load_module_only:
         if is_zero:
             dp_rank = engine.get_data_parallel_rank()

@@ -434,7 +434,17 @@ class DeepSpeedEngine(torch.nn.Module):
             self.optimizer.backward(loss, retain_graph=retain_graph)
         else:
             loss.backward(retain_graph=retain_graph)
-            if self.is_gradient_accumulation_boundary():
+            # This micro-step has not been counted yet (micro_steps += 1
+            # happens below), so the boundary test must look one ahead —
+            # otherwise the allreduce fires one micro-batch late and the
+            # boundary step consumes unreduced rank-local grads
+            # (ref engine.py:3284 uses (micro_steps + 1) % GAS == 0).
+            if self._is_gradient_accumulation_boundary is not None:
+                at_boundary = self._is_gradient_accumulation_boundary
+            else:
+                at_boundary = (self.micro_steps + 1) % \
+                    self.gradient_accumulation_steps() == 0
+            if at_boundary:
                 self.allreduce_gradients()
         if self.wall_clock_breakdown():
             self.timers("backward").stop()

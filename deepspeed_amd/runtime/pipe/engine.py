@@ -119,6 +119,13 @@ class PipelineEngine(DeepSpeedEngine):
         b = cmd.buffer_id
         self._n_backwards += 1
         is_boundary = self._n_backwards == self.micro_batches
+        if getattr(self.module, "tied_comms", None):
+            # Tied-weight grads must be allreduced across stages BEFORE the
+            # ZeRO bucket reduce consumes the grad16 views; keep the boundary
+            # off during the final backward and flush every bucket in
+            # _exec_reduce_grads after allreduce_tied_weight_gradients()
+            # (ref pipe/engine.py _exec_reduce_tied_grads ordering).
+            is_boundary = False
         if hasattr(self.optimizer, "set_accumulation_boundary"):
             self.optimizer.set_accumulation_boundary(is_boundary)
         if hasattr(self.optimizer, "ensure_grad_views"):

@@ -580,10 +580,14 @@ class ZeroStage3Optimizer:
             try:
                 from torch.distributed.distributed_c10d import \
                     _coalescing_manager
-                op = _avg_op(world, inputs[0]) if inputs[0].is_cuda else None
+                from .stage_1_and_2 import _premul_avg_op
+                op = _premul_avg_op(world) if inputs[0].is_cuda else None
                 if op is None:
-                    for inp in inputs:
-                        inp.div_(world)
+                    # no on-wire averaging: pre-divide EVERY input in the
+                    # coalescing group, not just the first
+                    if world > 1:
+                        for inp in inputs:
+                            inp.div_(world)
                     op = dist.ReduceOp.SUM
                 with _coalescing_manager(pg, self.device,
                                          async_ops=True) as cm:

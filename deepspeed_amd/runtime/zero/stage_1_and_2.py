@@ -35,19 +35,32 @@ def _pad_to(numel, multiple):
     return (numel + multiple - 1) // multiple * multiple
 
 
+def _premul_avg_op(world):
+    """RCCL PreMulSum(1/world) op, or None if unavailable. Never mutates
+    any tensor — callers that get None must pre-divide EVERY input they
+    are about to reduce, not just the first."""
+    if world <= 1:
+        return None
+    try:
+        import torch.distributed as td
+        return td._make_nccl_premul_sum(1.0 / world)
+    except Exception:
+        return None
+
+
 def _avg_op(world, tensor):
     """RCCL PreMulSum(1/world) — averages on the wire, no pre-divide pass.
 
     Falls back to an explicit in-place pre-divide where unsupported (gloo).
+    Only valid for reducing exactly `tensor`; for a batch of tensors use
+    _premul_avg_op and pre-divide each one on fallback.
     """
     if world <= 1:
         return dist.ReduceOp.SUM
     if tensor.is_cuda:
-        try:
-            import torch.distributed as td
-            return td._make_nccl_premul_sum(1.0 / world)
-        except Exception:
-            pass
+        op = _premul_avg_op(world)
+        if op is not None:
+            return op
     tensor.div_(world)
     return dist.ReduceOp.SUM
 
@@ -419,6 +432,12 @@ class ZeroStage12Optimizer:
                 "group_idx": b.group_idx,
                 "numel_padded": b.numel_padded,
                 "shard_numel": b.shard_numel,
+                # expert buckets are partitioned over the (smaller)
+                # expert-DP group and hold DIFFERENT experts per EP rank —
+                # offline reassembly must not concatenate them across the
+                # full DP world (source them from expert_ep_rank files)
+                "world": b.world,
+                "expert": b.pg is not None,
                 "params": [(names.get(id(p), f"param_{i}_{j}"),
                             b.offsets[p], p.numel(), list(p.shape))
                            for j, p in enumerate(b.params)],

@@ -41,6 +41,13 @@ def _reassemble(shards, key="flat"):
         flats_per_rank = [sd["single_partition_of_fp32_groups"]
                           for sd in shards]
         for bi, binfo in enumerate(layout["buckets"]):
+            if binfo.get("expert"):
+                # Expert buckets are partitioned over the expert-DP group
+                # (each EP rank holds DIFFERENT experts): concatenating
+                # across all dp-rank files would interleave unrelated
+                # experts. Those params are sourced from the
+                # expert_ep_rank model-state files instead.
+                continue
             full = torch.cat([flats[bi].detach().float()
                               for flats in flats_per_rank])
             for name, off, numel, shape in binfo["params"]:
@@ -72,6 +79,14 @@ def get_fp32_state_dict_from_zero_checkpoint(checkpoint_dir, tag=None):
         for bname in ms.get("buffer_names", []):
             if bname in ms["module"]:
                 state_dict[bname] = ms["module"][bname]
+    # MoE (stages 1/2): expert params come from the per-EP-rank expert
+    # files — each carries its own experts' 16-bit weights (ref
+    # _save_moe_checkpoint, engine.py:4921)
+    for efile in sorted(glob.glob(os.path.join(
+            dirpath, "expert_ep_rank_*_model_states.pt"))):
+        esd = torch.load(efile, map_location="cpu", weights_only=False)
+        for name, t in esd["module"].items():
+            state_dict[name] = t.detach().float()
     return state_dict
 
 
