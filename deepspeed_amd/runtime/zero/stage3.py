@@ -222,6 +222,7 @@ class ZeroStage3Optimizer:
         self._inflight = {}         # module -> AllGatherHandle
 
         self._global_grad_norm = 0.0
+        self._cached_norm_sq = None
         # opt-in cross-rank sanity asserts (DSAMD_SANITY=1): verify every
         # rank reduces the same params in the same order
         self._sanity = _os.environ.get("DSAMD_SANITY") == "1"
@@ -625,36 +626,41 @@ class ZeroStage3Optimizer:
     def is_gradient_accumulation_boundary(self):
         return self.micro_step % self.gradient_accumulation_steps == 0
 
-    def has_overflow(self):
-        found = False
-        for sg in self.sub_groups:
-            s = sg.grad32.sum()
-            if torch.isinf(s) or torch.isnan(s):
-                found = True
-                break
+    def _grad_norm_sq(self):
+        """One pass over the fp32 grad slabs: squared L2 norm (double),
+        all-reduced over DP. Serves BOTH the overflow check (inf/nan in any
+        grad makes the squared sum non-finite) and grad clipping — the
+        separate grad32.sum() scan the fp16 path used to do is folded in."""
+        grads = [sg.grad32 for sg in self.sub_groups]
+        if grads and grads[0].is_cuda:
+            from ...ops.loader import get_ext
+            total_sq = get_ext(required=True).l2norm_sq(grads).double()
+        else:
+            total_sq = torch.zeros(1, dtype=torch.float64)
+            for g in grads:
+                s = g.double()
+                total_sq += torch.dot(s, s)
+        total_sq = total_sq.reshape(1)
         if dist.is_initialized():
-            t = torch.tensor([1.0 if found else 0.0], device=self.device)
-            dist.all_reduce(t, op=dist.ReduceOp.MAX, group=self.dp_group)
-            found = bool(t.item())
-        return found
+            total_sq = total_sq.to(self.device)
+            dist.all_reduce(total_sq, group=self.dp_group)
+        return total_sq
+
+    def has_overflow(self):
+        total_sq = self._cached_norm_sq if self._cached_norm_sq is not None \
+            else self._grad_norm_sq()
+        self._cached_norm_sq = total_sq
+        return not bool(torch.isfinite(total_sq).item())
 
     def _combined_scale(self):
         """loss_scale x clip coefficient; folded into the Adam kernel as
-        grad_scale = 1/combined (zero extra passes over the 32 GB shards)."""
+        grad_scale = 1/combined (zero extra passes over the 32 GB shards).
+        Reuses the norm pass the fp16 overflow check already did."""
         scale = self.loss_scaler.loss_scale
         combined = scale
         if self.clip_grad > 0.0:
-            grads = [sg.grad32 for sg in self.sub_groups]
-            if grads and grads[0].is_cuda:
-                from ...ops.loader import get_ext
-                total_sq = get_ext(required=True).l2norm_sq(grads).double()
-            else:
-                total_sq = torch.zeros(1, dtype=torch.float64)
-                for g in grads:
-                    total_sq += float(torch.linalg.vector_norm(g))**2
-            if dist.is_initialized():
-                total_sq = total_sq.to(self.device)
-                dist.all_reduce(total_sq, group=self.dp_group)
+            total_sq = self._cached_norm_sq \
+                if self._cached_norm_sq is not None else self._grad_norm_sq()
             norm = total_sq.sqrt().item() / scale
             self._global_grad_norm = norm
             clip = norm / self.clip_grad
@@ -684,6 +690,7 @@ class ZeroStage3Optimizer:
         self._flush_ipg()
         self._sync_comm_streams()
         self._drain_inflight()
+        self._cached_norm_sq = None  # one norm pass per step, shared below
 
         if self.dtype == torch.float16:
             self.overflow = self.has_overflow()

@@ -26,7 +26,7 @@ def _worker(rank, world_size, fn, args, kwargs, init_file, result_dir, backend):
                                 init_method=f"file://{init_file}",
                                 rank=rank, world_size=world_size)
         if backend == "nccl":
-            torch.cuda.set_device(rank)
+            torch.cuda.set_device(rank % torch.cuda.device_count())
         result = fn(*args, **kwargs)
         with open(os.path.join(result_dir, f"result_{rank}.pkl"), "wb") as f:
             pickle.dump(("ok", result), f)
