@@ -19,7 +19,8 @@ DEFAULT_TIMEOUT_S = 300
 def _worker(rank, world_size, fn, args, kwargs, init_file, result_dir, backend):
     try:
         os.environ["RANK"] = str(rank)
-        os.environ["LOCAL_RANK"] = str(rank)
+        ndev = torch.cuda.device_count() if backend == "nccl" else 0
+        os.environ["LOCAL_RANK"] = str(rank % ndev if ndev else rank)
         os.environ["WORLD_SIZE"] = str(world_size)
         os.environ["MASTER_ADDR"] = "127.0.0.1"
         dist.init_process_group(backend=backend,

@@ -7,11 +7,13 @@ Two layers of evidence, both runnable on a single MI355X:
    reduce_and_partition_stream semantics) actually executes on a GPU and
    must produce the same training result as the single-stream path.
 
-2. world_size=2 with both ranks on one GPU over RCCL — exercises the real
-   collective code paths (reduce-scatter/all-gather/all-to-all on the
-   nccl backend). RCCL may refuse duplicate devices like NCCL does; in
-   that case these tests skip with the exact error, and the forced-stream
-   tests remain the hardware evidence.
+2. world_size=2 over RCCL — exercises the real collective code paths
+   (reduce-scatter/all-gather/all-to-all on the nccl backend). Measured
+   2026-09 on this pool: RCCL 2.26.6 REFUSES two ranks on one device
+   (ncclInvalidUsage "Duplicate GPU detected"), so these tests skip on a
+   1-GPU box and run for real whenever >=2 GPUs are visible (e.g. the
+   driver's 8-GPU scaling box); the forced-stream tests above are the
+   1-GPU hardware evidence.
 """
 import os
 
@@ -125,14 +127,11 @@ def test_zero3_forced_streams_fp16_overflow_path():
 
 # ------------------------------------------------- world-2 on one GPU
 
-def _skip_if_duplicate_gpu_unsupported(exc):
-    msg = str(exc)
-    for pat in ("Duplicate GPU", "duplicate GPU", "invalid usage",
-                "invalidUsage", "unhandled cuda error", "NCCL error"):
-        if pat in msg:
-            pytest.skip(f"RCCL refuses multiple ranks on one device: "
-                        f"{msg[-300:]}")
-    raise exc
+def _need_two_gpus():
+    if torch.cuda.device_count() < 2:
+        pytest.skip("needs >=2 GPUs: RCCL 2.26.6 refuses duplicate devices "
+                    "(ncclInvalidUsage 'Duplicate GPU detected', measured "
+                    "2026-09 on MI355X pool)")
 
 
 def _w2_zero3(stage=3, steps=4):
@@ -174,22 +173,18 @@ def test_world2_rccl_zero3_one_gpu():
     """2 ranks sharing cuda:0 over RCCL: full ZeRO-3 collective path
     (coalesced reduce-scatter, all-gather prefetch, overlap streams at
     world>1) executes on hardware."""
-    try:
-        results = run_distributed(_w2_zero3, world_size=2, backend="nccl",
-                                  args=(3,), timeout=420)
-    except AssertionError as e:
-        _skip_if_duplicate_gpu_unsupported(e)
+    _need_two_gpus()
+    results = run_distributed(_w2_zero3, world_size=2, backend="nccl",
+                              args=(3,), timeout=420)
     (l0, n0), (l1, n1) = results
     assert l0[-1] < l0[0], f"rank0 no progress: {l0}"
     assert n0 == n1
 
 
 def test_world2_rccl_zero2_one_gpu():
-    try:
-        results = run_distributed(_w2_zero3, world_size=2, backend="nccl",
-                                  args=(2,), timeout=420)
-    except AssertionError as e:
-        _skip_if_duplicate_gpu_unsupported(e)
+    _need_two_gpus()
+    results = run_distributed(_w2_zero3, world_size=2, backend="nccl",
+                              args=(2,), timeout=420)
     (l0, _), (l1, _) = results
     assert l0[-1] < l0[0], f"rank0 no progress: {l0}"
 
@@ -226,11 +221,9 @@ def _w2_ulysses(steps=2):
 
 def test_world2_rccl_ulysses_one_gpu():
     """Ulysses degree-2 all-to-all over RCCL matches full-sequence SDPA."""
-    try:
-        results = run_distributed(_w2_ulysses, world_size=2, backend="nccl",
-                                  timeout=300)
-    except AssertionError as e:
-        _skip_if_duplicate_gpu_unsupported(e)
+    _need_two_gpus()
+    results = run_distributed(_w2_ulysses, world_size=2, backend="nccl",
+                              timeout=300)
     assert all(r < 5e-2 for r in results)
 
 
@@ -284,9 +277,7 @@ def _w2_moe(steps=3):
 
 def test_world2_rccl_moe_ep2_one_gpu():
     """MoE expert-parallel all-to-all dispatch (EP=2) over RCCL."""
-    try:
-        results = run_distributed(_w2_moe, world_size=2, backend="nccl",
-                                  timeout=300)
-    except AssertionError as e:
-        _skip_if_duplicate_gpu_unsupported(e)
+    _need_two_gpus()
+    results = run_distributed(_w2_moe, world_size=2, backend="nccl",
+                              timeout=300)
     assert results[0][-1] < results[0][0] * 1.5
