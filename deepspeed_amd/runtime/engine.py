@@ -308,6 +308,7 @@ class DeepSpeedEngine(torch.nn.Module):
                 sub_group_size=zc.sub_group_size,
                 overlap_comm=zc.overlap_comm,
                 offload_optimizer=zc.offload_optimizer,
+                offload_param=zc.offload_param,
                 clip_grad=self.gradient_clipping(),
                 static_loss_scale=self._static_loss_scale(),
                 dynamic_loss_scale=self._dynamic_loss_scale(),

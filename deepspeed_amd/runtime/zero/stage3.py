@@ -40,35 +40,50 @@ class SubGroup:
     same pass as the fp32 update (no separate cast+copy over 16 GB)."""
 
     __slots__ = ("params", "offsets", "master32", "grad32", "flat16",
-                 "flat16_cpu", "grad_stage", "group_idx", "numel", "offload")
+                 "flat16_cpu", "grad_stage", "group_idx", "numel", "offload",
+                 "param_offload", "dtype16")
 
     def __init__(self, params, offsets, numel, group_idx, device,
-                 offload=False):
+                 offload=False, param_offload=False):
         self.params = params
         self.offsets = offsets
         self.numel = numel
         self.group_idx = group_idx
         self.offload = offload
+        self.param_offload = param_offload
         dtype16 = params[0].ds_tensor.dtype if params else torch.bfloat16
-        pin = offload and torch.cuda.is_available()
+        self.dtype16 = dtype16
+        pin = torch.cuda.is_available()
         host = torch.device("cpu")
         state_dev = host if offload else device
-        self.flat16 = torch.empty(numel, dtype=dtype16, device=device)
+        # param offload (ZeRO-Infinity tier): the 16-bit shard slab itself
+        # lives in pinned host memory; fetches stage it H2D on the gather
+        # stream (stage3_params._shard_on_device)
+        slab_dev = host if param_offload else device
+        self.flat16 = torch.empty(numel, dtype=dtype16, device=slab_dev,
+                                  pin_memory=pin and param_offload)
         self.master32 = torch.empty(numel, dtype=torch.float32,
-                                    device=state_dev, pin_memory=pin)
+                                    device=state_dev,
+                                    pin_memory=pin and offload)
         for p in params:
             off = offsets[p]
-            n = p.ds_tensor.numel()
+            n = p.ds_shard_numel
             self.flat16[off:off + n].copy_(p.ds_tensor)
             p.ds_tensor = self.flat16[off:off + n]
             self.master32[off:off + n].copy_(p.ds_tensor.float())
         self.master32 = self.master32.detach().requires_grad_(True)
         self.grad32 = torch.zeros(numel, dtype=torch.float32,
-                                  device=state_dev, pin_memory=pin)
+                                  device=state_dev,
+                                  pin_memory=pin and offload)
         if offload:
-            # pinned staging: bf16 shard out (H2D) + grad shard in (D2H)
-            self.flat16_cpu = torch.empty(numel, dtype=dtype16, device=host,
-                                          pin_memory=pin)
+            if param_offload:
+                # the host slab IS the 16-bit output buffer: the fused CPU
+                # Adam writes it in place, nothing to publish H2D
+                self.flat16_cpu = self.flat16
+            else:
+                # pinned staging: bf16 shard out (H2D)
+                self.flat16_cpu = torch.empty(numel, dtype=dtype16,
+                                              device=host, pin_memory=pin)
             self.grad_stage = torch.empty(numel, dtype=dtype16, device=host,
                                           pin_memory=pin)
         else:
@@ -77,7 +92,7 @@ class SubGroup:
 
     def grad_shard_view(self, p):
         off = self.offsets[p]
-        return self.grad32[off:off + p.ds_tensor.numel()]
+        return self.grad32[off:off + p.ds_shard_numel]
 
     def accumulate_grad(self, p, shard16_gpu):
         """shard16_gpu: this rank's reduced gradient shard (device)."""
@@ -97,16 +112,21 @@ class SubGroup:
 
     def copy_master_to_shards(self):
         if self.offload:
-            self.flat16_cpu.copy_(self.master32.detach().to(torch.bfloat16)
-                                  if self.flat16_cpu.dtype == torch.bfloat16
+            self.flat16_cpu.copy_(self.master32.detach().to(self.dtype16)
+                                  if self.flat16_cpu.dtype != torch.float32
                                   else self.master32.detach())
-            self.flat16.copy_(self.flat16_cpu, non_blocking=True)
+            if self.flat16_cpu is not self.flat16:
+                self.flat16.copy_(self.flat16_cpu, non_blocking=True)
         else:
+            # cross-device copy_ handles dtype+D2H in one pass when the
+            # param slab is host-resident (param offload, GPU optimizer)
             self.flat16.copy_(self.master32.detach())
 
     def publish_flat16(self):
-        """After a fused CPU-Adam step wrote flat16_cpu, push H2D."""
-        self.flat16.copy_(self.flat16_cpu, non_blocking=True)
+        """After a fused CPU-Adam step wrote flat16_cpu, push H2D (no-op
+        when the slab is host-resident: flat16_cpu IS flat16)."""
+        if self.flat16_cpu is not self.flat16:
+            self.flat16.copy_(self.flat16_cpu, non_blocking=True)
 
 
 class ZeroStage3Optimizer:
@@ -122,6 +142,7 @@ class ZeroStage3Optimizer:
                  sub_group_size=int(1e9),
                  overlap_comm=True,
                  offload_optimizer=None,
+                 offload_param=None,
                  clip_grad=0.0,
                  static_loss_scale=1.0,
                  dynamic_loss_scale=False,
@@ -154,6 +175,23 @@ class ZeroStage3Optimizer:
         self.offload_device = getattr(offload_optimizer, "device", "none") \
             if offload_optimizer is not None else "none"
         self.offload_optimizer = self.offload_device in ("cpu", "nvme")
+        # ZeRO-Infinity parameter tier (ref partitioned_param_swapper.py:37):
+        # cpu => 16-bit shard slabs live in pinned host RAM; nvme => slabs
+        # additionally spill to O_DIRECT files with an LRU host budget
+        self.param_offload_device = getattr(offload_param, "device", "none") \
+            if offload_param is not None else "none"
+        self.param_offload = self.param_offload_device in ("cpu", "nvme")
+        self.param_swapper = None
+        if self.param_offload_device == "nvme":
+            from ..swap_tensor.param_swapper import ParamSlabSwapper
+            import os as _os
+            ppath = getattr(offload_param, "nvme_path", None) \
+                or "/tmp/dsamd_nvme_swap"
+            ppath = _os.path.join(ppath, f"param_rank{self.rank}")
+            self.param_swapper = ParamSlabSwapper(
+                ppath,
+                max_in_cpu=getattr(offload_param, "max_in_cpu", int(1e9)),
+                pin_memory=getattr(offload_param, "pin_memory", True))
         self.nvme_swapper = None
         if self.offload_device == "nvme":
             from ..swap_tensor.optimizer_swapper import OptimizerStateSwapper
@@ -192,6 +230,11 @@ class ZeroStage3Optimizer:
             raise NotImplementedError("MiCS + expert parallelism")
         self._shard_module_params()
         self._build_sub_groups()
+        if self.param_swapper is not None:
+            for i, sg in enumerate(self.sub_groups):
+                self.param_swapper.register(i, sg)
+            self._sg_index = {id(sg): i for i, sg in
+                              enumerate(self.sub_groups)}
         if self.nvme_swapper is not None:
             self._init_nvme_state()
         self._gather_persistent_params()
@@ -240,6 +283,9 @@ class ZeroStage3Optimizer:
             if dist.is_initialized() else 1
         src = (dist.get_global_rank(bcast_group, 0)
                if bcast_group is not None and bcast_world > 1 else 0)
+        # param offload: shard straight to host pinned memory so a big
+        # model never stages its full shards through HBM at init
+        shard_dev = torch.device("cpu") if self.param_offload else self.device
         for p in params:
             if not is_zero_param(p):
                 p.data = p.data.to(self.device, self.dtype)
@@ -252,12 +298,12 @@ class ZeroStage3Optimizer:
                         dist.broadcast(p.data,
                                        dist.get_global_rank(pg, 0),
                                        group=pg)
-                    convert_to_zero_param(p, pg, self.device, self.dtype,
+                    convert_to_zero_param(p, pg, shard_dev, self.dtype,
                                           self.persist_threshold)
                     continue
                 if bcast_world > 1:
                     dist.broadcast(p.data, src, group=bcast_group)
-                convert_to_zero_param(p, self.dp_group, self.device,
+                convert_to_zero_param(p, self.dp_group, shard_dev,
                                       self.dtype, self.persist_threshold)
         for b in self.module.buffers():
             b.data = b.data.to(self.device)
@@ -281,15 +327,17 @@ class ZeroStage3Optimizer:
                 if numel >= self.sub_group_size and cur:
                     self.sub_groups.append(
                         SubGroup(cur, offsets, numel, gi, self.device,
-                                 offload=self.offload_optimizer))
+                                 offload=self.offload_optimizer,
+                                 param_offload=self.param_offload))
                     cur, offsets, numel = [], OrderedDict(), 0
                 offsets[p] = numel
                 cur.append(p)
-                numel += p.ds_tensor.numel()
+                numel += p.ds_shard_numel
             if cur:
                 self.sub_groups.append(
                     SubGroup(cur, offsets, numel, gi, self.device,
-                             offload=self.offload_optimizer))
+                             offload=self.offload_optimizer,
+                             param_offload=self.param_offload))
         self.param_to_subgroup = {}
         for sg in self.sub_groups:
             for p in sg.params:
@@ -319,6 +367,8 @@ class ZeroStage3Optimizer:
         for g in self.optimizer.param_groups:
             g["step"] = g.get("step", 0) + 1
         for i, sg in enumerate(self.sub_groups):
+            if self.param_swapper is not None:
+                self.param_swapper.ensure_resident(i)
             group = self.optimizer.param_groups[sg.group_idx]
             beta1, beta2 = group.get("betas", (0.9, 0.999))
             master, ea, eas = self.nvme_swapper.swap_in(i, sg.numel)
@@ -344,6 +394,42 @@ class ZeroStage3Optimizer:
             self.nvme_swapper.swap_out(i)
             self.nvme_swapper.release_buffers(i)
             sg.publish_flat16()
+            if self.param_swapper is not None:
+                self.param_swapper.mark_dirty(i)
+                self.param_swapper.evict_to_budget()
+
+    def _param_offload_step(self, combined, fused):
+        """NVMe param tier: per-sub-group step under the host-RAM budget.
+        Each slab is made resident, updated (fused CPU/GPU Adam writes the
+        16-bit host slab in the same pass), marked dirty, then eligible
+        for eviction back to its O_DIRECT file."""
+        bases = [g.get("step", 0) for g in self.optimizer.param_groups]
+        if fused:
+            self.optimizer.set_grad_scale(1.0 / combined)
+        for i, sg in enumerate(self.sub_groups):
+            self.param_swapper.ensure_resident(i)
+            # optimizer.step() bumps each group's step counter once per
+            # call; pin it so N sub-group calls count as ONE step
+            for g, b in zip(self.optimizer.param_groups, bases):
+                g["step"] = b
+            if fused:
+                self.optimizer.set_fused_out16(
+                    {sg.master32: (sg.flat16_cpu if sg.offload
+                                   else sg.flat16)})
+            elif combined != 1.0:
+                sg.grad32.mul_(1.0 / combined)
+            sg.master32.grad = sg.grad32
+            self.optimizer.step()
+            sg.master32.grad = None
+            if not fused:
+                sg.copy_master_to_shards()
+            elif sg.offload:
+                sg.publish_flat16()
+            self.param_swapper.mark_dirty(i)
+            self.param_swapper.evict_to_budget()
+        if fused:
+            self.optimizer.set_grad_scale(1.0)
+            self.optimizer.set_fused_out16({})
 
     def _param_pg(self, p):
         return getattr(p, "ds_group", None) or self.dp_group
@@ -352,6 +438,17 @@ class ZeroStage3Optimizer:
         """Launch one gather per distinct partitioning group (dense
         params share dp_group; expert params use their expert-DP group).
         Returns a handle with .wait()."""
+        if self.param_swapper is not None:
+            # NVMe param tier: bring evicted shard slabs back to pinned
+            # host RAM before the H2D + all-gather; the whole working set
+            # of this fetch is protected from eviction
+            need_ids = []
+            for pm in params:
+                sg = self.param_to_subgroup.get(pm)
+                if sg is not None and self._sg_index[id(sg)] not in need_ids:
+                    need_ids.append(self._sg_index[id(sg)])
+            for i in need_ids:
+                self.param_swapper.ensure_resident(i, exclude=need_ids)
         by_pg = {}
         for p in params:
             by_pg.setdefault(id(self._param_pg(p)), []).append(p)
@@ -564,7 +661,7 @@ class ZeroStage3Optimizer:
         from .stage_1_and_2 import _avg_op
         inputs = []
         for p in params:
-            shard_numel = p.ds_tensor.numel()
+            shard_numel = p.ds_shard_numel
             if shard_numel * world == p.ds_numel:
                 # aligned param: reduce-scatter the autograd grad in place
                 padded = p.grad.reshape(-1)
@@ -710,6 +807,11 @@ class ZeroStage3Optimizer:
             return
 
         fused = hasattr(self.optimizer, "set_grad_scale")
+        if self.param_swapper is not None:
+            self._param_offload_step(combined, fused)
+            self._clear_grads()
+            self._refresh_persistent_params()
+            return
         if fused:
             self.optimizer.set_grad_scale(1.0 / combined)
             self.optimizer.set_fused_out16(
@@ -768,7 +870,7 @@ class ZeroStage3Optimizer:
                 "group_idx": sg.group_idx,
                 "numel": sg.numel,
                 "params": [(names.get(id(p), f"param_{p.ds_id}"),
-                            sg.offsets[p], p.ds_tensor.numel(), p.ds_numel,
+                            sg.offsets[p], p.ds_shard_numel, p.ds_numel,
                             list(p.ds_shape)) for p in sg.params],
             })
         return {"stage": 3, "world": self.world, "kind": "subgroup",
@@ -815,7 +917,7 @@ class ZeroStage3Optimizer:
             for p in sg.params:
                 shapes[pmap.get(id(p), f"param_{p.ds_id}")] = {
                     "shape": p.ds_shape, "numel": p.ds_numel,
-                    "shard_numel": p.ds_tensor.numel(),
+                    "shard_numel": p.ds_shard_numel,
                     "subgroup_offset": sg.offsets[p]}
         return shapes
 
@@ -830,6 +932,8 @@ class ZeroStage3Optimizer:
             eas = sd.get("nvme_exp_avg")
             eass = sd.get("nvme_exp_avg_sq")
             for i, (sg, m) in enumerate(zip(self.sub_groups, saved)):
+                if self.param_swapper is not None:
+                    self.param_swapper.ensure_resident(i)
                 mb, eab, easb = self.nvme_swapper.swap_in(i, sg.numel)
                 mb.copy_(m.cpu())
                 if eas is not None:
@@ -839,11 +943,19 @@ class ZeroStage3Optimizer:
                 self.nvme_swapper.swap_out(i)
                 self.nvme_swapper.release_buffers(i)
                 sg.publish_flat16()
+                if self.param_swapper is not None:
+                    self.param_swapper.mark_dirty(i)
+                    self.param_swapper.evict_to_budget()
             self._refresh_persistent_params()
             return
-        for sg, s in zip(self.sub_groups, saved):
+        for i, (sg, s) in enumerate(zip(self.sub_groups, saved)):
+            if self.param_swapper is not None:
+                self.param_swapper.ensure_resident(i)
             sg.master32.data.copy_(s.data.to(sg.master32.device))
             sg.copy_master_to_shards()
+            if self.param_swapper is not None:
+                self.param_swapper.mark_dirty(i)
+                self.param_swapper.evict_to_budget()
         self._refresh_persistent_params()
 
     def empty_partition_cache(self):

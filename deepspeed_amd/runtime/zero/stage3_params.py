@@ -77,6 +77,8 @@ def convert_to_zero_param(p, dp_group, device, dtype,
     if end > start:
         ds_tensor[:end - start].copy_(flat[start:end])
     p.ds_tensor = ds_tensor
+    p.ds_shard_numel = shard_numel  # stable even when the shard is swapped
+    p.ds_dtype = dtype
     p.ds_group = dp_group  # partitioning group (expert-DP for MoE experts)
     p.ds_persist = p.ds_numel <= persist_threshold
     p.ds_status = ZeroParamStatus.NOT_AVAILABLE
@@ -87,10 +89,11 @@ def convert_to_zero_param(p, dp_group, device, dtype,
 
 
 def free_param(p):
-    """Drop the gathered full tensor; shard stays."""
+    """Drop the gathered full tensor; shard stays (possibly offloaded)."""
     if p.ds_status == ZeroParamStatus.AVAILABLE and not p.ds_persist:
-        p.data = torch.empty(0, dtype=p.ds_tensor.dtype,
-                             device=p.ds_tensor.device)
+        dev = torch.device("cuda") if torch.cuda.is_available() \
+            else torch.device("cpu")
+        p.data = torch.empty(0, dtype=p.ds_dtype, device=dev)
         p.ds_full_buffer = None
         p.ds_status = ZeroParamStatus.NOT_AVAILABLE
 
@@ -145,6 +148,27 @@ def all_gather_params(params, dp_group, async_op=True, stream=None):
     buffers = []
     works = []
     if world == 1:
+        if torch.cuda.is_available() and not todo[0].ds_tensor.is_cuda:
+            # param offload: shards live in pinned host memory — the
+            # "gather" at world 1 is one async H2D copy per param on the
+            # side stream (SDMA blit engines, no CUs; ref
+            # partitioned_param_swapper.py:291 swap_in role)
+            if stream is not None:
+                stream.wait_stream(torch.cuda.current_stream())
+                ctx = torch.cuda.stream(stream)
+            else:
+                import contextlib
+                ctx = contextlib.nullcontext()
+            with ctx:
+                for p in todo:
+                    buf = torch.empty(p.ds_shard_numel,
+                                      dtype=p.ds_tensor.dtype, device="cuda")
+                    buf.copy_(p.ds_tensor, non_blocking=True)
+                    buffers.append(buf)
+            h = AllGatherHandle(todo, [], buffers, dp_group, stream=stream)
+            if stream is None:
+                h.wait()
+            return h
         for p in todo:
             buffers.append(p.ds_tensor)
         h = AllGatherHandle(todo, [], buffers, dp_group)
@@ -158,6 +182,18 @@ def all_gather_params(params, dp_group, async_op=True, stream=None):
     return _launch_gathers(todo, dp_group, world, async_op, None)
 
 
+def _shard_on_device(p):
+    """RCCL gather input must be a device tensor; offloaded shards are
+    staged through one async pinned-H2D copy (enqueued on the caller's
+    active stream, so it pipelines ahead of the collective)."""
+    if p.ds_tensor.is_cuda or not torch.cuda.is_available():
+        return p.ds_tensor
+    dev = torch.empty(p.ds_shard_numel, dtype=p.ds_tensor.dtype,
+                      device="cuda")
+    dev.copy_(p.ds_tensor, non_blocking=True)
+    return dev
+
+
 def _launch_gathers(todo, dp_group, world, async_op, stream):
     buffers = []
     works = []
@@ -165,13 +201,13 @@ def _launch_gathers(todo, dp_group, world, async_op, stream):
     if use_coalescing:
         try:
             from torch.distributed.distributed_c10d import _coalescing_manager
-            device = todo[0].ds_tensor.device
+            shards = [_shard_on_device(p) for p in todo]
+            device = shards[0].device
             with _coalescing_manager(dp_group, device, async_ops=True) as cm:
-                for p in todo:
-                    buf = torch.empty(p.ds_tensor.numel() * world,
-                                      dtype=p.ds_tensor.dtype, device=device)
-                    dist.all_gather_into_tensor(buf, p.ds_tensor,
-                                                group=dp_group)
+                for p, sh in zip(todo, shards):
+                    buf = torch.empty(sh.numel() * world,
+                                      dtype=sh.dtype, device=device)
+                    dist.all_gather_into_tensor(buf, sh, group=dp_group)
                     buffers.append(buf)
             works = [cm]
         except Exception as e:
@@ -183,10 +219,11 @@ def _launch_gathers(todo, dp_group, world, async_op, stream):
             use_coalescing = False
     if not use_coalescing:
         for p in todo:
-            buf = torch.empty(p.ds_tensor.numel() * world,
-                              dtype=p.ds_tensor.dtype,
-                              device=p.ds_tensor.device)
-            w = dist.all_gather_into_tensor(buf, p.ds_tensor, group=dp_group,
+            sh = _shard_on_device(p)
+            buf = torch.empty(sh.numel() * world,
+                              dtype=sh.dtype,
+                              device=sh.device)
+            w = dist.all_gather_into_tensor(buf, sh, group=dp_group,
                                             async_op=async_op)
             buffers.append(buf)
             works.append(w)

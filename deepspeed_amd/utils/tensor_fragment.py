@@ -20,7 +20,7 @@ def _find_owner(optimizer, param):
     if hasattr(optimizer, "param_to_subgroup"):
         sg = optimizer.param_to_subgroup.get(param)
         if sg is not None:
-            return "subgroup", sg, sg.offsets[param], param.ds_tensor.numel()
+            return "subgroup", sg, sg.offsets[param], param.ds_shard_numel
     return None, None, None, None
 
 
