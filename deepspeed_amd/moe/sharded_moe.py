@@ -78,6 +78,52 @@ def top1gating(logits, capacity_factor, min_capacity=4, drop_tokens=True):
     return topkgating(logits, 1, capacity_factor, min_capacity, drop_tokens)
 
 
+def topkgating_indices(logits, k, capacity_factor, min_capacity=4,
+                       drop_tokens=True, ep_group=None):
+    """Index-form top-k gating: O(S*k) dispatch metadata instead of the
+    [S, E, C] one-hot tensors (ref MOELayer fast paths, sharded_moe.py:618;
+    at Mixtral scale the dense combine tensor would be S*E*C elements).
+
+    Returns (l_aux, topk_idx [S,k], weights [S,k] fp32, locations [S,k],
+    keep [S,k] bool, capacity).
+    """
+    gates = F.softmax(logits, dim=1)
+    num_tokens, num_experts = gates.shape
+    capacity = _capacity(gates, capacity_factor * k, min_capacity)
+
+    topk_vals, topk_idx = torch.topk(gates, k, dim=1)  # [S, k]
+    mask = torch.zeros_like(gates)
+    mask.scatter_(1, topk_idx, 1.0)
+
+    me = gates.mean(dim=0)
+    ce = mask.float().mean(dim=0)
+    l_aux = torch.sum(me * ce) * num_experts * num_experts / k
+
+    # position of each token within its expert queue (token order)
+    locations = (torch.cumsum(mask, dim=0) - 1)
+    loc_sj = locations.gather(1, topk_idx).long()  # [S, k]
+
+    if drop_tokens:
+        keep = loc_sj < capacity
+    else:
+        # capacity grows to the longest queue; must agree across the EP
+        # group so the all-to-all shapes match (ref top2gating no-drop)
+        local_max = int(loc_sj.max().item()) + 1 if num_tokens else 1
+        if ep_group is not None and dist.get_world_size(ep_group) > 1:
+            t = torch.tensor([max(capacity, local_max)],
+                             device=logits.device)
+            dist.all_reduce(t, op=dist.ReduceOp.MAX, group=ep_group)
+            capacity = int(t.item())
+        else:
+            capacity = max(capacity, local_max)
+        keep = torch.ones_like(loc_sj, dtype=torch.bool)
+
+    vals = topk_vals * keep  # drop, then renormalize over kept experts
+    denom = vals.sum(dim=1, keepdim=True).clamp(min=1e-9)
+    weights = vals / denom
+    return l_aux, topk_idx, weights, loc_sj, keep, capacity
+
+
 class TopKGate(torch.nn.Module):
     """Gate module (ref sharded_moe.py:528). Keeps wg in fp32."""
 
@@ -125,10 +171,30 @@ class MOELayer(torch.nn.Module):
     def forward(self, input_):
         d_model = input_.shape[-1]
         reshaped = input_.reshape(-1, d_model)
-        self.l_aux, combine, dispatch, C = self.gate(reshaped)
-        E = combine.shape[1]
-        dispatched = torch.einsum("sec,sm->ecm",
-                                  dispatch.to(input_.dtype), reshaped)
+        S = reshaped.shape[0]
+        gate = self.gate
+        logits = gate.wg(reshaped.float() if gate.wg.weight.dtype ==
+                         torch.float32 else reshaped).float()
+        cf = gate.capacity_factor if self.training \
+            else gate.eval_capacity_factor
+        self.l_aux, idx, w, loc, keep, C = topkgating_indices(
+            logits, gate.k, cf, gate.min_capacity, gate.drop_tokens,
+            ep_group=self.ep_group if self.ep_size > 1 else None)
+        E = logits.shape[1]
+        self.exp_counts = keep.sum(0) if gate.k > 1 else None
+
+        # index dispatch: each kept (token, choice) owns one unique slot
+        # e*C + loc in the flat [E*C, M] buffer — O(S*k) work/memory, no
+        # [S, E, C] one-hot tensors
+        flat_pos = idx * C + loc                      # [S, k]
+        keep_f = keep.reshape(-1)
+        kept_pos = flat_pos.reshape(-1)[keep_f]       # [N]
+        token_idx = torch.arange(S, device=reshaped.device) \
+            .unsqueeze(1).expand(S, gate.k).reshape(-1)[keep_f]
+        dispatched = reshaped.new_zeros(E * C, d_model).index_copy(
+            0, kept_pos, reshaped.index_select(0, token_idx))
+        dispatched = dispatched.reshape(E, C, d_model)
+
         if self.ep_size > 1:
             dispatched = _AllToAll.apply(self.ep_group, dispatched)
         # [E, C, M] -> [ep_size, local_experts, C, M]
@@ -138,6 +204,13 @@ class MOELayer(torch.nn.Module):
         expert_out = expert_out.reshape(E, C, d_model)
         if self.ep_size > 1:
             expert_out = _AllToAll.apply(self.ep_group, expert_out)
-        combined = torch.einsum("sec,ecm->sm", combine.to(input_.dtype),
-                                expert_out)
+
+        # index combine: gather each choice's expert output row and take
+        # the gate-weighted sum over k (dropped choices carry weight 0)
+        expert_flat = expert_out.reshape(E * C, d_model)
+        safe_pos = flat_pos.reshape(-1).clamp_(max=E * C - 1)
+        gathered = expert_flat.index_select(0, safe_pos) \
+            .reshape(S, gate.k, d_model)
+        wk = (w * keep).to(gathered.dtype).unsqueeze(-1)
+        combined = (wk * gathered).sum(dim=1)
         return combined.reshape(input_.shape)
