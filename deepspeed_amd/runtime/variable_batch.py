@@ -111,7 +111,7 @@ class VariableBatchSizeLR:
     def step(self, batch_id=None):
         if batch_id is None:
             batch_id = self._batch
-            self._batch += 1
+        self._batch = batch_id + 1
         if self.inner is not None:
             self.inner.step()
             self.base_lrs = self.inner.get_last_lr()
