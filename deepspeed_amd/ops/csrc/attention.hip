@@ -292,6 +292,15 @@ DEV_INLINE unsigned int cvtpk2(float lo, float hi) {
   return (unsigned int)a | ((unsigned int)b << 16);
 }
 
+// swizzled row-major byte offset for a [R][DH] bf16 tile
+template <int DH>
+DEV_INLINE int kswz_t(int row, int byte_off) {
+  return row * (DH * 2) + (byte_off ^ ((row & 7) << 4));
+}
+
+// kvmask: optional [B,S] additive float mask over kv columns (padding
+// masks, ALiBi-free BERT-class workloads); nullptr = none.
+template <int DH>
 __launch_bounds__(512, 2)
 __global__ void flash_fwd_v5_kernel(
     const short* __restrict__ q,  // [B,S,Hq,D]
@@ -299,17 +308,21 @@ __global__ void flash_fwd_v5_kernel(
     const short* __restrict__ v,  // [B,S,Hk,D]
     short* __restrict__ out,      // [B,S,Hq,D]
     float* __restrict__ lse_out,  // [B,Hq,S]
+    const float* __restrict__ kvmask,  // [B,S] or null
     int B, int S, int Hq, int Hk, float scale, int causal) {
   // K double-buffered: tile t+1's global_load_lds issues during tile t's
   // compute (async-STAGE, guide T14); V global loads land in registers one
   // tile early and scatter to LDS after the barrier.
-  __shared__ short k_lds[2][KT5 * DHEAD];              // swizzled row-major
-  __shared__ short vt_lds[DHEAD * (KT5 + VT_PAD)];     // transposed
+  __shared__ short k_lds[2][KT5 * DH];              // swizzled row-major
+  __shared__ short vt_lds[DH * (KT5 + VT_PAD)];     // transposed
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int l31 = lane & 31;
   const int hi = lane >> 5;  // half-wave: 0 or 1
+
+  constexpr int NQF = DH / 16;   // Q k-slices
+  constexpr int NOT = DH / 32;   // O^T d m-tiles
 
   const int qtile = blockIdx.x;
   const int bh = blockIdx.y;
@@ -320,19 +333,20 @@ __global__ void flash_fwd_v5_kernel(
   const int wave_q = qbase + wave * QB5;
   const int my_q = wave_q + l31;  // this lane's q row (lane-local softmax)
 
-  const long long qrs = (long long)Hq * DHEAD;
-  const long long kvrs = (long long)Hk * DHEAD;
-  const short* qp = q + ((long long)b * S) * qrs + h * DHEAD;
-  const short* kp = k + ((long long)b * S) * kvrs + hk * DHEAD;
-  const short* vp = v + ((long long)b * S) * kvrs + hk * DHEAD;
+  const long long qrs = (long long)Hq * DH;
+  const long long kvrs = (long long)Hk * DH;
+  const short* qp = q + ((long long)b * S) * qrs + h * DH;
+  const short* kp = k + ((long long)b * S) * kvrs + hk * DH;
+  const short* vp = v + ((long long)b * S) * kvrs + hk * DH;
+  const float* mp = kvmask ? kvmask + (long long)b * S : nullptr;
 
   // ---- Q fragments: qfrag[ks] = Q[my_q][16*ks + hi*8 .. +7] -------------
-  bf16x8_t qfrag[8];
+  bf16x8_t qfrag[NQF];
   {
     int srow = my_q < S ? my_q : S - 1;
     const short* src = qp + (long long)srow * qrs + hi * 8;
 #pragma unroll
-    for (int ks = 0; ks < 8; ++ks)
+    for (int ks = 0; ks < NQF; ++ks)
       qfrag[ks] = *reinterpret_cast<const bf16x8_t*>(src + 16 * ks);
   }
 
@@ -341,9 +355,9 @@ __global__ void flash_fwd_v5_kernel(
   // hidden *log2e multiply; LSE converts back with ln2 at the epilogue.
   const float sc2 = scale * 1.44269504f;
   float m_run = -INFINITY, l_run = 0.f;
-  f32x16_t ot[4];  // O^T accum: rows d = 32*mt + (reg&3)+8*(reg>>2)+4*hi
+  f32x16_t ot[NOT];  // O^T accum: rows d = 32*mt + (reg&3)+8*(reg>>2)+4*hi
 #pragma unroll
-  for (int mt = 0; mt < 4; ++mt)
+  for (int mt = 0; mt < NOT; ++mt)
 #pragma unroll
     for (int r = 0; r < 16; ++r) ot[mt][r] = 0.f;
 
@@ -351,17 +365,24 @@ __global__ void flash_fwd_v5_kernel(
   const int n_tiles = (kv_end + KT5 - 1) / KT5;
 
   // async-STAGE helpers -----------------------------------------------------
-  const int st_row = threadIdx.x >> 3;        // V staging: row this thread
-  const int st_c0 = (threadIdx.x & 7) * 16;   //            first col
+  // V staging: DH/16 threads per row, 16 cols each (guard rows >= KT5 for
+  // DH < 128 where 512 threads cover more than one tile)
+  constexpr int VTPR = DH / 16;
+  const int st_row = threadIdx.x / VTPR;
+  const int st_c0 = (threadIdx.x % VTPR) * 16;
+  // K tile = KT5*DH*2 bytes in 1024-byte wave segments (64 lanes x 16 B)
+  constexpr int KSEGS = KT5 * DH * 2 / 1024;
+  constexpr int KPASSES = (KSEGS + 7) / 8;
 
 #define ISSUE_K(t_)                                                          \
   {                                                                          \
     int kvb_ = (t_)*KT5;                                                     \
     short* kbuf_ = k_lds[(t_)&1];                                            \
-    for (int pass = 0; pass < 2; ++pass) {                                   \
-      int linear = (pass * 8 + wave) * 1024 + lane * 16;                     \
-      int row = linear >> 8;                                                 \
-      int colbyte = linear & 255;                                            \
+    for (int pass = 0; pass < KPASSES; ++pass) {                             \
+      int seg = pass * 8 + wave;                                             \
+      int linear = seg * 1024 + lane * 16;                                   \
+      int row = linear / (DH * 2);                                           \
+      int colbyte = linear % (DH * 2);                                       \
       int src_col = colbyte ^ ((row & 7) << 4);                              \
       int grow = kvb_ + row;                                                 \
       int srow = grow < S ? grow : S - 1;                                    \
@@ -369,7 +390,7 @@ __global__ void flash_fwd_v5_kernel(
       __builtin_amdgcn_global_load_lds(                                      \
           reinterpret_cast<const unsigned int*>(src),                        \
           reinterpret_cast<unsigned int*>(reinterpret_cast<char*>(kbuf_) +   \
-                                          (pass * 8 + wave) * 1024),         \
+                                          seg * 1024),                       \
           16, 0, 0);                                                         \
     }                                                                        \
   }
@@ -378,7 +399,7 @@ __global__ void flash_fwd_v5_kernel(
   {                                                                          \
     int grow = (t_)*KT5 + st_row;                                            \
     const short* vrow = vp + (long long)grow * kvrs;                         \
-    if (grow < S) {                                                          \
+    if (st_row < KT5 && grow < S) {                                          \
       dst_[0] = *reinterpret_cast<const bf16x8_t*>(vrow + st_c0);            \
       dst_[1] = *reinterpret_cast<const bf16x8_t*>(vrow + st_c0 + 8);        \
     } else {                                                                 \
@@ -397,7 +418,7 @@ __global__ void flash_fwd_v5_kernel(
     if (has_next) ISSUE_K(t + 1);  // lands in the other K buffer
     // scatter this tile's V registers into LDS (the compiler's wait for
     // vreg also retires this tile's older global_load_lds -> k_lds)
-    {
+    if (st_row < KT5) {
       char* vbase = reinterpret_cast<char*>(vt_lds);
 #pragma unroll
       for (int cc = 0; cc < 2; ++cc) {
@@ -421,10 +442,10 @@ __global__ void flash_fwd_v5_kernel(
       for (int r = 0; r < 16; ++r) acc[r] = 0.f;
       int krow = 32 * mt + l31;
 #pragma unroll
-      for (int ks = 0; ks < 8; ++ks) {
+      for (int ks = 0; ks < NQF; ++ks) {
         bf16x8_t kf = *reinterpret_cast<const bf16x8_t*>(
             reinterpret_cast<char*>(k_lds[t & 1]) +
-            kswz(krow, (16 * ks + hi * 8) * 2));
+            kswz_t<DH>(krow, (16 * ks + hi * 8) * 2));
         acc = MFMA_BF16_32x32x16(kf, qfrag[ks], acc, 0, 0, 0);
       }
       st[mt] = acc;
@@ -434,7 +455,7 @@ __global__ void flash_fwd_v5_kernel(
     // ---- online softmax (lane-local q row) ------------------------------
     // btile is wave-uniform: interior tiles take the cmp-free path
     const bool btile = (causal && (kvbase + KT5 > qbase)) ||
-                       (kvbase + KT5 > S);
+                       (kvbase + KT5 > S) || mp != nullptr;
     float tmax = -INFINITY;
     if (btile) {
 #pragma unroll
@@ -443,6 +464,7 @@ __global__ void flash_fwd_v5_kernel(
         for (int r = 0; r < 16; ++r) {
           float sv = st[mt][r] * sc2;
           int gkv = kvbase + 32 * mt + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          if (mp && gkv < S) sv += mp[gkv] * 1.44269504f;
           if ((causal && gkv > my_q) || gkv >= S) sv = -INFINITY;
           st[mt][r] = sv;
           tmax = fmaxf(tmax, sv);
@@ -466,7 +488,7 @@ __global__ void flash_fwd_v5_kernel(
       float c = (m_run == -INFINITY) ? 0.f : __builtin_amdgcn_exp2f(m_run - m_new);
       l_run *= c;
 #pragma unroll
-      for (int mt = 0; mt < 4; ++mt)
+      for (int mt = 0; mt < NOT; ++mt)
 #pragma unroll
         for (int r = 0; r < 16; ++r) ot[mt][r] *= c;
       m_run = m_new;
@@ -511,7 +533,7 @@ __global__ void flash_fwd_v5_kernel(
     // ---- O^T += V^T P^T  (4 d m-tiles x 4 kv k-slices) ------------------
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int mt = 0; mt < 4; ++mt) {
+    for (int mt = 0; mt < NOT; ++mt) {
       int drow = 32 * mt + l31;
 #pragma unroll
       for (int ks = 0; ks < 4; ++ks) {
@@ -532,9 +554,9 @@ __global__ void flash_fwd_v5_kernel(
   // ---- epilogue: O^T lane holds d rows for its q col --------------------
   if (my_q < S) {
     float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
-    short* orow = out + ((long long)b * S + my_q) * qrs + h * DHEAD;
+    short* orow = out + ((long long)b * S + my_q) * qrs + h * DH;
 #pragma unroll
-    for (int mt = 0; mt < 4; ++mt)
+    for (int mt = 0; mt < NOT; ++mt)
 #pragma unroll
       for (int g = 0; g < 4; ++g) {
         int d0 = 32 * mt + 8 * g + 4 * hi;  // 4 consecutive d
@@ -552,13 +574,22 @@ __global__ void flash_fwd_v5_kernel(
 
 std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
                                        at::Tensor v, bool causal,
-                                       double scale) {
+                                       double scale,
+                                       c10::optional<at::Tensor> kvmask) {
   TORCH_CHECK(q.scalar_type() == at::kBFloat16);
   TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
   int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
   int Hk = k.size(2);
-  TORCH_CHECK(D == 128, "flash_attn_fwd: head_dim must be 128");
+  TORCH_CHECK(D == 128 || D == 64,
+              "flash_attn_fwd: head_dim must be 64 or 128");
   TORCH_CHECK(Hq % Hk == 0, "GQA requires Hq % Hk == 0");
+  const float* mp = nullptr;
+  if (kvmask.has_value() && kvmask->defined()) {
+    TORCH_CHECK(kvmask->scalar_type() == at::kFloat &&
+                kvmask->is_contiguous() && kvmask->numel() == (long)B * S,
+                "kvmask must be contiguous float [B,S]");
+    mp = kvmask->data_ptr<float>();
+  }
   auto out = at::empty_like(q);
   auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
   auto stream = c10::hip::getCurrentHIPStream();
@@ -566,7 +597,7 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
     const char* e = getenv("DSAMD_ATTN_FWD_V4");
     return e && e[0] == '1';
   }();
-  if (use_v4) {
+  if (use_v4 && D == 128 && !mp) {
     dim3 grid((S + QTILE - 1) / QTILE, B * Hq);
     hipLaunchKernelGGL(flash_fwd_kernel, grid, dim3(512), 0, stream.stream(),
                        reinterpret_cast<const short*>(q.data_ptr()),
@@ -577,14 +608,25 @@ std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
                        causal ? 1 : 0);
   } else {
     dim3 grid((S + QT5 - 1) / QT5, B * Hq);
-    hipLaunchKernelGGL(flash_fwd_v5_kernel, grid, dim3(512), 0,
-                       stream.stream(),
-                       reinterpret_cast<const short*>(q.data_ptr()),
-                       reinterpret_cast<const short*>(k.data_ptr()),
-                       reinterpret_cast<const short*>(v.data_ptr()),
-                       reinterpret_cast<short*>(out.data_ptr()),
-                       lse.data_ptr<float>(), B, S, Hq, Hk, (float)scale,
-                       causal ? 1 : 0);
+    if (D == 128) {
+      hipLaunchKernelGGL(flash_fwd_v5_kernel<128>, grid, dim3(512), 0,
+                         stream.stream(),
+                         reinterpret_cast<const short*>(q.data_ptr()),
+                         reinterpret_cast<const short*>(k.data_ptr()),
+                         reinterpret_cast<const short*>(v.data_ptr()),
+                         reinterpret_cast<short*>(out.data_ptr()),
+                         lse.data_ptr<float>(), mp, B, S, Hq, Hk,
+                         (float)scale, causal ? 1 : 0);
+    } else {
+      hipLaunchKernelGGL(flash_fwd_v5_kernel<64>, grid, dim3(512), 0,
+                         stream.stream(),
+                         reinterpret_cast<const short*>(q.data_ptr()),
+                         reinterpret_cast<const short*>(k.data_ptr()),
+                         reinterpret_cast<const short*>(v.data_ptr()),
+                         reinterpret_cast<short*>(out.data_ptr()),
+                         lse.data_ptr<float>(), mp, B, S, Hq, Hk,
+                         (float)scale, causal ? 1 : 0);
+    }
   }
   HIP_CHECK_KERNEL();
   return {out, lse};

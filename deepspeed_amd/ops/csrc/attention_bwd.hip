@@ -24,9 +24,8 @@ using f32x4_t = __attribute__((ext_vector_type(4))) float;
 
 namespace bwd {
 
-constexpr int DHEAD = 128;
-
 // drow[b,h,s] = sum_d dout[b,s,h,d] * out[b,s,h,d]  (one wave per row)
+template <int DH>
 __global__ void drow_kernel(const short* __restrict__ dout,
                             const short* __restrict__ out,
                             float* __restrict__ drow, int S, int Hq,
@@ -36,13 +35,16 @@ __global__ void drow_kernel(const short* __restrict__ dout,
   long long row = (long long)blockIdx.x * 4 + wave;  // over B*S*Hq
   if (row >= n_rows) return;
   const long long nbsh = (long long)S * Hq;
-  const short* a = dout + row * DHEAD;
-  const short* b = out + row * DHEAD;
-  // 128 elems over 64 lanes: one 2-elem (4 B) load per lane
-  unsigned int au = reinterpret_cast<const unsigned int*>(a)[lane];
-  unsigned int bu = reinterpret_cast<const unsigned int*>(b)[lane];
-  float acc = bf2f((short)(au & 0xffff)) * bf2f((short)(bu & 0xffff)) +
-              bf2f((short)(au >> 16)) * bf2f((short)(bu >> 16));
+  const short* a = dout + row * DH;
+  const short* b = out + row * DH;
+  // DH/2 u32 words over 64 lanes (1 word/lane at DH=128)
+  float acc = 0.f;
+  for (int w = lane; w < DH / 2; w += 64) {
+    unsigned int au = reinterpret_cast<const unsigned int*>(a)[w];
+    unsigned int bu = reinterpret_cast<const unsigned int*>(b)[w];
+    acc += bf2f((short)(au & 0xffff)) * bf2f((short)(bu & 0xffff)) +
+           bf2f((short)(au >> 16)) * bf2f((short)(bu >> 16));
+  }
   acc = wave_reduce_sum(acc);
   if (lane == 0) {
     // row index [b, s, h] -> drow[b, h, s]
@@ -54,9 +56,10 @@ __global__ void drow_kernel(const short* __restrict__ dout,
   }
 }
 
-// row-major [R][128] bf16 with ((row&7)<<4) XOR
+// row-major [R][DH] bf16 with ((row&7)<<4) XOR
+template <int DH>
 DEV_INLINE int rmswz(int row, int byte_off) {
-  return row * (DHEAD * 2) + (byte_off ^ ((row & 7) << 4));
+  return row * (DH * 2) + (byte_off ^ ((row & 7) << 4));
 }
 
 // transposed [128][W + 8] bf16 with 16B-granule XOR on the inner index
@@ -71,6 +74,7 @@ DEV_INLINE short f2bf_(float f) { return f2bf(f); }
 // PASS 1: kv-outer. Block: 4 waves x 16 kv rows = 64 kv rows; iterates q
 // tiles of 32. Computes per-(q-head) dK, dV.
 // ---------------------------------------------------------------------------
+template <int DH>
 __launch_bounds__(256, 2)
 __global__ void flash_bwd_dkv_kernel(
     const short* __restrict__ q,     // [B,S,Hq,D]
@@ -79,14 +83,17 @@ __global__ void flash_bwd_dkv_kernel(
     const short* __restrict__ dout,  // [B,S,Hq,D]
     const float* __restrict__ lse,   // [B,Hq,S]
     const float* __restrict__ drow,  // [B,Hq,S]
+    const float* __restrict__ kvmask,  // [B,S] or null
     short* __restrict__ dk,          // [B,S,Hk,D]
     short* __restrict__ dv,          // [B,S,Hk,D]
     int B, int S, int Hq, int Hk, float scale, int causal) {
   constexpr int QIT = 32;
-  __shared__ short q_lds[QIT * DHEAD];               // row-major swz
-  __shared__ short do_lds[QIT * DHEAD];              // row-major swz
-  __shared__ short qt_lds[DHEAD * (QIT + 8)];        // transposed
-  __shared__ short dot_lds[DHEAD * (QIT + 8)];       // transposed
+  constexpr int NKS = DH / 32;  // A-operand k-slices
+  constexpr int NDT = DH / 16;  // d-tiles
+  __shared__ short q_lds[QIT * DH];               // row-major swz
+  __shared__ short do_lds[QIT * DH];              // row-major swz
+  __shared__ short qt_lds[DH * (QIT + 8)];        // transposed
+  __shared__ short dot_lds[DH * (QIT + 8)];       // transposed
   __shared__ short pw_lds[4][16 * (QIT + 8)];        // per-wave P^T stage
   __shared__ short dsw_lds[4][16 * (QIT + 8)];       // per-wave dS^T stage
 
@@ -103,28 +110,29 @@ __global__ void flash_bwd_dkv_kernel(
   const int kv_base = kv_block * 64;
   const int wave_kv = kv_base + wave * 16;
 
-  const long long qrs = (long long)Hq * DHEAD;
-  const long long kvrs = (long long)Hk * DHEAD;
-  const short* kp = k + ((long long)b * S) * kvrs + hk * DHEAD;
-  const short* vp = v + ((long long)b * S) * kvrs + hk * DHEAD;
+  const long long qrs = (long long)Hq * DH;
+  const long long kvrs = (long long)Hk * DH;
+  const short* kp = k + ((long long)b * S) * kvrs + hk * DH;
+  const short* vp = v + ((long long)b * S) * kvrs + hk * DH;
+  const float* mp = kvmask ? kvmask + (long long)b * S : nullptr;
 
-  // K/V fragments in registers (A-operand layout, 4 d-slices)
-  bf16x8_t kfrag[4], vfrag[4];
+  // K/V fragments in registers (A-operand layout, DH/32 d-slices)
+  bf16x8_t kfrag[NKS], vfrag[NKS];
   {
     int krow = wave_kv + l15;
     int srow = krow < S ? krow : S - 1;
     const short* ks = kp + (long long)srow * kvrs;
     const short* vs = vp + (long long)srow * kvrs;
 #pragma unroll
-    for (int ks_i = 0; ks_i < 4; ++ks_i) {
+    for (int ks_i = 0; ks_i < NKS; ++ks_i) {
       kfrag[ks_i] = *reinterpret_cast<const bf16x8_t*>(ks + ks_i * 32 + l4 * 8);
       vfrag[ks_i] = *reinterpret_cast<const bf16x8_t*>(vs + ks_i * 32 + l4 * 8);
     }
   }
 
-  f32x4_t dk_acc[8], dv_acc[8];
+  f32x4_t dk_acc[NDT], dv_acc[NDT];
 #pragma unroll
-  for (int dt = 0; dt < 8; ++dt) {
+  for (int dt = 0; dt < NDT; ++dt) {
     dk_acc[dt] = f32x4_t{0.f, 0.f, 0.f, 0.f};
     dv_acc[dt] = f32x4_t{0.f, 0.f, 0.f, 0.f};
   }
@@ -133,36 +141,40 @@ __global__ void flash_bwd_dkv_kernel(
   // accumulate over the GQA group: all q-heads sharing this kv head
   for (int g = 0; g < G; ++g) {
   const int h = hk * G + g;
-  const short* qp = q + ((long long)b * S) * qrs + h * DHEAD;
-  const short* dop = dout + ((long long)b * S) * qrs + h * DHEAD;
+  const short* qp = q + ((long long)b * S) * qrs + h * DH;
+  const short* dop = dout + ((long long)b * S) * qrs + h * DH;
   const float* lsep = lse + ((long long)b * Hq + h) * S;
   const float* drp = drow + ((long long)b * Hq + h) * S;
   for (int qb = q_start; qb < S; qb += QIT) {
     // ---- stage Q/dO (row-major swizzled + transposed) ------------------
     {
-      int row = threadIdx.x >> 3;        // 0..31
-      int c0 = (threadIdx.x & 7) * 16;
-      int grow = qb + row;
-      int srow = grow < S ? grow : S - 1;
-      const short* qs = qp + (long long)srow * qrs;
-      const short* ds = dop + (long long)srow * qrs;
-      char* qb_ = reinterpret_cast<char*>(q_lds);
-      char* db_ = reinterpret_cast<char*>(do_lds);
-      char* qtb = reinterpret_cast<char*>(qt_lds);
-      char* dtb = reinterpret_cast<char*>(dot_lds);
+      // DH/16 threads per row, 16 cols each (2 x bf16x8)
+      constexpr int TPR = DH / 16;
+      int row = threadIdx.x / TPR;
+      int c0 = (threadIdx.x % TPR) * 16;
+      if (row < QIT) {
+        int grow = qb + row;
+        int srow = grow < S ? grow : S - 1;
+        const short* qs = qp + (long long)srow * qrs;
+        const short* ds = dop + (long long)srow * qrs;
+        char* qb_ = reinterpret_cast<char*>(q_lds);
+        char* db_ = reinterpret_cast<char*>(do_lds);
+        char* qtb = reinterpret_cast<char*>(qt_lds);
+        char* dtb = reinterpret_cast<char*>(dot_lds);
 #pragma unroll
-      for (int cc = 0; cc < 2; ++cc) {
-        int col = c0 + cc * 8;
-        bf16x8_t qv = *reinterpret_cast<const bf16x8_t*>(qs + col);
-        bf16x8_t dv = *reinterpret_cast<const bf16x8_t*>(ds + col);
-        *reinterpret_cast<bf16x8_t*>(qb_ + rmswz(row, col * 2)) = qv;
-        *reinterpret_cast<bf16x8_t*>(db_ + rmswz(row, col * 2)) = dv;
+        for (int cc = 0; cc < 2; ++cc) {
+          int col = c0 + cc * 8;
+          bf16x8_t qv = *reinterpret_cast<const bf16x8_t*>(qs + col);
+          bf16x8_t dv = *reinterpret_cast<const bf16x8_t*>(ds + col);
+          *reinterpret_cast<bf16x8_t*>(qb_ + rmswz<DH>(row, col * 2)) = qv;
+          *reinterpret_cast<bf16x8_t*>(db_ + rmswz<DH>(row, col * 2)) = dv;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          *reinterpret_cast<short*>(qtb + trswz<QIT>(col + j, row * 2)) =
-              qv[j];
-          *reinterpret_cast<short*>(dtb + trswz<QIT>(col + j, row * 2)) =
-              dv[j];
+          for (int j = 0; j < 8; ++j) {
+            *reinterpret_cast<short*>(qtb + trswz<QIT>(col + j, row * 2)) =
+                qv[j];
+            *reinterpret_cast<short*>(dtb + trswz<QIT>(col + j, row * 2)) =
+                dv[j];
+          }
         }
       }
     }
@@ -177,13 +189,13 @@ __global__ void flash_bwd_dkv_kernel(
       f32x4_t acc2 = {0.f, 0.f, 0.f, 0.f};
       int qrow = l15 + 16 * ct;
 #pragma unroll
-      for (int ks_i = 0; ks_i < 4; ++ks_i) {
+      for (int ks_i = 0; ks_i < NKS; ++ks_i) {
         bf16x8_t bq = *reinterpret_cast<const bf16x8_t*>(
             reinterpret_cast<char*>(q_lds) +
-            rmswz(qrow, (l4 * 8 + 32 * ks_i) * 2));
+            rmswz<DH>(qrow, (l4 * 8 + 32 * ks_i) * 2));
         bf16x8_t bd = *reinterpret_cast<const bf16x8_t*>(
             reinterpret_cast<char*>(do_lds) +
-            rmswz(qrow, (l4 * 8 + 32 * ks_i) * 2));
+            rmswz<DH>(qrow, (l4 * 8 + 32 * ks_i) * 2));
         acc = MFMA_BF16_16x16x32(kfrag[ks_i], bq, acc, 0, 0, 0);
         acc2 = MFMA_BF16_16x16x32(vfrag[ks_i], bd, acc2, 0, 0, 0);
       }
@@ -204,7 +216,9 @@ __global__ void flash_bwd_dkv_kernel(
       for (int r = 0; r < 4; ++r) {
         int gkv = wave_kv + l4 * 4 + r;
         bool valid = gq < S && gkv < S && (!causal || gq >= gkv);
-        float p = valid ? __expf(st[ct][r] * scale - l) : 0.f;
+        float sval = st[ct][r] * scale;
+        if (mp && gkv < S) sval += mp[gkv];
+        float p = valid ? __expf(sval - l) : 0.f;
         float dsv = p * (dpt[ct][r] - dr) * scale;
         pw[(l4 * 4 + r) * (QIT + 8) + l15 + 16 * ct] = f2bf_(p);
         dw[(l4 * 4 + r) * (QIT + 8) + l15 + 16 * ct] = f2bf_(dsv);
@@ -219,7 +233,7 @@ __global__ void flash_bwd_dkv_kernel(
         dw + l15 * (QIT + 8) + l4 * 8);
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int dt = 0; dt < 8; ++dt) {
+    for (int dt = 0; dt < NDT; ++dt) {
       bf16x8_t bdo = *reinterpret_cast<const bf16x8_t*>(
           reinterpret_cast<char*>(dot_lds) +
           trswz<QIT>(l15 + 16 * dt, (l4 * 8) * 2));
@@ -239,10 +253,10 @@ __global__ void flash_bwd_dkv_kernel(
   for (int r = 0; r < 4; ++r) {
     int gkv = wave_kv + l4 * 4 + r;
     if (gkv >= S) continue;
-    short* dkr = dk + ((long long)b * S + gkv) * kvrs + hk * DHEAD;
-    short* dvr = dv + ((long long)b * S + gkv) * kvrs + hk * DHEAD;
+    short* dkr = dk + ((long long)b * S + gkv) * kvrs + hk * DH;
+    short* dvr = dv + ((long long)b * S + gkv) * kvrs + hk * DH;
 #pragma unroll
-    for (int dt = 0; dt < 8; ++dt) {
+    for (int dt = 0; dt < NDT; ++dt) {
       dkr[l15 + 16 * dt] = f2bf_(dk_acc[dt][r]);
       dvr[l15 + 16 * dt] = f2bf_(dv_acc[dt][r]);
     }
@@ -252,17 +266,21 @@ __global__ void flash_bwd_dkv_kernel(
 // ---------------------------------------------------------------------------
 // PASS 2: q-outer. Block: 4 waves x 16 q rows = 64; iterates kv tiles of 64.
 // ---------------------------------------------------------------------------
+template <int DH>
 __launch_bounds__(256, 2)
 __global__ void flash_bwd_dq_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ drow,
+    const float* __restrict__ kvmask,  // [B,S] or null
     short* __restrict__ dq,  // [B,S,Hq,D]
     int B, int S, int Hq, int Hk, float scale, int causal) {
   constexpr int KVT = 64;
-  __shared__ short k_lds[KVT * DHEAD];          // row-major swz
-  __shared__ short v_lds[KVT * DHEAD];          // row-major swz
-  __shared__ short kt_lds[DHEAD * (KVT + 8)];   // transposed
+  constexpr int NKS = DH / 32;
+  constexpr int NDT = DH / 16;
+  __shared__ short k_lds[KVT * DH];          // row-major swz
+  __shared__ short v_lds[KVT * DH];          // row-major swz
+  __shared__ short kt_lds[DH * (KVT + 8)];   // transposed
   __shared__ short dsw_lds[4][16 * (KVT + 8)];  // per-wave dS stage
 
   const int lane = threadIdx.x & 63;
@@ -278,24 +296,25 @@ __global__ void flash_bwd_dq_kernel(
   const int qbase = qtile * 64;
   const int wave_q = qbase + wave * 16;
 
-  const long long qrs = (long long)Hq * DHEAD;
-  const long long kvrs = (long long)Hk * DHEAD;
-  const short* qp = q + ((long long)b * S) * qrs + h * DHEAD;
-  const short* dop = dout + ((long long)b * S) * qrs + h * DHEAD;
-  const short* kp = k + ((long long)b * S) * kvrs + hk * DHEAD;
-  const short* vp = v + ((long long)b * S) * kvrs + hk * DHEAD;
+  const long long qrs = (long long)Hq * DH;
+  const long long kvrs = (long long)Hk * DH;
+  const short* qp = q + ((long long)b * S) * qrs + h * DH;
+  const short* dop = dout + ((long long)b * S) * qrs + h * DH;
+  const short* kp = k + ((long long)b * S) * kvrs + hk * DH;
+  const short* vp = v + ((long long)b * S) * kvrs + hk * DH;
   const float* lsep = lse + ((long long)b * Hq + h) * S;
   const float* drp = drow + ((long long)b * Hq + h) * S;
+  const float* mp = kvmask ? kvmask + (long long)b * S : nullptr;
 
   // Q and dO fragments in registers
-  bf16x8_t qfrag[4], dofrag[4];
+  bf16x8_t qfrag[NKS], dofrag[NKS];
   {
     int qrow = wave_q + l15;
     int srow = qrow < S ? qrow : S - 1;
     const short* qs = qp + (long long)srow * qrs;
     const short* ds = dop + (long long)srow * qrs;
 #pragma unroll
-    for (int ks_i = 0; ks_i < 4; ++ks_i) {
+    for (int ks_i = 0; ks_i < NKS; ++ks_i) {
       qfrag[ks_i] = *reinterpret_cast<const bf16x8_t*>(qs + ks_i * 32 + l4 * 8);
       dofrag[ks_i] =
           *reinterpret_cast<const bf16x8_t*>(ds + ks_i * 32 + l4 * 8);
@@ -309,9 +328,9 @@ __global__ void flash_bwd_dq_kernel(
     dr_r[r] = grow < S ? drp[grow] : 0.f;
   }
 
-  f32x4_t dq_acc[8];
+  f32x4_t dq_acc[NDT];
 #pragma unroll
-  for (int dt = 0; dt < 8; ++dt) dq_acc[dt] = f32x4_t{0.f, 0.f, 0.f, 0.f};
+  for (int dt = 0; dt < NDT; ++dt) dq_acc[dt] = f32x4_t{0.f, 0.f, 0.f, 0.f};
 
   const int kv_end = causal ? min(S, qbase + 64) : S;
   const int n_tiles = (kv_end + KVT - 1) / KVT;
@@ -319,26 +338,30 @@ __global__ void flash_bwd_dq_kernel(
     const int kvb = t * KVT;
     // ---- stage K/V row-major + Kt transposed ----------------------------
     {
-      int row = threadIdx.x >> 2;        // 0..63
-      int c0 = (threadIdx.x & 3) * 32;
-      int grow = kvb + row;
-      int srow = grow < S ? grow : S - 1;
-      const short* ks = kp + (long long)srow * kvrs;
-      const short* vs = vp + (long long)srow * kvrs;
-      char* kb_ = reinterpret_cast<char*>(k_lds);
-      char* vb_ = reinterpret_cast<char*>(v_lds);
-      char* ktb = reinterpret_cast<char*>(kt_lds);
+      // DH/32 threads per row, 32 cols each (4 x bf16x8)
+      constexpr int TPR = DH / 32;
+      int row = threadIdx.x / TPR;
+      int c0 = (threadIdx.x % TPR) * 32;
+      if (row < KVT) {
+        int grow = kvb + row;
+        int srow = grow < S ? grow : S - 1;
+        const short* ks = kp + (long long)srow * kvrs;
+        const short* vs = vp + (long long)srow * kvrs;
+        char* kb_ = reinterpret_cast<char*>(k_lds);
+        char* vb_ = reinterpret_cast<char*>(v_lds);
+        char* ktb = reinterpret_cast<char*>(kt_lds);
 #pragma unroll
-      for (int cc = 0; cc < 4; ++cc) {
-        int col = c0 + cc * 8;
-        bf16x8_t kv8 = *reinterpret_cast<const bf16x8_t*>(ks + col);
-        bf16x8_t vv8 = *reinterpret_cast<const bf16x8_t*>(vs + col);
-        *reinterpret_cast<bf16x8_t*>(kb_ + rmswz(row, col * 2)) = kv8;
-        *reinterpret_cast<bf16x8_t*>(vb_ + rmswz(row, col * 2)) = vv8;
+        for (int cc = 0; cc < 4; ++cc) {
+          int col = c0 + cc * 8;
+          bf16x8_t kv8 = *reinterpret_cast<const bf16x8_t*>(ks + col);
+          bf16x8_t vv8 = *reinterpret_cast<const bf16x8_t*>(vs + col);
+          *reinterpret_cast<bf16x8_t*>(kb_ + rmswz<DH>(row, col * 2)) = kv8;
+          *reinterpret_cast<bf16x8_t*>(vb_ + rmswz<DH>(row, col * 2)) = vv8;
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          *reinterpret_cast<short*>(ktb + trswz<KVT>(col + j, row * 2)) =
-              kv8[j];
+          for (int j = 0; j < 8; ++j)
+            *reinterpret_cast<short*>(ktb + trswz<KVT>(col + j, row * 2)) =
+                kv8[j];
+        }
       }
     }
     __syncthreads();
@@ -352,13 +375,13 @@ __global__ void flash_bwd_dq_kernel(
       f32x4_t a2 = {0.f, 0.f, 0.f, 0.f};
       int krow = l15 + 16 * ct;
 #pragma unroll
-      for (int ks_i = 0; ks_i < 4; ++ks_i) {
+      for (int ks_i = 0; ks_i < NKS; ++ks_i) {
         bf16x8_t bk = *reinterpret_cast<const bf16x8_t*>(
             reinterpret_cast<char*>(k_lds) +
-            rmswz(krow, (l4 * 8 + 32 * ks_i) * 2));
+            rmswz<DH>(krow, (l4 * 8 + 32 * ks_i) * 2));
         bf16x8_t bv = *reinterpret_cast<const bf16x8_t*>(
             reinterpret_cast<char*>(v_lds) +
-            rmswz(krow, (l4 * 8 + 32 * ks_i) * 2));
+            rmswz<DH>(krow, (l4 * 8 + 32 * ks_i) * 2));
         a1 = MFMA_BF16_16x16x32(qfrag[ks_i], bk, a1, 0, 0, 0);
         a2 = MFMA_BF16_16x16x32(dofrag[ks_i], bv, a2, 0, 0, 0);
       }
@@ -376,7 +399,9 @@ __global__ void flash_bwd_dq_kernel(
         int grow = wave_q + l4 * 4 + r;
         int gcol = kvb + l15 + 16 * ct;
         bool valid = grow < S && gcol < S && (!causal || gcol <= grow);
-        float p = valid ? __expf(s_acc[ct][r] * scale - lse_r[r]) : 0.f;
+        float sval = s_acc[ct][r] * scale;
+        if (mp && gcol < S) sval += mp[gcol];
+        float p = valid ? __expf(sval - lse_r[r]) : 0.f;
         float dsv = p * (dp_acc[ct][r] - dr_r[r]) * scale;
         dw[(l4 * 4 + r) * (KVT + 8) + l15 + 16 * ct] = f2bf_(dsv);
       }
@@ -390,7 +415,7 @@ __global__ void flash_bwd_dq_kernel(
       bf16x8_t a = *reinterpret_cast<const bf16x8_t*>(
           dw + l15 * (KVT + 8) + l4 * 8 + 32 * ks_i);
 #pragma unroll
-      for (int dt = 0; dt < 8; ++dt) {
+      for (int dt = 0; dt < NDT; ++dt) {
         bf16x8_t bkt = *reinterpret_cast<const bf16x8_t*>(
             reinterpret_cast<char*>(kt_lds) +
             trswz<KVT>(l15 + 16 * dt, (l4 * 8 + 32 * ks_i) * 2));
@@ -406,59 +431,83 @@ __global__ void flash_bwd_dq_kernel(
   for (int r = 0; r < 4; ++r) {
     int grow = wave_q + l4 * 4 + r;
     if (grow >= S) continue;
-    short* dqr = dq + ((long long)b * S + grow) * qrs + h * DHEAD;
+    short* dqr = dq + ((long long)b * S + grow) * qrs + h * DH;
 #pragma unroll
-    for (int dt = 0; dt < 8; ++dt)
+    for (int dt = 0; dt < NDT; ++dt)
       dqr[l15 + 16 * dt] = f2bf_(dq_acc[dt][r]);
   }
 }
 
 }  // namespace bwd
 
-std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
-                                       at::Tensor v, at::Tensor dout,
-                                       at::Tensor out, at::Tensor lse,
-                                       bool causal, double scale) {
-  TORCH_CHECK(q.scalar_type() == at::kBFloat16);
-  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous() &&
-              dout.is_contiguous() && out.is_contiguous());
-  int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
-  int Hk = k.size(2);
-  TORCH_CHECK(D == 128, "flash_attn_bwd: head_dim must be 128");
-  auto dq = at::empty_like(q);
-  auto dk = at::empty_like(k);
-  auto dv = at::empty_like(v);
-  auto drow = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+template <int DH>
+static void launch_bwd(at::Tensor& q, at::Tensor& k, at::Tensor& v,
+                       at::Tensor& dout, at::Tensor& out, at::Tensor& lse,
+                       at::Tensor& dq, at::Tensor& dk, at::Tensor& dv,
+                       const float* mp, int B, int S, int Hq, int Hk,
+                       float scale, int causal) {
   auto stream = c10::hip::getCurrentHIPStream();
+  auto drow = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
   long long rows = (long long)B * S * Hq;
   dim3 gridd((rows + 3) / 4);
-  hipLaunchKernelGGL(bwd::drow_kernel, gridd, dim3(256), 0, stream.stream(),
+  hipLaunchKernelGGL(bwd::drow_kernel<DH>, gridd, dim3(256), 0,
+                     stream.stream(),
                      reinterpret_cast<const short*>(dout.data_ptr()),
                      reinterpret_cast<const short*>(out.data_ptr()),
                      drow.data_ptr<float>(), S, Hq, rows);
   HIP_CHECK_KERNEL();
   dim3 grid1((S + 63) / 64, B * Hk);
-  hipLaunchKernelGGL(bwd::flash_bwd_dkv_kernel, grid1, dim3(256), 0,
+  hipLaunchKernelGGL(bwd::flash_bwd_dkv_kernel<DH>, grid1, dim3(256), 0,
                      stream.stream(),
                      reinterpret_cast<const short*>(q.data_ptr()),
                      reinterpret_cast<const short*>(k.data_ptr()),
                      reinterpret_cast<const short*>(v.data_ptr()),
                      reinterpret_cast<const short*>(dout.data_ptr()),
-                     lse.data_ptr<float>(), drow.data_ptr<float>(),
+                     lse.data_ptr<float>(), drow.data_ptr<float>(), mp,
                      reinterpret_cast<short*>(dk.data_ptr()),
                      reinterpret_cast<short*>(dv.data_ptr()), B, S, Hq,
-                     Hk, (float)scale, causal ? 1 : 0);
+                     Hk, scale, causal);
   HIP_CHECK_KERNEL();
   dim3 grid2((S + 63) / 64, B * Hq);
-  hipLaunchKernelGGL(bwd::flash_bwd_dq_kernel, grid2, dim3(256), 0,
+  hipLaunchKernelGGL(bwd::flash_bwd_dq_kernel<DH>, grid2, dim3(256), 0,
                      stream.stream(),
                      reinterpret_cast<const short*>(q.data_ptr()),
                      reinterpret_cast<const short*>(k.data_ptr()),
                      reinterpret_cast<const short*>(v.data_ptr()),
                      reinterpret_cast<const short*>(dout.data_ptr()),
-                     lse.data_ptr<float>(), drow.data_ptr<float>(),
+                     lse.data_ptr<float>(), drow.data_ptr<float>(), mp,
                      reinterpret_cast<short*>(dq.data_ptr()), B, S, Hq, Hk,
-                     (float)scale, causal ? 1 : 0);
+                     scale, causal);
   HIP_CHECK_KERNEL();
+}
+
+std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
+                                       at::Tensor v, at::Tensor dout,
+                                       at::Tensor out, at::Tensor lse,
+                                       bool causal, double scale,
+                                       c10::optional<at::Tensor> kvmask) {
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous() &&
+              dout.is_contiguous() && out.is_contiguous());
+  int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
+  int Hk = k.size(2);
+  TORCH_CHECK(D == 128 || D == 64,
+              "flash_attn_bwd: head_dim must be 64 or 128");
+  const float* mp = nullptr;
+  if (kvmask.has_value() && kvmask->defined()) {
+    TORCH_CHECK(kvmask->scalar_type() == at::kFloat &&
+                kvmask->is_contiguous() && kvmask->numel() == (long)B * S,
+                "kvmask must be contiguous float [B,S]");
+    mp = kvmask->data_ptr<float>();
+  }
+  auto dq = at::empty_like(q);
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  if (D == 128)
+    launch_bwd<128>(q, k, v, dout, out, lse, dq, dk, dv, mp, B, S, Hq, Hk,
+                    (float)scale, causal ? 1 : 0);
+  else
+    launch_bwd<64>(q, k, v, dout, out, lse, dq, dk, dv, mp, B, S, Hq, Hk,
+                   (float)scale, causal ? 1 : 0);
   return {dq, dk, dv};
 }

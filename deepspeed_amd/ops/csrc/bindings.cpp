@@ -13,11 +13,13 @@ void accum_bf16_to_f32(at::Tensor dst, at::Tensor src, double scale);
 at::Tensor l2norm_sq(std::vector<at::Tensor> tensors);
 std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
                                        at::Tensor v, bool causal,
-                                       double scale);
+                                       double scale,
+                                       c10::optional<at::Tensor> kvmask);
 std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
                                        at::Tensor v, at::Tensor dout,
                                        at::Tensor out, at::Tensor lse,
-                                       bool causal, double scale);
+                                       bool causal, double scale,
+                                       c10::optional<at::Tensor> kvmask);
 void multi_tensor_adagrad(std::vector<at::Tensor> params,
                           std::vector<at::Tensor> grads,
                           std::vector<at::Tensor> sq_accums, double lr,
@@ -112,8 +114,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("scales"), py::arg("qbits"), py::arg("group"),
         py::arg("shape"));
   m.def("flash_attn_fwd", &flash_attn_fwd, py::arg("q"), py::arg("k"),
-        py::arg("v"), py::arg("causal") = true, py::arg("scale") = 0.0);
-  m.def("flash_attn_bwd", &flash_attn_bwd);
+        py::arg("v"), py::arg("causal") = true, py::arg("scale") = 0.0,
+        py::arg("kvmask") = c10::nullopt);
+  m.def("flash_attn_bwd", &flash_attn_bwd, py::arg("q"), py::arg("k"),
+        py::arg("v"), py::arg("dout"), py::arg("out"), py::arg("lse"),
+        py::arg("causal") = true, py::arg("scale") = 0.0,
+        py::arg("kvmask") = c10::nullopt);
   m.def("multi_tensor_adagrad", &multi_tensor_adagrad, py::arg("params"),
         py::arg("grads"), py::arg("sq_accums"), py::arg("lr"),
         py::arg("eps") = 1e-8, py::arg("weight_decay") = 0.0,
