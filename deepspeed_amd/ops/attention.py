@@ -24,17 +24,21 @@ from .loader import get_ext
 
 class _FlashAttnFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, causal, scale):
+    def forward(ctx, q, k, v, causal, scale, kvmask=None):
         ext = get_ext(required=True)
-        out, lse = ext.flash_attn_fwd(q, k, v, causal, scale)
-        ctx.save_for_backward(q, k, v, out, lse)
+        out, lse = ext.flash_attn_fwd(q, k, v, causal, scale, kvmask)
+        ctx.save_for_backward(q, k, v, out, lse,
+                              kvmask if kvmask is not None
+                              else torch.empty(0))
         ctx.causal = causal
         ctx.scale = scale
         return out
 
     @staticmethod
     def backward(ctx, dout):
-        q, k, v, out, lse = ctx.saved_tensors
+        q, k, v, out, lse, kvmask = ctx.saved_tensors
+        if kvmask.numel() == 0:
+            kvmask = None
         causal, scale = ctx.causal, ctx.scale
         B, S, Hq, D = q.shape
         Hk = k.shape[2]
@@ -43,8 +47,8 @@ class _FlashAttnFn(torch.autograd.Function):
         if os.environ.get("DSAMD_FLASH_BWD", "1") == "1":
             ext = get_ext(required=True)
             dq, dk, dv = ext.flash_attn_bwd(
-                q, k, v, dout.contiguous(), out, lse, causal, scale)
-            return dq, dk, dv, None, None
+                q, k, v, dout.contiguous(), out, lse, causal, scale, kvmask)
+            return dq, dk, dv, None, None, None
 
         # head-major views [B,H,S,D]
         qh = q.permute(0, 2, 1, 3)
@@ -73,6 +77,8 @@ class _FlashAttnFn(torch.autograd.Function):
             vc = vh[:, :, c0:c1]
             # S_c = q @ k^T * scale  -> P via saved LSE
             s_c = torch.einsum("bhgsd,bhcd->bhgsc", qg, kc).float() * scale
+            if kvmask is not None:
+                s_c = s_c + kvmask[:, c0:c1].view(B, 1, 1, 1, c1 - c0)
             p = torch.exp(s_c - lse_g.unsqueeze(-1))
             if causal:
                 mask = rows.view(1, 1, 1, S, 1) >= (c0 + torch.arange(
@@ -95,7 +101,7 @@ class _FlashAttnFn(torch.autograd.Function):
         dk_out = dk.permute(0, 2, 1, 3).to(k.dtype)
         dv_out = dv.permute(0, 2, 1, 3).to(v.dtype)
         return dq_out.contiguous(), dk_out.contiguous(), \
-            dv_out.contiguous(), None, None
+            dv_out.contiguous(), None, None, None
 
 
 # Training default: the in-tree HIP kernels (fwd v5 beats SDPA; bwd beats
@@ -104,20 +110,48 @@ class _FlashAttnFn(torch.autograd.Function):
 _USE_HIP_FLASH = os.environ.get("DSAMD_FLASH", "1") == "1"
 
 
+def _as_kv_padding_mask(attn_mask, B, Sk):
+    """[B,S] float kv mask if attn_mask is a kv-column-only additive mask
+    ([B,S], [B,1,1,S], [1,1,1,S], [B,1,S]); else None."""
+    if attn_mask is None or attn_mask.dtype == torch.bool:
+        return None
+    m = attn_mask
+    if m.dim() == 4 and m.shape[1] == 1 and m.shape[2] == 1:
+        m = m[:, 0, 0]
+    elif m.dim() == 3 and m.shape[1] == 1:
+        m = m[:, 0]
+    elif m.dim() != 2:
+        return None
+    if m.shape[-1] != Sk:
+        return None
+    if m.shape[0] == 1 and B > 1:
+        m = m.expand(B, Sk)
+    if m.shape[0] != B:
+        return None
+    return m.contiguous().float()
+
+
 def flash_attention(q, k, v, causal=True, attn_mask=None):
     """q [B,S,Hq,D], k/v [B,S,Hkv,D] -> [B,S,Hq,D]. GQA-aware.
 
-    attn_mask: optional additive float mask broadcastable to
-    [B, Hq, Sq, Sk] (ragged-batch decode); routes through SDPA."""
+    attn_mask: optional additive float mask. KV-padding-shaped masks
+    ([B,1,1,Sk] etc., the BERT case) run in the HIP kernel; arbitrary
+    [B,Hq,Sq,Sk] masks route through SDPA."""
     scale = 1.0 / math.sqrt(q.shape[-1])
+    hip_ok = _USE_HIP_FLASH and q.is_cuda and q.dtype == torch.bfloat16 \
+        and q.shape[-1] in (64, 128) and q.shape[1] == k.shape[1]
     if attn_mask is not None:
+        kvm = _as_kv_padding_mask(attn_mask, q.shape[0], k.shape[1]) \
+            if hip_ok else None
+        if kvm is not None:
+            return _FlashAttnFn.apply(q.contiguous(), k.contiguous(),
+                                      v.contiguous(), causal, scale, kvm)
         out = F.scaled_dot_product_attention(
             q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2),
             attn_mask=attn_mask, is_causal=False,
             enable_gqa=(k.shape[2] != q.shape[2]))
         return out.transpose(1, 2)
-    if _USE_HIP_FLASH and q.is_cuda and q.dtype == torch.bfloat16 \
-            and q.shape[-1] == 128:
+    if hip_ok:
         return _FlashAttnFn.apply(q.contiguous(), k.contiguous(),
                                   v.contiguous(), causal, scale)
     # fallback (CPU tests / non-128 head dims): torch SDPA math

@@ -1,0 +1,108 @@
+"""fp8 (OCP e4m3/e5m2) Linear through hipBLASLt tensorwise _scaled_mm.
+
+Measured on MI355X (profiles/gemm_probe_fp8_vs_bf16.log): tensorwise fp8
+_scaled_mm runs at ~2x the bf16 GEMM rate on the Llama-8B bench shapes
+(2.4-3.3 PF vs 1.3-1.6 PF). Non-scaled fp8 MFMA is bf16-rate on CDNA4, so
+hipBLASLt's scaled path is the only fp8 win; torch's MX-block (e8m0) path
+measured SLOWER than bf16 — not used.
+
+Recipe (transformer-engine-style, dynamic scaling):
+  forward: x,w -> e4m3 with per-tensor amax scales; y = x8 @ w8^T (bf16 out)
+  backward: dy -> e5m2; dx = dy8 @ w8 ; dw = dy8^T @ x8 (bf16 out)
+Master weights stay bf16/fp32 in the optimizer — fp8 is a compute
+format only. Reference role: deepspeed/ops/fp_quantizer fp8 gemm
+(Triton there; hipBLASLt here), deepspeed/linear/quantization.py.
+"""
+import os
+
+import torch
+
+E4M3_MAX = 448.0
+E5M2_MAX = 57344.0
+
+_HAS_FP8 = hasattr(torch, "float8_e4m3fn")
+
+
+def _fp8_ok(x, w):
+    if not (_HAS_FP8 and x.is_cuda and x.dtype in (torch.bfloat16,
+                                                   torch.float16)):
+        return False
+    m = x.numel() // x.shape[-1]
+    k = x.shape[-1]
+    n = w.shape[0]
+    return m % 16 == 0 and k % 16 == 0 and n % 16 == 0
+
+
+def _quant(t, dtype, fmax):
+    """Per-tensor dynamic scale: t = t8 * scale."""
+    amax = t.abs().amax().float().clamp(min=1e-12)
+    scale = (amax / fmax)
+    t8 = (t.float() / scale).clamp(-fmax, fmax).to(dtype)
+    return t8, scale
+
+
+def _t8(t8):
+    """fp8 transpose via int8 view (fp8 copy kernels exist, but the int8
+    view is dependable across torch builds)."""
+    return t8.view(torch.int8).t().contiguous().view(t8.dtype)
+
+
+class _Fp8LinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, bias):
+        shp = x.shape
+        x2 = x.reshape(-1, shp[-1])
+        x8, sx = _quant(x2, torch.float8_e4m3fn, E4M3_MAX)
+        w8, sw = _quant(w, torch.float8_e4m3fn, E4M3_MAX)
+        y = torch._scaled_mm(x8, w8.t(), scale_a=sx, scale_b=sw,
+                             out_dtype=x.dtype)
+        if bias is not None:
+            y = y + bias
+        ctx.save_for_backward(x8, sx, w8, sw)
+        ctx.has_bias = bias is not None
+        ctx.in_dtype = x.dtype
+        return y.reshape(*shp[:-1], w.shape[0])
+
+    @staticmethod
+    def backward(ctx, dy):
+        x8, sx, w8, sw = ctx.saved_tensors
+        shp = dy.shape
+        dy2 = dy.reshape(-1, shp[-1])
+        dy8, sdy = _quant(dy2, torch.float8_e5m2, E5M2_MAX)
+        # dx[M,K] = dy[M,N] @ w[N,K]: b must be column-major => wT8.t()
+        wt8 = _t8(w8)  # [K,N] row-major
+        dx = torch._scaled_mm(dy8, wt8.t(), scale_a=sdy, scale_b=sw,
+                              out_dtype=ctx.in_dtype)
+        # dw[N,K] = dy^T[N,M] @ x[M,K]: a row-major dyT8; b col-major xT8.t()
+        dyt8 = _t8(dy8)  # [N,M]
+        xt8 = _t8(x8)    # [K,M]
+        dw = torch._scaled_mm(dyt8, xt8.t(), scale_a=sdy, scale_b=sx,
+                              out_dtype=ctx.in_dtype)
+        db = dy2.sum(0) if ctx.has_bias else None
+        return dx.reshape(shp[:-1] + (wt8.shape[0],)), dw, db
+
+
+class Fp8Linear(torch.nn.Linear):
+    """Drop-in nn.Linear computing fwd+bwd GEMMs in fp8 when eligible
+    (CUDA, 16-divisible M/N/K); falls back to the bf16 path otherwise.
+    `convert(module)` swaps every nn.Linear whose name matches `include`."""
+
+    def forward(self, x):
+        if _fp8_ok(x, self.weight) and not _disabled():
+            return _Fp8LinearFn.apply(x, self.weight, self.bias)
+        return super().forward(x)
+
+    @classmethod
+    def convert(cls, module, include=None):
+        """Re-class matching nn.Linear instances in place (no copies)."""
+        n_conv = 0
+        for name, child in module.named_modules():
+            if type(child) is torch.nn.Linear and \
+                    (include is None or any(tag in name for tag in include)):
+                child.__class__ = cls
+                n_conv += 1
+        return n_conv
+
+
+def _disabled():
+    return os.environ.get("DSAMD_FP8", "1") == "0"
