@@ -187,6 +187,8 @@ def load_checkpoint(engine, load_dir, tag=None, load_module_strict=True,
         engine.skipped_steps = state.get("skipped_steps", 0)
         engine.micro_steps = state.get("micro_steps", 0)
 
+    from ..ops.fp8_linear import bump_fp8_version
+    bump_fp8_version()  # loaded weights invalidate fp8 caches
     client_state = {k: v for k, v in state.items()
                     if k not in ("module", "optimizer", "lr_scheduler",
                                  "buffer_names", "ds_config", "ds_version")}

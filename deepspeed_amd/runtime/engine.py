@@ -498,6 +498,8 @@ class DeepSpeedEngine(torch.nn.Module):
 
     def _take_model_step(self, lr_kwargs=None):
         self.optimizer.step()
+        from ..ops.fp8_linear import bump_fp8_version
+        bump_fp8_version()  # invalidate cached fp8 weight copies
         overflow = getattr(self.optimizer, "overflow", False)
         if not isinstance(self.optimizer, torch.optim.Optimizer):
             pass
