@@ -106,8 +106,10 @@ class Fp8Linear(torch.nn.Linear):
     def forward(self, x):
         if _fp8_ok(x, self.weight) and not _disabled():
             if self._w8 is None or self._wv != _VERSION[0]:
-                w8, sw = _quant(self.weight, torch.float8_e4m3fn, E4M3_MAX)
-                self._w8, self._sw, self._wt8 = w8, sw, _t8(w8)
+                with torch.no_grad():  # cache is a constant; dw flows via
+                    w8, sw = _quant(self.weight.detach(),  # the Function
+                                    torch.float8_e4m3fn, E4M3_MAX)
+                    self._w8, self._sw, self._wt8 = w8, sw, _t8(w8)
                 self._wv = _VERSION[0]
             return _Fp8LinearFn.apply(x, self.weight, self.bias,
                                       self._w8, self._sw, self._wt8)
