@@ -11,6 +11,10 @@ void multi_tensor_adam(std::vector<at::Tensor> params,
                        double grad_scale);
 void accum_bf16_to_f32(at::Tensor dst, at::Tensor src, double scale);
 at::Tensor l2norm_sq(std::vector<at::Tensor> tensors);
+at::Tensor fp8_amax(at::Tensor x);
+at::Tensor fp8_cast(at::Tensor x, at::Tensor scale, bool e5m2);
+std::vector<at::Tensor> fp8_cast_transpose(at::Tensor x, at::Tensor scale,
+                                           bool e5m2);
 std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
                                        at::Tensor v, bool causal,
                                        double scale,
@@ -97,6 +101,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("accum_bf16_to_f32", &accum_bf16_to_f32, py::arg("dst"),
         py::arg("src"), py::arg("scale") = 1.0);
   m.def("l2norm_sq", &l2norm_sq);
+  m.def("fp8_amax", &fp8_amax);
+  m.def("fp8_cast", &fp8_cast, py::arg("x"), py::arg("scale"),
+        py::arg("e5m2") = false);
+  m.def("fp8_cast_transpose", &fp8_cast_transpose, py::arg("x"),
+        py::arg("scale"), py::arg("e5m2") = false);
   m.def("gemv_bf16", &gemv_bf16);
   m.def("spatial_bias_add", &spatial_bias_add, py::arg("a"), py::arg("bias"),
         py::arg("other") = c10::nullopt);

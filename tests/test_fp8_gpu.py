@@ -88,3 +88,33 @@ def test_fp8_linear_trains_tiny_llama_mlp():
     assert l_fp8[-1] < l_fp8[0] * 0.9, f"fp8 no progress: {l_fp8}"
     assert abs(l_fp8[-1] - l_bf16[-1]) < 0.5, \
         f"fp8 diverged from bf16: {l_fp8[-1]} vs {l_bf16[-1]}"
+
+
+def test_fp8_quant_kernels_match_torch():
+    """HIP amax/cast/cast_transpose vs the torch-op reference."""
+    from deepspeed_amd.ops.loader import get_ext
+    from deepspeed_amd.ops.fp8_linear import E4M3_MAX, E5M2_MAX, _quant
+    ext = get_ext()
+    torch.manual_seed(0)
+    for M, K in ((128, 256), (192, 4096), (64, 48)):
+        x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16) * 3
+        amax = ext.fp8_amax(x)
+        assert abs(amax.item() - x.abs().amax().item()) < 1e-3, \
+            (amax.item(), x.abs().amax().item())
+        for e5m2, fmax, dt in ((False, E4M3_MAX, torch.float8_e4m3fn),
+                               (True, E5M2_MAX, torch.float8_e5m2)):
+            scale = (amax[0] / fmax).clamp(min=1e-12)
+            ref8, _ = _quant(x, dt, fmax)
+            y = ext.fp8_cast(x, scale, e5m2).view(dt)
+            mismatch = (y.view(torch.int8) != ref8.view(torch.int8)) \
+                .float().mean().item()
+            # rounding mode at exact halfway points may differ; demand
+            # bitwise-near agreement and tiny value error
+            err = (y.float() - ref8.float()).abs().max().item() * scale.item()
+            assert err <= scale.item() * 2 ** 4, f"cast err {err}"
+            assert mismatch < 0.02, f"e5m2={e5m2} mismatch {mismatch}"
+            yt_pair = ext.fp8_cast_transpose(x, scale, e5m2)
+            y2, yt = yt_pair[0].view(dt), yt_pair[1].view(dt)
+            assert torch.equal(y2.view(torch.int8), y.view(torch.int8))
+            assert torch.equal(yt.view(torch.int8),
+                               y.view(torch.int8).t().contiguous())
