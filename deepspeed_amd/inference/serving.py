@@ -50,7 +50,16 @@ class RaggedKVCache:
         self.lens = torch.zeros(max_batch, dtype=torch.long, device=device)
         self.rows = None        # active slot ids for this forward
         self.attn_mask = None
+        self.ragged = None      # (kpool, vpool, rows, lens) for HIP decode
         self._prefill_slot = None
+        # ragged decode HIP kernel reads straight from the pool — no
+        # per-token gather copies (ref inference/v2 ragged_ops role)
+        from ..ops.loader import get_ext
+        self._use_ragged = (device is not None
+                            and torch.device(device).type == "cuda"
+                            and dtype == torch.bfloat16
+                            and d in (64, 128)
+                            and get_ext(required=False) is not None)
 
     # -- model-facing update ------------------------------------------------
     def update(self, k, v):
@@ -61,6 +70,7 @@ class RaggedKVCache:
             self.v[s, :S] = v[0]
             self.lens[s] = S
             self.attn_mask = None
+            self.ragged = None
             return k, v                      # causal prefill over itself
         rows = self.rows                     # decode: [n, 1, Hk, D]
         pos = self.lens[rows]
@@ -68,6 +78,11 @@ class RaggedKVCache:
         self.v[rows, pos] = v[:, 0]
         self.lens[rows] = pos + 1
         lens = self.lens[rows]
+        if self._use_ragged:
+            self.attn_mask = None
+            self.ragged = (self.k, self.v, rows.contiguous(),
+                           lens.contiguous())
+            return k, v  # attention reads from the pool via the kernel
         maxlen = int(lens.max())
         ar = torch.arange(maxlen, device=k.device)
         valid = ar.unsqueeze(0) < lens.unsqueeze(1)          # [n, maxlen]

@@ -85,10 +85,18 @@ class LlamaAttention(nn.Module):
         k = apply_rope(k, cos, sin)
         if kv_cache is not None:
             k, v = kv_cache.update(k, v)
-            mask = getattr(kv_cache, "attn_mask", None)
-            causal = S > 1  # prefill chunk is causal; decode sees all past
-            o = flash_attention(q, k, v, causal=causal and mask is None,
-                                attn_mask=mask)
+            rg = getattr(kv_cache, "ragged", None)
+            if rg is not None:
+                # ragged decode straight from the slot pool (no copies)
+                from ..ops.loader import get_ext
+                o = get_ext(required=True).ragged_decode(
+                    q.contiguous(), *rg)
+            else:
+                mask = getattr(kv_cache, "attn_mask", None)
+                causal = S > 1  # prefill is causal; decode sees all past
+                o = flash_attention(q, k, v,
+                                    causal=causal and mask is None,
+                                    attn_mask=mask)
         elif self._dist_attn is not None:
             o = self._dist_attn(q, k, v, causal=True)
         else:

@@ -12,6 +12,8 @@ void multi_tensor_adam(std::vector<at::Tensor> params,
 void accum_bf16_to_f32(at::Tensor dst, at::Tensor src, double scale);
 at::Tensor l2norm_sq(std::vector<at::Tensor> tensors);
 at::Tensor fp8_amax(at::Tensor x);
+at::Tensor ragged_decode(at::Tensor q, at::Tensor kpool, at::Tensor vpool,
+                         at::Tensor rows, at::Tensor lens, long chunk);
 at::Tensor fp8_cast(at::Tensor x, at::Tensor scale, bool e5m2);
 std::vector<at::Tensor> fp8_cast_transpose(at::Tensor x, at::Tensor scale,
                                            bool e5m2);
@@ -102,6 +104,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("src"), py::arg("scale") = 1.0);
   m.def("l2norm_sq", &l2norm_sq);
   m.def("fp8_amax", &fp8_amax);
+  m.def("ragged_decode", &ragged_decode, py::arg("q"), py::arg("kpool"),
+        py::arg("vpool"), py::arg("rows"), py::arg("lens"),
+        py::arg("chunk") = 512);
   m.def("fp8_cast", &fp8_cast, py::arg("x"), py::arg("scale"),
         py::arg("e5m2") = false);
   m.def("fp8_cast_transpose", &fp8_cast_transpose, py::arg("x"),
