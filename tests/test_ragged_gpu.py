@@ -69,7 +69,7 @@ def test_serving_engine_uses_ragged_kernel():
     from deepspeed_amd.inference.serving import ContinuousBatchingEngine
     from deepspeed_amd.models.llama import LLAMA_CONFIGS, LlamaForCausalLM
     torch.manual_seed(0)
-    cfg = LLAMA_CONFIGS["llama-tiny"]
+    cfg = LLAMA_CONFIGS["llama-small"]  # head_dim 64 (tiny is 32)
     with torch.device("cuda"):
         model = LlamaForCausalLM(cfg).eval()
 
