@@ -36,7 +36,12 @@ class InferenceConfig:
     def __init__(self, config=None, **kwargs):
         config = dict(config or {})
         config.update(kwargs)
-        self.dtype = config.get("dtype", torch.bfloat16)
+        dt = config.get("dtype", torch.bfloat16)
+        self.dtype = {"fp32": torch.float32, "float32": torch.float32,
+                      "fp16": torch.float16, "half": torch.float16,
+                      "float16": torch.float16, "bf16": torch.bfloat16,
+                      "bfloat16": torch.bfloat16}.get(dt, dt) \
+            if isinstance(dt, str) else dt
         self.max_out_tokens = config.get("max_out_tokens", 1024)
         self.tensor_parallel = config.get("tensor_parallel",
                                           {"tp_size": 1})
@@ -64,6 +69,15 @@ class InferenceEngine(torch.nn.Module):
                 "tensor_parallel.tp_size > 1 requires torch.distributed"
             group = tp.get("tp_group") if isinstance(tp, dict) else None
             apply_tensor_parallel(self.module, group)
+        # HF models (config, no native .cfg): swap attention/norms/MLP
+        # onto the HIP ops (ref replace_with_kernel_inject,
+        # module_inject/replace_module.py:189)
+        if self._config.replace_with_kernel_inject \
+                and not hasattr(model, "cfg") \
+                and hasattr(model, "config"):
+            from ..module_inject.replace_module import \
+                replace_transformer_layer
+            replace_transformer_layer(self.module)
         self.module.to(self._config.dtype).to(self.device)
         self.module.eval()
         self._caches = None

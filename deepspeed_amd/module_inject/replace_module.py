@@ -1,0 +1,200 @@
+"""HF kernel injection: swap HuggingFace model internals onto the
+in-tree MI355X HIP ops.
+
+Parity: reference `module_inject/replace_module.py:189`
+(replace_transformer_layer), `replace_policy.py`, `containers/` — there
+the reference rebuilds whole layers as fused DeepSpeedTransformerInference
+modules. MI355X-native redesign: HF models since v4.40 route attention
+through a registry (AttentionInterface) and keep norms/MLPs as small
+modules, so injection is surgical:
+
+  * attention  -> "dsamd_flash" interface entry (HIP flash kernel for
+    bf16 D in {64,128}, kv-padding masks run in-kernel; SDPA otherwise)
+  * RMSNorm    -> fused HIP rms_norm (any module exposing .weight and
+    variance_epsilon/eps, e.g. Llama/Mistral/Qwen RMSNorm)
+  * LayerNorm  -> fused HIP layer_norm (BERT/GPT-2/OPT classes)
+  * SwiGLU MLP -> fused swiglu for gate/up/down MLPs with silu
+
+Per-family policies mirror the reference's policy surface; AutoPolicy
+detects by module shape so unlisted families still inject.
+"""
+import torch
+
+from ..ops.attention import flash_attention
+from ..ops.functional import layer_norm, rms_norm, swiglu
+from ..utils.logging import log_dist
+
+_IMPL_NAME = "dsamd_attn"
+
+
+def _dsamd_hf_attention(module, query, key, value, attention_mask,
+                        scaling=None, dropout=0.0, **kwargs):
+    """transformers AttentionInterface entry: [B,H,S,D] in/out-of-registry
+    convention ([B,S,H,D] back to the caller)."""
+    q = query.transpose(1, 2)
+    k = key.transpose(1, 2)
+    v = value.transpose(1, 2)
+    is_causal = getattr(module, "is_causal", True)
+    if dropout and module.training:
+        # dropout stays on the SDPA path (HIP kernel is inference/train
+        # without attn-dropout; BERT-class finetune uses p=0.1)
+        import torch.nn.functional as F
+        out = F.scaled_dot_product_attention(
+            query, key, value, attn_mask=attention_mask,
+            dropout_p=dropout, scale=scaling,
+            is_causal=is_causal and attention_mask is None,
+            enable_gqa=(key.shape[1] != query.shape[1]))
+        return out.transpose(1, 2), None
+    out = flash_attention(q.contiguous(), k.contiguous(), v.contiguous(),
+                          causal=is_causal, attn_mask=attention_mask)
+    return out, None
+
+
+def _register_attention():
+    from transformers.modeling_utils import AttentionInterface
+    if _IMPL_NAME not in AttentionInterface._global_mapping:
+        AttentionInterface.register(_IMPL_NAME, _dsamd_hf_attention)
+
+
+def _is_rmsnorm(mod):
+    return type(mod).__name__.endswith("RMSNorm") and \
+        hasattr(mod, "weight") and \
+        (hasattr(mod, "variance_epsilon") or hasattr(mod, "eps"))
+
+
+def _is_swiglu_mlp(mod):
+    return all(hasattr(mod, a) for a in
+               ("gate_proj", "up_proj", "down_proj")) and \
+        "silu" in str(getattr(mod, "act_fn", "")).lower()
+
+
+class _FusedRMSNorm(torch.nn.Module):
+    def __init__(self, weight, eps):
+        super().__init__()
+        self.weight = weight
+        self.eps = eps
+
+    def forward(self, x):
+        return rms_norm(x, self.weight, self.eps)
+
+
+class _FusedLayerNorm(torch.nn.Module):
+    def __init__(self, src):
+        super().__init__()
+        self.weight = src.weight
+        self.bias = src.bias
+        self.eps = src.eps
+
+    def forward(self, x):
+        return layer_norm(x, self.weight, self.bias, self.eps)
+
+
+class _FusedSwiGLUMLP(torch.nn.Module):
+    def __init__(self, src):
+        super().__init__()
+        self.gate_proj = src.gate_proj
+        self.up_proj = src.up_proj
+        self.down_proj = src.down_proj
+
+    def forward(self, x):
+        return self.down_proj(swiglu(self.gate_proj(x), self.up_proj(x)))
+
+
+# ---------------------------------------------------------------- policies
+class InjectionPolicy:
+    """What to swap for a model family (ref replace_policy.py surface)."""
+    attention = True
+    rmsnorm = True
+    layernorm = True
+    swiglu_mlp = True
+
+
+class LlamaPolicy(InjectionPolicy):
+    layernorm = False
+
+
+class MistralPolicy(LlamaPolicy):
+    pass
+
+
+class QwenPolicy(LlamaPolicy):
+    pass
+
+
+class GPT2Policy(InjectionPolicy):
+    rmsnorm = False
+    swiglu_mlp = False
+
+
+class BertPolicy(GPT2Policy):
+    pass
+
+
+class OPTPolicy(GPT2Policy):
+    pass
+
+
+class AutoPolicy(InjectionPolicy):
+    """Shape-detected: injects whatever matches."""
+
+
+_POLICY_BY_ARCH = {
+    "llama": LlamaPolicy, "mistral": MistralPolicy, "mixtral": MistralPolicy,
+    "qwen": QwenPolicy, "gpt2": GPT2Policy, "bert": BertPolicy,
+    "opt": OPTPolicy,
+}
+
+
+def policy_for(model):
+    name = type(model).__name__.lower()
+    for k, p in _POLICY_BY_ARCH.items():
+        if k in name:
+            return p
+    mt = getattr(getattr(model, "config", None), "model_type", "") or ""
+    for k, p in _POLICY_BY_ARCH.items():
+        if k in mt:
+            return p
+    return AutoPolicy
+
+
+def replace_transformer_layer(model, policy=None, dtype=None):
+    """Inject HIP kernels into an HF model in place. Returns counts."""
+    if policy is None:
+        policy = policy_for(model)
+    counts = {"attention": 0, "rmsnorm": 0, "layernorm": 0, "mlp": 0}
+
+    if policy.attention and hasattr(model, "config"):
+        try:
+            _register_attention()
+            if hasattr(model, "set_attn_implementation"):
+                model.set_attn_implementation(_IMPL_NAME)
+            model.config._attn_implementation = _IMPL_NAME
+            counts["attention"] = 1
+        except Exception as e:  # non-HF or old transformers
+            log_dist(f"attention injection skipped: {e}", ranks=[0])
+
+    def swap(parent, name, new):
+        setattr(parent, name, new)
+
+    for parent in list(model.modules()):
+        for name, child in list(parent.named_children()):
+            if policy.rmsnorm and _is_rmsnorm(child) and \
+                    not isinstance(child, _FusedRMSNorm):
+                eps = getattr(child, "variance_epsilon",
+                              getattr(child, "eps", 1e-6))
+                swap(parent, name, _FusedRMSNorm(child.weight, eps))
+                counts["rmsnorm"] += 1
+            elif policy.layernorm and type(child) is torch.nn.LayerNorm \
+                    and child.elementwise_affine:
+                swap(parent, name, _FusedLayerNorm(child))
+                counts["layernorm"] += 1
+            elif policy.swiglu_mlp and _is_swiglu_mlp(child) and \
+                    not isinstance(child, _FusedSwiGLUMLP):
+                swap(parent, name, _FusedSwiGLUMLP(child))
+                counts["mlp"] += 1
+    log_dist(f"kernel injection ({policy.__name__}): {counts}", ranks=[0])
+    return counts
+
+
+# reference-compatible alias
+replace_module = replace_transformer_layer
