@@ -20,9 +20,103 @@ def _num_to_string(num, precision=2):
     return str(num)
 
 
-def _module_flops(module, inputs, output):
-    """fwd MACs*2 for common module types (0 for containers)."""
+# ---------------------------------------------------------------- functional
+# patching (ref profiler.py:893 _patch_functionals, :952
+# _patch_tensor_methods): counts FLOPs at the torch.nn.functional /
+# torch-op level so non-module math (SDPA, einsum in MoE dispatch, bare
+# matmuls) is attributed to the module executing it.
+_module_stack = []
+_orphan = [0]
+_patched = {}
+
+
+def _add_flops(f):
+    if _module_stack:
+        _module_stack[-1].__flops__ += f
+    else:
+        _orphan[0] += f
+
+
+def _wrap(mod_obj, name, flop_fn):
+    orig = getattr(mod_obj, name)
+    _patched[(mod_obj, name)] = orig
+
+    def wrapper(*args, **kwargs):
+        out = orig(*args, **kwargs)
+        try:
+            _add_flops(flop_fn(out, *args, **kwargs))
+        except Exception:
+            pass
+        return out
+
+    setattr(mod_obj, name, wrapper)
+
+
+def _linear_flops(out, x, w, bias=None, *a, **k):
+    n = x.numel() // x.shape[-1]
+    f = 2 * n * w.shape[0] * w.shape[1]
+    if bias is not None:
+        f += n * w.shape[0]
+    return f
+
+
+def _matmul_flops(out, a, b, *args, **k):
+    # out shape [..., m, n]; contraction dim = a.shape[-1]
+    return 2 * out.numel() * a.shape[-1]
+
+
+def _baddbmm_flops(out, inp, a, b, *args, **k):
+    return 2 * out.numel() * a.shape[-1] + out.numel()
+
+
+def _sdpa_flops(out, q, k, v, *a, **kw):
+    B_H = q.numel() // (q.shape[-1] * q.shape[-2])
+    sq, skv, d = q.shape[-2], k.shape[-2], q.shape[-1]
+    return B_H * (4 * sq * skv * d + 5 * sq * skv)
+
+
+def _einsum_flops(out, eq, *ops, **k):
+    if not isinstance(eq, str) or len(ops) != 2:
+        return 0
+    lhs, _, _ = eq.partition("->")
+    terms = lhs.split(",")
+    dims = {}
+    for t, op in zip(terms, ops):
+        for ch, n in zip(t.replace(" ", ""), op.shape):
+            dims[ch] = n
+    total = 1
+    for n in dims.values():
+        total *= n
+    return 2 * total
+
+
+def _patch_functionals():
+    import torch.nn.functional as F
+    _wrap(F, "linear", _linear_flops)
+    _wrap(F, "scaled_dot_product_attention", _sdpa_flops)
+    _wrap(torch, "matmul", _matmul_flops)
+    _wrap(torch, "bmm", _matmul_flops)
+    _wrap(torch, "baddbmm", _baddbmm_flops)
+    _wrap(torch, "einsum", _einsum_flops)
+    _wrap(torch.Tensor, "__matmul__", _matmul_flops)
+
+
+def _unpatch_functionals():
+    for (mod_obj, name), orig in list(_patched.items()):
+        setattr(mod_obj, name, orig)
+    _patched.clear()
+
+
+def _module_flops(module, inputs, output, functional_counting=False):
+    """fwd MACs*2 for common module types (0 for containers). With
+    functional counting active, GEMM-backed modules return 0 (their math
+    is counted at the functional level — no double count)."""
     x = inputs[0] if inputs else None
+    if functional_counting and isinstance(
+            module, (nn.Linear, nn.Conv1d, nn.Conv2d)):
+        return 0
+    if functional_counting and module.__class__.__name__ in             ("LlamaAttention",):
+        return 0
     if isinstance(module, nn.Linear):
         n = x.numel() // x.shape[-1]
         f = 2 * n * module.in_features * module.out_features
@@ -57,8 +151,13 @@ class FlopsProfiler:
         self.started = False
         self._hooks = []
 
-    def start_profile(self, ignore_list=None):
+    def start_profile(self, ignore_list=None, patch_functionals=True):
         self.reset()
+        self._functional = patch_functionals
+        if patch_functionals:
+            _orphan[0] = 0
+            _module_stack.clear()
+            _patch_functionals()
         for name, mod in self.model.named_modules():
             if ignore_list and type(mod) in ignore_list:
                 continue
@@ -73,6 +172,8 @@ class FlopsProfiler:
         self.started = True
 
     def _pre_hook(self, mod, inputs):
+        if getattr(self, "_functional", False):
+            _module_stack.append(mod)
         if torch.cuda.is_available():
             torch.cuda.synchronize()
         mod.__start__ = time.time()
@@ -82,12 +183,21 @@ class FlopsProfiler:
             torch.cuda.synchronize()
         mod.__latency__ += time.time() - getattr(mod, "__start__", time.time())
         mod.__calls__ += 1
-        mod.__flops__ += _module_flops(mod, inputs, output)
+        mod.__flops__ += _module_flops(
+            mod, inputs, output,
+            functional_counting=getattr(self, "_functional", False))
+        if getattr(self, "_functional", False) and _module_stack and \
+                _module_stack[-1] is mod:
+            _module_stack.pop()
 
     def stop_profile(self):
         for h in self._hooks:
             h.remove()
         self._hooks = []
+        if getattr(self, "_functional", False):
+            _unpatch_functionals()
+            _module_stack.clear()
+            self._functional = False
 
     def reset(self):
         self.stop_profile()

@@ -94,3 +94,48 @@ def test_numa_helpers():
     assert n >= 1
     cores = get_cores_for_node(0)
     assert len(cores) >= 1
+
+
+def test_flops_profiler_counts_functional_ops():
+    """SDPA / bare matmul / einsum flops are counted (ref profiler.py:893
+    patches functionals; module hooks alone undercount non-module math)."""
+    import torch
+    from deepspeed_amd.profiling.flops_profiler import FlopsProfiler
+
+    class M(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.lin = torch.nn.Linear(32, 64)
+
+        def forward(self, x):
+            h = self.lin(x)
+            a = torch.matmul(h, h.transpose(-1, -2))
+            e = torch.einsum("bij,bjk->bik", a, a)
+            s = torch.nn.functional.scaled_dot_product_attention(
+                h.view(4, 2, 8, 32), h.view(4, 2, 8, 32),
+                h.view(4, 2, 8, 32))
+            return e.sum() + s.sum()
+
+    m = M()
+    prof = FlopsProfiler(m)
+    prof.start_profile()
+    m(torch.randn(4, 8, 32))
+    total = prof.get_total_flops()
+    prof.stop_profile()
+    lin = 2 * 4 * 8 * 32 * 64 + 4 * 8 * 64
+    mm = 2 * (4 * 8 * 8) * 64
+    ein = 2 * 4 * 8 * 8 * 8
+    assert total > lin + mm + ein, (total, lin + mm + ein)
+    # module-hook-only counting would see just the Linear
+    assert total > 1.3 * lin
+
+    # patching fully reverted
+    import torch.nn.functional as F
+    ref = F.linear(torch.randn(2, 32), m.lin.weight, m.lin.bias)
+    assert ref.shape == (2, 64)
+    prof2 = FlopsProfiler(m)
+    prof2.start_profile(patch_functionals=False)
+    m(torch.randn(4, 8, 32))
+    t2 = prof2.get_total_flops()
+    prof2.stop_profile()
+    assert t2 >= lin  # legacy module-hook path still works
