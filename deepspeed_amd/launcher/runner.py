@@ -26,7 +26,9 @@ def parse_args(args=None):
     p.add_argument("--master_port", type=int, default=29500)
     p.add_argument("--master_addr", type=str, default="")
     p.add_argument("--launcher", type=str, default="pdsh",
-                   choices=["pdsh", "ssh", "local"])
+                   choices=["pdsh", "ssh", "local", "openmpi", "mpich",
+                            "impi", "slurm", "mvapich"])
+    p.add_argument("--launcher_args", type=str, default="")
     p.add_argument("user_script", type=str)
     p.add_argument("user_args", nargs=argparse.REMAINDER)
     return p.parse_args(args)
@@ -89,6 +91,21 @@ def main(args=None):
     if args.num_nodes > 0:
         world = dict(list(world.items())[:args.num_nodes])
     master_addr = args.master_addr or list(world.keys())[0]
+
+    if args.launcher in ("openmpi", "mpich", "impi", "slurm", "mvapich",
+                         "pdsh"):
+        from .multinode_runner import get_runner
+        runner = get_runner(args.launcher, args, world)
+        if not runner.backend_exists():
+            logger.error(f"{args.launcher} backend not found on PATH")
+            sys.exit(1)
+        runner.add_export("MASTER_ADDR", master_addr)
+        runner.add_export("MASTER_PORT", str(args.master_port))
+        cmd = runner.get_cmd()
+        logger.info(f"launching: {' '.join(map(str, cmd))}")
+        sys.exit(subprocess.call(cmd))
+
+    # ssh fallback: fan launch.py out per node ourselves
     world_info = json.dumps({h: g for h, g in world.items()})
     procs = []
     for node_rank, host in enumerate(world):
@@ -99,10 +116,7 @@ def main(args=None):
             f"--master_port={args.master_port} "
             f"--world_info={shlex.quote(world_info)} "
             f"{args.user_script} {' '.join(args.user_args)}")
-        if args.launcher == "pdsh":
-            cmd = ["pdsh", "-w", host, launch_cmd]
-        else:
-            cmd = ["ssh", host, launch_cmd]
+        cmd = ["ssh", host, launch_cmd]
         procs.append(subprocess.Popen(cmd))
     rc = 0
     for p in procs:

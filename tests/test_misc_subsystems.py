@@ -242,3 +242,46 @@ def test_op_builder_shim():
     assert hasattr(ext, "multi_tensor_adam")
     assert CPUAdamBuilder().is_compatible()
     assert CPUAdamBuilder().jit_load() is ext
+
+
+def test_multinode_runner_commands():
+    """Runner classes build correct cross-node commands (ref
+    multinode_runner.py:126-393) — no cluster needed to verify."""
+    import sys
+    from types import SimpleNamespace
+    from deepspeed_amd.launcher.multinode_runner import (RUNNERS, get_runner)
+    args = SimpleNamespace(hostfile="/job/hostfile", master_port=29500,
+                           master_addr="node0", user_script="train.py",
+                           user_args=["--foo", "1"])
+    world = {"node0": [0, 1, 2, 3], "node1": [0, 1, 2, 3]}
+    assert set(RUNNERS) == {"pdsh", "openmpi", "mpich", "impi", "slurm",
+                            "mvapich"}
+
+    r = get_runner("openmpi", args, world)
+    r.add_export("MASTER_ADDR", "node0")
+    cmd = r.get_cmd()
+    assert cmd[:3] == ["mpirun", "-n", "8"]
+    assert "-x" in cmd and "MASTER_ADDR=node0" in cmd
+    assert cmd[-2:] == ["--foo", "1"] and "train.py" in cmd
+
+    r = get_runner("slurm", args, world)
+    cmd = r.get_cmd()
+    assert cmd[0] == "srun" and "--ntasks" in cmd
+    assert cmd[cmd.index("--ntasks") + 1] == "8"
+    assert cmd[cmd.index("--ntasks-per-node") + 1] == "4"
+
+    r = get_runner("mpich", args, world)
+    cmd = r.get_cmd()
+    assert cmd[cmd.index("-ppn") + 1] == "4"
+
+    r = get_runner("mvapich", args, world)
+    assert r.get_cmd()[:3] == ["mpirun_rsh", "-np", "8"]
+
+    r = get_runner("pdsh", args, world)
+    cmd = r.get_cmd()
+    assert cmd[0] == "pdsh" and "node0,node1" in cmd
+    assert "deepspeed_amd.launcher.launch" in cmd[-1]
+
+    import pytest as _pytest
+    with _pytest.raises(ValueError):
+        get_runner("nope", args, world)
