@@ -41,16 +41,31 @@ class _PositionalLM(torch.nn.Module):
 
 class _GraphedLM(torch.nn.Module):
     """Training calls replay the captured graphs; label-free calls
-    (generation) fall back to the eager module."""
+    (generation), extra kwargs, or inputs whose SHAPE differs from the
+    captured static shape fall back to the eager module (replaying a
+    hipGraph on mismatched shapes reads stale capture buffers — silent
+    corruption, not an error)."""
 
-    def __init__(self, graphed, eager):
+    def __init__(self, graphed, eager, static_shape, static_label_shape):
         super().__init__()
         self.graphed = graphed
         self.eager = eager
+        self._shape = tuple(static_shape)
+        self._lshape = tuple(static_label_shape) \
+            if static_label_shape is not None else None
+        self._warned = False
 
     def forward(self, input_ids, labels=None, **kw):
-        if labels is not None and not kw:
+        if labels is not None and not kw \
+                and tuple(input_ids.shape) == self._shape \
+                and (self._lshape is None
+                     or tuple(labels.shape) == self._lshape):
             return self.graphed(input_ids, labels)
+        if labels is not None and not self._warned:
+            self._warned = True
+            log_dist(
+                f"hipGraph fallback to eager: input {tuple(input_ids.shape)}"
+                f" != captured {self._shape}", ranks=[0])
         return self.eager(input_ids, labels=labels, **kw)
 
 
@@ -67,10 +82,11 @@ def graph_compile(module, sample_input, sample_labels=None,
     dev = next(module.parameters()).device
     sample = sample_input.to(dev)
     if sample_labels is not None:
+        lbl = sample_labels.to(dev)
         graphed = torch.cuda.make_graphed_callables(
-            _PositionalLM(module), (sample, sample_labels.to(dev)),
+            _PositionalLM(module), (sample, lbl),
             num_warmup_iters=num_warmup_iters)
-        out = _GraphedLM(graphed, module)
+        out = _GraphedLM(graphed, module, sample.shape, lbl.shape)
     else:
         out = torch.cuda.make_graphed_callables(
             module, (sample,), num_warmup_iters=num_warmup_iters)

@@ -202,3 +202,35 @@ def test_zero3_param_offload_nvme_gpu():
     assert losses[-1] < losses[0], losses
     assert len(sw._on_disk) > 0, "no slab was ever written to NVMe"
     engine.destroy()
+
+
+def test_engine_compile_shape_fallback():
+    """A batch whose shape differs from the captured static shape must
+    run EAGER (graph replay on wrong shapes is silent corruption)."""
+    _init_env()
+    import deepspeed_amd as ds
+    from deepspeed_amd.models.llama import LLAMA_CONFIGS, LlamaForCausalLM
+    cfg = LLAMA_CONFIGS["llama-tiny"]
+    torch.manual_seed(0)
+    with torch.device("cuda:0"):
+        model = LlamaForCausalLM(cfg)
+    config = {
+        "train_micro_batch_size_per_gpu": 2,
+        "optimizer": {"type": "AdamW", "params": {"lr": 3e-4}},
+        "zero_optimization": {"stage": 1},
+        "bf16": {"enabled": True},
+    }
+    engine, _, _, _ = ds.initialize(model=model, config=config)
+    sample = torch.randint(0, cfg.vocab_size, (2, 64), device="cuda:0")
+    engine.compile(sample_input=sample, sample_labels=sample)
+    # captured shape trains
+    loss = engine(sample, labels=sample)
+    engine.backward(loss)
+    engine.step()
+    # shorter batch falls back to eager and still trains correctly
+    short = torch.randint(0, cfg.vocab_size, (2, 32), device="cuda:0")
+    loss2 = engine(short, labels=short)
+    engine.backward(loss2)
+    engine.step()
+    assert torch.isfinite(loss2).item()
+    engine.destroy()
