@@ -204,3 +204,67 @@ def load_metric_index(output_path, metric_name):
                            f"{metric_name}_metric_values.txt")) as f:
         values = [int(x) for x in f.read().split()]
     return s2m, values, m2s
+
+
+class CurriculumMetricSampler(torch.utils.data.Sampler):
+    """Difficulty-gated index sampler driven by a DataAnalyzer metric
+    index (ref data_sampler.py:349 DeepSpeedDataSampler's metric path).
+
+    Each epoch step the eligible pool = all samples whose metric value is
+    <= the curriculum scheduler's current difficulty; indices are drawn
+    (shuffled, DP-sharded) from that pool, so training sees easy samples
+    first and the pool grows with the schedule.
+    """
+
+    def __init__(self, metric_index_path, metric_name, scheduler,
+                 total_samples=None, shuffle=True, seed=0,
+                 dp_rank=0, dp_world=1):
+        s2m, values, m2s = load_metric_index(metric_index_path, metric_name)
+        self.values = values                      # sorted distinct metrics
+        self.buckets = [m2s[i].tolist() for i in range(len(values))]
+        self.scheduler = scheduler
+        self.total = total_samples or sum(len(b) for b in self.buckets)
+        self.shuffle = shuffle
+        self.seed = seed
+        self.dp_rank = dp_rank
+        self.dp_world = dp_world
+        self.epoch = 0
+        self.global_step = 0
+
+    def set_epoch(self, epoch):
+        self.epoch = epoch
+
+    def _eligible(self):
+        diff = self.scheduler.update_difficulty(self.global_step)
+        pool = []
+        for v, b in zip(self.values, self.buckets):
+            if v <= diff:
+                pool.extend(b)
+        return pool or list(self.buckets[0])
+
+    def __iter__(self):
+        g = torch.Generator().manual_seed(self.seed + self.epoch)
+        produced = 0
+        while produced < self.total:
+            pool = self._eligible()
+            order = torch.randperm(len(pool), generator=g).tolist() \
+                if self.shuffle else range(len(pool))
+            for j in order:
+                if produced >= self.total:
+                    return
+                if produced % self.dp_world == self.dp_rank:
+                    yield pool[j]
+                produced += 1
+                self.global_step += 1
+
+    def __len__(self):
+        return (self.total + self.dp_world - 1) // self.dp_world
+
+    def state_dict(self):
+        return {"epoch": self.epoch, "global_step": self.global_step,
+                "scheduler": self.scheduler.state_dict()}
+
+    def load_state_dict(self, sd):
+        self.epoch = sd["epoch"]
+        self.global_step = sd["global_step"]
+        self.scheduler.load_state_dict(sd["scheduler"])

@@ -95,3 +95,35 @@ def _curriculum_engine():
     d1 = engine.curriculum_scheduler.get_current_difficulty()
     assert d1 > d0, (d0, d1)
     assert d1 == 32
+
+
+def test_curriculum_metric_sampler_gates_by_difficulty(tmp_path):
+    """Samples with metric above the current difficulty are excluded; the
+    pool grows as the schedule advances."""
+    import torch
+    from deepspeed_amd.runtime.data_pipeline import CurriculumScheduler
+    from deepspeed_amd.runtime.data_sampling import (CurriculumMetricSampler,
+                                                     DataAnalyzer)
+    data = [torch.arange(n) for n in
+            (3, 3, 5, 5, 8, 8, 12, 12, 20, 20)]
+    an = DataAnalyzer(data, ["seqlen"],
+                      [lambda batch: [len(x) for x in batch]],
+                      str(tmp_path))
+    an.run_map_reduce()
+    sched = CurriculumScheduler({
+        "curriculum_type": "seqlen", "min_difficulty": 5,
+        "max_difficulty": 20,
+        "schedule_type": "fixed_linear",
+        "schedule_config": {"total_curriculum_step": 20,
+                            "difficulty_step": 1}})
+    samp = CurriculumMetricSampler(str(tmp_path), "seqlen", sched,
+                                   total_samples=40)
+    idxs = list(samp)
+    assert len(idxs) == 40
+    # draws before the schedule reaches difficulty 8 (step 4 on this
+    # fixed_linear curve) must come from the easy pool
+    early = idxs[:4]
+    assert all(len(data[i]) <= 5 for i in early), \
+        [len(data[i]) for i in early]
+    # late draws include hard samples once difficulty reaches 20
+    assert any(len(data[i]) >= 12 for i in idxs[20:])
