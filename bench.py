@@ -29,6 +29,12 @@ def parse_args():
                    "--sp 8 --seq-len 32768)")
     p.add_argument("--ep", type=int, default=1,
                    help="expert-parallel degree for MoE models")
+    p.add_argument("--offload-param", type=str, default="none",
+                   choices=["none", "cpu", "nvme"],
+                   help="ZeRO-Infinity parameter tier")
+    p.add_argument("--no-pin", action="store_true",
+                   help="pageable host offload buffers (skip pinning "
+                        "cost at 70B scale)")
     p.add_argument("--offload", type=str, default="none",
                    choices=["none", "cpu", "nvme"])
     p.add_argument("--micro-batch", type=int, default=4)
@@ -109,7 +115,11 @@ def main():
     }
     if args.offload != "none":
         ds_config["zero_optimization"]["offload_optimizer"] = {
-            "device": args.offload}
+            "device": args.offload, "pin_memory": not args.no_pin}
+    if args.offload_param != "none":
+        ds_config["zero_optimization"]["offload_param"] = {
+            "device": args.offload_param, "pin_memory": not args.no_pin}
+        ds_config["zero_optimization"]["sub_group_size"] = int(5e8)
     if args.sp > 1:
         ds_config["sequence_parallel"] = {
             "sequence_parallel_size": args.sp}
@@ -203,6 +213,11 @@ def main():
                        "seq_len": S,
                        "parallelism": (f"zero{args.zero_stage}_"
                                        f"dp{world // sp_world}" +
+                                       (f"_offload-{args.offload}"
+                                        if args.offload != "none" else "") +
+                                       (f"_param-{args.offload_param}"
+                                        if args.offload_param != "none"
+                                        else "") +
                                        (f"_sp{sp_world}" if sp_world > 1
                                         else "") +
                                        (f"_ep{args.ep}" if args.ep > 1

@@ -44,7 +44,8 @@ class SubGroup:
                  "param_offload", "dtype16")
 
     def __init__(self, params, offsets, numel, group_idx, device,
-                 offload=False, param_offload=False):
+                 offload=False, param_offload=False, pin_opt=True,
+                 pin_param=True):
         self.params = params
         self.offsets = offsets
         self.numel = numel
@@ -53,18 +54,21 @@ class SubGroup:
         self.param_offload = param_offload
         dtype16 = params[0].ds_tensor.dtype if params else torch.bfloat16
         self.dtype16 = dtype16
+        # pinning giant slabs costs ~1 GB/s at registration: configs
+        # (offload_*.pin_memory) choose; 70B-scale host tiers run pageable
         pin = torch.cuda.is_available()
         host = torch.device("cpu")
         state_dev = host if offload else device
         # param offload (ZeRO-Infinity tier): the 16-bit shard slab itself
-        # lives in pinned host memory; fetches stage it H2D on the gather
-        # stream (stage3_params._shard_on_device)
+        # lives in (optionally pinned) host memory; fetches stage it H2D
+        # on the gather stream (stage3_params._shard_on_device)
         slab_dev = host if param_offload else device
         self.flat16 = torch.empty(numel, dtype=dtype16, device=slab_dev,
-                                  pin_memory=pin and param_offload)
+                                  pin_memory=pin and param_offload
+                                  and pin_param)
         self.master32 = torch.empty(numel, dtype=torch.float32,
                                     device=state_dev,
-                                    pin_memory=pin and offload)
+                                    pin_memory=pin and offload and pin_opt)
         for p in params:
             off = offsets[p]
             n = p.ds_shard_numel
@@ -74,7 +78,7 @@ class SubGroup:
         self.master32 = self.master32.detach().requires_grad_(True)
         self.grad32 = torch.zeros(numel, dtype=torch.float32,
                                   device=state_dev,
-                                  pin_memory=pin and offload)
+                                  pin_memory=pin and offload and pin_opt)
         if offload:
             if param_offload:
                 # the host slab IS the 16-bit output buffer: the fused CPU
@@ -83,9 +87,10 @@ class SubGroup:
             else:
                 # pinned staging: bf16 shard out (H2D)
                 self.flat16_cpu = torch.empty(numel, dtype=dtype16,
-                                              device=host, pin_memory=pin)
+                                              device=host,
+                                              pin_memory=pin and pin_opt)
             self.grad_stage = torch.empty(numel, dtype=dtype16, device=host,
-                                          pin_memory=pin)
+                                          pin_memory=pin and pin_opt)
         else:
             self.flat16_cpu = None
             self.grad_stage = None
@@ -180,6 +185,10 @@ class ZeroStage3Optimizer:
         # additionally spill to O_DIRECT files with an LRU host budget
         self.param_offload_device = getattr(offload_param, "device", "none") \
             if offload_param is not None else "none"
+        self._pin_opt = getattr(offload_optimizer, "pin_memory", True) \
+            if offload_optimizer is not None else True
+        self._pin_param = getattr(offload_param, "pin_memory", True) \
+            if offload_param is not None else True
         self.param_offload = self.param_offload_device in ("cpu", "nvme")
         self.param_swapper = None
         if self.param_offload_device == "nvme":
@@ -328,7 +337,9 @@ class ZeroStage3Optimizer:
                     self.sub_groups.append(
                         SubGroup(cur, offsets, numel, gi, self.device,
                                  offload=self.offload_optimizer,
-                                 param_offload=self.param_offload))
+                                 param_offload=self.param_offload,
+                                 pin_opt=self._pin_opt,
+                                 pin_param=self._pin_param))
                     cur, offsets, numel = [], OrderedDict(), 0
                 offsets[p] = numel
                 cur.append(p)
@@ -337,7 +348,9 @@ class ZeroStage3Optimizer:
                 self.sub_groups.append(
                     SubGroup(cur, offsets, numel, gi, self.device,
                              offload=self.offload_optimizer,
-                             param_offload=self.param_offload))
+                             param_offload=self.param_offload,
+                             pin_opt=self._pin_opt,
+                             pin_param=self._pin_param))
         self.param_to_subgroup = {}
         for sg in self.sub_groups:
             for p in sg.params:
