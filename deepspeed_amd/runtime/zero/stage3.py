@@ -444,6 +444,76 @@ class ZeroStage3Optimizer:
             self.optimizer.set_grad_scale(1.0)
             self.optimizer.set_fused_out16({})
 
+    @torch.no_grad()
+    def _muon_step(self, combined):
+        """Distributed Muon inside ZeRO-3 (ref stage3.py:1619
+        _apply_distributed_muon_update). Momentum stays SHARDED (the
+        momentum update is elementwise, so it commutes with sharding);
+        only Newton-Schulz needs the full 2-D matrix: one all-gather per
+        2-D param rebuilds the nesterov-effective update, every rank runs
+        NS on it (replicated, deterministic), and applies its own shard
+        of U to its master slice. Non-2D params take the sharded AdamW
+        path."""
+        from ...ops.muon import zeropower_via_newtonschulz5
+        inv = 1.0 / combined
+        if not hasattr(self, "_muon_state"):
+            self._muon_state = {}
+        for sg in self.sub_groups:
+            group = self.optimizer.param_groups[sg.group_idx]
+            group["step"] = group.get("step", 0) + 1
+            mom = group["momentum"]
+            for p in sg.params:
+                sn = p.ds_shard_numel
+                off = sg.offsets[p]
+                gs = sg.grad32[off:off + sn]
+                master = sg.master32[off:off + sn]
+                st = self._muon_state.setdefault(p.ds_id, {})
+                use_muon = len(p.ds_shape) == 2 and min(p.ds_shape) > 1
+                if use_muon:
+                    g = gs * inv
+                    buf = st.setdefault("momentum_buffer",
+                                        torch.zeros_like(gs))
+                    buf.mul_(mom).add_(g)
+                    eff = g.add(buf, alpha=mom) if group["nesterov"]                         else buf.clone()
+                    pg = self._param_pg(p)
+                    world = dist.get_world_size(pg)
+                    eff_dev = eff.to(self.device)
+                    if world > 1:
+                        full = torch.empty(world * sn, dtype=eff_dev.dtype,
+                                           device=eff_dev.device)
+                        dist.all_gather_into_tensor(full, eff_dev, group=pg)
+                    else:
+                        full = eff_dev
+                    mat = full[:p.ds_numel].view(p.ds_shape)
+                    u = zeropower_via_newtonschulz5(
+                        mat, group["ns_steps"]).reshape(-1)
+                    rank = dist.get_rank(pg)
+                    start = rank * sn
+                    end = min(start + sn, p.ds_numel)
+                    scale = max(1.0, p.ds_shape[0] / p.ds_shape[1]) ** 0.5
+                    if group["weight_decay"]:
+                        master.mul_(1 - group["lr"] *
+                                    group["weight_decay"])
+                    if end > start:
+                        master[:end - start].add_(
+                            u[start:end].to(master.device),
+                            alpha=-group["lr"] * scale)
+                else:
+                    g = gs * inv
+                    if "exp_avg" not in st:
+                        st["exp_avg"] = torch.zeros_like(g)
+                        st["exp_avg_sq"] = torch.zeros_like(g)
+                    b1, b2 = group["adamw_betas"]
+                    t = group["step"]
+                    m, v = st["exp_avg"], st["exp_avg_sq"]
+                    m.mul_(b1).add_(g, alpha=1 - b1)
+                    v.mul_(b2).addcmul_(g, g, value=1 - b2)
+                    denom = (v / (1 - b2 ** t)).sqrt_()                         .add_(group["adamw_eps"])
+                    master.addcdiv_(m, denom,
+                                    value=-group["adamw_lr"] /
+                                    (1 - b1 ** t))
+            sg.copy_master_to_shards()
+
     def _param_pg(self, p):
         return getattr(p, "ds_group", None) or self.dp_group
 
@@ -815,6 +885,13 @@ class ZeroStage3Optimizer:
 
         if self.nvme_swapper is not None:
             self._nvme_step(combined)
+            self._clear_grads()
+            self._refresh_persistent_params()
+            return
+
+        from ...ops.muon import Muon as _Muon
+        if isinstance(self.optimizer, _Muon):
+            self._muon_step(combined)
             self._clear_grads()
             self._refresh_persistent_params()
             return
