@@ -71,7 +71,7 @@ def test_serving_engine_uses_ragged_kernel():
     torch.manual_seed(0)
     cfg = LLAMA_CONFIGS["llama-small"]  # head_dim 64 (tiny is 32)
     with torch.device("cuda"):
-        model = LlamaForCausalLM(cfg).eval()
+        model = LlamaForCausalLM(cfg).bfloat16().eval()
 
     def run(force_fallback):
         eng = ContinuousBatchingEngine(model, max_batch=4)
