@@ -70,6 +70,8 @@ class ZeroConfig(DSConfigModel):
     offload_optimizer: Optional[OffloadOptimizerConfig] = None
     zenflow: Optional[dict] = None  # ZenFlow selective-offload (stage 1/2)
     zero_hpz_partition_size: int = 1
+    # ZeRO++ qwZ: int8 blockwise weight gathers (half the AG bytes)
+    zero_quantized_weights: bool = False
     mics_shard_size: int = -1
     round_robin_gradients: bool = False
     ignore_unused_parameters: bool = True

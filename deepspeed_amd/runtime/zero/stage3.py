@@ -148,6 +148,7 @@ class ZeroStage3Optimizer:
                  overlap_comm=True,
                  offload_optimizer=None,
                  offload_param=None,
+                 zero_quantized_weights=False,
                  clip_grad=0.0,
                  static_loss_scale=1.0,
                  dynamic_loss_scale=False,
@@ -174,6 +175,9 @@ class ZeroStage3Optimizer:
         self.gradient_accumulation_steps = gradient_accumulation_steps
         self.micro_step = 0
         self.overlap_comm = overlap_comm
+        # ZeRO++ qwZ (ref coalesced qwZ path): int8 blockwise shard
+        # gathers + fp16 scales — halves all-gather bytes over xGMI
+        self.quantized_weights = bool(zero_quantized_weights)
 
         self.device = (torch.device("cuda", torch.cuda.current_device())
                        if torch.cuda.is_available() else torch.device("cpu"))
@@ -536,7 +540,8 @@ class ZeroStage3Optimizer:
         for p in params:
             by_pg.setdefault(id(self._param_pg(p)), []).append(p)
         handles = [all_gather_params(ps, self._param_pg(ps[0]),
-                                     async_op=async_op, stream=stream)
+                                     async_op=async_op, stream=stream,
+                                     quantized=self.quantized_weights)
                    for ps in by_pg.values()]
         if len(handles) == 1:
             return handles[0]

@@ -45,6 +45,29 @@ def _pad_to(numel, multiple):
 
 _COALESCE_OK = True
 
+# ZeRO++ qwZ (ref partition_parameters.py:846 CUDAQuantizer,
+# all_gather_coalesced quantized path): gather int8 blockwise-quantized
+# shards + fp16 scales instead of bf16 — halves gather bytes on the wire.
+QUANT_BLOCK = 64  # shard_numel is always a multiple of ALIGN=64
+
+
+def quantize_shard(sh):
+    """[n] bf16/fp32 -> (int8 [n], fp16 scales [n/64]) blockwise absmax."""
+    x = sh.float().view(-1, QUANT_BLOCK)
+    scale = x.abs().amax(1, keepdim=True).clamp(min=1e-8) / 127.0
+    q = torch.clamp(torch.round(x / scale), -127, 127).to(torch.int8)
+    return q.view(-1), scale.to(torch.float16).view(-1)
+
+
+def dequantize_gathered(q, scales, world, dtype, out=None):
+    """int8 [world*n] + scales [world*n/64] -> dtype [world*n]."""
+    x = q.view(world, -1, QUANT_BLOCK).float() *         scales.float().view(world, -1, 1)
+    x = x.reshape(-1).to(dtype)
+    if out is not None:
+        out.copy_(x)
+        return out
+    return x.contiguous()
+
 
 def _supports_coalescing(group):
     if not _COALESCE_OK or not torch.cuda.is_available():
@@ -101,12 +124,15 @@ def free_param(p):
 class AllGatherHandle:
     """Waits on an in-flight coalesced gather and publishes p.data."""
 
-    def __init__(self, params, works, buffers, group, stream=None):
+    def __init__(self, params, works, buffers, group, stream=None,
+                 quant=None):
         self.params = params
         self.works = works  # list of work objs or a coalescing-manager
         self.buffers = buffers
         self.group = group
         self.stream = stream  # side comm stream the gather was enqueued on
+        # qwZ: list of (q8_buf, scale_buf, world) per param, or None
+        self.quant = quant
         self.complete = False
 
     def wait(self):
@@ -123,6 +149,11 @@ class AllGatherHandle:
             cur.wait_stream(self.stream)
             for buf in self.buffers:
                 buf.record_stream(cur)
+        if self.quant is not None:
+            # dequantize gathered int8 shards into the bf16 buffers
+            for buf, (q8, sc, world) in zip(self.buffers, self.quant):
+                if q8 is not None:
+                    dequantize_gathered(q8, sc, world, buf.dtype, out=buf)
         for p, buf in zip(self.params, self.buffers):
             p.data = buf.narrow(0, 0, p.ds_numel).view(p.ds_shape)
             p.ds_full_buffer = buf
@@ -130,14 +161,16 @@ class AllGatherHandle:
         self.complete = True
 
 
-def all_gather_params(params, dp_group, async_op=True, stream=None):
+def all_gather_params(params, dp_group, async_op=True, stream=None,
+                      quantized=False):
     """Launch coalesced all-gathers for NOT_AVAILABLE params.
 
     Returns an AllGatherHandle (already complete when nothing to do).
     Must be called identically on all ranks of the group. When `stream` is
     given, the collectives enqueue on that side stream (overlap with
     compute on the default stream); handle.wait() installs the
-    stream-order dependency.
+    stream-order dependency. `quantized` = ZeRO++ qwZ: int8 blockwise
+    shards + fp16 scales on the wire (half the gather bytes).
     """
     todo = [p for p in params if p.ds_status == ZeroParamStatus.NOT_AVAILABLE]
     for p in todo:
@@ -178,8 +211,9 @@ def all_gather_params(params, dp_group, async_op=True, stream=None):
         # side stream must see the up-to-date shards
         stream.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(stream):
-            return _launch_gathers(todo, dp_group, world, async_op, stream)
-    return _launch_gathers(todo, dp_group, world, async_op, None)
+            return _launch_gathers(todo, dp_group, world, async_op, stream,
+                                   quantized)
+    return _launch_gathers(todo, dp_group, world, async_op, None, quantized)
 
 
 def _shard_on_device(p):
@@ -194,7 +228,11 @@ def _shard_on_device(p):
     return dev
 
 
-def _launch_gathers(todo, dp_group, world, async_op, stream):
+def _launch_gathers(todo, dp_group, world, async_op, stream,
+                    quantized=False):
+    if quantized:
+        return _launch_gathers_quant(todo, dp_group, world, async_op,
+                                     stream)
     buffers = []
     works = []
     use_coalescing = _supports_coalescing(dp_group) and len(todo) > 1
@@ -228,6 +266,32 @@ def _launch_gathers(todo, dp_group, world, async_op, stream):
             buffers.append(buf)
             works.append(w)
     handle = AllGatherHandle(todo, works, buffers, dp_group, stream=stream)
+    if not async_op:
+        handle.wait()
+    return handle
+
+
+def _launch_gathers_quant(todo, dp_group, world, async_op, stream):
+    """qwZ gather: per param, all-gather (int8 shard, fp16 block scales);
+    dequant into the bf16 full buffer at wait() time."""
+    buffers, works, quant = [], [], []
+    for p in todo:
+        sh = _shard_on_device(p)
+        q, sc = quantize_shard(sh)
+        dev = q.device
+        qbuf = torch.empty(q.numel() * world, dtype=torch.int8, device=dev)
+        sbuf = torch.empty(sc.numel() * world, dtype=torch.float16,
+                           device=dev)
+        w1 = dist.all_gather_into_tensor(qbuf, q, group=dp_group,
+                                         async_op=async_op)
+        w2 = dist.all_gather_into_tensor(sbuf, sc, group=dp_group,
+                                         async_op=async_op)
+        buffers.append(torch.empty(p.ds_shard_numel * world,
+                                   dtype=p.ds_dtype, device=dev))
+        works += [w1, w2]
+        quant.append((qbuf, sbuf, world))
+    handle = AllGatherHandle(todo, works, buffers, dp_group, stream=stream,
+                             quant=quant)
     if not async_op:
         handle.wait()
     return handle
