@@ -72,6 +72,8 @@ class ZeroConfig(DSConfigModel):
     zero_hpz_partition_size: int = 1
     # ZeRO++ qwZ: int8 blockwise weight gathers (half the AG bytes)
     zero_quantized_weights: bool = False
+    # ZeRO++ qgZ: int8 all-to-all gradient reduction (half the RS bytes)
+    zero_quantized_gradients: bool = False
     mics_shard_size: int = -1
     round_robin_gradients: bool = False
     ignore_unused_parameters: bool = True

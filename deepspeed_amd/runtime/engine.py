@@ -310,6 +310,7 @@ class DeepSpeedEngine(torch.nn.Module):
                 offload_optimizer=zc.offload_optimizer,
                 offload_param=zc.offload_param,
                 zero_quantized_weights=zc.zero_quantized_weights,
+                zero_quantized_gradients=zc.zero_quantized_gradients,
                 clip_grad=self.gradient_clipping(),
                 static_loss_scale=self._static_loss_scale(),
                 dynamic_loss_scale=self._dynamic_loss_scale(),

@@ -149,6 +149,7 @@ class ZeroStage3Optimizer:
                  offload_optimizer=None,
                  offload_param=None,
                  zero_quantized_weights=False,
+                 zero_quantized_gradients=False,
                  clip_grad=0.0,
                  static_loss_scale=1.0,
                  dynamic_loss_scale=False,
@@ -178,6 +179,7 @@ class ZeroStage3Optimizer:
         # ZeRO++ qwZ (ref coalesced qwZ path): int8 blockwise shard
         # gathers + fp16 scales — halves all-gather bytes over xGMI
         self.quantized_weights = bool(zero_quantized_weights)
+        self.quantized_gradients = bool(zero_quantized_gradients)
 
         self.device = (torch.device("cuda", torch.cuda.current_device())
                        if torch.cuda.is_available() else torch.device("cpu"))
@@ -414,6 +416,37 @@ class ZeroStage3Optimizer:
             if self.param_swapper is not None:
                 self.param_swapper.mark_dirty(i)
                 self.param_swapper.evict_to_budget()
+
+    def _flush_ipg_quant(self, params, world, pg):
+        """ZeRO++ qgZ (single-node form, ref coalesced_collectives.py:31
+        all_to_all_quant_reduce): grads quantize to int8 blockwise BEFORE
+        the wire; one all-to-all moves each rank's shard-chunks (half the
+        reduce-scatter bytes), then dequant + sum + average locally."""
+        from .stage3_params import dequantize_gathered, quantize_shard
+        for p in params:
+            sn = p.ds_shard_numel
+            if sn * world == p.ds_numel:
+                padded = p.grad.reshape(-1)
+            else:
+                padded = torch.empty(sn * world, dtype=p.grad.dtype,
+                                     device=p.grad.device)
+                padded[:p.ds_numel].copy_(p.grad.reshape(-1))
+                padded[p.ds_numel:].zero_()
+            q, sc = quantize_shard(padded)          # int8 [world*sn]
+            qr = torch.empty_like(q)
+            sr = torch.empty_like(sc)
+            dist.all_to_all_single(qr, q, group=pg)
+            dist.all_to_all_single(sr, sc, group=pg)
+            # received: world chunks of MY shard -> dequant, sum, average
+            deq = dequantize_gathered(qr, sr, world, torch.float32)
+            shard = deq.view(world, sn).sum(0).div_(world)
+            sg = self.param_to_subgroup[p]
+            if self.replica_world > 1:
+                from .stage_1_and_2 import _avg_op
+                op = _avg_op(self.replica_world, shard)
+                dist.all_reduce(shard, op=op, group=self.replica_group)
+            sg.accumulate_grad(p, shard.to(p.grad.dtype))
+            p.grad = None
 
     def _param_offload_step(self, combined, fused):
         """NVMe param tier: per-sub-group step under the host-RAM budget.
@@ -745,6 +778,9 @@ class ZeroStage3Optimizer:
                     dist.all_reduce(g, op=op, group=self.replica_group)
                 sg.accumulate_grad(p, g)
                 p.grad = None
+            return
+        if self.quantized_gradients:
+            self._flush_ipg_quant(params, world, pg)
             return
         from .stage_1_and_2 import _avg_op
         inputs = []
