@@ -456,11 +456,22 @@ class DeepSpeedEngine(torch.nn.Module):
         return loss
 
     def allreduce_gradients(self, bucket_size=MEMORY_OPT_ALLREDUCE_SIZE):
-        """ZeRO-0 fp32 fallback: bucketed allreduce (ref engine.py:3784)."""
+        """ZeRO-0 fp32 fallback: bucketed allreduce (ref engine.py:3784);
+        sparse grads (sparse embeddings) average via (indices, values)
+        exchange instead of densifying (ref sparse_allreduce:3815)."""
         if self.dp_world_size <= 1:
             return
+        sparse = [p for p in self.module.parameters()
+                  if p.grad is not None and p.grad.is_sparse]
+        if sparse:
+            from .sparse_tensor import SparseTensor, sparse_allreduce
+            for p in sparse:
+                st = sparse_allreduce(SparseTensor(p.grad),
+                                      dp_group=self.dp_group,
+                                      dp_world_size=self.dp_world_size)
+                p.grad = st.to_coo_tensor().coalesce()
         grads = [p.grad for p in self.module.parameters()
-                 if p.grad is not None]
+                 if p.grad is not None and not p.grad.is_sparse]
         bucket, numel = [], 0
         for g in grads:
             bucket.append(g)

@@ -285,3 +285,65 @@ def test_multinode_runner_commands():
     import pytest as _pytest
     with _pytest.raises(ValueError):
         get_runner("nope", args, world)
+
+
+def _sparse_allreduce_body():
+    import torch
+    import torch.distributed as tdist
+    from deepspeed_amd.runtime.sparse_tensor import (SparseTensor,
+                                                     sparse_allreduce)
+    rank = tdist.get_rank()
+    emb = torch.nn.Embedding(10, 4, sparse=True)
+    with torch.no_grad():
+        emb.weight.zero_()
+    ids = torch.tensor([1, 3] if rank == 0 else [3, 7])
+    emb(ids).sum().backward()
+    st = sparse_allreduce(SparseTensor(emb.weight.grad))
+    return st.to_dense()
+
+
+def test_sparse_tensor_allreduce_world2():
+    """Sparse embedding grads average via (indices, values) exchange."""
+    import torch
+    from tests.common import run_distributed
+    outs = run_distributed(_sparse_allreduce_body, world_size=2)
+    want = torch.zeros(10, 4)
+    want[1] = 0.5   # touched by rank 0 only -> 1/2
+    want[3] = 1.0   # touched by both -> (1+1)/2
+    want[7] = 0.5
+    for o in outs:
+        assert torch.allclose(o, want), o
+
+
+def test_sd_loader_megatron_merge_split(tmp_path):
+    """MegatronSDLoader merges MP=2 -> 1 and splits MP=1 -> 2 with the
+    right axes per tensor family (ref state_dict_factory.py:21)."""
+    import torch
+    from deepspeed_amd.runtime.state_dict_factory import SDLoaderFactory
+    col = torch.arange(8.0).view(4, 2)   # query_key_value: cat dim 0
+    row = torch.arange(8.0).view(2, 4)   # attention.dense: cat dim 1
+    sd0 = {"module": {"attn.query_key_value.weight": col[:2],
+                      "attn.attention.dense.weight": row[:, :2],
+                      "ln.weight": torch.ones(2)}}
+    sd1 = {"module": {"attn.query_key_value.weight": col[2:],
+                      "attn.attention.dense.weight": row[:, 2:],
+                      "ln.weight": torch.ones(2)}}
+    f0, f1 = str(tmp_path / "mp0.pt"), str(tmp_path / "mp1.pt")
+    torch.save(sd0, f0)
+    torch.save(sd1, f1)
+
+    loader = SDLoaderFactory.get_sd_loader([f0, f1])
+    _, merged = loader.load(mp_world_size=1, mp_rank=0)
+    assert torch.equal(merged["attn.query_key_value.weight"], col)
+    assert torch.equal(merged["attn.attention.dense.weight"], row)
+
+    # split back: save merged as MP=1 then load at MP=2
+    fm = str(tmp_path / "merged.pt")
+    torch.save({"module": merged}, fm)
+    loader2 = SDLoaderFactory.get_sd_loader([fm])
+    _, part1 = loader2.load(mp_world_size=2, mp_rank=1)
+    assert torch.equal(part1["attn.query_key_value.weight"], col[2:])
+    assert torch.equal(part1["attn.attention.dense.weight"], row[:, 2:])
+
+    data = {"type": "ds_model", "checkpoints": [fm], "version": 1}
+    assert SDLoaderFactory.get_sd_loader_json(data) is data
