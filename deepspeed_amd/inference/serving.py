@@ -52,6 +52,7 @@ class RaggedKVCache:
         self.attn_mask = None
         self.ragged = None      # (kpool, vpool, rows, lens) for HIP decode
         self._prefill_slot = None
+        self._prefill_off = 0   # chunked (SplitFuse-style) prefill offset
         # ragged decode HIP kernel reads straight from the pool — no
         # per-token gather copies (ref inference/v2 ragged_ops role)
         from ..ops.loader import get_ext
@@ -66,12 +67,23 @@ class RaggedKVCache:
         if self._prefill_slot is not None:   # [1, S, Hk, D] prompt chunk
             s = self._prefill_slot
             S = k.shape[1]
-            self.k[s, :S] = k[0]
-            self.v[s, :S] = v[0]
-            self.lens[s] = S
-            self.attn_mask = None
+            o = self._prefill_off
+            self.k[s, o:o + S] = k[0]
+            self.v[s, o:o + S] = v[0]
+            self.lens[s] = o + S
             self.ragged = None
-            return k, v                      # causal prefill over itself
+            if o == 0:
+                self.attn_mask = None
+                return k, v                  # causal prefill over itself
+            # chunk at offset o attends to the cached prefix 0..o plus a
+            # causal window within itself (Dynamic SplitFuse-style step)
+            total = o + S
+            q_pos = torch.arange(o, total, device=k.device).view(S, 1)
+            k_pos = torch.arange(total, device=k.device).view(1, total)
+            self.attn_mask = torch.where(k_pos <= q_pos, 0.0,
+                                         float("-inf")) \
+                .view(1, 1, S, total).to(torch.float32)
+            return (self.k[s:s + 1, :total], self.v[s:s + 1, :total])
         rows = self.rows                     # decode: [n, 1, Hk, D]
         pos = self.lens[rows]
         self.k[rows, pos] = k[:, 0]
@@ -94,7 +106,8 @@ class RaggedKVCache:
 class ContinuousBatchingEngine:
     """Token-level continuous batching over a native model."""
 
-    def __init__(self, model, max_batch=8, max_seq=None, device=None):
+    def __init__(self, model, max_batch=8, max_seq=None, device=None,
+                 prefill_chunk=None):
         self.model = model
         cfg = model.cfg if hasattr(model, "cfg") else model.config
         self.cfg = cfg
@@ -106,6 +119,12 @@ class ContinuousBatchingEngine:
                                      cfg.num_key_value_heads, cfg.head_dim,
                                      dtype, self.device)
                        for _ in range(cfg.num_hidden_layers)]
+        # Dynamic SplitFuse-style scheduling (ref FastGen): bound the
+        # prompt tokens processed per step so long prompts stream in
+        # chunks instead of stalling the decode batch. None = whole
+        # prompt per step (legacy behavior).
+        self.prefill_chunk = prefill_chunk
+        self.prefilling: List[Request] = []
         self.free_slots = list(range(max_batch))
         self.pending: List[Request] = []
         self.running: List[Request] = []
@@ -126,7 +145,8 @@ class ContinuousBatchingEngine:
     # -- internals ----------------------------------------------------------
     @torch.no_grad()
     def _prefill(self, req):
-        req.slot = self.free_slots.pop()
+        if req.slot < 0:
+            req.slot = self.free_slots.pop()
         for c in self.caches:
             c._prefill_slot = req.slot
         ids = req.prompt.view(1, -1).to(self.device)
@@ -135,6 +155,29 @@ class ContinuousBatchingEngine:
             c._prefill_slot = None
         req.generated.append(self._sample(logits[0, -1], req))
         self._check_done(req)
+
+    @torch.no_grad()
+    def _prefill_chunk_step(self, req):
+        """Advance one request's prompt by at most prefill_chunk tokens;
+        returns True when the prompt is fully ingested (and the first
+        token sampled)."""
+        pos = getattr(req, "prefill_pos", 0)
+        n = len(req.prompt)
+        take = min(self.prefill_chunk, n - pos)
+        for c in self.caches:
+            c._prefill_slot = req.slot
+            c._prefill_off = pos
+        ids = req.prompt[pos:pos + take].view(1, -1).to(self.device)
+        logits = self.model(ids, kv_caches=self.caches, seq_offset=pos)
+        for c in self.caches:
+            c._prefill_slot = None
+            c._prefill_off = 0
+        req.prefill_pos = pos + take
+        if req.prefill_pos >= n:
+            req.generated.append(self._sample(logits[0, -1], req))
+            self._check_done(req)
+            return True
+        return False
 
     @torch.no_grad()
     def _decode(self, active):
@@ -182,11 +225,27 @@ class ContinuousBatchingEngine:
     # -- public loop --------------------------------------------------------
     @torch.no_grad()
     def step(self):
-        """Admit + one decode step. Returns requests finished this step."""
-        while self.pending and self.free_slots:
-            req = self.pending.pop(0)
-            self._prefill(req)
-            self.running.append(req)
+        """Admit + one decode step. Returns requests finished this step.
+
+        With prefill_chunk set, at most ONE chunk of ONE prompt is
+        ingested per step (bounded prefill work), interleaved with the
+        decode batch — long prompts no longer stall running decodes."""
+        if self.prefill_chunk is None:
+            while self.pending and self.free_slots:
+                req = self.pending.pop(0)
+                self._prefill(req)
+                self.running.append(req)
+        else:
+            while self.pending and self.free_slots:
+                req = self.pending.pop(0)
+                req.slot = self.free_slots.pop()
+                req.prefill_pos = 0
+                self.prefilling.append(req)
+            if self.prefilling:
+                req = self.prefilling[0]
+                if self._prefill_chunk_step(req):
+                    self.prefilling.pop(0)
+                    self.running.append(req)
         active = [r for r in self.running if not r.done]
         if active:
             self._decode(active)
