@@ -123,9 +123,13 @@ def init_compression(model, compression_config):
     n = 0
     for name, module in list(model.named_modules()):
         for child_name, child in list(module.named_children()):
-            if type(child) is torch.nn.Linear:
+            if type(child) is torch.nn.Linear or \
+                    isinstance(child, LinearLayerCompress):
                 full = f"{name}.{child_name}" if name else child_name
-                new = None
+                # already-compressible modules take new techniques in
+                # place (scheduler activates methods at different steps)
+                new = child if isinstance(child, LinearLayerCompress) \
+                    else None
                 for gname, g in wq.get("different_groups", {}).items():
                     if any(re.search(pat, full)
                            for pat in g.get("modules", [".*"])):
@@ -212,3 +216,66 @@ def apply_layer_reduction(model, layer_reduction_config):
     log_dist(f"layer_reduction: kept {len(new_layers)} layers {keep}",
              ranks=[0])
     return model
+
+
+class CompressionScheduler:
+    """Step-scheduled compression (ref compression/scheduler.py:12):
+    each technique activates once training reaches its
+    shared_parameters.schedule_offset (and deactivates at
+    schedule_offset_end if given), instead of from step 0."""
+
+    METHODS = ("weight_quantization", "activation_quantization",
+               "sparse_pruning", "row_pruning", "head_pruning")
+
+    def __init__(self, model, compression_config):
+        self.model = model
+        self.config = dict(compression_config or {})
+        self.training_steps = 0
+        self._activated = set()
+        self._deactivated = set()
+        # convert eagerly only methods with offset 0 (classic behavior)
+        self.step(0)
+
+    def _offsets(self, method):
+        sp = self.config.get(method, {}).get("shared_parameters", {})
+        return (sp.get("schedule_offset", 0),
+                sp.get("schedule_offset_end", None),
+                sp.get("enabled", True))
+
+    def step(self, increment=1):
+        self.training_steps += increment
+        for m in self.METHODS:
+            if m not in self.config:
+                continue
+            off, off_end, enabled = self._offsets(m)
+            if not enabled:
+                continue
+            if m not in self._activated and self.training_steps >= off:
+                init_compression(self.model, {m: self.config[m]})
+                self._activated.add(m)
+                log_dist(f"compression: {m} enabled at step "
+                         f"{self.training_steps}", ranks=[0])
+            if off_end is not None and m in self._activated and \
+                    m not in self._deactivated and \
+                    self.training_steps >= off_end:
+                for mod in self.model.modules():
+                    if isinstance(mod, LinearLayerCompress):
+                        if m == "weight_quantization":
+                            mod.weight_quant_enabled = False
+                        elif m == "activation_quantization":
+                            mod.act_quant_enabled = False
+                        else:
+                            mod.prune_mask = None
+                self._deactivated.add(m)
+                log_dist(f"compression: {m} disabled at step "
+                         f"{self.training_steps}", ranks=[0])
+
+    def state_dict(self):
+        return {"training_steps": self.training_steps,
+                "activated": sorted(self._activated),
+                "deactivated": sorted(self._deactivated)}
+
+    def load_state_dict(self, sd):
+        self.training_steps = sd["training_steps"]
+        self._activated = set(sd["activated"])
+        self._deactivated = set(sd["deactivated"])

@@ -122,3 +122,44 @@ def test_structured_pruning_and_layer_reduction():
                               "teacher_layer": [0, 2, 4]})
     assert len(m.layers) == 3
     assert m(torch.randn(2, 16)).shape == (2, 16)
+
+
+def test_compression_scheduler_offsets():
+    """Techniques activate at their schedule_offset, not step 0, and
+    deactivate at schedule_offset_end (ref compression/scheduler.py)."""
+    import torch
+    from deepspeed_amd.compression import (CompressionScheduler,
+                                           LinearLayerCompress)
+    model = torch.nn.Sequential(torch.nn.Linear(8, 8),
+                                torch.nn.Linear(8, 8))
+    cfg = {
+        "weight_quantization": {
+            "shared_parameters": {"enabled": True, "schedule_offset": 3},
+            "different_groups": {"wq": {"params": {"target_bits": 8},
+                                        "modules": [".*"]}}},
+        "sparse_pruning": {
+            "shared_parameters": {"enabled": True, "schedule_offset": 5,
+                                  "schedule_offset_end": 7},
+            "different_groups": {"sp": {"params": {"dense_ratio": 0.5},
+                                        "modules": [".*"]}}},
+    }
+    sched = CompressionScheduler(model, cfg)
+    assert not any(isinstance(m, LinearLayerCompress)
+                   for m in model.modules())
+    sched.step(3)  # -> step 3: weight quantization turns on
+    wq_mods = [m for m in model.modules()
+               if isinstance(m, LinearLayerCompress)]
+    assert wq_mods and all(m.weight_quant_enabled for m in wq_mods)
+    assert all(m.prune_mask is None for m in wq_mods)
+    sched.step(2)  # -> step 5: sparse pruning turns on
+    assert all(m.prune_mask is not None
+               for m in model.modules()
+               if isinstance(m, LinearLayerCompress))
+    sched.step(2)  # -> step 7: sparse pruning end
+    assert all(m.prune_mask is None
+               for m in model.modules()
+               if isinstance(m, LinearLayerCompress))
+    sd = sched.state_dict()
+    sched2 = CompressionScheduler(model, cfg)
+    sched2.load_state_dict(sd)
+    assert sched2.training_steps == sched.training_steps
