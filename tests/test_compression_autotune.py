@@ -163,3 +163,42 @@ def test_compression_scheduler_offsets():
     sched2 = CompressionScheduler(model, cfg)
     sched2.load_state_dict(sd)
     assert sched2.training_steps == sched.training_steps
+
+
+def test_autotuner_full_cost_model_and_pruning(tmp_path):
+    """Cost model fits sec = a + b*mb; infeasible configs are pruned by
+    the ZeRO memory model; experiment records land on disk."""
+    from deepspeed_amd.autotuning import (AutotunerFull, ModelBasedTuner,
+                                          estimate_memory_per_gpu)
+    t = ModelBasedTuner()
+    t.record(1, 0.1 + 0.02 * 1)
+    t.record(8, 0.1 + 0.02 * 8)
+    a, b = t.fit()
+    assert abs(a - 0.1) < 1e-6 and abs(b - 0.02) < 1e-6
+    assert t.propose([1, 2, 4, 8]) == 4   # best predicted untried tput
+    # memory model: stage 3 shards everything; stage 0 replicates
+    m0 = estimate_memory_per_gpu(7e9, 0, world=8)
+    m3 = estimate_memory_per_gpu(7e9, 3, world=8)
+    assert m3 < m0 / 3
+
+    import torch
+    from tests.simple_model import SimpleModel, make_batches
+
+    def data_fn(cfg):
+        mb = cfg["train_micro_batch_size_per_gpu"]
+        x = torch.randn(mb, 32)
+        return (x, x)
+
+    tuner = AutotunerFull(
+        lambda: SimpleModel(32), data_fn,
+        {"optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+         "train_micro_batch_size_per_gpu": 1},
+        steps=2, warmup=1, results_dir=str(tmp_path),
+        model_info={"num_params": 3000, "hidden_size": 32,
+                    "num_layers": 3, "seq_len": 1})
+    best, results = tuner.tune(micro_batches=(1, 4), stages=(1,),
+                               max_experiments=3)
+    assert best["zero_optimization"]["stage"] == 1
+    assert any("samples_per_sec" in r for r in results)
+    import os
+    assert len(os.listdir(tmp_path)) >= 1
