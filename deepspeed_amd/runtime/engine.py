@@ -547,6 +547,18 @@ class DeepSpeedEngine(torch.nn.Module):
         self.module.eval()
         return self
 
+    def offload_states(self, include=None, device="cpu",
+                       pin_memory=False, non_blocking=False):
+        """ZeRO-3: push engine states to host to free HBM (ref
+        engine.offload_states)."""
+        assert hasattr(self.optimizer, "offload_states"), \
+            "offload_states requires ZeRO stage 3"
+        self.optimizer.offload_states(include, device, pin_memory,
+                                      non_blocking)
+
+    def reload_states(self, non_blocking=False):
+        self.optimizer.reload_states(non_blocking)
+
     def zero_grad(self, set_to_none=True):
         if hasattr(self.optimizer, "zero_grad"):
             self.optimizer.zero_grad(set_to_none=set_to_none)

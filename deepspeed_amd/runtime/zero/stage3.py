@@ -1089,6 +1089,50 @@ class ZeroStage3Optimizer:
                 self.param_swapper.evict_to_budget()
         self._refresh_persistent_params()
 
+    @torch.no_grad()
+    def offload_states(self, include=None, device="cpu",
+                       pin_memory=False, non_blocking=False):
+        """Temporarily move engine-held states to host (ref
+        stage3.py:3583 offload_states / offload_states.py): frees HBM for
+        a generation phase (RLHF) or a different model. States:
+        lp_params (16-bit shard slabs), hp_params (fp32 masters),
+        lp_grads (grad accumulators), optim_states (Adam moments)."""
+        inc = set(include or ("lp_params", "hp_params", "lp_grads",
+                              "optim_states"))
+        dev = torch.device(device)
+
+        def _move(t):
+            if t is None or t.device == dev:
+                return t
+            new = torch.empty_like(t, device=dev,
+                                   pin_memory=pin_memory and
+                                   dev.type == "cpu")
+            new.copy_(t, non_blocking=non_blocking)
+            return new
+
+        for sg in self.sub_groups:
+            if "lp_params" in inc and sg.flat16 is not None:
+                sg.flat16 = _move(sg.flat16)
+                for p in sg.params:
+                    off = sg.offsets[p]
+                    p.ds_tensor = sg.flat16[off:off + p.ds_shard_numel]
+            if "hp_params" in inc:
+                sg.master32.data = _move(sg.master32.data)
+            if "lp_grads" in inc:
+                sg.grad32 = _move(sg.grad32)
+        if "optim_states" in inc:
+            for st in self.optimizer.state.values():
+                for k, v in list(st.items()):
+                    if torch.is_tensor(v):
+                        st[k] = _move(v)
+        if torch.cuda.is_available():
+            torch.cuda.empty_cache()
+
+    def reload_states(self, non_blocking=False):
+        """Bring offloaded states back to the training device."""
+        self.offload_states(device=self.device,
+                            non_blocking=non_blocking)
+
     def empty_partition_cache(self):
         """Release every gathered (non-persistent) param and return the
         HBM to the allocator (ref engine.empty_partition_cache)."""

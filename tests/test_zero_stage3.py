@@ -396,3 +396,62 @@ def _zero3_qgz_train(steps=4):
 def test_zero3_quantized_gradient_reduce():
     from tests.common import run_distributed
     run_distributed(_zero3_qgz_train, world_size=2)
+
+
+def test_tiled_linear_matches_dense():
+    import torch
+    from deepspeed_amd.runtime.zero.tiling import TiledLinear
+    torch.manual_seed(0)
+    ref = torch.nn.Linear(30, 20)
+    tl = TiledLinear(30, 20, in_splits=3, out_splits=2, init_linear=ref)
+    x = torch.randn(5, 30, requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+    y_ref = ref(x)
+    y = tl(x2)
+    assert torch.allclose(y, y_ref, atol=1e-6), (y - y_ref).abs().max()
+    g = torch.randn_like(y)
+    y.backward(g)
+    y_ref.backward(g)
+    assert torch.allclose(x.grad, x2.grad, atol=1e-6)
+
+
+def _offload_states_body():
+    import torch
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    from tests.simple_model import SimpleModel, make_batches
+    groups.reset_groups()
+    torch.manual_seed(11)
+    model = SimpleModel(32)
+    config = {
+        "train_micro_batch_size_per_gpu": 4,
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+        "zero_optimization": {"stage": 3},
+        "bf16": {"enabled": True},
+    }
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config=config)
+    batches = make_batches(4, 4, 32, dtype=torch.bfloat16)
+    for x, y in batches[:2]:
+        loss = engine(x, y)
+        engine.backward(loss)
+        engine.step()
+    masters = [sg.master32.detach().clone()
+               for sg in engine.optimizer.sub_groups]
+    engine.offload_states()
+    for sg in engine.optimizer.sub_groups:
+        assert sg.master32.device.type == "cpu"
+        assert sg.flat16.device.type == "cpu"
+    engine.reload_states()
+    for sg, m in zip(engine.optimizer.sub_groups, masters):
+        assert torch.allclose(sg.master32.cpu(), m.cpu())
+    # training continues after a round trip
+    x, y = batches[2]
+    loss = engine(x, y)
+    engine.backward(loss)
+    engine.step()
+    return True
+
+
+def test_offload_reload_states():
+    from tests.common import run_distributed
+    run_distributed(_offload_states_body, world_size=1)
