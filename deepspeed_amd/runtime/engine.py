@@ -81,6 +81,7 @@ class DeepSpeedEngine(torch.nn.Module):
             self.progressive_layer_drop = None
         self.skipped_steps = 0
         self._is_gradient_accumulation_boundary = None
+        self.random_ltd_scheduler = None
 
         self.device = (torch.device("cuda", dist.get_local_rank())
                        if torch.cuda.is_available() else torch.device("cpu"))
@@ -113,6 +114,10 @@ class DeepSpeedEngine(torch.nn.Module):
 
         self.training_dataloader = (self.deepspeed_io(training_data)
                                     if training_data is not None else None)
+
+        rltd = getattr(self._config, "random_ltd", None)
+        if rltd and rltd.get("enabled", True):
+            self._configure_random_ltd(rltd)
 
         self.losses = None
         self.flops_profiler = None
@@ -425,6 +430,30 @@ class DeepSpeedEngine(torch.nn.Module):
             self.timers("forward").stop()
         return loss
 
+    def _configure_random_ltd(self, cfg):
+        """Config-driven random-LTD (ref data_routing): wraps decoder
+        layers and steps the token schedule each optimizer step.
+
+        config: {"random_ltd": {"enabled": true, "layer_class":
+        "LlamaDecoderLayer", "min_tokens": 128, "max_tokens": 4096,
+        "schedule_steps": 1000, "layer_ids": [1,2,...]}}"""
+        from .random_ltd import convert_to_random_ltd
+        cls_name = cfg.get("layer_class", "LlamaDecoderLayer")
+        layer_cls = None
+        for m in self.module.modules():
+            if type(m).__name__ == cls_name:
+                layer_cls = type(m)
+                break
+        if layer_cls is None:
+            logger.warning(f"random_ltd: no {cls_name} modules found")
+            return
+        convert_to_random_ltd(self.module, layer_cls,
+                              cfg.get("min_tokens", 128),
+                              cfg.get("max_tokens", 100000),
+                              cfg.get("schedule_steps", 1000),
+                              layer_ids=cfg.get("layer_ids"))
+        self.random_ltd_scheduler = self.module.random_ltd_scheduler
+
     def wall_clock_breakdown(self):
         return self._config.wall_clock_breakdown
 
@@ -511,6 +540,8 @@ class DeepSpeedEngine(torch.nn.Module):
 
     def _take_model_step(self, lr_kwargs=None):
         self.optimizer.step()
+        if self.random_ltd_scheduler is not None:
+            self.random_ltd_scheduler.update(self.global_steps + 1)
         from ..ops.fp8_linear import bump_fp8_version
         bump_fp8_version()  # invalidate cached fp8 weight copies
         overflow = getattr(self.optimizer, "overflow", False)

@@ -27,6 +27,26 @@ class RandomLTDScheduler:
         self.current = max(self.min_tokens, min(n, self.max_tokens))
         return self.current
 
+    def get_current_seq(self):
+        return self.current
+
+    def get_total_layer_tokens(self, train_iters):
+        """Total tokens a wrapped layer processes over train_iters steps
+        (ref data_routing/scheduler.py:55) — for budget accounting."""
+        total = 0
+        for it in range(int(train_iters)):
+            frac = min(1.0, it / self.schedule_steps)
+            n = self.min_tokens + frac * (self.max_tokens -
+                                          self.min_tokens)
+            total += int(n // self.step_size * self.step_size)
+        return total
+
+    def state_dict(self):
+        return {"current": self.current}
+
+    def load_state_dict(self, sd):
+        self.current = sd["current"]
+
 
 class RandomLTDLayer(torch.nn.Module):
     """Wraps a decoder layer: forward(x, *args) with token dropping.
@@ -58,8 +78,10 @@ class RandomLTDLayer(torch.nn.Module):
 
 
 def convert_to_random_ltd(model, layer_cls, min_tokens, max_tokens,
-                          schedule_steps):
-    """Wrap every `layer_cls` submodule with random-LTD."""
+                          schedule_steps, layer_ids=None):
+    """Wrap `layer_cls` submodules with random-LTD. `layer_ids` selects
+    which occurrences (ref random_ltd_layer_id: usually the middle
+    layers — first/last stay full-sequence)."""
     sched = RandomLTDScheduler(min_tokens, max_tokens, schedule_steps)
     targets = []
     for parent in model.modules():
@@ -69,7 +91,9 @@ def convert_to_random_ltd(model, layer_cls, min_tokens, max_tokens,
             if isinstance(child, layer_cls):
                 targets.append((parent, name, child))
     n = 0
-    for parent, name, child in targets:
+    for i, (parent, name, child) in enumerate(targets):
+        if layer_ids is not None and i not in layer_ids:
+            continue
         setattr(parent, name, RandomLTDLayer(child, sched))
         n += 1
     log_dist(f"random-LTD: wrapped {n} layers "

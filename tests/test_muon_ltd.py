@@ -128,3 +128,45 @@ def test_muon_zero3_matches_single_process():
                           for p in ref_model.parameters()]):
         err = (got - want).abs().max().item()
         assert err < 5e-2, f"muon zero3 diverged: {err}"
+
+
+def test_random_ltd_config_driven():
+    """data_efficiency.data_routing.random_ltd wraps decoder layers at
+    engine init (middle layers only via layer_ids) and the token
+    schedule advances with optimizer steps."""
+    import os
+    import torch
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    from deepspeed_amd.models.llama import LLAMA_CONFIGS, LlamaForCausalLM
+    from deepspeed_amd.runtime.random_ltd import RandomLTDLayer
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29533")
+    groups.reset_groups()
+    torch.manual_seed(0)
+    cfg = LLAMA_CONFIGS["llama-tiny"]
+    model = LlamaForCausalLM(cfg)
+    config = {
+        "train_micro_batch_size_per_gpu": 2,
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+        "data_efficiency": {"data_routing": {"random_ltd": {
+            "enabled": True, "layer_class": "LlamaDecoderLayer",
+            "min_tokens": 16, "max_tokens": 64, "schedule_steps": 4,
+            "layer_ids": [1, 2]}}},
+    }
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config=config)
+    wrapped = [m for m in engine.module.modules()
+               if isinstance(m, RandomLTDLayer)]
+    assert len(wrapped) == 2  # only the middle layers
+    assert engine.random_ltd_scheduler is not None
+    assert engine.random_ltd_scheduler.current == 16
+    data = torch.randint(0, cfg.vocab_size, (2, 64))
+    for _ in range(4):
+        loss = engine(data, labels=data)
+        engine.backward(loss)
+        engine.step()
+    assert engine.random_ltd_scheduler.current == 64, \
+        engine.random_ltd_scheduler.current
+    # budget accounting helper
+    total = engine.random_ltd_scheduler.get_total_layer_tokens(4)
+    assert 16 * 4 <= total <= 64 * 4
