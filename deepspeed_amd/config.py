@@ -74,6 +74,8 @@ class ZeroConfig(DSConfigModel):
     zero_quantized_weights: bool = False
     # ZeRO++ qgZ: int8 all-to-all gradient reduction (half the RS bytes)
     zero_quantized_gradients: bool = False
+    # class names whose whole subtree gathers as one unit (MoE experts)
+    leaf_module: dict = {}
     mics_shard_size: int = -1
     round_robin_gradients: bool = False
     ignore_unused_parameters: bool = True
@@ -239,6 +241,7 @@ class DeepSpeedConfig:
         ev = config.get("eigenvalue", {})
         self.eigenvalue = ev if ev.get("enabled") else None
         self.wandb = config.get("wandb", {})
+        self.comet = config.get("comet", {})
         self.compression_training = config.get("compression_training")
 
         if self.fp16.enabled and self.bf16.enabled:

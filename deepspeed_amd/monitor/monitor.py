@@ -60,6 +60,30 @@ class TensorBoardMonitor(Monitor):
         self.writer.flush()
 
 
+class CometMonitor(Monitor):
+    """Comet sink (ref deepspeed/monitor/comet.py:23); no-ops when
+    comet_ml is not installed."""
+
+    def __init__(self, config):
+        super().__init__(config)
+        self.enabled = getattr(config, "enabled", False) and \
+            dist.get_rank() == 0
+        self._exp = None
+        if self.enabled:
+            try:
+                import comet_ml
+                self._exp = comet_ml.Experiment(
+                    project_name=getattr(config, "project", "dsamd"))
+            except ImportError:
+                self.enabled = False
+
+    def write_events(self, event_list):
+        if not self.enabled or self._exp is None:
+            return
+        for name, value, step in event_list:
+            self._exp.log_metric(name, value, step=step)
+
+
 class MonitorMaster(Monitor):
     def __init__(self, ds_config):
         self.monitors = []
@@ -73,6 +97,12 @@ class MonitorMaster(Monitor):
             from types import SimpleNamespace
             cfg = SimpleNamespace(**wb) if isinstance(wb, dict) else wb
             self.monitors.append(WandbMonitor(cfg))
+        cm = getattr(ds_config, "comet", None)
+        if cm and (cm.get("enabled") if isinstance(cm, dict)
+                   else getattr(cm, "enabled", False)):
+            from types import SimpleNamespace
+            cfg = SimpleNamespace(**cm) if isinstance(cm, dict) else cm
+            self.monitors.append(CometMonitor(cfg))
         self.enabled = len(self.monitors) > 0
 
     def write_events(self, event_list):

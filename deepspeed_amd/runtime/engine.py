@@ -316,6 +316,7 @@ class DeepSpeedEngine(torch.nn.Module):
                 offload_param=zc.offload_param,
                 zero_quantized_weights=zc.zero_quantized_weights,
                 zero_quantized_gradients=zc.zero_quantized_gradients,
+                leaf_module_names=(zc.leaf_module or {}).get("classes"),
                 clip_grad=self.gradient_clipping(),
                 static_loss_scale=self._static_loss_scale(),
                 dynamic_loss_scale=self._dynamic_loss_scale(),

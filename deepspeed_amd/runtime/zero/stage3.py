@@ -150,6 +150,7 @@ class ZeroStage3Optimizer:
                  offload_param=None,
                  zero_quantized_weights=False,
                  zero_quantized_gradients=False,
+                 leaf_module_names=None,
                  clip_grad=0.0,
                  static_loss_scale=1.0,
                  dynamic_loss_scale=False,
@@ -180,6 +181,7 @@ class ZeroStage3Optimizer:
         # gathers + fp16 scales — halves all-gather bytes over xGMI
         self.quantized_weights = bool(zero_quantized_weights)
         self.quantized_gradients = bool(zero_quantized_gradients)
+        self.leaf_module_names = list(leaf_module_names or [])
 
         self.device = (torch.device("cuda", torch.cuda.current_device())
                        if torch.cuda.is_available() else torch.device("cpu"))
@@ -595,13 +597,37 @@ class ZeroStage3Optimizer:
             self._gather_grouped(persist, async_op=False).wait()
 
     # ------------------------------------------------------------- hooks
+    def _leaf_module_classes(self):
+        """Class names configured as ZeRO leaf modules (ref
+        runtime/zero/leaf_module_config.py): their whole parameter
+        subtree gathers/releases as ONE unit — e.g. an MoE expert bank
+        whose data-dependent submodule order would defeat trace
+        prefetch."""
+        return set(getattr(self, "leaf_module_names", []) or [])
+
     def _install_module_hooks(self):
         """fetch/release hooks on every module owning direct params (or
-        registered external ones, see register_external_parameter)."""
+        registered external ones, see register_external_parameter).
+        Modules whose class name is in leaf_module_names hook at their
+        own level with their ENTIRE subtree's params; descendants are
+        skipped."""
         self._module_hooks = []
+        leaf_names = self._leaf_module_classes()
+        leaf_descendants = set()
         for mod in self.module.modules():
-            direct = [p for p in mod.parameters(recurse=False)
-                      if is_zero_param(p)]
+            if type(mod).__name__ in leaf_names:
+                for sub in mod.modules():
+                    if sub is not mod:
+                        leaf_descendants.add(id(sub))
+        for mod in self.module.modules():
+            if id(mod) in leaf_descendants:
+                continue
+            if type(mod).__name__ in leaf_names:
+                direct = [p for p in mod.parameters(recurse=True)
+                          if is_zero_param(p)]
+            else:
+                direct = [p for p in mod.parameters(recurse=False)
+                          if is_zero_param(p)]
             external = [p for p in getattr(mod, "_ds_external_params", [])
                         if is_zero_param(p)]
             if not direct and not external:

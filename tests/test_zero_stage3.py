@@ -455,3 +455,58 @@ def _offload_states_body():
 def test_offload_reload_states():
     from tests.common import run_distributed
     run_distributed(_offload_states_body, world_size=1)
+
+
+def _leaf_module_body():
+    """Leaf-module classes gather their whole subtree as one unit."""
+    import torch
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    groups.reset_groups()
+
+    class Expert(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.a = torch.nn.Linear(16, 32)
+            self.b = torch.nn.Linear(32, 16)
+
+        def forward(self, x):
+            return self.b(torch.nn.functional.gelu(self.a(x)))
+
+    class Net(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.inp = torch.nn.Linear(16, 16)
+            self.expert = Expert()
+
+        def forward(self, x, y):
+            return torch.nn.functional.mse_loss(
+                self.expert(self.inp(x)), y)
+
+    torch.manual_seed(0)
+    model = Net()
+    config = {
+        "train_micro_batch_size_per_gpu": 4,
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+        "zero_optimization": {"stage": 3,
+                              "stage3_param_persistence_threshold": 0,
+                              "leaf_module": {"classes": ["Expert"]}},
+        "bf16": {"enabled": True},
+    }
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config=config)
+    assert engine.optimizer.leaf_module_names == ["Expert"]
+    x = torch.randn(4, 16, dtype=torch.bfloat16)
+    y = torch.randn(4, 16, dtype=torch.bfloat16)
+    losses = []
+    for _ in range(3):
+        loss = engine(x, y)
+        engine.backward(loss)
+        engine.step()
+        losses.append(loss.item())
+    assert losses[-1] < losses[0]
+    return True
+
+
+def test_zero3_leaf_module():
+    from tests.common import run_distributed
+    run_distributed(_leaf_module_body, world_size=2)
