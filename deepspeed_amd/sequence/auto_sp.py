@@ -125,3 +125,61 @@ def _wrap_generic(model, sp_group):
                 mod.core_attention = wrapped
             else:
                 mod.attn_fn = wrapped
+
+
+# ---------------------------------------------------------------- compile pass
+def _sp_sdpa(query, key, value, *args, **kwargs):
+    """SDPA with Ulysses all-to-alls spliced around it: inputs arrive
+    sequence-sharded [B, H, s_local, D]; heads scatter / sequence
+    gathers before the attention, inverse after."""
+    import torch.nn.functional as F
+
+    from ..comm import groups
+    from .layer import _SeqAllToAll
+    spg = groups.get_sequence_parallel_group()
+    # [B,H,s,D] -> [B,s,H,D] for the a2a layout
+    q = query.transpose(1, 2).contiguous()
+    k = key.transpose(1, 2).contiguous()
+    v = value.transpose(1, 2).contiguous()
+    q = _SeqAllToAll.apply(spg, q, 2, 1)
+    k = _SeqAllToAll.apply(spg, k, 2, 1)
+    v = _SeqAllToAll.apply(spg, v, 2, 1)
+    out = F.scaled_dot_product_attention(
+        q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2),
+        *args, **kwargs)
+    out = out.transpose(1, 2).contiguous()
+    out = _SeqAllToAll.apply(spg, out, 1, 2)
+    return out.transpose(1, 2)
+
+
+def autosp_backend(gm, example_inputs):
+    """torch.compile backend: rewrite every scaled_dot_product_attention
+    node in the captured graph to the sequence-parallel form (ref
+    compile/passes/sp_compile.py — AutoSP as a graph pass, not a module
+    wrapper). Use: torch.compile(model, backend=autosp_backend) on a
+    model fed sequence-sharded inputs, with the SP group initialized."""
+    import torch
+    import torch.nn.functional as F
+    n = 0
+    targets = {F.scaled_dot_product_attention,
+               torch.ops.aten.scaled_dot_product_attention.default}
+    for node in gm.graph.nodes:
+        if node.op == "call_function" and node.target in targets:
+            node.target = _sp_sdpa
+            n += 1
+    if n:
+        gm.recompile()
+        log_dist(f"AutoSP compile pass: rewrote {n} attention nodes",
+                 ranks=[0])
+    return gm
+
+
+def autosp_compile(model, sp_size=None):
+    """Convenience: initialize the SP group and return
+    torch.compile(model, backend=autosp_backend)."""
+    import torch
+
+    from ..comm import groups
+    if sp_size is not None and not groups.sequence_parallel_is_initialized():
+        groups.initialize_sequence_parallel(sp_size)
+    return torch.compile(model, backend=autosp_backend)
