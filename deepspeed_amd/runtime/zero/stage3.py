@@ -1056,7 +1056,7 @@ class ZeroStage3Optimizer:
                 "partition_count": self.world,
                 "shard_layout": self.shard_layout(),
             }
-        return {
+        sd = {
             "loss_scaler": self.loss_scaler,
             "base_optimizer_state": self.optimizer.state_dict(),
             "fp32_flat_groups": [sg.master32 for sg in self.sub_groups],
@@ -1065,6 +1065,12 @@ class ZeroStage3Optimizer:
             "partition_count": self.world,
             "shard_layout": self.shard_layout(),
         }
+        if getattr(self, "_muon_state", None):
+            # distributed-Muon momentum/Adam shards (keyed by ds_id)
+            sd["muon_state"] = {
+                k: {n: v for n, v in st.items()}
+                for k, st in self._muon_state.items()}
+        return sd
 
     def _param_shapes(self):
         """name -> full shape, in module order (for zero_to_fp32)."""
@@ -1083,6 +1089,9 @@ class ZeroStage3Optimizer:
             self.loss_scaler = sd["loss_scaler"]
         if load_optimizer_states and "base_optimizer_state" in sd:
             self.optimizer.load_state_dict(sd["base_optimizer_state"])
+        if load_optimizer_states and "muon_state" in sd:
+            self._muon_state = {k: dict(st)
+                                for k, st in sd["muon_state"].items()}
         saved = sd.get("fp32_flat_groups", [])
         assert len(saved) == len(self.sub_groups)
         if self.nvme_swapper is not None:

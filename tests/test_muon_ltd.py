@@ -170,3 +170,49 @@ def test_random_ltd_config_driven():
     # budget accounting helper
     total = engine.random_ltd_scheduler.get_total_layer_tokens(4)
     assert 16 * 4 <= total <= 64 * 4
+
+
+def _muon_ckpt_body(tmpdir):
+    """Muon momentum shards survive a checkpoint round trip."""
+    import torch
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    from tests.simple_model import SimpleModel, make_batches
+    groups.reset_groups()
+    torch.manual_seed(11)
+    model = SimpleModel(32)
+    config = {
+        "train_micro_batch_size_per_gpu": 4,
+        "optimizer": {"type": "Muon",
+                      "params": {"lr": 0.02, "momentum": 0.9,
+                                 "adamw_lr": 1e-3}},
+        "zero_optimization": {"stage": 3},
+        "bf16": {"enabled": True},
+    }
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config=config)
+    batches = make_batches(6, 4, 32, dtype=torch.bfloat16)
+    for x, y in batches[:3]:
+        loss = engine(x, y)
+        engine.backward(loss)
+        engine.step()
+    bufs = {k: {n: v.clone() for n, v in st.items()}
+            for k, st in engine.optimizer._muon_state.items()}
+    engine.save_checkpoint(tmpdir, tag="m")
+
+    groups.reset_groups()
+    torch.manual_seed(11)
+    model2 = SimpleModel(32)
+    engine2, _, _, _ = deepspeed_amd.initialize(model=model2, config=config)
+    engine2.load_checkpoint(tmpdir, tag="m")
+    st2 = engine2.optimizer._muon_state
+    assert set(st2) == set(bufs)
+    for k in bufs:
+        for n, v in bufs[k].items():
+            assert torch.allclose(st2[k][n], v), (k, n)
+    return True
+
+
+def test_muon_state_checkpoint_roundtrip(tmp_path):
+    from tests.common import run_distributed
+    run_distributed(_muon_ckpt_body, world_size=1,
+                    args=(str(tmp_path),))
