@@ -45,6 +45,8 @@ def parse_args():
                         "hipBLASLt _scaled_mm (bf16 master numerics)")
     p.add_argument("--fp8-attn-proj", action="store_true",
                    help="also run qkv/o projections in fp8")
+    p.add_argument("--fp8-head", action="store_true",
+                   help="also run the lm_head GEMM in fp8")
     p.add_argument("--act-ckpt", action="store_true",
                    help="enable activation checkpointing (default off: "
                    "288 GB HBM3E fits full activations at these configs)")
@@ -87,13 +89,15 @@ def main():
     # build directly on device: 8B bf16 = 16 GB, fits trivially in 288 GB
     with torch.device(device):
         model = model_cls(cfg)
-    if args.fp8_mlp or args.fp8_attn_proj:
+    if args.fp8_mlp or args.fp8_attn_proj or args.fp8_head:
         from deepspeed_amd.ops.fp8_linear import Fp8Linear
         tags = []
         if args.fp8_mlp:
             tags += ["gate_proj", "up_proj", "down_proj"]
         if args.fp8_attn_proj:
             tags += ["q_proj", "k_proj", "v_proj", "o_proj"]
+        if args.fp8_head:
+            tags += ["lm_head"]
         n_conv = Fp8Linear.convert(model, include=tags)
         if rank == 0:
             print(f"# fp8 linear: {n_conv} layers converted", flush=True)
@@ -206,8 +210,8 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": ("bf16+fp8gemm" if (args.fp8_mlp or
-                                          args.fp8_attn_proj) else "bf16"),
+            "dtype": ("bf16+fp8gemm" if (args.fp8_mlp or args.fp8_attn_proj
+                                          or args.fp8_head) else "bf16"),
             "data": "synthetic",
             "config": {"model": args.model, "global_batch": global_batch,
                        "seq_len": S,
