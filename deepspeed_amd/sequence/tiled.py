@@ -73,3 +73,45 @@ def tiled_logits_loss(hidden, lm_weight, labels, num_shards=8,
             part = chunk_loss(h, lab, nv)
         total = total + part.float()
     return (total / total_valid).to(hidden.dtype)
+
+
+def agree_num_shards(num_shards, group=None):
+    """All SP ranks must tile identically or collectives inside fn
+    deadlock (ref ulysses_sp.py TiledMLP example): take the group MAX."""
+    import torch.distributed as dist
+    if not dist.is_initialized() or dist.get_world_size(group) == 1:
+        return num_shards
+    t = torch.tensor(int(num_shards))
+    if torch.cuda.is_available():
+        t = t.cuda()
+    dist.all_reduce(t, op=dist.ReduceOp.MAX, group=group)
+    return int(t.item())
+
+
+def enable_tiled_mlp_for_hf(model, num_shards=None, sp_group=None):
+    """Monkey-patch gate/up/down MLP modules of an HF model to run
+    sequence-tiled (ref ulysses_sp.py:943 TiledMLP HF example).
+
+    Design note vs the reference: the reference re-runs backward per
+    shard via a custom autograd.Function and gates ZeRO's per-param
+    reduction with ds_grad_is_ready; here each shard is a checkpointed
+    segment of ONE autograd graph, so grads accumulate exactly once and
+    the ZeRO hooks need no gating."""
+    import math
+    n_patched = 0
+    for mod in model.modules():
+        if all(hasattr(mod, a) for a in ("gate_proj", "up_proj",
+                                         "down_proj")):
+            inner = mod.forward
+
+            def tiled_forward(x, _inner=inner, _shards=num_shards,
+                              _spg=sp_group):
+                s = _shards or max(1, math.ceil(x.shape[-2] /
+                                                max(x.shape[-1], 1)))
+                s = agree_num_shards(s, _spg)
+                return sequence_tiled_compute(_inner, x, s, dim=-2,
+                                              compute_params=True)
+
+            mod.forward = tiled_forward
+            n_patched += 1
+    return n_patched
