@@ -50,3 +50,103 @@ def apply_tensor_parallel(model, tp_group=None):
     log_dist(f"AutoTP: sharded {n_shard} linears across tp={world}",
              ranks=[0])
     return model
+
+
+# name-pattern classification for arbitrary (HF) model families
+# (ref auto_tp.py:294 tp_parser / :357 _replace)
+COLUMN_PATTERNS = ("q_proj", "k_proj", "v_proj", "gate_proj", "up_proj",
+                   "query_key_value", "dense_h_to_4h", "fc1", "wi_0",
+                   "wi_1", "c_attn", "query", "key", "value")
+ROW_PATTERNS = ("o_proj", "down_proj", "dense_4h_to_h", "out_proj",
+                "fc2", "wo", "c_proj", "attention.dense")
+# per-rank head attributes HF attention modules carry
+HEAD_ATTRS = ("num_heads", "num_attention_heads", "num_key_value_heads",
+              "num_kv_heads", "embed_dim", "hidden_size", "split_size",
+              "all_head_size")
+
+
+def tp_parser(model):
+    """Classify every Linear by name into column/row/replicated.
+    Returns {qualified_name: "column"|"row"} (ref AutoTP.tp_parser)."""
+    plan = {}
+    for name, mod in model.named_modules():
+        if not isinstance(mod, torch.nn.Linear):
+            continue
+        if any(p in name for p in ROW_PATTERNS):
+            plan[name] = "row"
+        elif any(p in name for p in COLUMN_PATTERNS):
+            plan[name] = "column"
+    return plan
+
+
+def apply_tensor_parallel_hf(model, tp_group=None, plan=None):
+    """Generic (HF-compatible) AutoTP: shard Linears per the name-pattern
+    plan and divide per-rank head-count attributes on the parent
+    attention modules."""
+    group = tp_group if tp_group is not None \
+        else grp.get_tensor_parallel_group()
+    world = dist.get_world_size(group)
+    rank = dist.get_rank(group)
+    if world == 1:
+        return model
+    plan = plan or tp_parser(model)
+    mods = dict(model.named_modules())
+    n = 0
+    for name, kind in plan.items():
+        parent_name, _, child = name.rpartition(".")
+        parent = mods.get(parent_name, model)
+        lin = getattr(parent, child)
+        if kind == "column":
+            assert lin.out_features % world == 0, name
+            new = LinearLayer.from_linear(lin, group, rank, world)
+        else:
+            assert lin.in_features % world == 0, name
+            new = LinearAllreduce.from_linear(lin, group, rank, world)
+        setattr(parent, child, new)
+        n += 1
+    # per-rank head bookkeeping on modules that own sharded linears
+    touched_parents = {name.rpartition(".")[0] for name in plan}
+    for pname in touched_parents:
+        pm = mods.get(pname)
+        if pm is None:
+            continue
+        for attr in HEAD_ATTRS:
+            v = getattr(pm, attr, None)
+            if isinstance(v, int) and v % world == 0:
+                setattr(pm, attr, v // world)
+    log_dist(f"AutoTP(HF): sharded {n} linears across tp={world}",
+             ranks=[0])
+    return model
+
+
+def add_tp_training_hooks(model, tp_group=None):
+    """Training AutoTP (ref module_inject training path): replicated
+    (non-sharded) params — norms, embeddings, lm_head — see the SAME
+    batch on every TP rank, so their grads must all-reduce (average)
+    over the TP group each backward; sharded layers' grads stay
+    rank-local by construction."""
+    group = tp_group if tp_group is not None \
+        else grp.get_tensor_parallel_group()
+    world = dist.get_world_size(group)
+    if world == 1:
+        return []
+    sharded_params = set()
+    for mod in model.modules():
+        if isinstance(mod, (LinearLayer, LinearAllreduce)):
+            for p in mod.parameters(recurse=False):
+                sharded_params.add(id(p))
+
+    def make_hook(pg):
+        def hook(param):
+            param.grad.div_(dist.get_world_size(pg))
+            dist.all_reduce(param.grad, group=pg)
+        return hook
+
+    handles = []
+    for p in model.parameters():
+        if id(p) not in sharded_params and p.requires_grad:
+            handles.append(
+                p.register_post_accumulate_grad_hook(make_hook(group)))
+    log_dist(f"AutoTP training: {len(handles)} replicated params "
+             f"all-reduce grads over tp={world}", ranks=[0])
+    return handles
