@@ -117,11 +117,85 @@ class FastCheckpointEngine(CheckpointEngine):
         return meta["__obj__"]
 
 
+class DecoupledCheckpointEngine(CheckpointEngine):
+    """Background-process writer (ref
+    runtime/checkpoint_engine/decoupled_checkpoint_engine.py): save()
+    hands the CPU-materialized state to a spawned writer process and
+    returns immediately; commit(tag) blocks until every file of that tag
+    is durably on disk. Training overlaps with serialization."""
+
+    def __init__(self, config_params=None):
+        super().__init__(config_params)
+        import multiprocessing as mp
+        ctx = mp.get_context("spawn")
+        self._queue = ctx.Queue()
+        self._done = ctx.Queue()
+        self._proc = ctx.Process(target=self._writer_loop,
+                                 args=(self._queue, self._done),
+                                 daemon=True)
+        self._proc.start()
+        self._pending = 0
+
+    @staticmethod
+    def _writer_loop(q, done):
+        import torch as _torch
+        while True:
+            item = q.get()
+            if item is None:
+                break
+            path, payload = item
+            _torch.save(payload, path)
+            done.put(path)
+
+    @staticmethod
+    def _to_cpu(obj):
+        if torch.is_tensor(obj):
+            return obj.detach().cpu().clone()
+        if isinstance(obj, dict):
+            return {k: DecoupledCheckpointEngine._to_cpu(v)
+                    for k, v in obj.items()}
+        if isinstance(obj, (list, tuple)):
+            t = [DecoupledCheckpointEngine._to_cpu(v) for v in obj]
+            return type(obj)(t) if not isinstance(obj, tuple) else tuple(t)
+        return obj
+
+    def save(self, state_dict, path):
+        # snapshot to host BEFORE returning: the trainer may mutate the
+        # live tensors on the very next step
+        payload = self._to_cpu(state_dict)
+        self._queue.put((path, payload))
+        self._pending += 1
+
+    def commit(self, tag):
+        while self._pending:
+            self._done.get()
+            self._pending -= 1
+        return True
+
+    def load(self, path, map_location=None):
+        self.commit(None)  # drain in-flight writes first
+        return torch.load(path, map_location=map_location,
+                          weights_only=False)
+
+    def close(self):
+        if self._proc.is_alive():
+            self._queue.put(None)
+            self._proc.join(10)
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
 def make_checkpoint_engine(name="torch", config_params=None):
     if name in (None, "torch"):
         return TorchCheckpointEngine(config_params)
     if name == "fast":
         return FastCheckpointEngine(config_params)
+    if name == "decoupled":
+        return DecoupledCheckpointEngine(config_params)
     raise ValueError(f"unknown checkpoint engine {name} "
                      "(nebula/datastates are external services)")
 
