@@ -275,7 +275,7 @@ def inference_all_reduce(tensor, op=ReduceOp.SUM, group=None):
 
 
 def log_summary(show_straggler=False):
-    comms_logger.log_all()
+    comms_logger.log_all(show_straggler=show_straggler)
 
 
 def destroy_process_group():

@@ -58,7 +58,26 @@ class CommsLogger:
             log_dist(f"comm op: {op_name} | size {size_bytes} B | "
                      f"{latency_s*1000:.3f} ms", ranks=[0])
 
-    def log_all(self):
+    def log_all(self, show_straggler=False):
+        """Summary table; with show_straggler, also reports per-op
+        straggler delay = avg(rank-max latency) - avg(own latency)
+        across the world (ref comms_logging get_bw + log_summary
+        straggler effect)."""
+        if show_straggler:
+            import torch
+            import torch.distributed as dist
+            if dist.is_initialized() and dist.get_world_size() > 1:
+                log_dist("straggler effect (max-over-ranks minus own "
+                         "mean latency):", ranks=[0])
+                for op_name, sizes in sorted(self.comms_dict.items()):
+                    own = [sum(l) / max(len(l), 1)
+                           for _, (c, l, b) in sorted(sizes.items())]
+                    t = torch.tensor(own, dtype=torch.float64)
+                    mx = t.clone()
+                    dist.all_reduce(mx, op=dist.ReduceOp.MAX)
+                    delay = (mx - t).mean().item() * 1000 if len(own) \
+                        else 0.0
+                    log_dist(f"  {op_name}: {delay:.3f} ms", ranks=[0])
         for op_name, sizes in sorted(self.comms_dict.items()):
             log_dist(f"Op: {op_name}", ranks=[0])
             log_dist(f"{'size(B)':>14} {'count':>8} {'avg lat(ms)':>12} "
