@@ -691,8 +691,86 @@ class DeepSpeedEngine(torch.nn.Module):
     def get_lr(self):
         return [g["lr"] for g in self.optimizer.param_groups]
 
+    def set_lr(self, lr):
+        for g in self.optimizer.param_groups:
+            g["lr"] = lr
+
+    def get_mom(self):
+        """Momentum/betas per group (ref engine.get_mom)."""
+        key = "betas" if any("betas" in g for g in
+                             self.optimizer.param_groups) else "momentum"
+        return [g.get(key) for g in self.optimizer.param_groups]
+
     def get_global_grad_norm(self):
         return getattr(self.optimizer, "get_global_grad_norm", lambda: 0.0)()
+
+    def get_batch_info(self):
+        """(train_batch_size, micro_batch, gradient_accumulation_steps)."""
+        return (self.train_batch_size(),
+                self.train_micro_batch_size_per_gpu(),
+                self.gradient_accumulation_steps())
+
+    def set_train_batch_size(self, train_batch_size):
+        """Adjust the global batch by changing GAS (DP world and micro
+        batch stay fixed — ref engine.set_train_batch_size)."""
+        mb = self.train_micro_batch_size_per_gpu()
+        denom = mb * self.dp_world_size
+        if train_batch_size % denom != 0:
+            raise ValueError(
+                f"train_batch_size {train_batch_size} not divisible by "
+                f"micro_batch*dp_world {denom}")
+        self._config.gradient_accumulation_steps = train_batch_size // denom
+        self._config.train_batch_size = train_batch_size
+
+    def set_train_micro_batch_size(self, micro_batch_size):
+        self._config.train_micro_batch_size_per_gpu = micro_batch_size
+        self._config.train_batch_size = micro_batch_size * \
+            self.dp_world_size * self.gradient_accumulation_steps()
+
+    @property
+    def loss_scale(self):
+        return getattr(self.optimizer, "loss_scale", 1.0)
+
+    def was_step_applied(self):
+        """False when the last step was skipped by overflow backoff."""
+        return not getattr(self.optimizer, "overflow", False)
+
+    def no_sync(self):
+        """Context manager suppressing the ZeRO-0 boundary allreduce for
+        local accumulation (ref engine.no_sync). ZeRO>=1 owns its
+        reduction schedule and rejects it."""
+        assert self.zero_optimization_stage() == 0, \
+            "no_sync is only meaningful for the ZeRO-0 fallback path"
+        import contextlib
+
+        @contextlib.contextmanager
+        def ctx():
+            prev = self._is_gradient_accumulation_boundary
+            self.set_gradient_accumulation_boundary(False)
+            try:
+                yield
+            finally:
+                self._is_gradient_accumulation_boundary = prev
+        return ctx()
+
+    def get_sequence_parallel_group(self):
+        return groups.get_sequence_parallel_group()
+
+    def get_model_parallel_rank(self):
+        try:
+            return dist.get_rank(groups.get_tensor_parallel_group())
+        except Exception:
+            return 0
+
+    get_tensor_parallel_rank = get_model_parallel_rank
+
+    def sparse_allreduce(self, sparse, dp_group=None):
+        from .sparse_tensor import sparse_allreduce as _sa
+        return _sa(sparse, dp_group or self.dp_group)
+
+    def sparse_allreduce_bucket(self, bucket, dp_group=None):
+        from .sparse_tensor import sparse_allreduce_bucket as _sab
+        return _sab(bucket, dp_group or self.dp_group)
 
     def destroy(self):
         if hasattr(self.optimizer, "destroy"):
