@@ -349,6 +349,7 @@ class DeepSpeedEngine(torch.nn.Module):
                 dynamic_loss_scale=self._dynamic_loss_scale(),
                 dynamic_loss_args=self._dynamic_loss_args(),
                 dtype=dtype,
+                zero_quantized_gradients=zc.zero_quantized_gradients,
                 gradient_accumulation_steps=self.gradient_accumulation_steps())
         else:
             self.optimizer = basic

@@ -130,6 +130,7 @@ class ZeroStage12Optimizer:
                  postscale_gradients=True,
                  gradient_accumulation_steps=1,
                  ignore_unused_parameters=True,
+                 zero_quantized_gradients=False,
                  mpu=None):
         self.optimizer = init_optimizer
         self.engine = engine
@@ -144,6 +145,8 @@ class ZeroStage12Optimizer:
         self.gradient_accumulation_steps = gradient_accumulation_steps
         self.micro_step = 0
         self.ignore_unused_parameters = ignore_unused_parameters
+        # ZeRO++ qgZ: int8 blockwise grads on the wire (half the RS bytes)
+        self.quantized_gradients = bool(zero_quantized_gradients)
 
         self.device = (torch.device("cuda", torch.cuda.current_device())
                        if torch.cuda.is_available() else torch.device("cpu"))
@@ -302,9 +305,20 @@ class ZeroStage12Optimizer:
         else:
             ctx = _nullctx()
         with ctx:
+            pg = b.pg if b.pg is not None else self.dp_group
+            if self.quantized_gradients and b.world > 1:
+                shard = self._quant_reduce(b, pg)
+                if shard.is_cuda and shard.dtype == torch.bfloat16:
+                    from ...ops.loader import get_ext
+                    get_ext(required=True).accum_bf16_to_f32(
+                        b.grad32, shard, 1.0)
+                else:
+                    b.grad32.add_(shard.float())
+                if stream is not None:
+                    shard.record_stream(stream)
+                return
             shard = torch.empty(b.shard_numel, dtype=self.dtype,
                                 device=b.grad16.device)
-            pg = b.pg if b.pg is not None else self.dp_group
             op = _avg_op(b.world, b.grad16)
             dist.reduce_scatter_tensor(shard, b.grad16, op=op,
                                        group=pg)
@@ -315,6 +329,19 @@ class ZeroStage12Optimizer:
                 b.grad32.add_(shard.float())
             if stream is not None:
                 shard.record_stream(stream)
+
+    def _quant_reduce(self, b, pg):
+        """qgZ bucket reduction: int8 blockwise quantize -> all-to-all ->
+        dequant + sum + average (ref coalesced qgZ; single-node 1-hop)."""
+        from .stage3_params import dequantize_gathered, quantize_shard
+        q, sc = quantize_shard(b.grad16)
+        qr = torch.empty_like(q)
+        sr = torch.empty_like(sc)
+        dist.all_to_all_single(qr, q, group=pg)
+        dist.all_to_all_single(sr, sc, group=pg)
+        deq = dequantize_gathered(qr, sr, b.world, torch.float32)
+        return deq.view(b.world, b.shard_numel).sum(0) \
+            .div_(b.world).to(self.dtype)
 
     def _sync_comm(self):
         if self.comm_stream is not None:

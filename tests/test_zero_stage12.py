@@ -130,3 +130,55 @@ def _gas_equivalence():
 def test_gas_equivalence_single_rank():
     results = run_distributed(_gas_equivalence, world_size=1)
     assert all(results)
+
+
+def _qgz12_body(stage, steps=4):
+    """Stage 1/2 qgZ: int8 bucket reduction tracks exact reduce-scatter."""
+    import torch
+    import torch.distributed as tdist
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    from tests.simple_model import SimpleModel, make_batches
+    rank = tdist.get_rank()
+    world = tdist.get_world_size()
+
+    def run(qgz):
+        groups.reset_groups()
+        torch.manual_seed(11)
+        model = SimpleModel(32)
+        config = {
+            "train_micro_batch_size_per_gpu": 4,
+            "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+            "zero_optimization": {"stage": stage,
+                                  "zero_quantized_gradients": qgz},
+            "bf16": {"enabled": True},
+        }
+        engine, _, _, _ = deepspeed_amd.initialize(model=model,
+                                                   config=config)
+        assert engine.optimizer.quantized_gradients == qgz
+        batches = make_batches(steps * world, 4, 32, seed=5,
+                               dtype=torch.bfloat16)
+        for i in range(steps):
+            x, y = batches[i * world + rank]
+            loss = engine(x, y)
+            engine.backward(loss)
+            engine.step()
+        return [b.master32.detach().cpu().clone()
+                for b in engine.optimizer.buckets]
+
+    s_q = run(True)
+    s_f = run(False)
+    for a, b in zip(s_q, s_f):
+        err = (a - b).abs().max().item()
+        assert err < 2e-2, f"stage{stage} qgZ diverged: {err}"
+    return True
+
+
+def test_zero1_quantized_gradients():
+    from tests.common import run_distributed
+    run_distributed(_qgz12_body, world_size=2, args=(1,))
+
+
+def test_zero2_quantized_gradients():
+    from tests.common import run_distributed
+    run_distributed(_qgz12_body, world_size=2, args=(2,))
