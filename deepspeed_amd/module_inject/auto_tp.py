@@ -104,9 +104,17 @@ def apply_tensor_parallel_hf(model, tp_group=None, plan=None):
             new = LinearAllreduce.from_linear(lin, group, rank, world)
         setattr(parent, child, new)
         n += 1
-    # per-rank head bookkeeping on modules that own sharded linears
-    touched_parents = {name.rpartition(".")[0] for name in plan}
-    for pname in touched_parents:
+    # per-rank head bookkeeping — only on ATTENTION modules (ones whose
+    # sharded children look like q/k/v); dividing size attrs on MLP
+    # modules would corrupt families that read them in forward
+    attn_markers = ("q_proj", "query", "c_attn", "query_key_value")
+    touched_parents = {}
+    for name in plan:
+        pname, _, child = name.rpartition(".")
+        touched_parents.setdefault(pname, []).append(child)
+    for pname, children in touched_parents.items():
+        if not any(any(m in c for m in attn_markers) for c in children):
+            continue
         pm = mods.get(pname)
         if pm is None:
             continue

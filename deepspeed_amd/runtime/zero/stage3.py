@@ -250,6 +250,8 @@ class ZeroStage3Optimizer:
         if self.param_swapper is not None:
             for i, sg in enumerate(self.sub_groups):
                 self.param_swapper.register(i, sg)
+                for p in sg.params:
+                    p.ds_swap = (self.param_swapper, i)
             self._sg_index = {id(sg): i for i, sg in
                               enumerate(self.sub_groups)}
         if self.nvme_swapper is not None:

@@ -174,6 +174,12 @@ def all_gather_params(params, dp_group, async_op=True, stream=None,
     """
     todo = [p for p in params if p.ds_status == ZeroParamStatus.NOT_AVAILABLE]
     for p in todo:
+        # NVMe-evicted shard (param swapper repointed ds_tensor to None):
+        # restore it before any direct gather — covers callers that don't
+        # go through the optimizer (GatheredParameters, hybrid engine)
+        if p.ds_tensor is None and hasattr(p, "ds_swap"):
+            sw, i = p.ds_swap
+            sw.ensure_resident(i)
         p.ds_status = ZeroParamStatus.INFLIGHT
     if not todo:
         return AllGatherHandle([], [], [], dp_group)
