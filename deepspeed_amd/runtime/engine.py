@@ -309,6 +309,7 @@ class DeepSpeedEngine(torch.nn.Module):
                 reduce_bucket_size=zc.reduce_bucket_size,
                 prefetch_bucket_size=zc.prefetch_bucket_size,
                 param_persistence_threshold=zc.param_persistence_threshold,
+                model_persistence_threshold=zc.model_persistence_threshold,
                 max_live_parameters=zc.max_live_parameters,
                 sub_group_size=zc.sub_group_size,
                 overlap_comm=zc.overlap_comm,

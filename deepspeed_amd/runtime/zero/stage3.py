@@ -143,6 +143,7 @@ class ZeroStage3Optimizer:
                  reduce_bucket_size=int(5e8),
                  prefetch_bucket_size=int(5e7),
                  param_persistence_threshold=int(1e5),
+                 model_persistence_threshold=int(1e14),
                  max_live_parameters=int(1e9),
                  sub_group_size=int(1e9),
                  overlap_comm=True,
@@ -173,6 +174,7 @@ class ZeroStage3Optimizer:
         self.reduce_bucket_size = int(reduce_bucket_size)
         self.prefetch_bucket_size = int(prefetch_bucket_size)
         self.persist_threshold = int(param_persistence_threshold)
+        self.model_persistence_threshold = int(model_persistence_threshold)
         self.max_live_parameters = int(max_live_parameters)
         self.sub_group_size = int(sub_group_size)
         self.gradient_accumulation_steps = gradient_accumulation_steps
@@ -331,6 +333,17 @@ class ZeroStage3Optimizer:
         self._all_params = [p for p in self.module.parameters()]
         for i, p in enumerate(self._all_params):
             p.ds_param_index = i
+        # model_persistence_threshold (ref partition_parameters.py:1264):
+        # cap TOTAL persisted elements per partition — stop persisting
+        # once the cumulative budget is spent
+        budget = self.model_persistence_threshold // max(self.world, 1)
+        persisted = 0
+        for p in self._all_params:
+            if p.ds_persist:
+                if persisted + p.ds_numel > budget:
+                    p.ds_persist = False
+                else:
+                    persisted += p.ds_numel
 
     def _build_sub_groups(self):
         """Pack param shards into flat fp32 sub-groups per optimizer group."""

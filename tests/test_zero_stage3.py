@@ -557,3 +557,42 @@ def _max_live_body():
 def test_zero3_max_live_parameters_caps_prefetch():
     from tests.common import run_distributed
     run_distributed(_max_live_body, world_size=1)
+
+
+def _model_persist_body():
+    """model_persistence_threshold caps total persisted elements."""
+    import torch
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    from tests.simple_model import SimpleModel
+
+    def persisted_elems(threshold):
+        groups.reset_groups()
+        torch.manual_seed(0)
+        model = SimpleModel(32)
+        config = {
+            "train_micro_batch_size_per_gpu": 4,
+            "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+            "zero_optimization": {
+                "stage": 3,
+                "stage3_param_persistence_threshold": int(1e9),
+                "stage3_model_persistence_threshold": threshold},
+            "bf16": {"enabled": True},
+        }
+        engine, _, _, _ = deepspeed_amd.initialize(model=model,
+                                                   config=config)
+        n = sum(p.ds_numel for p in engine.optimizer._all_params
+                if p.ds_persist)
+        engine.optimizer.destroy()
+        return n
+
+    unlimited = persisted_elems(int(1e14))
+    capped = persisted_elems(2000)
+    assert unlimited > 2000
+    assert capped <= 2000, capped
+    return True
+
+
+def test_zero3_model_persistence_threshold():
+    from tests.common import run_distributed
+    run_distributed(_model_persist_body, world_size=1)
