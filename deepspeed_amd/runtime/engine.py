@@ -311,6 +311,7 @@ class DeepSpeedEngine(torch.nn.Module):
                 param_persistence_threshold=zc.param_persistence_threshold,
                 model_persistence_threshold=zc.model_persistence_threshold,
                 max_live_parameters=zc.max_live_parameters,
+                max_reuse_distance=zc.max_reuse_distance,
                 sub_group_size=zc.sub_group_size,
                 overlap_comm=zc.overlap_comm,
                 offload_optimizer=zc.offload_optimizer,

@@ -145,6 +145,7 @@ class ZeroStage3Optimizer:
                  param_persistence_threshold=int(1e5),
                  model_persistence_threshold=int(1e14),
                  max_live_parameters=int(1e9),
+                 max_reuse_distance=int(1e9),
                  sub_group_size=int(1e9),
                  overlap_comm=True,
                  offload_optimizer=None,
@@ -176,6 +177,7 @@ class ZeroStage3Optimizer:
         self.persist_threshold = int(param_persistence_threshold)
         self.model_persistence_threshold = int(model_persistence_threshold)
         self.max_live_parameters = int(max_live_parameters)
+        self.max_reuse_distance = int(max_reuse_distance)
         self.sub_group_size = int(sub_group_size)
         self.gradient_accumulation_steps = gradient_accumulation_steps
         self.micro_step = 0
@@ -665,6 +667,22 @@ class ZeroStage3Optimizer:
         self._trace_pos = 0
         if self._trace and not self._trace_complete:
             self._trace_complete = True
+            self._compute_reuse_keep()
+
+    def _compute_reuse_keep(self):
+        """max_reuse_distance (ref zero/config.py:238, coordinator
+        release-by-reuse-distance): a module whose backward re-fetch is
+        closer than the threshold (in parameter elements walked through
+        the tail of forward + head of backward) keeps its params gathered
+        across the turn instead of releasing + re-gathering."""
+        sizes = [sum(p.ds_numel for p in m._ds_direct_params)
+                 for m in self._trace]
+        suffix = 0
+        for i in range(len(self._trace) - 1, -1, -1):
+            # distance fwd-exit(i) -> bwd-entry(i): rest of forward plus
+            # the same modules walked back = 2 * suffix(i+1)
+            self._trace[i]._ds_keep_for_backward =                 2 * suffix < self.max_reuse_distance
+            suffix += sizes[i]
 
     # -- forward path
     def _pre_forward_hook(self, mod, inputs):
@@ -758,9 +776,11 @@ class ZeroStage3Optimizer:
             self._trace_pos = min(max(pos, 0), n)
 
     def release_sub_module(self, mod):
+        keep = getattr(mod, "_ds_keep_for_backward", False)
         for p in mod._ds_direct_params:
             p.ds_active_sub_modules.discard(id(mod))
-            if not p.ds_active_sub_modules and not p.ds_persist:
+            if not keep and not p.ds_active_sub_modules \
+                    and not p.ds_persist:
                 free_param(p)
 
     def _install_grad_hooks(self):
