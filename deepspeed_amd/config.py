@@ -55,6 +55,10 @@ class ZeroConfig(DSConfigModel):
     reduce_bucket_size: int = int(5e8)
     allgather_bucket_size: int = int(5e8)
     reduce_scatter: bool = True
+    # contiguous_gradients / allgather_partitions / round_robin_gradients:
+    # accepted for config compatibility; this design ALWAYS uses flat
+    # contiguous grad slabs, partitioned all-gathers, and balanced flat
+    # buckets, so these switches are inherently on
     overlap_comm: Optional[bool] = None  # default True for stage>=2
     contiguous_gradients: bool = True
     allgather_partitions: bool = True

@@ -352,6 +352,9 @@ class DeepSpeedEngine(torch.nn.Module):
                 dynamic_loss_args=self._dynamic_loss_args(),
                 dtype=dtype,
                 zero_quantized_gradients=zc.zero_quantized_gradients,
+                gradient_predivide_factor=self._config
+                .gradient_predivide_factor,
+                communication_data_type=self._comm_dtype(),
                 gradient_accumulation_steps=self.gradient_accumulation_steps())
         else:
             self.optimizer = basic
@@ -457,6 +460,13 @@ class DeepSpeedEngine(torch.nn.Module):
                               cfg.get("schedule_steps", 1000),
                               layer_ids=cfg.get("layer_ids"))
         self.random_ltd_scheduler = self.module.random_ltd_scheduler
+
+    def _comm_dtype(self):
+        cdt = self._config.communication_data_type
+        if cdt is None:
+            return None
+        return {"fp32": torch.float32, "fp16": torch.float16,
+                "bf16": torch.bfloat16}.get(cdt, None)
 
     def wall_clock_breakdown(self):
         return self._config.wall_clock_breakdown

@@ -128,6 +128,7 @@ class ZeroStage12Optimizer:
                  dtype=torch.bfloat16,
                  gradient_predivide_factor=1.0,
                  postscale_gradients=True,
+                 communication_data_type=None,
                  gradient_accumulation_steps=1,
                  ignore_unused_parameters=True,
                  zero_quantized_gradients=False,
@@ -147,6 +148,14 @@ class ZeroStage12Optimizer:
         self.ignore_unused_parameters = ignore_unused_parameters
         # ZeRO++ qgZ: int8 blockwise grads on the wire (half the RS bytes)
         self.quantized_gradients = bool(zero_quantized_gradients)
+        # large-world fp16 overflow mitigation (ref engine
+        # gradient_predivide_factor): grads pre-divide by F before the
+        # SUM reduce and post-divide by world/F
+        self.gradient_predivide_factor = float(gradient_predivide_factor)
+        self.postscale_gradients = bool(postscale_gradients)
+        # optional on-wire dtype for grad reduction (e.g. fp32 comm for
+        # bf16 grads — ref communication_data_type)
+        self.communication_data_type = communication_data_type
 
         self.device = (torch.device("cuda", torch.cuda.current_device())
                        if torch.cuda.is_available() else torch.device("cpu"))
@@ -317,11 +326,31 @@ class ZeroStage12Optimizer:
                 if stream is not None:
                     shard.record_stream(stream)
                 return
-            shard = torch.empty(b.shard_numel, dtype=self.dtype,
-                                device=b.grad16.device)
-            op = _avg_op(b.world, b.grad16)
-            dist.reduce_scatter_tensor(shard, b.grad16, op=op,
-                                       group=pg)
+            comm_dtype = self.communication_data_type
+            f = self.gradient_predivide_factor
+            if f != 1.0 and self.postscale_gradients and b.world > 1:
+                src = b.grad16 if comm_dtype is None \
+                    else b.grad16.to(comm_dtype)
+                src.div_(f)
+                shard = torch.empty(b.shard_numel, dtype=src.dtype,
+                                    device=src.device)
+                dist.reduce_scatter_tensor(shard, src,
+                                           op=dist.ReduceOp.SUM, group=pg)
+                shard.div_(b.world / f)
+                shard = shard.to(self.dtype)
+            elif comm_dtype is not None and comm_dtype != self.dtype:
+                src = b.grad16.to(comm_dtype)
+                shard = torch.empty(b.shard_numel, dtype=comm_dtype,
+                                    device=src.device)
+                op = _avg_op(b.world, src)
+                dist.reduce_scatter_tensor(shard, src, op=op, group=pg)
+                shard = shard.to(self.dtype)
+            else:
+                shard = torch.empty(b.shard_numel, dtype=self.dtype,
+                                    device=b.grad16.device)
+                op = _avg_op(b.world, b.grad16)
+                dist.reduce_scatter_tensor(shard, b.grad16, op=op,
+                                           group=pg)
             if shard.is_cuda and shard.dtype == torch.bfloat16:
                 from ...ops.loader import get_ext
                 get_ext(required=True).accum_bf16_to_f32(b.grad32, shard, 1.0)

@@ -182,3 +182,51 @@ def test_zero1_quantized_gradients():
 def test_zero2_quantized_gradients():
     from tests.common import run_distributed
     run_distributed(_qgz12_body, world_size=2, args=(2,))
+
+
+def _predivide_body(steps=3):
+    """gradient_predivide_factor + fp32 communication_data_type produce
+    the same training result as the default on-wire averaging."""
+    import torch
+    import torch.distributed as tdist
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    from tests.simple_model import SimpleModel, make_batches
+    rank = tdist.get_rank()
+    world = tdist.get_world_size()
+
+    def run(extra):
+        groups.reset_groups()
+        torch.manual_seed(11)
+        model = SimpleModel(32)
+        config = {
+            "train_micro_batch_size_per_gpu": 4,
+            "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+            "zero_optimization": {"stage": 2},
+            "bf16": {"enabled": True},
+        }
+        config.update(extra)
+        engine, _, _, _ = deepspeed_amd.initialize(model=model,
+                                                   config=config)
+        batches = make_batches(steps * world, 4, 32, seed=5,
+                               dtype=torch.bfloat16)
+        for i in range(steps):
+            x, y = batches[i * world + rank]
+            loss = engine(x, y)
+            engine.backward(loss)
+            engine.step()
+        return [b.master32.detach().cpu().clone()
+                for b in engine.optimizer.buckets]
+
+    base = run({})
+    pre = run({"gradient_predivide_factor": 2.0,
+               "communication_data_type": "fp32"})
+    for a, b in zip(base, pre):
+        err = (a - b).abs().max().item()
+        assert err < 5e-3, f"predivide path diverged: {err}"
+    return True
+
+
+def test_zero2_predivide_and_comm_dtype():
+    from tests.common import run_distributed
+    run_distributed(_predivide_body, world_size=2)
