@@ -74,16 +74,23 @@ def main():
     if torch.cuda.is_available():
         torch.cuda.set_device(device)
 
+    from deepspeed_amd.models.bert import BERT_CONFIGS, BertForPreTraining
+    is_bert = args.model in BERT_CONFIGS
     if args.model in MIXTRAL_CONFIGS:
         cfg = MIXTRAL_CONFIGS[args.model]
         cfg.ep_size = min(args.ep, world) if args.ep > 1 else 1
         model_cls = MixtralForCausalLM
+    elif is_bert:
+        cfg = BERT_CONFIGS[args.model]
+        model_cls = BertForPreTraining
+        args.seq_len = min(args.seq_len, cfg.max_position_embeddings)
     else:
         cfg = LLAMA_CONFIGS[args.model]
         model_cls = LlamaForCausalLM
-    cfg.activation_checkpointing = args.act_ckpt
-    if args.seq_len > cfg.max_position_embeddings:
-        cfg.max_position_embeddings = args.seq_len
+    if not is_bert:
+        cfg.activation_checkpointing = args.act_ckpt
+        if args.seq_len > cfg.max_position_embeddings:
+            cfg.max_position_embeddings = args.seq_len
     torch.manual_seed(1234 + rank)
     t0 = time.time()
     # build directly on device: 8B bf16 = 16 GB, fits trivially in 288 GB
