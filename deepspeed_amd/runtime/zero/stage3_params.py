@@ -149,6 +149,13 @@ class AllGatherHandle:
             cur.wait_stream(self.stream)
             for buf in self.buffers:
                 buf.record_stream(cur)
+            if self.quant is not None:
+                # dequant below reads these on the compute stream; tell
+                # the allocator so comm-stream reuse can't race it
+                for q8, sc, _w in self.quant:
+                    if q8 is not None:
+                        q8.record_stream(cur)
+                        sc.record_stream(cur)
         if self.quant is not None:
             # dequantize gathered int8 shards into the bf16 buffers
             for buf, (q8, sc, world) in zip(self.buffers, self.quant):
