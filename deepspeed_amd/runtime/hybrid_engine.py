@@ -41,8 +41,27 @@ def gathered_for_generation(engine):
 
 @torch.no_grad()
 def generate(engine, input_ids, max_new_tokens=32, temperature=0.0,
-             top_k=0, eos_token_id=None):
-    """Generate with the (possibly ZeRO-3-sharded) training weights."""
+             top_k=0, eos_token_id=None,
+             offload_states_during_generate=False):
+    """Generate with the (possibly ZeRO-3-sharded) training weights.
+
+    offload_states_during_generate: push optimizer masters/moments/grad
+    accumulators to host for the generation phase (engine.offload_states)
+    so long-rollout KV caches get the HBM — reloaded before returning."""
+    if offload_states_during_generate:
+        engine.offload_states(include=("hp_params", "lp_grads",
+                                       "optim_states"))
+    try:
+        return _generate_inner(engine, input_ids, max_new_tokens,
+                               temperature, top_k, eos_token_id)
+    finally:
+        if offload_states_during_generate:
+            engine.reload_states()
+
+
+@torch.no_grad()
+def _generate_inner(engine, input_ids, max_new_tokens, temperature,
+                    top_k, eos_token_id):
     module = engine.module
     cfg = getattr(module, "cfg", None)
     assert cfg is not None, "model must expose .cfg"
