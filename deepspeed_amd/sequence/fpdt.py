@@ -100,3 +100,30 @@ def fpdt_attention(q, k, v, chunk_size=4096, causal=True,
             out, lse = update_out_and_lse(out, lse, po, plse)
         outs.append(out.to(q.dtype))
     return torch.cat(outs, dim=1)
+
+
+# ---- chunked FFN + logits-loss (ref fpdt_layer.py FPDT_FFN:1127,
+# FPDT_LogitsLoss:1208) — reuse the tiled-compute machinery: shards are
+# checkpointed segments of one autograd graph (recompute-in-backward).
+def fpdt_ffn(ffn_module, x, chunk_size=4096):
+    """Run an FFN over the sequence in chunks (activation memory bounded
+    by one chunk)."""
+    import math
+
+    from .tiled import sequence_tiled_compute
+    shards = max(1, math.ceil(x.shape[1] / chunk_size))
+    return sequence_tiled_compute(ffn_module, x, shards, dim=1,
+                                  compute_params=True)
+
+
+def fpdt_logits_loss(hidden, lm_weight, labels, chunk_size=4096,
+                     shift_labels=True, ignore_index=-100):
+    """CE loss over chunked logits — the [B, S, V] tensor never
+    materializes at once."""
+    import math
+
+    from .tiled import tiled_logits_loss
+    shards = max(1, math.ceil(hidden.shape[1] / chunk_size))
+    return tiled_logits_loss(hidden, lm_weight, labels,
+                             num_shards=shards, shift_labels=shift_labels,
+                             ignore_index=ignore_index)

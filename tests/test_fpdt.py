@@ -58,3 +58,36 @@ def test_fpdt_offload_matches():
     ref = _full_attention(q, k, v)
     got = fpdt_attention(q, k, v, chunk_size=16, offload_to_host=True)
     assert torch.allclose(got, ref, atol=1e-4)
+
+
+def test_fpdt_ffn_and_logits_loss_match_dense():
+    import math
+    import torch
+    from deepspeed_amd.sequence.fpdt import fpdt_ffn, fpdt_logits_loss
+    torch.manual_seed(0)
+    ffn = torch.nn.Sequential(torch.nn.Linear(32, 64), torch.nn.GELU(),
+                              torch.nn.Linear(64, 32))
+    x = torch.randn(2, 64, 32, requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+    y_ref = ffn(x)
+    y = fpdt_ffn(ffn, x2, chunk_size=16)
+    assert torch.allclose(y, y_ref, atol=1e-6)
+    g = torch.randn_like(y)
+    y_ref.backward(g)
+    y.backward(g)
+    assert torch.allclose(x.grad, x2.grad, atol=1e-6)
+
+    V = 128
+    w = torch.randn(V, 32) * 0.05
+    h = torch.randn(2, 64, 32, requires_grad=True)
+    h2 = h.detach().clone().requires_grad_(True)
+    labels = torch.randint(0, V, (2, 64))
+    ref = torch.nn.functional.cross_entropy(
+        (h[:, :-1] @ w.t()).reshape(-1, V).float(),
+        labels[:, 1:].reshape(-1))
+    got = fpdt_logits_loss(h2, w, labels, chunk_size=16)
+    assert abs(ref.item() - got.item()) < 1e-4, (ref.item(), got.item())
+    ref.backward()
+    got.backward()
+    assert torch.allclose(h.grad, h2.grad, atol=1e-5), \
+        (h.grad - h2.grad).abs().max()
