@@ -78,6 +78,8 @@ class ZeroConfig(DSConfigModel):
     zero_quantized_weights: bool = False
     # ZeRO++ qgZ: int8 all-to-all gradient reduction (half the RS bytes)
     zero_quantized_gradients: bool = False
+    # frozen params kept as int8 blockwise residency copies
+    zero_quantized_nontrainable_weights: bool = False
     # class names whose whole subtree gathers as one unit (MoE experts)
     leaf_module: dict = {}
     mics_shard_size: int = -1

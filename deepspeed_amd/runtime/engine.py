@@ -318,6 +318,8 @@ class DeepSpeedEngine(torch.nn.Module):
                 offload_param=zc.offload_param,
                 zero_quantized_weights=zc.zero_quantized_weights,
                 zero_quantized_gradients=zc.zero_quantized_gradients,
+                zero_quantized_nontrainable_weights=
+                zc.zero_quantized_nontrainable_weights,
                 leaf_module_names=(zc.leaf_module or {}).get("classes"),
                 clip_grad=self.gradient_clipping(),
                 static_loss_scale=self._static_loss_scale(),

@@ -152,6 +152,7 @@ class ZeroStage3Optimizer:
                  offload_param=None,
                  zero_quantized_weights=False,
                  zero_quantized_gradients=False,
+                 zero_quantized_nontrainable_weights=False,
                  leaf_module_names=None,
                  clip_grad=0.0,
                  static_loss_scale=1.0,
@@ -186,6 +187,8 @@ class ZeroStage3Optimizer:
         # gathers + fp16 scales — halves all-gather bytes over xGMI
         self.quantized_weights = bool(zero_quantized_weights)
         self.quantized_gradients = bool(zero_quantized_gradients)
+        self.quantized_nontrainable = bool(
+            zero_quantized_nontrainable_weights)
         self.leaf_module_names = list(leaf_module_names or [])
 
         self.device = (torch.device("cuda", torch.cuda.current_device())
@@ -251,6 +254,19 @@ class ZeroStage3Optimizer:
         if self._expert_pg and self.replica_world > 1:
             raise NotImplementedError("MiCS + expert parallelism")
         self._shard_module_params()
+        if self.quantized_nontrainable:
+            # frozen params (LoRA bases etc.) keep only an int8 blockwise
+            # residency copy — halves their memory (ref
+            # zero_quantized_nontrainable_weights)
+            from .stage3_params import quantize_frozen_param
+            nq = 0
+            for p in self._all_params:
+                if not p.requires_grad and not p.ds_persist \
+                        and p.ds_tensor is not None:
+                    quantize_frozen_param(p)
+                    nq += 1
+            log_dist(f"ZeRO-3: {nq} frozen params int8-quantized",
+                     ranks=[0])
         self._build_sub_groups()
         if self.param_swapper is not None:
             for i, sg in enumerate(self.sub_groups):

@@ -194,7 +194,9 @@ def all_gather_params(params, dp_group, async_op=True, stream=None,
     buffers = []
     works = []
     if world == 1:
-        if torch.cuda.is_available() and not todo[0].ds_tensor.is_cuda:
+        host_shards = any(p.ds_tensor is None or not p.ds_tensor.is_cuda
+                          for p in todo)
+        if torch.cuda.is_available() and host_shards:
             # param offload: shards live in pinned host memory — the
             # "gather" at world 1 is one async H2D copy per param on the
             # side stream (SDMA blit engines, no CUs; ref
@@ -207,16 +209,13 @@ def all_gather_params(params, dp_group, async_op=True, stream=None,
                 ctx = contextlib.nullcontext()
             with ctx:
                 for p in todo:
-                    buf = torch.empty(p.ds_shard_numel,
-                                      dtype=p.ds_tensor.dtype, device="cuda")
-                    buf.copy_(p.ds_tensor, non_blocking=True)
-                    buffers.append(buf)
+                    buffers.append(_shard_on_device(p))
             h = AllGatherHandle(todo, [], buffers, dp_group, stream=stream)
             if stream is None:
                 h.wait()
             return h
         for p in todo:
-            buffers.append(p.ds_tensor)
+            buffers.append(_shard_source(p))
         h = AllGatherHandle(todo, [], buffers, dp_group)
         h.wait()
         return h
@@ -229,16 +228,34 @@ def all_gather_params(params, dp_group, async_op=True, stream=None,
     return _launch_gathers(todo, dp_group, world, async_op, None, quantized)
 
 
+def _shard_source(p):
+    """The 16-bit shard to gather: ds_tensor, or a transient dequant of
+    the int8 residency copy (zero_quantized_nontrainable_weights — frozen
+    params live as blockwise int8 + fp16 scales, halving their memory)."""
+    if p.ds_tensor is not None:
+        return p.ds_tensor
+    q8, sc = p.ds_quant
+    return dequantize_gathered(q8, sc, 1, p.ds_dtype)
+
+
 def _shard_on_device(p):
     """RCCL gather input must be a device tensor; offloaded shards are
     staged through one async pinned-H2D copy (enqueued on the caller's
     active stream, so it pipelines ahead of the collective)."""
-    if p.ds_tensor.is_cuda or not torch.cuda.is_available():
-        return p.ds_tensor
-    dev = torch.empty(p.ds_shard_numel, dtype=p.ds_tensor.dtype,
+    src = _shard_source(p)
+    if src.is_cuda or not torch.cuda.is_available():
+        return src
+    dev = torch.empty(p.ds_shard_numel, dtype=src.dtype,
                       device="cuda")
-    dev.copy_(p.ds_tensor, non_blocking=True)
+    dev.copy_(src, non_blocking=True)
     return dev
+
+
+def quantize_frozen_param(p):
+    """Replace a frozen param's 16-bit shard with int8 + scales."""
+    q8, sc = quantize_shard(p.ds_tensor)
+    p.ds_quant = (q8, sc)
+    p.ds_tensor = None
 
 
 def _launch_gathers(todo, dp_group, world, async_op, stream,

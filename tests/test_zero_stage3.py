@@ -643,3 +643,71 @@ def _reuse_distance_body():
 def test_zero3_max_reuse_distance():
     from tests.common import run_distributed
     run_distributed(_reuse_distance_body, world_size=1)
+
+
+def _frozen_quant_body(steps=3):
+    """zero_quantized_nontrainable_weights: frozen params live as int8,
+    LoRA-style training still converges and matches within quant error."""
+    import torch
+    import torch.distributed as tdist
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+
+    class LoraNet(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            torch.manual_seed(3)
+            self.base = torch.nn.Linear(32, 32, bias=False)
+            self.base.weight.requires_grad_(False)
+            self.lora_a = torch.nn.Linear(32, 4, bias=False)
+            self.lora_b = torch.nn.Linear(4, 32, bias=False)
+            torch.nn.init.zeros_(self.lora_b.weight)
+
+        def forward(self, x, y):
+            h = self.base(x) + self.lora_b(self.lora_a(x))
+            return torch.nn.functional.mse_loss(h, y)
+
+    rank = tdist.get_rank()
+    world = tdist.get_world_size()
+
+    def run(flag):
+        groups.reset_groups()
+        torch.manual_seed(11)
+        model = LoraNet()
+        config = {
+            "train_micro_batch_size_per_gpu": 4,
+            "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+            "zero_optimization": {
+                "stage": 3,
+                "stage3_param_persistence_threshold": 0,
+                "zero_quantized_nontrainable_weights": flag},
+            "bf16": {"enabled": True},
+        }
+        engine, _, _, _ = deepspeed_amd.initialize(model=model,
+                                                   config=config)
+        if flag:
+            frozen = [p for p in engine.optimizer._all_params
+                      if not p.requires_grad]
+            assert frozen and all(p.ds_tensor is None and
+                                  hasattr(p, "ds_quant") for p in frozen)
+        g = torch.Generator().manual_seed(5 + rank)
+        losses = []
+        for _ in range(steps):
+            x = torch.randn(4, 32, generator=g).bfloat16()
+            y = torch.randn(4, 32, generator=g).bfloat16()
+            loss = engine(x, y)
+            engine.backward(loss)
+            engine.step()
+            losses.append(loss.item())
+        return losses
+
+    l_q = run(True)
+    l_f = run(False)
+    for a, b in zip(l_q, l_f):
+        assert abs(a - b) < 0.1, (l_q, l_f)
+    return True
+
+
+def test_zero3_quantized_nontrainable_weights():
+    from tests.common import run_distributed
+    run_distributed(_frozen_quant_body, world_size=2)
