@@ -230,6 +230,9 @@ class DeepSpeedConfig:
             **config.get("torch_autocast", {}))
         self.pipeline = PipelineConfig(**config.get("pipeline", {}))
         self.moe = MoEConfig(**config.get("moe", {}))
+        # data_types.grad_accum_dtype: accepted for compatibility; the
+        # ZeRO paths here ALWAYS accumulate gradients into fp32 slabs
+        # (grad32), i.e. the safest reference setting is inherent
         self.data_types = config.get("data_types", {})
         # curriculum learning: legacy top-level key or
         # data_efficiency.data_sampling.curriculum_learning (reference
@@ -248,6 +251,13 @@ class DeepSpeedConfig:
         self.eigenvalue = ev if ev.get("enabled") else None
         self.wandb = config.get("wandb", {})
         self.comet = config.get("comet", {})
+        # reference "amp" (apex O1/O2) maps onto torch autocast here —
+        # same mixed-precision semantics without the apex dependency
+        amp = config.get("amp", {})
+        if amp.get("enabled") and not config.get("torch_autocast"):
+            self.torch_autocast = TorchAutocastConfig(
+                enabled=True,
+                dtype=amp.get("dtype", "float16"))
         self.compression_training = config.get("compression_training")
 
         if self.fp16.enabled and self.bf16.enabled:
