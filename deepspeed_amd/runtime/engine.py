@@ -621,6 +621,12 @@ class DeepSpeedEngine(torch.nn.Module):
     def load_checkpoint(self, load_dir, tag=None, load_module_strict=True,
                         load_optimizer_states=True, load_lr_scheduler_states=True,
                         load_module_only=False):
+        if getattr(self._config, "load_universal_checkpoint", False):
+            # checkpoint.load_universal: true -> load_dir points at a
+            # ds_to_universal output (per-param fp32 fragments), loadable
+            # into ANY topology (ref universal_checkpoint.py:149)
+            self.load_universal_checkpoint(load_dir)
+            return load_dir, {}
         from .checkpointing import load_checkpoint as _load
         return _load(self, load_dir, tag=tag,
                      load_module_strict=load_module_strict,
