@@ -641,10 +641,21 @@ class DeepSpeedEngine(torch.nn.Module):
         backend "hipgraph": capture the module's fwd+bwd as hipGraphs
         (see deepspeed_amd/compile.py — ZeRO comm stays eager/overlapped).
         """
+        if backend == "inductor":
+            # plain torch.compile of the module; composes with the ZeRO
+            # grad hooks (tested: test_torch_compile_with_zero2)
+            self.module = torch.compile(self.module)
+            return self
+        if backend == "autosp":
+            from ..sequence.auto_sp import autosp_backend
+            self.module = torch.compile(self.module,
+                                        backend=autosp_backend)
+            return self
         if backend != "hipgraph":
             raise ValueError(
-                "only the 'hipgraph' backend exists on this stack "
-                "(no Triton/inductor codegen in ROCm torch here)")
+                f"unknown compile backend {backend!r}: choices are "
+                "'hipgraph' (fwd+bwd graph capture), 'inductor' "
+                "(torch.compile), 'autosp' (SDPA->Ulysses graph pass)")
         if sample_input is None:
             raise ValueError("compile() needs a static-shape sample_input")
         from ..compile import engine_compile
