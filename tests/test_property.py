@@ -1,0 +1,53 @@
+"""Property-based invariants (hypothesis) for quantization + packing."""
+import torch
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+
+@settings(max_examples=30, deadline=None)
+@given(nblocks=st.integers(1, 32), scale=st.floats(1e-3, 1e3),
+       seed=st.integers(0, 10000))
+def test_quantize_shard_roundtrip_bound(nblocks, scale, seed):
+    from deepspeed_amd.runtime.zero.stage3_params import (
+        QUANT_BLOCK, dequantize_gathered, quantize_shard)
+    g = torch.Generator().manual_seed(seed)
+    x = torch.randn(nblocks * QUANT_BLOCK, generator=g) * scale
+    q, sc = quantize_shard(x)
+    rec = dequantize_gathered(q, sc, 1, torch.float32)
+    blocks = x.view(-1, QUANT_BLOCK)
+    bound = blocks.abs().amax(1, keepdim=True) / 127.0 * 1.01 + 1e-6
+    err = (rec.view(-1, QUANT_BLOCK) - blocks).abs()
+    assert (err <= bound).all(), float(err.max())
+
+
+@settings(max_examples=30, deadline=None)
+@given(lens=st.lists(st.integers(1, 300), min_size=1, max_size=40),
+       budget=st.integers(50, 600),
+       order=st.sampled_from(["dataloader", "random", "seqlen"]))
+def test_packing_invariants(lens, budget, order):
+    from deepspeed_amd.runtime.variable_batch import batch_by_seqlens
+    mbs, sizes, maxlens = batch_by_seqlens(lens, budget,
+                                           sequence_picking_order=order,
+                                           seed=0)
+    eligible = [i for i, n in enumerate(lens) if n <= budget]
+    packed = [i for _, ids in mbs for i in ids]
+    # every eligible sample exactly once, none oversized
+    assert sorted(packed) == sorted(eligible)
+    for (_, ids), mx, sz in zip(mbs, maxlens, sizes):
+        assert sum(lens[i] for i in ids) <= budget
+        assert mx == max(lens[i] for i in ids)
+        assert sz == len(ids)
+
+
+@settings(max_examples=20, deadline=None)
+@given(n=st.integers(1, 2048), seed=st.integers(0, 1000))
+def test_fp8_torch_quant_bound(n, seed):
+    from deepspeed_amd.ops.fp8_linear import E4M3_MAX, _quant
+    if not hasattr(torch, "float8_e4m3fn"):
+        return
+    g = torch.Generator().manual_seed(seed)
+    x = (torch.randn(8 * n, generator=g) * 3).bfloat16()
+    q, s = _quant(x, torch.float8_e4m3fn, E4M3_MAX)
+    rec = q.float() * s
+    assert (rec - x.float()).abs().max() <= \
+        x.float().abs().max() * 2 ** -3 + 1e-3
