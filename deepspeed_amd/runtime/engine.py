@@ -300,8 +300,15 @@ class DeepSpeedEngine(torch.nn.Module):
                     zc.mics_shard_size)
                 log_dist(f"MiCS: shard_size={zc.mics_shard_size}, "
                          "grads average across replica groups", ranks=[0])
-            self.optimizer = ZeroStage3Optimizer(
+            s3_cls = ZeroStage3Optimizer
+            s3_extra = {}
+            if zc.zenflow is not None:
+                from .zenflow import ZenFlowZeroStage3Optimizer
+                s3_cls = ZenFlowZeroStage3Optimizer
+                s3_extra["zenflow_config"] = zc.zenflow
+            self.optimizer = s3_cls(
                 basic,
+                **s3_extra,
                 module=self.module,
                 engine=self,
                 dp_process_group=shard_group,
