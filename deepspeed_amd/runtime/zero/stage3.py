@@ -196,6 +196,12 @@ class ZeroStage3Optimizer:
         self.offload_device = getattr(offload_optimizer, "device", "none") \
             if offload_optimizer is not None else "none"
         self.offload_optimizer = self.offload_device in ("cpu", "nvme")
+        # partial offload (ref offload_optimizer.ratio / zero_partial_
+        # offload): keep (1-ratio) of the optimizer-state elements on the
+        # GPU with a fused device Adam, offload the rest
+        self.offload_ratio = float(getattr(offload_optimizer, "ratio",
+                                           1.0)) \
+            if offload_optimizer is not None else 1.0
         # ZeRO-Infinity parameter tier (ref partitioned_param_swapper.py:37):
         # cpu => 16-bit shard slabs live in pinned host RAM; nvme => slabs
         # additionally spill to O_DIRECT files with an LRU host budget
@@ -378,7 +384,7 @@ class ZeroStage3Optimizer:
                 if numel >= self.sub_group_size and cur:
                     self.sub_groups.append(
                         SubGroup(cur, offsets, numel, gi, self.device,
-                                 offload=self.offload_optimizer,
+                                 offload=self._sg_offload(numel),
                                  param_offload=self.param_offload,
                                  pin_opt=self._pin_opt,
                                  pin_param=self._pin_param))
@@ -389,7 +395,7 @@ class ZeroStage3Optimizer:
             if cur:
                 self.sub_groups.append(
                     SubGroup(cur, offsets, numel, gi, self.device,
-                             offload=self.offload_optimizer,
+                             offload=self._sg_offload(numel),
                              param_offload=self.param_offload,
                              pin_opt=self._pin_opt,
                              pin_param=self._pin_param))
@@ -400,6 +406,26 @@ class ZeroStage3Optimizer:
         for gi, g in enumerate(self.optimizer.param_groups):
             g["params"] = [sg.master32 for sg in self.sub_groups
                            if sg.group_idx == gi]
+
+    def _sg_offload(self, numel):
+        """Partial offload: greedily offload sub-groups while the
+        offloaded fraction trails `ratio` — converges to ratio of total
+        elements without knowing the total upfront."""
+        if not self.offload_optimizer:
+            return False
+        if self.offload_ratio >= 1.0:
+            return True
+        if self.offload_ratio <= 0.0:
+            return False
+        if not hasattr(self, "_off_elems"):
+            self._off_elems = 0
+            self._tot_elems = 0
+        behind = self._off_elems <= self.offload_ratio * self._tot_elems
+        self._tot_elems += numel
+        if behind:
+            self._off_elems += numel
+            return True
+        return False
 
     def _init_nvme_state(self):
         """Write initial fp32 masters to NVMe; drop resident copies."""
@@ -483,6 +509,55 @@ class ZeroStage3Optimizer:
                 dist.all_reduce(shard, op=op, group=self.replica_group)
             sg.accumulate_grad(p, shard.to(p.grad.dtype))
             p.grad = None
+
+    def _partial_offload_step(self, combined):
+        """Mixed placement: host sub-groups step through the (CPU) base
+        optimizer, device-resident sub-groups through a dedicated fused
+        device Adam (ref zero_partial_offload semantics)."""
+        if not hasattr(self, "_gpu_adam"):
+            from ...ops.adam import FusedAdam
+            g0 = self.optimizer.param_groups[0]
+            dev_masters = [sg.master32 for sg in self.sub_groups
+                           if not sg.offload]
+            self._gpu_adam = FusedAdam(
+                dev_masters, lr=g0["lr"],
+                betas=g0.get("betas", (0.9, 0.999)),
+                eps=g0.get("eps", 1e-8),
+                weight_decay=g0.get("weight_decay", 0.0)) \
+                if dev_masters else None
+        if self._gpu_adam is not None:
+            # follow any LR schedule applied to the base optimizer
+            self._gpu_adam.param_groups[0]["lr"] = \
+                self.optimizer.param_groups[0]["lr"]
+        for which, opt in (("host", self.optimizer),
+                           ("dev", self._gpu_adam)):
+            if opt is None:
+                continue
+            fused = hasattr(opt, "set_grad_scale")
+            sel = [sg for sg in self.sub_groups
+                   if sg.offload == (which == "host")]
+            if not sel:
+                continue
+            if fused:
+                opt.set_grad_scale(1.0 / combined)
+                opt.set_fused_out16(
+                    {sg.master32: (sg.flat16_cpu if sg.offload
+                                   else sg.flat16) for sg in sel})
+            elif combined != 1.0:
+                for sg in sel:
+                    sg.grad32.mul_(1.0 / combined)
+            for sg in sel:
+                sg.master32.grad = sg.grad32
+            opt.step()
+            for sg in sel:
+                sg.master32.grad = None
+                if not fused:
+                    sg.copy_master_to_shards()
+                elif sg.offload:
+                    sg.publish_flat16()
+            if fused:
+                opt.set_grad_scale(1.0)
+                opt.set_fused_out16({})
 
     def _param_offload_step(self, combined, fused):
         """NVMe param tier: per-sub-group step under the host-RAM budget.
@@ -1022,6 +1097,12 @@ class ZeroStage3Optimizer:
             return
 
         fused = hasattr(self.optimizer, "set_grad_scale")
+        if getattr(self, "offload_ratio", 1.0) < 1.0 \
+                and self.offload_optimizer:
+            self._partial_offload_step(combined)
+            self._clear_grads()
+            self._refresh_persistent_params()
+            return
         if self.param_swapper is not None:
             self._param_offload_step(combined, fused)
             self._clear_grads()

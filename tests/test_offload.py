@@ -311,3 +311,57 @@ def _param_offload_ckpt(tmpdir):
 def test_param_offload_checkpoint_roundtrip(tmp_path):
     run_distributed(_param_offload_ckpt, world_size=1,
                     args=(str(tmp_path),))
+
+
+def _partial_offload_body(steps=4):
+    """offload_optimizer.ratio: a fraction of sub-groups keeps GPU-fused
+    state while the rest offloads; trains to the same result."""
+    import torch
+    import torch.distributed as tdist
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    from tests.simple_model import SimpleModel, make_batches
+    rank = tdist.get_rank()
+    world = tdist.get_world_size()
+
+    def run(ratio):
+        groups.reset_groups()
+        torch.manual_seed(11)
+        model = SimpleModel(HIDDEN)
+        config = {
+            "train_micro_batch_size_per_gpu": 4,
+            "optimizer": {"type": "AdamW", "params": {"lr": LR}},
+            "zero_optimization": {
+                "stage": 3, "sub_group_size": 800,
+                "offload_optimizer": {"device": "cpu", "ratio": ratio}},
+            "bf16": {"enabled": True},
+        }
+        engine, _, _, _ = deepspeed_amd.initialize(model=model,
+                                                   config=config)
+        zopt = engine.optimizer
+        if 0 < ratio < 1:
+            kinds = {sg.offload for sg in zopt.sub_groups}
+            assert kinds == {True, False}, \
+                f"expected mixed placement, got {kinds}"
+        batches = make_batches(steps * world, 4, HIDDEN,
+                               dtype=torch.bfloat16)
+        for i in range(steps):
+            x, y = batches[i * world + rank]
+            loss = engine(x, y)
+            engine.backward(loss)
+            engine.step()
+        shards = [sg.master32.detach().cpu().clone()
+                  for sg in zopt.sub_groups]
+        engine.optimizer.destroy()
+        return shards
+
+    s_half = run(0.5)
+    s_full = run(1.0)
+    for a, b in zip(s_half, s_full):
+        err = (a - b).abs().max().item()
+        assert err < 1e-4, f"partial offload diverged: {err}"
+    return True
+
+
+def test_zero3_partial_offload_ratio():
+    run_distributed(_partial_offload_body, world_size=2)
