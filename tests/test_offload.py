@@ -313,6 +313,11 @@ def test_param_offload_checkpoint_roundtrip(tmp_path):
                     args=(str(tmp_path),))
 
 
+def test_param_offload_checkpoint_roundtrip_world2(tmp_path):
+    run_distributed(_param_offload_ckpt, world_size=2,
+                    args=(str(tmp_path),))
+
+
 def _partial_offload_body(steps=4):
     """offload_optimizer.ratio: a fraction of sub-groups keeps GPU-fused
     state while the rest offloads; trains to the same result."""
