@@ -505,6 +505,10 @@ class DeepSpeedEngine(torch.nn.Module):
             self.timers("backward").stop()
         self.micro_steps += 1
         self.global_samples += self.train_micro_batch_size_per_gpu()
+        try:  # remembered for the step-boundary monitor events
+            self.losses = float(loss.detach().float())
+        except Exception:
+            pass
         return loss
 
     def allreduce_gradients(self, bucket_size=MEMORY_OPT_ALLREDUCE_SIZE):
@@ -580,6 +584,21 @@ class DeepSpeedEngine(torch.nn.Module):
         self.global_steps += 1
         if self.global_steps % self.steps_per_print() == 0:
             self._report_progress()
+        if self.monitor.enabled:
+            events = [("Train/lr", self.get_lr()[0] if self.get_lr()
+                       else 0.0, self.global_steps)]
+            if self.losses is not None:
+                events.append(("Train/loss", self.losses,
+                               self.global_steps))
+            scale = getattr(self.optimizer, "loss_scale", None)
+            if scale is not None:
+                events.append(("Train/loss_scale", float(scale),
+                               self.global_steps))
+            norm = self.get_global_grad_norm()
+            if norm:
+                events.append(("Train/grad_norm", float(norm),
+                               self.global_steps))
+            self.monitor.write_events(events)
 
     def _report_progress(self):
         lr = [g["lr"] for g in self.optimizer.param_groups] \

@@ -347,3 +347,35 @@ def test_sd_loader_megatron_merge_split(tmp_path):
 
     data = {"type": "ds_model", "checkpoints": [fm], "version": 1}
     assert SDLoaderFactory.get_sd_loader_json(data) is data
+
+
+def test_monitor_step_events(tmp_path):
+    """Step boundaries write loss/lr/scale events to enabled sinks
+    (ref engine.py:3586 _write_monitor)."""
+    import os
+    import torch
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    from tests.simple_model import SimpleModel, make_batches
+    groups.reset_groups()
+    torch.manual_seed(0)
+    model = SimpleModel(32)
+    config = {
+        "train_micro_batch_size_per_gpu": 4,
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+        "bf16": {"enabled": True},
+        "zero_optimization": {"stage": 1},
+        "csv_monitor": {"enabled": True, "output_path": str(tmp_path),
+                        "job_name": "t"},
+    }
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config=config)
+    x, y = make_batches(1, 4, 32, dtype=torch.bfloat16)[0]
+    for _ in range(3):
+        loss = engine(x, y)
+        engine.backward(loss)
+        engine.step()
+    files = []
+    for root, _, fs in os.walk(tmp_path):
+        files += [f for f in fs if f.endswith(".csv")]
+    assert any("loss" in f.lower() for f in files), files
+    assert any("lr" in f.lower() for f in files), files
