@@ -156,6 +156,82 @@ class DeepSpeedEngine(torch.nn.Module):
     def steps_per_print(self):
         return self._config.steps_per_print
 
+    # --- config getter shims (reference engine API surface) ---
+    def dynamic_loss_scale(self):
+        return self._config.fp16.loss_scale == 0.0
+
+    def initial_dynamic_scale(self):
+        return 2.0 ** self._config.fp16.initial_scale_power
+
+    def dynamic_loss_scale_args(self):
+        f = self._config.fp16
+        return {"init_scale": 2.0 ** f.initial_scale_power,
+                "scale_window": f.loss_scale_window,
+                "min_scale": f.min_loss_scale,
+                "hysteresis": f.hysteresis}
+
+    def zero_reduce_bucket_size(self):
+        return self._config.zero_config.reduce_bucket_size
+
+    def zero_allgather_bucket_size(self):
+        return self._config.zero_config.allgather_bucket_size
+
+    def zero_overlap_comm(self):
+        return self._config.zero_config.overlap_comm
+
+    def zero_prefetch_bucket_size(self):
+        return self._config.zero_config.prefetch_bucket_size
+
+    def zero_param_persistence_threshold(self):
+        return self._config.zero_config.param_persistence_threshold
+
+    def zero_max_live_parameters(self):
+        return self._config.zero_config.max_live_parameters
+
+    def zero_max_reuse_distance(self):
+        return self._config.zero_config.max_reuse_distance
+
+    def zero_sub_group_size(self):
+        return self._config.zero_config.sub_group_size
+
+    def zero_offload_optimizer(self):
+        return self._config.zero_config.offload_optimizer
+
+    def zero_offload_param(self):
+        return self._config.zero_config.offload_param
+
+    def zero_quantized_weights(self):
+        return self._config.zero_config.zero_quantized_weights
+
+    def zero_quantized_gradients(self):
+        return self._config.zero_config.zero_quantized_gradients
+
+    def zero_gather_16bit_weights_on_model_save(self):
+        return self._config.zero_config \
+            .gather_16bit_weights_on_model_save
+
+    def flops_profiler_enabled(self):
+        return self._config.flops_profiler.enabled
+
+    def wall_clock_breakdown_enabled(self):
+        return self._config.wall_clock_breakdown
+
+    def optimizer_name(self):
+        return self._config.optimizer.type \
+            if self._config.optimizer else None
+
+    def optimizer_params(self):
+        return dict(self._config.optimizer.params) \
+            if self._config.optimizer else None
+
+    def scheduler_name(self):
+        return self._config.scheduler.type \
+            if self._config.scheduler else None
+
+    def scheduler_params(self):
+        return dict(self._config.scheduler.params) \
+            if self._config.scheduler else None
+
     @property
     def config(self):
         return self._config
