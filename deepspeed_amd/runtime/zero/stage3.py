@@ -660,7 +660,14 @@ class ZeroStage3Optimizer:
                     master.addcdiv_(m, denom,
                                     value=-group["adamw_lr"] /
                                     (1 - b1 ** t))
-            sg.copy_master_to_shards()
+            if self.param_swapper is not None:
+                i = self._sg_index[id(sg)]
+                self.param_swapper.ensure_resident(i)
+                sg.copy_master_to_shards()
+                self.param_swapper.mark_dirty(i)
+                self.param_swapper.evict_to_budget()
+            else:
+                sg.copy_master_to_shards()
 
     def _param_pg(self, p):
         return getattr(p, "ds_group", None) or self.dp_group
@@ -1099,6 +1106,11 @@ class ZeroStage3Optimizer:
         fused = hasattr(self.optimizer, "set_grad_scale")
         if getattr(self, "offload_ratio", 1.0) < 1.0 \
                 and self.offload_optimizer:
+            if self.param_swapper is not None:
+                raise NotImplementedError(
+                    "offload_optimizer.ratio < 1 with offload_param nvme "
+                    "is unsupported: use ratio 1.0 (full offload) with "
+                    "the NVMe parameter tier")
             self._partial_offload_step(combined)
             self._clear_grads()
             self._refresh_persistent_params()
