@@ -257,6 +257,9 @@ class ZenFlowZeroStage3Optimizer(ZeroStage3Optimizer):
 
     def __init__(self, init_optimizer, zenflow_config=None, **kw):
         super().__init__(init_optimizer, **kw)
+        assert self.param_swapper is None, \
+            "ZenFlow manages its own host state; combine it with " \
+            "offload_param cpu, not the NVMe parameter tier"
         zf = zenflow_config or {}
         self.zf = zf if isinstance(zf, ZenFlowConfig) else ZenFlowConfig(**zf)
         self._zf_step = 0
