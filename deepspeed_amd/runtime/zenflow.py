@@ -417,3 +417,7 @@ class ZenFlowZeroStage3Optimizer(ZeroStage3Optimizer):
                 st["m_cpu"].copy_(m)
                 st["v_cpu"].copy_(v)
                 st["hot_idx"] = None
+
+    def destroy(self):
+        self._drain_threads()
+        super().destroy()
