@@ -711,3 +711,39 @@ def _frozen_quant_body(steps=3):
 def test_zero3_quantized_nontrainable_weights():
     from tests.common import run_distributed
     run_distributed(_frozen_quant_body, world_size=2)
+
+
+def _qwz_fp16_body(steps=4):
+    """qwZ under fp16 dynamic loss scaling (fp16 shard quantization)."""
+    import torch
+    import torch.distributed as tdist
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    from tests.simple_model import SimpleModel, make_batches
+    groups.reset_groups()
+    rank = tdist.get_rank()
+    world = tdist.get_world_size()
+    torch.manual_seed(11)
+    model = SimpleModel(32)
+    config = {
+        "train_micro_batch_size_per_gpu": 4,
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+        "zero_optimization": {"stage": 3, "zero_quantized_weights": True,
+                              "stage3_param_persistence_threshold": 0},
+        "fp16": {"enabled": True, "initial_scale_power": 8},
+    }
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config=config)
+    batches = make_batches(steps * world, 4, 32, seed=5,
+                           dtype=torch.float16)
+    for i in range(steps):
+        x, y = batches[i * world + rank]
+        loss = engine(x, y)
+        engine.backward(loss)
+        engine.step()
+        assert torch.isfinite(torch.tensor(loss.item()))
+    return True
+
+
+def test_zero3_qwz_fp16():
+    from tests.common import run_distributed
+    run_distributed(_qwz_fp16_body, world_size=2)
