@@ -202,3 +202,31 @@ def test_autotuner_full_cost_model_and_pruning(tmp_path):
     assert any("samples_per_sec" in r for r in results)
     import os
     assert len(os.listdir(tmp_path)) >= 1
+
+
+def test_moq_quantizer_schedule():
+    """MoQ bit widths anneal 16->8 per layer period; eigenvalue-sensitive
+    layers anneal slower; params really quantize."""
+    import torch
+    from deepspeed_amd.runtime.quantize import Quantizer
+    q = Quantizer(layer_num=2, start_bits=16, target_bits=8,
+                  quantize_period=5, q_verbose=False)
+    p0 = torch.randn(32, 32)
+    p1 = torch.randn(32, 32)
+    orig0 = p0.clone()
+    for step in range(20):
+        q.quantize([[p0], [p1]])
+    assert q.bits == [8, 8]
+    assert not q.any_precision_switch()
+    assert not torch.equal(p0, orig0)  # fake-quantized in place
+    # distinct values collapse to <= 2^8 levels per row-ish group
+    assert p0.unique().numel() < orig0.unique().numel()
+
+    # eigenvalue modulation: layer 1 twice as sensitive -> longer period
+    q2 = Quantizer(layer_num=2, start_bits=16, target_bits=8,
+                   quantize_period=5, q_eigenvalue=True)
+    a, b = torch.randn(8, 8), torch.randn(8, 8)
+    for step in range(7):
+        q2.quantize([[a], [b]], eigenvalue_enabled=True,
+                    block_eigenvalue={0: 0.1, 1: 10.0})
+    assert q2.bits[0] == 8 and q2.bits[1] == 16, q2.bits
