@@ -351,3 +351,44 @@ def test_index_dispatch_large_ec_trains():
     assert x.grad is not None and x.grad.abs().sum() > 0
     gate_w = moe.deepspeed_moe.gate.wg.weight
     assert gate_w.grad is not None and gate_w.grad.abs().sum() > 0
+
+
+def _tp_mappings_body():
+    """drop_tokens/gather_tokens round-trip across the TP group with
+    correct autograd (ref moe/mappings.py parity)."""
+    import torch
+    import torch.distributed as tdist
+    from deepspeed_amd.comm import groups
+    from deepspeed_amd.moe.mappings import drop_tokens, gather_tokens
+    groups.reset_groups()
+    groups.initialize_tensor_parallel(tdist.get_world_size())
+    rank = groups.get_tensor_parallel_rank()
+    tp = groups.get_tensor_parallel_world_size()
+    torch.manual_seed(7)  # same seed everywhere: replicated activations
+    x = torch.randn(2, 8, 4, requires_grad=True)
+    dropped = drop_tokens(x, dim=1)
+    assert dropped.shape[1] == 8 // tp
+    assert torch.equal(dropped, x.detach().chunk(tp, dim=1)[rank])
+    y = gather_tokens(dropped, dim=1)
+    assert torch.allclose(y, x.detach())
+    w = torch.randn(2, 8, 4)
+    (y * w).sum().backward()
+    # gather bwd slices, drop bwd all-gathers: full grad == w
+    assert torch.allclose(x.grad, w)
+    groups.reset_groups()
+    return True
+
+
+def test_tp_token_mappings_world2():
+    from tests.common import run_distributed
+    run_distributed(_tp_mappings_body, world_size=2)
+
+
+def test_tp_token_mappings_identity_no_tp():
+    """Without TP configured, drop/gather are identities."""
+    from deepspeed_amd.comm import groups
+    from deepspeed_amd.moe.mappings import drop_tokens, gather_tokens
+    groups.reset_groups()
+    import torch
+    x = torch.randn(2, 6, 4)
+    assert drop_tokens(x) is x and gather_tokens(x) is x
