@@ -41,7 +41,7 @@ def gathered_for_generation(engine):
 
 @torch.no_grad()
 def generate(engine, input_ids, max_new_tokens=32, temperature=0.0,
-             top_k=0, eos_token_id=None,
+             top_k=0, top_p=1.0, eos_token_id=None,
              offload_states_during_generate=False):
     """Generate with the (possibly ZeRO-3-sharded) training weights.
 
@@ -53,15 +53,25 @@ def generate(engine, input_ids, max_new_tokens=32, temperature=0.0,
                                        "optim_states"))
     try:
         return _generate_inner(engine, input_ids, max_new_tokens,
-                               temperature, top_k, eos_token_id)
+                               temperature, top_k, top_p, eos_token_id)
     finally:
         if offload_states_during_generate:
             engine.reload_states()
 
 
+def _top_p_filter(logits, top_p):
+    """Mask tokens outside the smallest nucleus with cumulative prob
+    >= top_p (the highest-prob token always survives)."""
+    srt, idx = torch.sort(logits, descending=True, dim=-1)
+    cum = torch.softmax(srt, -1).cumsum(-1)
+    drop_sorted = cum - torch.softmax(srt, -1) >= top_p
+    drop = torch.zeros_like(drop_sorted).scatter(-1, idx, drop_sorted)
+    return logits.masked_fill(drop, float("-inf"))
+
+
 @torch.no_grad()
 def _generate_inner(engine, input_ids, max_new_tokens, temperature,
-                    top_k, eos_token_id):
+                    top_k, top_p, eos_token_id):
     module = engine.module
     cfg = getattr(module, "cfg", None)
     assert cfg is not None, "model must expose .cfg"
@@ -86,6 +96,8 @@ def _generate_inner(engine, input_ids, max_new_tokens, temperature,
                     if top_k > 0:
                         kth = torch.topk(nl, top_k, dim=-1).values[:, -1:]
                         nl = nl.masked_fill(nl < kth, float("-inf"))
+                    if top_p < 1.0:
+                        nl = _top_p_filter(nl, top_p)
                     nxt = torch.multinomial(torch.softmax(nl, -1), 1)
                 else:
                     nxt = nl.argmax(-1, keepdim=True)
