@@ -105,16 +105,18 @@ class AioHandle {
 
   long get_block_size() const { return block_size_; }
 
-  void _rw(at::Tensor buffer, const std::string& fname, bool write) {
+  void _rw(at::Tensor buffer, const std::string& fname, bool write,
+           long file_offset = 0, bool truncate = true) {
     TORCH_CHECK(buffer.is_contiguous() && buffer.device().is_cpu(),
                 "aio buffers must be contiguous host tensors");
     char* data = reinterpret_cast<char*>(buffer.data_ptr());
     size_t nbytes = buffer.numel() * buffer.element_size();
-    bool direct = aligned_for_direct(data, nbytes);
+    bool direct = aligned_for_direct(data, nbytes) &&
+                  (file_offset % kAlign == 0);
     int fd = open_file(fname, write, direct);
     TORCH_CHECK(fd >= 0, "aio: cannot open ", fname);
-    if (write) {
-      int rc = ftruncate(fd, nbytes);
+    if (write && truncate) {
+      int rc = ftruncate(fd, file_offset + nbytes);
       (void)rc;
     }
     size_t nblocks = (nbytes + block_size_ - 1) / block_size_;
@@ -122,14 +124,14 @@ class AioHandle {
     for (size_t b = 0; b < nblocks; ++b) {
       size_t off = b * block_size_;
       size_t len = std::min((size_t)block_size_, nbytes - off);
-      pool_.submit([fd, data, off, len, write, err] {
+      pool_.submit([fd, data, off, len, write, err, file_offset] {
         ssize_t done = 0;
         while (done < (ssize_t)len) {
           ssize_t r = write
                           ? ::pwrite(fd, data + off + done, len - done,
-                                     off + done)
+                                     file_offset + off + done)
                           : ::pread(fd, data + off + done, len - done,
-                                    off + done);
+                                    file_offset + off + done);
           if (r <= 0) {
             err->store(errno ? errno : -1);
             return;
@@ -173,6 +175,28 @@ class AioHandle {
     return n;
   }
 
+  // offset variants (streaming fast-file-writer): write `buffer` at
+  // `file_offset` without truncating — the writer truncates at close.
+  long sync_pwrite_at(at::Tensor buffer, const std::string& fname,
+                      long file_offset) {
+    _rw(buffer, fname, true, file_offset, /*truncate=*/false);
+    return buffer.numel() * buffer.element_size();
+  }
+
+  long sync_pread_at(at::Tensor buffer, const std::string& fname,
+                     long file_offset) {
+    _rw(buffer, fname, false, file_offset, false);
+    return buffer.numel() * buffer.element_size();
+  }
+
+  long async_pwrite_at(at::Tensor buffer, const std::string& fname,
+                       long file_offset) {
+    pending_.emplace_back(std::thread([this, buffer, fname, file_offset] {
+      _rw(buffer, fname, true, file_offset, false);
+    }));
+    return 0;
+  }
+
  private:
   long block_size_;
   ThreadPool pool_;
@@ -194,6 +218,11 @@ void bind_aio(py::module_& m) {
            py::call_guard<py::gil_scoped_release>())
       .def("async_pread", &AioHandle::async_pread)
       .def("async_pwrite", &AioHandle::async_pwrite)
+      .def("sync_pwrite_at", &AioHandle::sync_pwrite_at,
+           py::call_guard<py::gil_scoped_release>())
+      .def("sync_pread_at", &AioHandle::sync_pread_at,
+           py::call_guard<py::gil_scoped_release>())
+      .def("async_pwrite_at", &AioHandle::async_pwrite_at)
       .def("wait", &AioHandle::wait,
            py::call_guard<py::gil_scoped_release>());
 }
