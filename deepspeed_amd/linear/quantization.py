@@ -52,4 +52,5 @@ class QuantizedLinear(torch.nn.Module):
 
     def forward(self, x):
         w = self.weight.dequantized().to(x.dtype)
-        return torch.nn.functional.linear(x, w, self.bias)
+        b = self.bias.to(x.dtype) if self.bias is not None else None
+        return torch.nn.functional.linear(x, w, b)
