@@ -19,6 +19,7 @@ from .runtime.pipe import PipelineModule, LayerSpec  # noqa: F401
 from .moe.layer import MoE  # noqa: F401
 from .ops.adam import FusedAdam  # noqa: F401
 from .utils.logging import logger, log_dist  # noqa: F401
+from .utils.init_on_device import OnDevice  # noqa: F401
 from .runtime.zero import Init as zero_Init  # noqa: F401 (zero.Init)
 from .elasticity import compute_elastic_config  # noqa: F401
 

@@ -379,3 +379,16 @@ def test_monitor_step_events(tmp_path):
         files += [f for f in fs if f.endswith(".csv")]
     assert any("loss" in f.lower() for f in files), files
     assert any("lr" in f.lower() for f in files), files
+
+
+def test_on_device_meta_init():
+    """deepspeed_amd.OnDevice builds shape-only meta modules at a chosen
+    dtype and restores defaults on exit (ref utils/init_on_device.py)."""
+    import torch
+    import deepspeed_amd
+    with deepspeed_amd.OnDevice(dtype=torch.bfloat16, device="meta"):
+        m = torch.nn.Linear(16, 16)
+    assert m.weight.device.type == "meta"
+    assert m.weight.dtype == torch.bfloat16
+    assert torch.get_default_dtype() == torch.float32
+    assert torch.zeros(1).device.type == "cpu"
