@@ -134,14 +134,27 @@ class OPTPolicy(GPT2Policy):
     pass
 
 
+class BloomPolicy(GPT2Policy):
+    # alibi biases ride inside bloom's attention math; leave the
+    # attention fn alone and fuse the layernorms only
+    attention = False
+
+
+class GPTNeoXPolicy(GPT2Policy):
+    pass
+
+
 class AutoPolicy(InjectionPolicy):
     """Shape-detected: injects whatever matches."""
 
 
 _POLICY_BY_ARCH = {
     "llama": LlamaPolicy, "mistral": MistralPolicy, "mixtral": MistralPolicy,
-    "qwen": QwenPolicy, "gpt2": GPT2Policy, "bert": BertPolicy,
-    "opt": OPTPolicy,
+    "qwen": QwenPolicy, "internlm": LlamaPolicy, "gpt2": GPT2Policy,
+    "bert": BertPolicy, "distilbert": BertPolicy, "opt": OPTPolicy,
+    "bloom": BloomPolicy, "gpt_neox": GPTNeoXPolicy,
+    "gptneox": GPTNeoXPolicy, "gptj": GPTNeoXPolicy,
+    "gpt_neo": GPT2Policy, "falcon": GPT2Policy,
 }
 
 

@@ -69,3 +69,42 @@ def test_init_inference_injects_hf():
     ids = torch.randint(0, 512, (1, 8))
     out = eng.generate(ids, max_new_tokens=4)
     assert out.shape[1] >= 12 or out.shape[1] == 12
+
+
+def test_inject_bloom_layernorms_no_attn():
+    """Bloom: alibi attention left alone, layernorms fused."""
+    from transformers import BloomConfig, BloomForCausalLM
+    from deepspeed_amd.module_inject.replace_module import (
+        BloomPolicy, policy_for, replace_transformer_layer)
+    cfg = BloomConfig(hidden_size=64, n_layer=2, n_head=4, vocab_size=512)
+    torch.manual_seed(0)
+    model = BloomForCausalLM(cfg).eval()
+    assert policy_for(model) is BloomPolicy
+    ids = torch.randint(0, 512, (2, 10))
+    with torch.no_grad():
+        ref = model(ids).logits
+    counts = replace_transformer_layer(model)
+    assert counts["attention"] == 0 and counts["layernorm"] >= 2 * 2
+    with torch.no_grad():
+        got = model(ids).logits
+    assert (got - ref).abs().max().item() < 2e-4
+
+
+def test_inject_gpt_neox():
+    from transformers import GPTNeoXConfig, GPTNeoXForCausalLM
+    from deepspeed_amd.module_inject.replace_module import (
+        GPTNeoXPolicy, policy_for, replace_transformer_layer)
+    cfg = GPTNeoXConfig(hidden_size=64, num_hidden_layers=2,
+                        num_attention_heads=4, intermediate_size=128,
+                        vocab_size=512, max_position_embeddings=64)
+    torch.manual_seed(0)
+    model = GPTNeoXForCausalLM(cfg).eval()
+    assert policy_for(model) is GPTNeoXPolicy
+    ids = torch.randint(0, 512, (2, 10))
+    with torch.no_grad():
+        ref = model(ids).logits
+    counts = replace_transformer_layer(model)
+    assert counts["layernorm"] >= 4
+    with torch.no_grad():
+        got = model(ids).logits
+    assert (got - ref).abs().max().item() < 2e-3
