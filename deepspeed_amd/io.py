@@ -140,6 +140,11 @@ class FastFileWriter(BaseFileWriter):
             self._handle.async_pwrite_at(buf, self.file_path,
                                          self._file_off)
             self._draining = True
+            if len(self._bufs) == 1:
+                # single-buffer mode: the same memory is refilled next,
+                # so the drain must complete before write() continues
+                self._handle.wait()
+                self._draining = False
             self._stats["aio_bytes"] += nb
         else:
             os.pwrite(self._fd, self._views[self._fill], self._file_off)

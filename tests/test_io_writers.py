@@ -66,3 +66,21 @@ def test_torch_save_through_fast_writer(tmp_path):
     assert torch.equal(back["w"], obj["w"])
     assert torch.equal(back["nested"]["b"], obj["nested"]["b"])
     assert stats["write_bytes"] == os.path.getsize(p)
+
+
+def test_fast_writer_single_buffer_aio(tmp_path):
+    """double_buffer=False: drain must complete before the (single)
+    buffer is refilled — data integrity across many refills."""
+    from deepspeed_amd.ops.loader import get_ext
+    ext = get_ext(required=False)
+    if ext is None:
+        import pytest
+        pytest.skip("ext not built")
+    h = ext.aio_handle(1 << 16, 4, False, False, 2)
+    p = str(tmp_path / "s.bin")
+    data = os.urandom(50000)
+    w = FastFileWriter(p, handle=h, buffer_bytes=8192, double_buffer=False)
+    for i in range(0, len(data), 1234):
+        w.write(data[i:i + 1234])
+    w.close()
+    assert open(p, "rb").read() == data
