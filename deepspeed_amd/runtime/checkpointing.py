@@ -50,6 +50,36 @@ def _expert_param_names(module):
     return out
 
 
+def _expert_global_name_map(module):
+    """{local_param_name: global_param_name} for every expert param:
+    `deepspeed_experts.<local>` -> `deepspeed_experts.<ep_rank*nl+local>`
+    (ref _get_moe_state_dict's global expert ids). Globally-unique names
+    let offline tools (zero_to_fp32, universal) merge the per-EP-rank
+    expert files without collisions."""
+    from ..comm import groups
+    from ..moe.experts import Experts
+    out = {}
+    for mod_name, mod in module.named_modules():
+        if not isinstance(mod, Experts):
+            continue
+        nl = mod.num_local_experts
+        gn = None
+        for p in mod.parameters():
+            gn = getattr(p, "group_name", None)
+            if gn is not None:
+                break
+        if gn is None:
+            continue
+        _ensure_expert_groups(gn)
+        ep_rank = groups.get_expert_parallel_rank(gn)
+        for i, exp in enumerate(mod.deepspeed_experts):
+            g = ep_rank * nl + i
+            for pn, _ in exp.named_parameters():
+                local = f"{mod_name}.deepspeed_experts.{i}.{pn}"
+                out[local] = f"{mod_name}.deepspeed_experts.{g}.{pn}"
+    return out
+
+
 def save_checkpoint(engine, save_dir, tag=None, client_state=None,
                     save_latest=True, exclude_frozen_parameters=False):
     tag = _tag(engine, tag)
@@ -76,7 +106,9 @@ def save_checkpoint(engine, save_dir, tag=None, client_state=None,
         ep_rank = groups.get_expert_parallel_rank(gn)
         edp_rank = dist.get_rank(groups.get_expert_data_parallel_group(gn))
         if edp_rank == 0:
-            esd = {k: v for k, v in full_sd.items() if k in expert_names}
+            gmap = _expert_global_name_map(engine.module)
+            esd = {gmap.get(k, k): v for k, v in full_sd.items()
+                   if k in expert_names}
             torch.save({"module": esd, "ds_version": VERSION},
                        os.path.join(ckpt_dir, _expert_ckpt_name(ep_rank)))
 
@@ -151,7 +183,12 @@ def load_checkpoint(engine, load_dir, tag=None, load_module_strict=True,
         efile = os.path.join(ckpt_dir, _expert_ckpt_name(ep_rank))
         if os.path.exists(efile):
             esd = torch.load(efile, map_location="cpu", weights_only=False)
-            engine.module.load_state_dict(esd["module"], strict=False)
+            # expert files carry GLOBAL expert ids; map back to this
+            # rank's local names (old local-named files pass through)
+            inv = {g: l for l, g in
+                   _expert_global_name_map(engine.module).items()}
+            sd = {inv.get(k, k): v for k, v in esd["module"].items()}
+            engine.module.load_state_dict(sd, strict=False)
 
     if load_module_only and is_zero:
         # Module-only load under ZeRO still has to reconcile the optimizer's

@@ -392,3 +392,76 @@ def test_tp_token_mappings_identity_no_tp():
     import torch
     x = torch.randn(2, 6, 4)
     assert drop_tokens(x) is x and gather_tokens(x) is x
+
+
+def _moe_expert_global_names():
+    """Expert ckpt files carry GLOBAL expert ids: no cross-EP-rank name
+    collisions, offline fp32 reassembly keeps every expert, round-trip
+    load restores local weights."""
+    import os
+    import re
+    import tempfile
+    import torch
+    import torch.distributed as dist
+    import deepspeed_amd as ds
+    from deepspeed_amd.moe.layer import MoE
+    torch.manual_seed(dist.get_rank())
+    M = 16
+
+    class Net(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.inp = torch.nn.Linear(M, M)
+            self.moe = MoE(M, torch.nn.Linear(M, M), num_experts=4,
+                           ep_size=2, k=1, capacity_factor=8.0)
+
+        def forward(self, x):
+            h, _, _ = self.moe(self.inp(x))
+            return h
+
+    cfg = {"train_micro_batch_size_per_gpu": 2,
+           "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+           "bf16": {"enabled": True},
+           "zero_optimization": {"stage": 2}}
+    engine, _, _, _ = ds.initialize(model=Net(), config=cfg)
+    x = torch.randn(2, M).bfloat16()
+    loss = engine(x).float().pow(2).mean()
+    engine.backward(loss)
+    engine.step()
+    want = {n: p.detach().float().clone()
+            for n, p in engine.module.named_parameters()}
+    tmp = tempfile.mkdtemp()
+    obj = [tmp]
+    dist.broadcast_object_list(obj, src=0)
+    tmp = obj[0]
+    engine.save_checkpoint(tmp, tag="t0")
+    if dist.get_rank() == 0:
+        e0 = torch.load(os.path.join(
+            tmp, "t0", "expert_ep_rank_0_mp_rank_00_model_states.pt"),
+            weights_only=False)
+        e1 = torch.load(os.path.join(
+            tmp, "t0", "expert_ep_rank_1_mp_rank_00_model_states.pt"),
+            weights_only=False)
+        k0, k1 = set(e0["module"]), set(e1["module"])
+        assert k0.isdisjoint(k1), ("collision", k0 & k1)
+        idxs = sorted({int(re.search(r"deepspeed_experts\.(\d+)\.",
+                                     k).group(1)) for k in (k0 | k1)})
+        assert idxs == [0, 1, 2, 3], idxs
+        from deepspeed_amd.utils.zero_to_fp32 import (
+            get_fp32_state_dict_from_zero_checkpoint)
+        sd = get_fp32_state_dict_from_zero_checkpoint(tmp)
+        got = {int(re.search(r"deepspeed_experts\.(\d+)\.", k).group(1))
+               for k in sd if "deepspeed_experts" in k}
+        assert got == {0, 1, 2, 3}, got
+    loss = engine(x).float().pow(2).mean()
+    engine.backward(loss)
+    engine.step()
+    engine.load_checkpoint(tmp, tag="t0")
+    for n, p in engine.module.named_parameters():
+        assert torch.allclose(p.detach().float(), want[n], atol=1e-2), n
+    return True
+
+
+def test_moe_expert_global_names_world2():
+    from tests.common import run_distributed
+    assert all(run_distributed(_moe_expert_global_names, world_size=2))
