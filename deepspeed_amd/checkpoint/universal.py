@@ -40,19 +40,35 @@ def _iter_params_with_states(shards):
         return st.get(key)
 
     if layout["kind"] == "bucket":
+        W = len(shards)
         flats_per_rank = [sd["single_partition_of_fp32_groups"]
                           for sd in shards]
-        for bi, binfo in enumerate(layout["buckets"]):
-            full = {"fp32": torch.cat([f[bi].detach().float()
-                                       for f in flats_per_rank])}
+
+        def assemble(bi, ranks, rbinfo, qual=""):
+            full = {"fp32": torch.cat([flats_per_rank[r][bi].detach()
+                                       .float() for r in ranks])}
             for key in state_keys:
-                parts = [master_state(r, bi, key)
-                         for r in range(len(shards))]
+                parts = [master_state(r, bi, key) for r in ranks]
                 if all(p is not None for p in parts):
                     full[key] = torch.cat([p.float() for p in parts])
-            for name, off, numel, shape in binfo["params"]:
-                yield name, {k: v[off:off + numel].reshape(shape).clone()
-                             for k, v in full.items()}
+            for name, off, numel, shape in rbinfo["params"]:
+                yield name + qual, {k: v[off:off + numel].reshape(shape)
+                                    .clone() for k, v in full.items()}
+
+        for bi, binfo in enumerate(layout["buckets"]):
+            if binfo.get("expert"):
+                # expert buckets partition over the expert-DP group
+                # (stride ep); each EP offset holds DIFFERENT experts
+                # under the SAME module-local names, so entries are
+                # qualified "@ep<off>" (load resolves by rank % ep)
+                edp = binfo.get("world", W)
+                ep = max(W // edp, 1)
+                for off in range(ep):
+                    ranks = [off + k * ep for k in range(edp)]
+                    rbinfo = shards[ranks[0]]["shard_layout"]["buckets"][bi]
+                    yield from assemble(bi, ranks, rbinfo, f"@ep{off}")
+            else:
+                yield from assemble(bi, list(range(W)), binfo)
     else:  # subgroup / stage 3
         flats_per_rank = [sd["fp32_flat_groups"] for sd in shards]
         for gi, ginfo in enumerate(layout["subgroups"]):
@@ -132,15 +148,26 @@ def load_universal_into_optimizer(optimizer, universal_dir):
         rank = optimizer.rank
         for bi, (b, binfo) in enumerate(zip(optimizer.buckets,
                                             layout["buckets"])):
+            shard_rank, qual = rank, ""
+            if binfo.get("expert"):
+                # this bucket shards over the expert-DP group: pick the
+                # "@ep<off>" entries for this rank's EP offset and slice
+                # by its position WITHIN the expert-DP group
+                edp = binfo.get("world", world)
+                ep = max(world // edp, 1)
+                qual = f"@ep{rank % ep}"
+                shard_rank = rank // ep
             full = {"fp32": torch.zeros(binfo["numel_padded"])}
             for key in state_keys:
                 full[key] = torch.zeros(binfo["numel_padded"])
             for name, off, numel, shape in binfo["params"]:
-                t = _load_param(universal_dir, name, state_keys)
+                t = _load_param(universal_dir, name + qual, state_keys)
+                if not t and qual:  # pre-qualifier universal dirs
+                    t = _load_param(universal_dir, name, state_keys)
                 for key in full:
                     if key in t:
                         full[key][off:off + numel] = t[key].reshape(-1)
-            lo = rank * binfo["shard_numel"]
+            lo = shard_rank * binfo["shard_numel"]
             hi = lo + binfo["shard_numel"]
             b.master32.data.copy_(full["fp32"][lo:hi])
             master = b.master32
@@ -149,8 +176,9 @@ def load_universal_into_optimizer(optimizer, universal_dir):
             b.shard16.copy_(b.master32.detach().to(b.shard16.dtype))
         from .. import comm as dist
         for b in optimizer.buckets:
+            # expert buckets gather over their expert-DP group
             dist.all_gather_into_tensor(b.flat16, b.shard16,
-                                        group=optimizer.dp_group)
+                                        group=b.pg or optimizer.dp_group)
     else:  # stage 3 subgroups
         rank = optimizer.rank
         for gi, (sg, ginfo) in enumerate(zip(optimizer.sub_groups,
