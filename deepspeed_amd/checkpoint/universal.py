@@ -70,26 +70,42 @@ def _iter_params_with_states(shards):
             else:
                 yield from assemble(bi, list(range(W)), binfo)
     else:  # subgroup / stage 3
+        W = len(shards)
         flats_per_rank = [sd["fp32_flat_groups"] for sd in shards]
-        for gi, ginfo in enumerate(layout["subgroups"]):
-            states = {"fp32": flats_per_rank}
-            extra = {}
+
+        def emit(gi, ranks, entry, qual=""):
+            name, off, shard_numel, full_numel, shape = entry[:5]
+            out = {"fp32": torch.cat(
+                [flats_per_rank[r][gi].detach()
+                 .float()[off:off + shard_numel]
+                 for r in ranks])[:full_numel].reshape(shape).clone()}
             for key in state_keys:
-                parts = [master_state(r, gi, key)
-                         for r in range(len(shards))]
+                parts = [master_state(r, gi, key) for r in ranks]
                 if all(p is not None for p in parts):
-                    extra[key] = parts
-            for name, off, shard_numel, full_numel, shape in ginfo["params"]:
-                out = {}
-                out["fp32"] = torch.cat(
-                    [f[gi].detach().float()[off:off + shard_numel]
-                     for f in flats_per_rank])[:full_numel] \
-                    .reshape(shape).clone()
-                for key, parts in extra.items():
                     out[key] = torch.cat(
                         [p.float()[off:off + shard_numel]
-                         for p in parts])[:full_numel].reshape(shape).clone()
-                yield name, out
+                         for p in parts])[:full_numel] \
+                        .reshape(shape).clone()
+            return name + qual, out
+
+        for r, sd in enumerate(shards):
+            lay = sd["shard_layout"]
+            for gi, ginfo in enumerate(lay["subgroups"]):
+                for entry in ginfo["params"]:
+                    gw = entry[5] if len(entry) > 5 else W
+                    if gw >= W:  # dense param: emit once, from rank 0
+                        if r == 0:
+                            yield emit(gi, range(W), entry)
+                    else:
+                        # expert param: sharded over the expert-DP group
+                        # (stride ep); qualified "@ep<off>" like the
+                        # stage-1/2 expert buckets
+                        ep = W // gw
+                        epoff = r % ep
+                        if r // ep != 0:
+                            continue
+                        ranks = [epoff + k * ep for k in range(gw)]
+                        yield emit(gi, ranks, entry, f"@ep{epoff}")
 
 
 def ds_to_universal(checkpoint_dir, output_dir, tag=None):
@@ -185,10 +201,21 @@ def load_universal_into_optimizer(optimizer, universal_dir):
                                              layout["subgroups"])):
             fp32 = torch.zeros(sg.numel)
             states = {key: torch.zeros(sg.numel) for key in state_keys}
-            for name, off, shard_numel, full_numel, shape in ginfo["params"]:
-                t = _load_param(universal_dir, name, state_keys)
+            for entry in ginfo["params"]:
+                name, off, shard_numel, full_numel, shape = entry[:5]
+                gw = entry[5] if len(entry) > 5 else world
+                shard_rank, qual = rank, ""
+                if gw < world:  # expert param: slice by expert-DP pos
+                    ep = world // gw
+                    qual = f"@ep{rank % ep}"
+                    shard_rank = rank // ep
+                t = _load_param(universal_dir, name + qual, state_keys)
+                if not t and qual:
+                    t = _load_param(universal_dir, name, state_keys)
+                if "fp32" not in t:
+                    continue
                 flat = t["fp32"].reshape(-1)
-                lo = rank * shard_numel
+                lo = shard_rank * shard_numel
                 hi = min(lo + shard_numel, full_numel)
                 if hi > lo:
                     fp32[off:off + hi - lo] = flat[lo:hi]

@@ -1177,9 +1177,15 @@ class ZeroStage3Optimizer:
             subgroups.append({
                 "group_idx": sg.group_idx,
                 "numel": sg.numel,
+                # 6th element: the param's SHARD world — expert params
+                # partition over the (smaller) expert-DP group, so
+                # offline reassembly must concatenate only that group's
+                # ranks (stride ep = world // shard_world)
                 "params": [(names.get(id(p), f"param_{p.ds_id}"),
                             sg.offsets[p], p.ds_shard_numel, p.ds_numel,
-                            list(p.ds_shape)) for p in sg.params],
+                            list(p.ds_shape),
+                            dist.get_world_size(p.ds_group))
+                           for p in sg.params],
             })
         return {"stage": 3, "world": self.world, "kind": "subgroup",
                 "subgroups": subgroups}

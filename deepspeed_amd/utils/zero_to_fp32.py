@@ -53,13 +53,51 @@ def _reassemble(shards, key="flat"):
             for name, off, numel, shape in binfo["params"]:
                 yield name, full[off:off + numel].reshape(shape).clone()
     elif layout["kind"] == "subgroup":  # stage 3
-        flats_per_rank = [sd["fp32_flat_groups"] for sd in shards]
-        for gi, ginfo in enumerate(layout["subgroups"]):
-            for name, off, shard_numel, full_numel, shape in ginfo["params"]:
-                pieces = [flats[gi].detach().float()[off:off + shard_numel]
-                          for flats in flats_per_rank]
-                yield name, torch.cat(pieces)[:full_numel] \
-                    .reshape(shape).clone()
+        import re
+        W = len(shards)
+        expat = re.compile(r"(.*deepspeed_experts\.)(\d+)(\..*)")
+        emitted = set()
+        for r, sd in enumerate(shards):
+            lay = sd["shard_layout"]
+            # local-expert count per Experts path (for global expert ids)
+            counts = {}
+            for g in lay["subgroups"]:
+                for e in g["params"]:
+                    m = expat.match(e[0])
+                    if m:
+                        counts.setdefault(m.group(1),
+                                          set()).add(int(m.group(2)))
+            counts = {k: len(v) for k, v in counts.items()}
+            for gi, ginfo in enumerate(lay["subgroups"]):
+                for e in ginfo["params"]:
+                    name, off, shard_numel, full_numel, shape = e[:5]
+                    gw = e[5] if len(e) > 5 else W
+                    if gw >= W:  # dense: same on every rank, emit once
+                        if r != 0:
+                            continue
+                        ranks = range(W)
+                        out = name
+                    else:
+                        # expert param: sharded over the expert-DP group
+                        # [epoff, epoff+ep, ...]; module-local expert ids
+                        # map to GLOBAL ids (epoff*num_local + local)
+                        ep = W // gw
+                        epoff = r % ep
+                        if r // ep != 0:
+                            continue  # first member of the group emits
+                        ranks = [epoff + k * ep for k in range(gw)]
+                        m = expat.match(name)
+                        out = (f"{m.group(1)}"
+                               f"{epoff * counts[m.group(1)] + int(m.group(2))}"
+                               f"{m.group(3)}") if m else f"{name}@ep{epoff}"
+                    if out in emitted:
+                        continue
+                    emitted.add(out)
+                    pieces = [shards[rr]["fp32_flat_groups"][gi].detach()
+                              .float()[off:off + shard_numel]
+                              for rr in ranks]
+                    yield out, torch.cat(pieces)[:full_numel] \
+                        .reshape(shape).clone()
     else:
         raise ValueError(f"unknown layout kind {layout['kind']}")
 
