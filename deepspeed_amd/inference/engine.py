@@ -109,75 +109,11 @@ class InferenceEngine(torch.nn.Module):
         (inference/engine.py:_create_cuda_graph) — 8B decode is otherwise
         launch-bound (~300 kernels/token).
         """
-        from ..models.llama import StaticKVCache, llama_decode_step
-        assert torch.cuda.is_available(), "hipGraph decode needs a GPU"
-        input_ids = input_ids.to(self.device)
-        B, S = input_ids.shape
-        cfg = self._model_cfg
-        max_seq = min(cfg.max_position_embeddings, S + max_new_tokens)
-        dev = self.device
-        caches = [StaticKVCache(B, max_seq, cfg.num_key_value_heads,
-                                cfg.head_dim, self._config.dtype, dev)
-                  for _ in range(cfg.num_hidden_layers)]
-        was_ckpt = cfg.activation_checkpointing
-        cfg.activation_checkpointing = False
-        try:
-            # ---- prefill (eager) ----
-            prefill_caches = []
-
-            class _Adapter:
-                def __init__(self, sc):
-                    self.sc = sc
-                    self.pos = 0
-
-                def update(self, k, v):
-                    self.sc.prefill(k, v)
-                    self.pos = k.shape[1]
-                    return (self.sc.k[:, :self.pos].contiguous(),
-                            self.sc.v[:, :self.pos].contiguous())
-
-            adapters = [_Adapter(c) for c in caches]
-            logits = self.module(input_ids, kv_caches=adapters)
-
-            # ---- static state ----
-            cos_t = self.module.model.rope_cos.float().to(dev)
-            sin_t = self.module.model.rope_sin.float().to(dev)
-            pos_idx = torch.tensor([S - 1], device=dev)
-            attn_mask = torch.full((1, 1, 1, max_seq), float("-inf"),
-                                   device=dev, dtype=self._config.dtype)
-            attn_mask[..., :S] = 0.0
-            id_buf = logits[:, -1, :].argmax(-1, keepdim=True)
-            out_tokens = torch.zeros(B, max_new_tokens, dtype=torch.long,
-                                     device=dev)
-            step_idx = torch.zeros(1, dtype=torch.long, device=dev)
-
-            def one_step():
-                out_tokens.index_copy_(1, step_idx, id_buf)
-                step_idx.add_(1)
-                pos_idx.add_(1)
-                attn_mask.index_fill_(3, pos_idx, 0.0)
-                lg = llama_decode_step(self.module, id_buf, caches, pos_idx,
-                                       attn_mask, cos_t, sin_t)
-                id_buf.copy_(lg[:, -1, :].argmax(-1, keepdim=True))
-
-            # warmup on a side stream (allocator settles), then capture
-            s = torch.cuda.Stream()
-            s.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(s):
-                one_step()
-            torch.cuda.current_stream().wait_stream(s)
-            n_graphed = max_new_tokens - 1
-            if n_graphed > 0:
-                graph = torch.cuda.CUDAGraph()
-                with torch.cuda.graph(graph):
-                    one_step()
-                for _ in range(n_graphed - 1):
-                    graph.replay()
-            # final token from the last id_buf
-            out_tokens.index_copy_(1, step_idx, id_buf)
-            return torch.cat([input_ids, out_tokens], dim=1)
-        finally:
-            cfg.activation_checkpointing = was_ckpt
+        from .graph_decode import hipgraph_greedy_decode
+        return hipgraph_greedy_decode(self.module, self._model_cfg,
+                                      self._config.dtype,
+                                      input_ids.to(self.device),
+                                      max_new_tokens)
 
     @torch.no_grad()
     def generate(self, input_ids, max_new_tokens=32, temperature=0.0,

@@ -42,21 +42,47 @@ def gathered_for_generation(engine):
 @torch.no_grad()
 def generate(engine, input_ids, max_new_tokens=32, temperature=0.0,
              top_k=0, top_p=1.0, eos_token_id=None,
-             offload_states_during_generate=False):
+             offload_states_during_generate=False, use_hipgraph=False):
     """Generate with the (possibly ZeRO-3-sharded) training weights.
 
     offload_states_during_generate: push optimizer masters/moments/grad
     accumulators to host for the generation phase (engine.offload_states)
-    so long-rollout KV caches get the HBM — reloaded before returning."""
+    so long-rollout KV caches get the HBM — reloaded before returning.
+
+    use_hipgraph: capture the per-token decode step as ONE hipGraph and
+    replay it (ref runtime/hybrid_engine_graph.py role; greedy only —
+    decode is launch-bound, ~300 kernels/token on 8B)."""
+    if use_hipgraph and temperature > 0:
+        raise ValueError("hipGraph rollout decode is greedy-only")
     if offload_states_during_generate:
         engine.offload_states(include=("hp_params", "lp_grads",
                                        "optim_states"))
     try:
+        if use_hipgraph:
+            return _generate_hipgraph(engine, input_ids, max_new_tokens)
         return _generate_inner(engine, input_ids, max_new_tokens,
                                temperature, top_k, top_p, eos_token_id)
     finally:
         if offload_states_during_generate:
             engine.reload_states()
+
+
+@torch.no_grad()
+def _generate_hipgraph(engine, input_ids, max_new_tokens):
+    """Graph-captured greedy decode straight off the gathered training
+    shards (the gather context holds every weight live across replays)."""
+    from ..inference.graph_decode import hipgraph_greedy_decode
+    module = engine.module
+    cfg = getattr(module, "cfg", None)
+    assert cfg is not None, "model must expose .cfg"
+    was_training = module.training
+    module.eval()
+    try:
+        with gathered_for_generation(engine):
+            return hipgraph_greedy_decode(module, cfg, engine.config.dtype,
+                                          input_ids, max_new_tokens)
+    finally:
+        module.train(was_training)
 
 
 def _top_p_filter(logits, top_p):

@@ -117,6 +117,9 @@ class HybridEngineRollout(RolloutEngine):
         by_len = {}
         for b in range(B):
             by_len.setdefault(int(lens[b]), []).append(b)
+        use_graph = (self.config.use_graph_capture
+                     and sampling.temperature == 0
+                     and torch.cuda.is_available())
         for plen, idxs in by_len.items():
             prompts = torch.stack(
                 [ids[b, -plen:] for b in idxs]).repeat_interleave(n, 0)
@@ -125,7 +128,8 @@ class HybridEngineRollout(RolloutEngine):
                 temperature=sampling.temperature,
                 top_k=max(sampling.top_k, 0),
                 top_p=sampling.top_p,
-                eos_token_id=self.eos_token_id)
+                eos_token_id=self.eos_token_id,
+                use_hipgraph=use_graph)
             for j, b in enumerate(idxs):
                 for s in range(n):
                     rows[b * n + s] = (out[j * n + s], plen)
