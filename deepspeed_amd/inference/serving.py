@@ -107,7 +107,7 @@ class ContinuousBatchingEngine:
     """Token-level continuous batching over a native model."""
 
     def __init__(self, model, max_batch=8, max_seq=None, device=None,
-                 prefill_chunk=None):
+                 prefill_chunk=None, prefill_budget=None):
         self.model = model
         cfg = model.cfg if hasattr(model, "cfg") else model.config
         self.cfg = cfg
@@ -122,8 +122,11 @@ class ContinuousBatchingEngine:
         # Dynamic SplitFuse-style scheduling (ref FastGen): bound the
         # prompt tokens processed per step so long prompts stream in
         # chunks instead of stalling the decode batch. None = whole
-        # prompt per step (legacy behavior).
+        # prompt per step (legacy behavior). prefill_budget additionally
+        # lets SEVERAL requests advance per step until the token budget
+        # is spent (defaults to one chunk of one request).
         self.prefill_chunk = prefill_chunk
+        self.prefill_budget = prefill_budget
         self.prefilling: List[Request] = []
         self.free_slots = list(range(max_batch))
         self.pending: List[Request] = []
@@ -241,11 +244,16 @@ class ContinuousBatchingEngine:
                 req.slot = self.free_slots.pop()
                 req.prefill_pos = 0
                 self.prefilling.append(req)
-            if self.prefilling:
+            budget = self.prefill_budget or self.prefill_chunk
+            while self.prefilling and budget > 0:
                 req = self.prefilling[0]
+                remaining = len(req.prompt) - getattr(req, "prefill_pos", 0)
+                budget -= min(self.prefill_chunk, remaining)
                 if self._prefill_chunk_step(req):
                     self.prefilling.pop(0)
                     self.running.append(req)
+                if self.prefill_budget is None:
+                    break  # legacy: one chunk of one request per step
         active = [r for r in self.running if not r.done]
         if active:
             self._decode(active)

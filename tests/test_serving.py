@@ -136,3 +136,40 @@ def test_splitfuse_chunked_prefill_matches_whole_prompt():
         al = a.tolist() if hasattr(a, "tolist") else list(a)
         bl = b.tolist() if hasattr(b, "tolist") else list(b)
         assert al == bl, f"chunked prefill diverged: {al} vs {bl}"
+
+
+def test_splitfuse_token_budget_multi_request():
+    """prefill_budget lets multiple requests advance per step within a
+    token budget; outputs match whole-prompt prefill and fewer steps
+    are needed than one-chunk-per-step."""
+    import torch
+    from deepspeed_amd.inference.serving import ContinuousBatchingEngine
+    from deepspeed_amd.models.llama import LLAMA_CONFIGS, LlamaForCausalLM
+    torch.manual_seed(0)
+    cfg = LLAMA_CONFIGS["llama-tiny"]
+    model = LlamaForCausalLM(cfg).eval()
+    g = torch.Generator().manual_seed(5)
+    prompts = [torch.randint(0, cfg.vocab_size, (n,), generator=g)
+               for n in (30, 9, 17, 5)]
+
+    def run(chunk, budget):
+        eng = ContinuousBatchingEngine(model, max_batch=4,
+                                       prefill_chunk=chunk,
+                                       prefill_budget=budget)
+        for p in prompts:
+            eng.add_request(p, max_new_tokens=6)
+        steps = 0
+        out = {}
+        while eng.pending or eng.running or eng.prefilling:
+            for r in eng.step():
+                out[r.rid] = r.tokens
+            steps += 1
+            assert steps < 500
+        return [out[k] for k in sorted(out)], steps
+
+    whole, _ = run(None, None)
+    budgeted, s_budget = run(8, 32)
+    onechunk, s_one = run(8, None)
+    for a, b in zip(whole, budgeted):
+        assert a.tolist() == b.tolist()
+    assert s_budget <= s_one, (s_budget, s_one)
