@@ -19,7 +19,11 @@ class HybridEngineMixin:
 
 @contextlib.contextmanager
 def gathered_for_generation(engine):
-    """Gather all ZeRO-3 shards for the duration of generation."""
+    """Gather all ZeRO-3 shards for the duration of generation.
+
+    While everything is gathered, the per-submodule fetch/release hooks
+    are paused: re-fetch/free per decode step is pure churn, and frees
+    or collectives inside a hipGraph capture would poison the graph."""
     if engine.zero_optimization_stage() != 3:
         yield
         return
@@ -31,9 +35,13 @@ def gathered_for_generation(engine):
     handle = all_gather_params(need, engine.optimizer.dp_group,
                                async_op=False)
     handle.wait()
+    opt = engine.optimizer
+    was_paused = getattr(opt, "_hooks_paused", False)
+    opt._hooks_paused = True
     try:
         yield
     finally:
+        opt._hooks_paused = was_paused
         for p in need:
             if not p.ds_persist:
                 free_param(p)

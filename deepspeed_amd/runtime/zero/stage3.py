@@ -309,6 +309,10 @@ class ZeroStage3Optimizer:
         self._trace_complete = False
         self._trace_pos = 0
         self._inflight = {}         # module -> AllGatherHandle
+        # hybrid-engine generation holds ALL params gathered: fetch/free
+        # per submodule is pure churn there (and collectives/frees inside
+        # hipGraph capture), so the hooks can be paused
+        self._hooks_paused = False
 
         self._global_grad_norm = 0.0
         self._cached_norm_sq = None
@@ -784,9 +788,13 @@ class ZeroStage3Optimizer:
 
     # -- forward path
     def _pre_forward_hook(self, mod, inputs):
+        if self._hooks_paused:
+            return
         self.fetch_sub_module(mod, forward=True)
 
     def _post_forward_hook(self, mod, inputs, output):
+        if self._hooks_paused:
+            return
         self.release_sub_module(mod)
 
     # -- backward path
