@@ -51,3 +51,25 @@ def test_fp8_torch_quant_bound(n, seed):
     rec = q.float() * s
     assert (rec - x.float()).abs().max() <= \
         x.float().abs().max() * 2 ** -3 + 1e-3
+
+
+@settings(max_examples=25, deadline=None)
+@given(rows=st.integers(1, 64), cols=st.integers(1, 64),
+       bits=st.sampled_from([4, 6, 8]), gpow=st.integers(0, 4),
+       seed=st.integers(0, 999))
+def test_weight_quantizer_bound(rows, cols, bits, gpow, seed):
+    """WeightQuantization: symmetric int-N error <= scale/2 per group,
+    dequant(quant(x)) fixed point after one round."""
+    from deepspeed_amd.runtime.weight_quantizer import WeightQuantization
+    groups = 2 ** gpow
+    if (rows * cols) % groups:
+        return
+    g = torch.Generator().manual_seed(seed)
+    w = torch.randn(rows, cols, generator=g) * 2
+    wq = WeightQuantization()
+    q, s = wq.quantize_data(w, bits, groups)
+    back = wq.dequantize_data(q, s, groups)
+    err = (back - w).abs().reshape(groups, -1).amax(1)
+    assert (err <= s / 2 + 1e-6).all()
+    q2, s2 = wq.quantize_data(back, bits, groups)
+    assert torch.equal(q.reshape(-1), q2.reshape(-1))
