@@ -408,3 +408,19 @@ def test_moe_zero3_offline_reassembly_and_universal():
         for r in range(2):
             for n, v in saved[r].items():
                 assert torch.allclose(got[r][n], v, atol=1e-2), (r, n)
+
+
+def test_deepspeed_checkpoint_inspection():
+    """DeepSpeedCheckpoint answers topology/content questions offline
+    (ref checkpoint/deepspeed_checkpoint.py)."""
+    from deepspeed_amd.checkpoint.inspect import DeepSpeedCheckpoint
+    with tempfile.TemporaryDirectory() as d:
+        run_distributed(_save_and_export, world_size=2, args=(2, d))
+        ck = DeepSpeedCheckpoint(d)
+        s = ck.summary()
+        assert s["zero_stage"] == 2 and s["dp_degree"] == 2
+        names = ck.parameter_names()
+        assert names and s["n_params"] == len(names)
+        got = dict(ck.fp32_parameters())
+        assert set(got) == set(names)
+        assert "global_steps" in ck.client_state()
