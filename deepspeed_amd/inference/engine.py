@@ -47,6 +47,13 @@ class InferenceConfig:
                                           {"tp_size": 1})
         self.replace_with_kernel_inject = config.get(
             "replace_with_kernel_inject", True)
+        # int8 weight residency (ref init_inference dtype=torch.int8 +
+        # quantization_setting): dequant-on-the-fly group-wise linears
+        q = dict(config.get("quant", {}) or {})
+        if self.dtype in (torch.int8, "int8"):
+            q.setdefault("enabled", True)
+            self.dtype = torch.bfloat16  # compute dtype stays bf16 MFMA
+        self.quant = q
 
 
 class InferenceEngine(torch.nn.Module):
@@ -79,6 +86,14 @@ class InferenceEngine(torch.nn.Module):
                 replace_transformer_layer
             replace_transformer_layer(self.module)
         self.module.to(self._config.dtype).to(self.device)
+        if self._config.quant.get("enabled"):
+            from .quantization import init_quantization
+            q = self._config.quant
+            init_quantization(
+                self.module,
+                quantize_bits=q.get("bits", 8),
+                groups=q.get("groups", 64),
+                mlp_extra_grouping=q.get("mlp_extra_grouping", False))
         self.module.eval()
         self._caches = None
 

@@ -116,3 +116,22 @@ def test_init_inference_hf_model_delegates_generate():
     ids = torch.randint(0, 128, (1, 8))
     out = eng.generate(ids, max_new_tokens=4)
     assert out.shape[1] == 12
+
+
+def test_init_inference_int8_weight_residency():
+    """dtype int8 (or quant.enabled) swaps linears for group-wise int8
+    dequant-on-the-fly modules; generation still works (ref
+    init_inference quantization_setting)."""
+    import torch
+    from deepspeed_amd import init_inference
+    from deepspeed_amd.linear.quantization import QuantizedLinear
+    from deepspeed_amd.models.llama import LLAMA_CONFIGS, LlamaForCausalLM
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(LLAMA_CONFIGS["llama-tiny"])
+    eng = init_inference(model, config={"dtype": "int8"})
+    nq = sum(isinstance(m, QuantizedLinear) for m in eng.module.modules())
+    assert nq > 0, "no linears quantized"
+    assert eng._config.dtype == torch.bfloat16  # compute stays bf16
+    ids = torch.randint(0, 512, (1, 8))
+    out = eng.generate(ids, max_new_tokens=4)
+    assert out.shape == (1, 12)
