@@ -313,6 +313,10 @@ class ZeroStage3Optimizer:
         # per submodule is pure churn there (and collectives/frees inside
         # hipGraph capture), so the hooks can be paused
         self._hooks_paused = False
+        # partition-traffic counters (ref partitioned_param_profiler)
+        self._partition_stats = {"fetches": 0, "prefetch_hits": 0,
+                                 "demand_gathers": 0, "gathered_numel": 0,
+                                 "releases": 0}
 
         self._global_grad_norm = 0.0
         self._cached_norm_sq = None
@@ -811,6 +815,8 @@ class ZeroStage3Optimizer:
 
     def fetch_sub_module(self, mod, forward=True):
         params = mod._ds_direct_params
+        st = self._partition_stats
+        st["fetches"] += 1
         for p in params:
             p.ds_active_sub_modules.add(id(mod))
         # record trace on first iteration
@@ -818,10 +824,13 @@ class ZeroStage3Optimizer:
             self._trace.append(mod)
         h = self._inflight.pop(mod, None)
         if h is not None:
+            st["prefetch_hits"] += 1
             h.wait()
         need = [p for p in params
                 if p.ds_status == ZeroParamStatus.NOT_AVAILABLE]
         if need:
+            st["demand_gathers"] += 1
+            st["gathered_numel"] += sum(p.ds_numel for p in need)
             self._gather_grouped(need,
                               stream=self.ag_stream).wait()
         # tied params may be INFLIGHT under another module's handle
@@ -887,7 +896,15 @@ class ZeroStage3Optimizer:
             p.ds_active_sub_modules.discard(id(mod))
             if not keep and not p.ds_active_sub_modules \
                     and not p.ds_persist:
+                self._partition_stats["releases"] += 1
                 free_param(p)
+
+    def partition_stats(self):
+        """Fetch/prefetch/release counters since construction (role of
+        ref zero/partitioned_param_profiler.py). prefetch_hits /
+        demand_gathers shows how much the trace-replay prefetch covers;
+        gathered_numel is total elements pulled on demand."""
+        return dict(self._partition_stats)
 
     def _install_grad_hooks(self):
         self._grad_hooks = []
