@@ -684,6 +684,16 @@ class DeepSpeedEngine(torch.nn.Module):
         log_dist(f"step={self.global_steps}, skipped={self.skipped_steps}, "
                  f"lr={lr}, scale={loss_scale}, grad_norm={norm:.4f}",
                  ranks=[0])
+        if self.wall_clock_breakdown() and \
+                hasattr(self.optimizer, "partition_stats"):
+            st = self.optimizer.partition_stats()
+            if st["fetches"]:
+                log_dist(
+                    f"zero3 partition traffic: {st['fetches']} fetches, "
+                    f"{st['prefetch_hits']} prefetch hits, "
+                    f"{st['demand_gathers']} demand gathers "
+                    f"({st['gathered_numel'] / 1e6:.1f}M elems), "
+                    f"{st['releases']} releases", ranks=[0])
         self.monitor.write_events([
             ("Train/lr", lr[0] if lr else 0.0, self.global_steps),
         ])
