@@ -29,6 +29,7 @@ class LlamaConfig:
     tie_word_embeddings: bool = False
     initializer_range: float = 0.02
     activation_checkpointing: bool = True
+    attention_bias: bool = False  # Qwen2-style qkv bias
 
     @property
     def head_dim(self):
@@ -49,6 +50,20 @@ LLAMA_CONFIGS = {
                                num_hidden_layers=8, num_attention_heads=16,
                                num_key_value_heads=4, vocab_size=32000,
                                max_position_embeddings=4096),
+    # Qwen2 family: llama architecture + qkv bias + tied tiny variants
+    "qwen2-7b": LlamaConfig(hidden_size=3584, intermediate_size=18944,
+                            num_hidden_layers=28, num_attention_heads=28,
+                            num_key_value_heads=4, vocab_size=152064,
+                            max_position_embeddings=32768,
+                            rope_theta=1e6, rms_norm_eps=1e-6,
+                            attention_bias=True),
+    "qwen2-tiny": LlamaConfig(hidden_size=256, intermediate_size=688,
+                              num_hidden_layers=4, num_attention_heads=8,
+                              num_key_value_heads=4, vocab_size=2048,
+                              max_position_embeddings=512,
+                              attention_bias=True,
+                              tie_word_embeddings=True,
+                              activation_checkpointing=False),
 }
 
 
@@ -67,9 +82,10 @@ class LlamaAttention(nn.Module):
         super().__init__()
         self.cfg = cfg
         h, d = cfg.hidden_size, cfg.head_dim
-        self.q_proj = nn.Linear(h, cfg.num_attention_heads * d, bias=False)
-        self.k_proj = nn.Linear(h, cfg.num_key_value_heads * d, bias=False)
-        self.v_proj = nn.Linear(h, cfg.num_key_value_heads * d, bias=False)
+        ab = getattr(cfg, "attention_bias", False)
+        self.q_proj = nn.Linear(h, cfg.num_attention_heads * d, bias=ab)
+        self.k_proj = nn.Linear(h, cfg.num_key_value_heads * d, bias=ab)
+        self.v_proj = nn.Linear(h, cfg.num_key_value_heads * d, bias=ab)
         self.o_proj = nn.Linear(cfg.num_attention_heads * d, h, bias=False)
         self.sp_group = None  # set by enable_ulysses()
         self._dist_attn = None
