@@ -35,6 +35,12 @@ def config_from_hf(hf_config):
         max_position_embeddings=hf_config.max_position_embeddings,
         rms_norm_eps=getattr(hf_config, "rms_norm_eps", 1e-5),
         rope_theta=_hf_rope_theta(hf_config),
+        # Qwen2 configs set attention bias (qkv) on the same tree
+        attention_bias=getattr(hf_config, "attention_bias",
+                               getattr(hf_config, "model_type", "")
+                               .startswith("qwen2")),
+        tie_word_embeddings=getattr(hf_config, "tie_word_embeddings",
+                                    False),
     )
 
 

@@ -135,3 +135,28 @@ def test_init_inference_int8_weight_residency():
     ids = torch.randint(0, 512, (1, 8))
     out = eng.generate(ids, max_new_tokens=4)
     assert out.shape == (1, 12)
+
+
+def test_hf_qwen2_import_logits_match():
+    """Qwen2 HF checkpoint (qkv bias, tied embeddings) imports onto the
+    native modules with logits parity."""
+    import pytest
+    import torch
+    transformers = pytest.importorskip("transformers")
+    from transformers import Qwen2Config, Qwen2ForCausalLM
+    from deepspeed_amd.models.hf import load_hf_llama
+    hf_cfg = Qwen2Config(hidden_size=128, intermediate_size=256,
+                         num_hidden_layers=2, num_attention_heads=4,
+                         num_key_value_heads=2, vocab_size=512,
+                         max_position_embeddings=128,
+                         tie_word_embeddings=True)
+    torch.manual_seed(0)
+    hf = Qwen2ForCausalLM(hf_cfg).eval()
+    native = load_hf_llama(hf).eval()
+    assert native.model.layers[0].self_attn.q_proj.bias is not None
+    ids = torch.randint(0, 512, (2, 16))
+    with torch.no_grad():
+        ref = hf(ids).logits.float()
+        got = native(ids).float()
+    err = (got - ref).abs().max().item()
+    assert err < 2e-3, f"qwen2 logits diverged: {err}"
