@@ -160,3 +160,27 @@ def test_hf_qwen2_import_logits_match():
         got = native(ids).float()
     err = (got - ref).abs().max().item()
     assert err < 2e-3, f"qwen2 logits diverged: {err}"
+
+
+def test_hf_mistral_import_logits_match():
+    """Mistral shares the llama tree; import + parity (sliding window
+    is a no-op below the window size)."""
+    import pytest
+    import torch
+    transformers = pytest.importorskip("transformers")
+    from transformers import MistralConfig, MistralForCausalLM
+    from deepspeed_amd.models.hf import load_hf_llama
+    hf_cfg = MistralConfig(hidden_size=128, intermediate_size=256,
+                           num_hidden_layers=2, num_attention_heads=4,
+                           num_key_value_heads=2, vocab_size=512,
+                           max_position_embeddings=128,
+                           sliding_window=4096)
+    torch.manual_seed(0)
+    hf = MistralForCausalLM(hf_cfg).eval()
+    native = load_hf_llama(hf).eval()
+    ids = torch.randint(0, 512, (2, 16))
+    with torch.no_grad():
+        ref = hf(ids).logits.float()
+        got = native(ids).float()
+    err = (got - ref).abs().max().item()
+    assert err < 2e-3, f"mistral logits diverged: {err}"
