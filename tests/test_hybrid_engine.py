@@ -208,3 +208,24 @@ def _hybrid_hipgraph_gpu_body():
     engine.step()
     assert torch.isfinite(torch.tensor(loss.item()))
     return True
+
+
+def test_rollout_dataclass_validation():
+    """RolloutRequest/RolloutBatch reject malformed shapes (ref
+    runtime/rollout/base.py __post_init__ checks)."""
+    import pytest
+    import torch
+    from deepspeed_amd.runtime.rollout import RolloutBatch, RolloutRequest
+    ids = torch.zeros(2, 5, dtype=torch.long)
+    with pytest.raises(ValueError):
+        RolloutRequest(prompt_ids=torch.zeros(5, dtype=torch.long),
+                       prompt_attention_mask=torch.ones(5))
+    with pytest.raises(ValueError):
+        RolloutRequest(prompt_ids=ids,
+                       prompt_attention_mask=torch.ones(2, 4))
+    b = RolloutBatch(input_ids=ids, attention_mask=torch.ones(2, 5),
+                     response_start_idx=torch.tensor([3, 2]))
+    assert b.batch_size == 2 and b.seq_len == 5
+    with pytest.raises(ValueError):
+        RolloutBatch(input_ids=ids, attention_mask=torch.ones(2, 5),
+                     response_start_idx=torch.tensor([3]))
