@@ -169,6 +169,13 @@ class UlyssesConfig(DSConfigModel):
     sequence_parallel_size: int = 1
 
 
+class TensorParallelConfig(DSConfigModel):
+    """Training AutoTP (ref runtime/tensor_parallel/config.py):
+    autotp_size > 1 shards the module's linears over a TP group at
+    initialize() and installs replicated-grad hooks."""
+    autotp_size: int = 1
+
+
 class PipelineConfig(DSConfigModel):
     stages: str = "auto"
     partition: str = "best"
@@ -226,6 +233,10 @@ class DeepSpeedConfig:
         self.csv_monitor = MonitorCSVConfig(**config.get("csv_monitor", {}))
         self.tensorboard = MonitorTensorBoardConfig(**config.get("tensorboard", {}))
         self.ulysses = UlyssesConfig(**config.get("sequence_parallel", {}))
+        tp_raw = config.get("tensor_parallel", {})
+        self.tensor_parallel = TensorParallelConfig(
+            autotp_size=tp_raw.get("autotp_size", tp_raw.get("tp_size", 1))
+            if isinstance(tp_raw, dict) else 1)
         self.torch_autocast = TorchAutocastConfig(
             **config.get("torch_autocast", {}))
         self.pipeline = PipelineConfig(**config.get("pipeline", {}))

@@ -97,6 +97,19 @@ class DeepSpeedEngine(torch.nn.Module):
         if sp > 1 and not groups.sequence_parallel_is_initialized():
             groups.initialize_sequence_parallel(sp)
 
+        # training AutoTP (ref runtime/tensor_parallel/tp_manager.py):
+        # shard the module over a TP group and make ZeRO/DDP partition
+        # over the orthogonal DP group
+        tp = self._config.tensor_parallel.autotp_size
+        if tp > 1:
+            assert sp <= 1, "autotp_size with sequence parallelism: pick one"
+            if groups.get_tensor_parallel_group() is None:
+                groups.initialize_tensor_parallel(tp)
+            from ..module_inject.auto_tp import (add_tp_training_hooks,
+                                                 apply_tensor_parallel_hf)
+            apply_tensor_parallel_hf(model)
+            add_tp_training_hooks(model)
+
         self.dp_group = groups.get_sequence_data_parallel_group() \
             if sp > 1 else groups.get_data_parallel_group()
         self.dp_world_size = dist.get_world_size(self.dp_group)

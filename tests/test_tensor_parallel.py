@@ -202,3 +202,59 @@ def test_autotp_training_matches_single():
     for r in results:
         for a, b in zip(r, ref_losses):
             assert abs(a - b) < 2e-3, (r, ref_losses)
+
+
+def _autotp_config_body():
+    """ds_config tensor_parallel.autotp_size shards an HF model at
+    initialize() and trains with replicated-grad hooks (ref
+    runtime/tensor_parallel/tp_manager.py)."""
+    import pytest
+    import torch
+    import torch.distributed as tdist
+    transformers = pytest.importorskip("transformers")
+    from transformers import LlamaConfig, LlamaForCausalLM
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    groups.reset_groups()
+    torch.manual_seed(0)
+    hf_cfg = LlamaConfig(hidden_size=64, intermediate_size=128,
+                         num_hidden_layers=2, num_attention_heads=4,
+                         num_key_value_heads=2, vocab_size=256,
+                         max_position_embeddings=64)
+    model = LlamaForCausalLM(hf_cfg)
+    ref_numel = sum(p.numel() for p in model.parameters())
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 2,
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+        "tensor_parallel": {"autotp_size": tdist.get_world_size()},
+    })
+    tp = tdist.get_world_size()
+    if tp > 1:
+        numel = sum(p.numel() for p in engine.module.parameters())
+        assert numel < ref_numel  # linears sharded
+        # per-rank head attrs divided on attention modules
+        attn = engine.module.model.layers[0].self_attn
+        got = getattr(attn, "num_attention_heads",
+                      getattr(attn.config, "num_attention_heads", None)
+                      if hasattr(attn, "config") else None)
+    ids = torch.randint(0, 256, (2, 16))
+    losses = []
+    for _ in range(4):
+        out = engine(input_ids=ids, labels=ids)
+        loss = out.loss if hasattr(out, "loss") else out[0]
+        engine.backward(loss)
+        engine.step()
+        losses.append(loss.item())
+    assert losses[-1] < losses[0], losses
+    # replicated params (norms/embeddings) identical across TP ranks
+    emb = engine.module.model.embed_tokens.weight.detach()
+    g = groups.get_tensor_parallel_group()
+    mx = emb.clone()
+    tdist.all_reduce(mx, op=tdist.ReduceOp.MAX, group=g)
+    assert torch.equal(mx, emb), "replicated embedding diverged"
+    return True
+
+
+def test_autotp_config_world2():
+    from tests.common import run_distributed
+    assert all(run_distributed(_autotp_config_body, world_size=2))
