@@ -83,6 +83,24 @@ def init_inference(model, config=None, **kwargs):
     return InferenceEngine(model, config=config, **kwargs)
 
 
+def tp_model_init(model, tp_size, dtype=None, config=None, **kwargs):
+    """Shard `model` over a TP group of `tp_size` for training (ref
+    deepspeed/__init__.py:408; sharding here happens immediately — the
+    config path `tensor_parallel.autotp_size` does the same at
+    initialize())."""
+    import torch
+    from .comm import groups
+    from .module_inject.auto_tp import (add_tp_training_hooks,
+                                        apply_tensor_parallel_hf)
+    if groups.get_tensor_parallel_group() is None:
+        groups.initialize_tensor_parallel(tp_size)
+    apply_tensor_parallel_hf(model)
+    add_tp_training_hooks(model)
+    if dtype is not None:
+        model.to(dtype)
+    return model
+
+
 def add_config_arguments(parser):
     """Ref deepspeed/__init__.py:305."""
     group = parser.add_argument_group("DeepSpeed-AMD",

@@ -258,3 +258,24 @@ def _autotp_config_body():
 def test_autotp_config_world2():
     from tests.common import run_distributed
     assert all(run_distributed(_autotp_config_body, world_size=2))
+
+
+def _tp_model_init_body():
+    import torch
+    import torch.distributed as tdist
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    from deepspeed_amd.models.llama import LLAMA_CONFIGS, LlamaForCausalLM
+    groups.reset_groups()
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(LLAMA_CONFIGS["llama-tiny"])
+    full = sum(p.numel() for p in model.parameters())
+    model = deepspeed_amd.tp_model_init(model, tp_size=2,
+                                        dtype=torch.bfloat16)
+    assert sum(p.numel() for p in model.parameters()) < full
+    return True
+
+
+def test_tp_model_init_world2():
+    from tests.common import run_distributed
+    assert all(run_distributed(_tp_model_init_body, world_size=2))
