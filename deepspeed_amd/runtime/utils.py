@@ -74,6 +74,20 @@ def clip_tensors_by_global_norm(tensors, max_norm, global_norm, eps=1e-6):
     return global_norm
 
 
+def clip_grad_norm_(parameters, max_norm, norm_type=2, mpu=None,
+                    group=None):
+    """Clip the gradients of `parameters` by their global norm across
+    ranks (reference-compatible name, ref runtime/utils.py:359)."""
+    if isinstance(parameters, torch.Tensor):
+        parameters = [parameters]
+    grads = [p.grad for p in parameters if p.grad is not None]
+    if not grads:
+        return 0.0
+    total = get_flat_grad_norm(grads, norm_type=norm_type, group=group)
+    clip_tensors_by_global_norm(grads, max_norm, total)
+    return total
+
+
 class CheckOverflow:
     """Detect inf/nan in gradients across ranks (fp16 dynamic loss scaling)."""
 
