@@ -331,9 +331,12 @@ class Init:
     """Context manager: parameters created inside are sharded at module
     construction (ref partition_parameters.py:940).
 
-    Implementation: patches nn.Module.register_parameter so each fresh
-    parameter is converted immediately after registration — construction of
-    an N-param model never holds more than one full parameter per rank.
+    Implementation: wraps the `__init__` of every nn.Module subclass so
+    each submodule's DIRECT parameters are converted as soon as its own
+    constructor finishes (ref InsertPostInitMethodToModuleSubClasses) —
+    default initializers (reset_parameters) still see full tensors, and
+    construction of an N-param model never holds more than one full
+    submodule per rank.
     """
 
     def __init__(self, module=None, data_parallel_group=None, dtype=None,
@@ -346,7 +349,7 @@ class Init:
             torch.device("cuda", torch.cuda.current_device())
             if torch.cuda.is_available() else torch.device("cpu"))
         self.persist_threshold = param_persistence_threshold
-        self._orig_register = None
+        self._orig_inits = {}
         if config_dict_or_path is not None:
             from ...config import DeepSpeedConfig
             cfg = DeepSpeedConfig(config_dict_or_path,
@@ -370,21 +373,33 @@ class Init:
             dist.init_distributed()
         init = self
 
-        self._orig_register = torch.nn.Module.register_parameter
+        def make_wrapper(cls_init):
+            def wrapper(mod, *a, **k):
+                cls_init(mod, *a, **k)
+                for p in mod._parameters.values():
+                    if p is not None and not is_zero_param(p):
+                        convert_to_zero_param(p, init.dp_group,
+                                              init.device, init.dtype,
+                                              init.persist_threshold)
+            wrapper._ds_init_wrapper = True
+            return wrapper
 
-        def wrapped_register(mod, name, param):
-            init._orig_register(mod, name, param)
-            if param is not None and not is_zero_param(param):
-                convert_to_zero_param(param, init.dp_group, init.device,
-                                      init.dtype, init.persist_threshold)
+        def subclasses(cls):
+            for s in cls.__subclasses__():
+                yield s
+                yield from subclasses(s)
 
-        torch.nn.Module.register_parameter = wrapped_register
+        for cls in list(set(subclasses(torch.nn.Module))):
+            ci = cls.__dict__.get("__init__")
+            if ci is not None:
+                self._orig_inits[cls] = ci
+                cls.__init__ = make_wrapper(ci)
         return self
 
     def __exit__(self, *exc):
-        if self._orig_register is not None:
-            torch.nn.Module.register_parameter = self._orig_register
-            self._orig_register = None
+        for cls, ci in self._orig_inits.items():
+            cls.__init__ = ci
+        self._orig_inits = {}
         return False
 
 

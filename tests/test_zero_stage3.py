@@ -784,3 +784,70 @@ def test_partition_stats_counters():
     kicks in, demand gathers counted with element volume."""
     from tests.common import run_distributed
     run_distributed(_partition_stats_body, world_size=1)
+
+
+def _multiple_engines_body():
+    """Two live engines in one process (RLHF actor+critic pattern,
+    ref tests/unit/runtime/test_multiple_models.py): independent ZeRO
+    state, interleaved train steps, both converge."""
+    import torch
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    from tests.simple_model import SimpleModel
+    groups.reset_groups()
+    torch.manual_seed(0)
+    cfg = {"train_micro_batch_size_per_gpu": 4,
+           "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+           "bf16": {"enabled": True}}
+    e3, _, _, _ = deepspeed_amd.initialize(
+        model=SimpleModel(32), config={**cfg,
+                                       "zero_optimization": {"stage": 3}})
+    e2, _, _, _ = deepspeed_amd.initialize(
+        model=SimpleModel(32), config={**cfg,
+                                       "zero_optimization": {"stage": 2}})
+    x = torch.randn(4, 32).bfloat16()
+    y = torch.randn(4, 32).bfloat16()
+    l3, l2 = [], []
+    for _ in range(6):  # interleaved: actor step then critic step
+        loss = e3(x, y)
+        e3.backward(loss)
+        e3.step()
+        l3.append(loss.item())
+        loss = e2(x, y)
+        e2.backward(loss)
+        e2.step()
+        l2.append(loss.item())
+    assert l3[-1] < l3[0], l3
+    assert l2[-1] < l2[0], l2
+    return True
+
+
+def test_multiple_engines_one_process():
+    from tests.common import run_distributed
+    run_distributed(_multiple_engines_body, world_size=1)
+
+
+def test_zero_init_nesting():
+    """Nested zero.Init contexts restore the register_parameter chain
+    correctly (ref test_zero_nesting_init)."""
+    import torch
+    import torch.distributed as tdist
+    from deepspeed_amd.comm import groups
+    from deepspeed_amd.runtime.zero.stage3_params import (Init,
+                                                          is_zero_param)
+    groups.reset_groups()
+    if not tdist.is_initialized():
+        tdist.init_process_group("gloo",
+                                 init_method="tcp://127.0.0.1:29639",
+                                 rank=0, world_size=1)
+    orig = torch.nn.Module.register_parameter
+    with Init():
+        a = torch.nn.Linear(8, 8)
+        with Init():
+            b = torch.nn.Linear(8, 8)
+        c = torch.nn.Linear(8, 8)  # outer context still active
+    d = torch.nn.Linear(8, 8)      # fully restored
+    assert all(is_zero_param(p) for m in (a, b, c)
+               for p in m.parameters())
+    assert not any(is_zero_param(p) for p in d.parameters())
+    assert torch.nn.Module.register_parameter is orig
