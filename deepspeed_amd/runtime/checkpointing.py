@@ -19,6 +19,13 @@ def _tag(engine, tag):
     return tag if tag is not None else f"global_step{engine.global_steps}"
 
 
+def _mp_rank():
+    # tensor-parallel ranks hold DIFFERENT module shards: each writes
+    # its own mp_rank_XX file (ref _get_ckpt_name mp_rank placement)
+    from ..comm import groups
+    return groups.get_tensor_parallel_rank()
+
+
 def _model_states_name(mp_rank=0):
     return f"mp_rank_{mp_rank:02d}_model_states.pt"
 
@@ -112,7 +119,7 @@ def save_checkpoint(engine, save_dir, tag=None, client_state=None,
             torch.save({"module": esd, "ds_version": VERSION},
                        os.path.join(ckpt_dir, _expert_ckpt_name(ep_rank)))
 
-    # model states: rank 0 of each DP group (mp_rank 0 — no TP yet)
+    # model states: dp-rank 0 of each TP shard writes its mp_rank file
     if dp_rank == 0:
         state = {
             "module": engine.module_state_dict(
@@ -135,13 +142,16 @@ def save_checkpoint(engine, save_dir, tag=None, client_state=None,
         if expert_names and state.get("module"):
             state["module"] = {k: v for k, v in state["module"].items()
                                if k not in expert_names}
-        torch.save(state, os.path.join(ckpt_dir, _model_states_name()))
+        torch.save(state, os.path.join(ckpt_dir,
+                                       _model_states_name(_mp_rank())))
 
     # zero shards: every dp rank
     if is_zero:
         zstate = {"optimizer_state_dict": engine.optimizer.state_dict(),
                   "ds_version": VERSION}
-        torch.save(zstate, os.path.join(ckpt_dir, _zero_ckpt_name(dp_rank)))
+        torch.save(zstate,
+                   os.path.join(ckpt_dir,
+                                _zero_ckpt_name(dp_rank, _mp_rank())))
 
     if dist.is_initialized():
         dist.barrier()
@@ -163,7 +173,7 @@ def load_checkpoint(engine, load_dir, tag=None, load_module_strict=True,
         with open(latest) as f:
             tag = f.read().strip()
     ckpt_dir = os.path.join(load_dir, str(tag))
-    model_file = os.path.join(ckpt_dir, _model_states_name())
+    model_file = os.path.join(ckpt_dir, _model_states_name(_mp_rank()))
     state = torch.load(model_file, map_location="cpu", weights_only=False)
 
     is_zero = hasattr(engine.optimizer, "load_state_dict") and \
@@ -198,7 +208,8 @@ def load_checkpoint(engine, load_dir, tag=None, load_module_strict=True,
             # Stage 3: module weights live only in the zero shards — restore
             # them from the per-dp-rank zero file without optimizer state.
             dp_rank = engine.get_data_parallel_rank()
-            zfile = os.path.join(ckpt_dir, _zero_ckpt_name(dp_rank))
+            zfile = os.path.join(ckpt_dir,
+                                 _zero_ckpt_name(dp_rank, _mp_rank()))
             zstate = torch.load(zfile, map_location="cpu", weights_only=False)
             engine.optimizer.load_state_dict(zstate["optimizer_state_dict"],
                                              load_optimizer_states=False)
@@ -208,7 +219,8 @@ def load_checkpoint(engine, load_dir, tag=None, load_module_strict=True,
     if not load_module_only:
         if is_zero:
             dp_rank = engine.get_data_parallel_rank()
-            zfile = os.path.join(ckpt_dir, _zero_ckpt_name(dp_rank))
+            zfile = os.path.join(ckpt_dir,
+                                 _zero_ckpt_name(dp_rank, _mp_rank()))
             zstate = torch.load(zfile, map_location="cpu", weights_only=False)
             engine.optimizer.load_state_dict(
                 zstate["optimizer_state_dict"],

@@ -279,3 +279,51 @@ def _tp_model_init_body():
 def test_tp_model_init_world2():
     from tests.common import run_distributed
     assert all(run_distributed(_tp_model_init_body, world_size=2))
+
+
+def _tp_checkpoint_body(ckpt_dir):
+    """TP-sharded engines write per-mp-rank checkpoint files and resume
+    their own shards (ref _get_ckpt_name mp_rank placement)."""
+    import os
+    import torch
+    import torch.distributed as tdist
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    from deepspeed_amd.models.llama import LLAMA_CONFIGS, LlamaForCausalLM
+    groups.reset_groups()
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(LLAMA_CONFIGS["llama-tiny"])
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 2,
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+        "zero_optimization": {"stage": 1},
+        "bf16": {"enabled": True},
+        "tensor_parallel": {"autotp_size": tdist.get_world_size()}})
+    ids = torch.randint(0, 2048, (2, 16))
+    loss = engine(ids, labels=ids)
+    engine.backward(loss)
+    engine.step()
+    want = {n: p.detach().float().clone()
+            for n, p in engine.module.named_parameters()}
+    engine.save_checkpoint(ckpt_dir, tag="t0")
+    tdist.barrier()
+    if tdist.get_rank() == 0:
+        files = sorted(os.listdir(os.path.join(ckpt_dir, "t0")))
+        assert "mp_rank_00_model_states.pt" in files, files
+        assert "mp_rank_01_model_states.pt" in files, files
+    # perturb, then restore this rank's shard
+    loss = engine(ids, labels=ids)
+    engine.backward(loss)
+    engine.step()
+    engine.load_checkpoint(ckpt_dir, tag="t0")
+    for n, p in engine.module.named_parameters():
+        assert torch.allclose(p.detach().float(), want[n], atol=1e-2), n
+    return True
+
+
+def test_tp_checkpoint_per_mp_rank_world2():
+    import tempfile
+    from tests.common import run_distributed
+    with tempfile.TemporaryDirectory() as d:
+        assert all(run_distributed(_tp_checkpoint_body, world_size=2,
+                                   args=(d,)))
