@@ -24,10 +24,24 @@ def _read_tag(ckpt_dir, tag):
     return os.path.join(ckpt_dir, str(tag))
 
 
-def _load_zero_shards(dirpath):
-    files = sorted(glob.glob(os.path.join(
-        dirpath, "zero_pp_rank_*_mp_rank_*_optim_states.pt")),
-        key=lambda f: int(os.path.basename(f).split("_")[3]))
+def _load_zero_shards(dirpath, mp_rank=None):
+    all_files = glob.glob(os.path.join(
+        dirpath, "zero_pp_rank_*_mp_rank_*_optim_states.pt"))
+    mp_ranks = sorted({os.path.basename(f).split("_")[6]
+                       for f in all_files})
+    if mp_rank is None:
+        if len(mp_ranks) > 1:
+            # TP shards hold DIFFERENT parameters: dp-concat across mp
+            # ranks would interleave garbage. Offline TP-slice merging
+            # is not implemented — reconstruct per mp rank.
+            raise NotImplementedError(
+                f"checkpoint has {len(mp_ranks)} tensor-parallel shards;"
+                " pass mp_rank= to reassemble one TP shard at a time")
+        mp_rank = int(mp_ranks[0]) if mp_ranks else 0
+    files = sorted((f for f in all_files
+                    if os.path.basename(f).split("_")[6]
+                    == f"{mp_rank:02d}"),
+                   key=lambda f: int(os.path.basename(f).split("_")[3]))
     if not files:
         raise FileNotFoundError(f"no zero shard files in {dirpath}")
     return [torch.load(f, map_location="cpu", weights_only=False)
