@@ -804,6 +804,12 @@ class DeepSpeedEngine(torch.nn.Module):
 
         Ref engine.py:5567 / _zero3_consolidated_16bit_state_dict:5491.
         """
+        from ..comm import groups as _grp
+        if _grp.get_tensor_parallel_world_size() > 1:
+            raise NotImplementedError(
+                "save_16bit_model under tensor parallelism would write "
+                "one TP shard as if it were the full model; use "
+                "save_checkpoint (per-mp-rank files) instead")
         sd = self._consolidated_16bit_state_dict()
         if dist.get_rank() == 0 and sd is not None:
             os.makedirs(save_dir, exist_ok=True)
