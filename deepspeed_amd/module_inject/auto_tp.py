@@ -56,9 +56,10 @@ def apply_tensor_parallel(model, tp_group=None):
 # (ref auto_tp.py:294 tp_parser / :357 _replace)
 COLUMN_PATTERNS = ("q_proj", "k_proj", "v_proj", "gate_proj", "up_proj",
                    "query_key_value", "dense_h_to_4h", "fc1", "wi_0",
-                   "wi_1", "c_attn", "query", "key", "value")
+                   "wi_1", "c_attn", "query", "key", "value",
+                   ".w1", ".w3", "c_fc")  # mixtral experts / gpt2 mlp
 ROW_PATTERNS = ("o_proj", "down_proj", "dense_4h_to_h", "out_proj",
-                "fc2", "wo", "c_proj", "attention.dense")
+                "fc2", "wo", "c_proj", "attention.dense", ".w2")
 # per-rank head attributes HF attention modules carry
 HEAD_ATTRS = ("num_heads", "num_attention_heads", "num_key_value_heads",
               "num_kv_heads", "embed_dim", "hidden_size", "split_size",
