@@ -216,6 +216,13 @@ def initialize_tensor_parallel(tp_size):
 
 
 def get_tensor_parallel_group():
+    if _TENSOR_PARALLEL_GROUP is None and _mpu is not None:
+        # Megatron-style mpu passed to initialize(): its model-parallel
+        # group IS the tensor-parallel group
+        for name in ("get_tensor_model_parallel_group",
+                     "get_model_parallel_group"):
+            if hasattr(_mpu, name):
+                return getattr(_mpu, name)()
     return _TENSOR_PARALLEL_GROUP
 
 
