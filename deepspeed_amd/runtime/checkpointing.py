@@ -26,6 +26,23 @@ def _mp_rank():
     return groups.get_tensor_parallel_rank()
 
 
+def _pp_stage(engine):
+    """Pipeline stage id when the module is a multi-stage
+    PipelineModule, else None — stages hold DIFFERENT layers, so each
+    writes stage-qualified files (ref PipelineModule layer_* files)."""
+    m = engine.module
+    if getattr(m, "num_stages", 1) > 1:
+        return m.stage_id
+    return None
+
+
+def _qualify(name, stage):
+    if stage is None:
+        return name
+    base, ext = name.rsplit(".", 1)
+    return f"{base}_pp_rank_{stage}.{ext}"
+
+
 def _model_states_name(mp_rank=0):
     return f"mp_rank_{mp_rank:02d}_model_states.pt"
 
@@ -142,16 +159,17 @@ def save_checkpoint(engine, save_dir, tag=None, client_state=None,
         if expert_names and state.get("module"):
             state["module"] = {k: v for k, v in state["module"].items()
                                if k not in expert_names}
-        torch.save(state, os.path.join(ckpt_dir,
-                                       _model_states_name(_mp_rank())))
+        torch.save(state, os.path.join(
+            ckpt_dir, _qualify(_model_states_name(_mp_rank()),
+                               _pp_stage(engine))))
 
     # zero shards: every dp rank
     if is_zero:
         zstate = {"optimizer_state_dict": engine.optimizer.state_dict(),
                   "ds_version": VERSION}
-        torch.save(zstate,
-                   os.path.join(ckpt_dir,
-                                _zero_ckpt_name(dp_rank, _mp_rank())))
+        torch.save(zstate, os.path.join(
+            ckpt_dir, _qualify(_zero_ckpt_name(dp_rank, _mp_rank()),
+                               _pp_stage(engine))))
 
     if dist.is_initialized():
         dist.barrier()
@@ -173,7 +191,9 @@ def load_checkpoint(engine, load_dir, tag=None, load_module_strict=True,
         with open(latest) as f:
             tag = f.read().strip()
     ckpt_dir = os.path.join(load_dir, str(tag))
-    model_file = os.path.join(ckpt_dir, _model_states_name(_mp_rank()))
+    model_file = os.path.join(
+        ckpt_dir, _qualify(_model_states_name(_mp_rank()),
+                           _pp_stage(engine)))
     state = torch.load(model_file, map_location="cpu", weights_only=False)
 
     is_zero = hasattr(engine.optimizer, "load_state_dict") and \
@@ -208,8 +228,9 @@ def load_checkpoint(engine, load_dir, tag=None, load_module_strict=True,
             # Stage 3: module weights live only in the zero shards — restore
             # them from the per-dp-rank zero file without optimizer state.
             dp_rank = engine.get_data_parallel_rank()
-            zfile = os.path.join(ckpt_dir,
-                                 _zero_ckpt_name(dp_rank, _mp_rank()))
+            zfile = os.path.join(
+                ckpt_dir, _qualify(_zero_ckpt_name(dp_rank, _mp_rank()),
+                                   _pp_stage(engine)))
             zstate = torch.load(zfile, map_location="cpu", weights_only=False)
             engine.optimizer.load_state_dict(zstate["optimizer_state_dict"],
                                              load_optimizer_states=False)
@@ -219,8 +240,9 @@ def load_checkpoint(engine, load_dir, tag=None, load_module_strict=True,
     if not load_module_only:
         if is_zero:
             dp_rank = engine.get_data_parallel_rank()
-            zfile = os.path.join(ckpt_dir,
-                                 _zero_ckpt_name(dp_rank, _mp_rank()))
+            zfile = os.path.join(
+                ckpt_dir, _qualify(_zero_ckpt_name(dp_rank, _mp_rank()),
+                                   _pp_stage(engine)))
             zstate = torch.load(zfile, map_location="cpu", weights_only=False)
             engine.optimizer.load_state_dict(
                 zstate["optimizer_state_dict"],

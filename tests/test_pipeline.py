@@ -277,3 +277,46 @@ def _pipe_train_zero2(steps=2):
 def test_pipeline_zero2_runs():
     results = run_distributed(_pipe_train_zero2, world_size=2)
     assert abs(results[0][0] - results[1][0]) < 1e-6
+
+
+def _pipe_ckpt_body(ckpt_dir):
+    """Each pipeline stage writes stage-qualified checkpoint files
+    (stages hold DIFFERENT layers) and resumes its own weights."""
+    import os
+    import torch
+    import torch.distributed as tdist
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    from deepspeed_amd.runtime.pipe.module import PipelineModule
+    groups.reset_groups()
+    model = PipelineModule(layers=_make_specs(), num_stages=2,
+                           loss_fn=_loss_fn, partition_method="uniform")
+    config = {"train_micro_batch_size_per_gpu": MICRO,
+              "gradient_accumulation_steps": GAS,
+              "optimizer": {"type": "AdamW", "params": {"lr": LR}},
+              "bf16": {"enabled": False}}
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config=config)
+    batches = _data(3 * GAS)
+    it = iter(batches)
+    engine.train_batch(it)
+    want = {n: p.detach().clone()
+            for n, p in engine.module.named_parameters()}
+    engine.save_checkpoint(ckpt_dir, tag="t0")
+    tdist.barrier()
+    if tdist.get_rank() == 0:
+        files = sorted(os.listdir(os.path.join(ckpt_dir, "t0")))
+        assert "mp_rank_00_model_states_pp_rank_0.pt" in files, files
+        assert "mp_rank_00_model_states_pp_rank_1.pt" in files, files
+    engine.train_batch(it)  # perturb
+    engine.load_checkpoint(ckpt_dir, tag="t0")
+    for n, p in engine.module.named_parameters():
+        assert torch.allclose(p.detach(), want[n], atol=1e-6), n
+    return True
+
+
+def test_pipeline_checkpoint_per_stage():
+    import tempfile
+    from tests.common import run_distributed
+    with tempfile.TemporaryDirectory() as d:
+        assert all(run_distributed(_pipe_ckpt_body, world_size=2,
+                                   args=(d,)))
