@@ -43,6 +43,11 @@ def _load_zero_shards(dirpath, mp_rank=None):
                     == f"{mp_rank:02d}"),
                    key=lambda f: int(os.path.basename(f).split("_")[3]))
     if not files:
+        if glob.glob(os.path.join(dirpath, "*_pp_rank_*.pt")):
+            raise NotImplementedError(
+                "pipeline-parallel checkpoint (stage-qualified files): "
+                "offline cross-stage reassembly is not implemented — "
+                "resume with the same pipeline topology instead")
         raise FileNotFoundError(f"no zero shard files in {dirpath}")
     return [torch.load(f, map_location="cpu", weights_only=False)
             ["optimizer_state_dict"] for f in files]
