@@ -188,8 +188,34 @@ class MoEConfig(DSConfigModel):
     ep_size: int = 1
 
 
+_KNOWN_TOP_KEYS = frozenset((
+    "train_batch_size", "train_micro_batch_size_per_gpu",
+    "gradient_accumulation_steps", "gradient_clipping",
+    "gradient_predivide_factor", "prescale_gradients", "steps_per_print",
+    "wall_clock_breakdown", "memory_breakdown", "dump_state",
+    "zero_optimization", "zero_allow_untested_optimizer", "bf16",
+    "bfloat16", "fp16", "amp", "torch_autocast", "data_types",
+    "communication_data_type", "optimizer", "scheduler",
+    "activation_checkpointing", "flops_profiler", "comms_logger",
+    "csv_monitor", "tensorboard", "wandb", "comet", "curriculum_learning",
+    "data_efficiency", "compression_training", "progressive_layer_drop",
+    "eigenvalue", "elasticity", "autotuning", "pipeline", "moe",
+    "sequence_parallel", "tensor_parallel", "checkpoint",
+    "checkpoint_tag_validation", "quantize_training", "monitor_config",
+))
+
+
 class DeepSpeedConfig:
     """Parsed top-level config. Accepts a dict or a JSON file path."""
+
+    @staticmethod
+    def _warn_unknown_keys(config):
+        unknown = [k for k in config
+                   if k not in _KNOWN_TOP_KEYS and not k.startswith("_")]
+        if unknown:
+            from .utils.logging import logger
+            logger.warning(
+                f"ds_config keys not recognized (typo?): {unknown}")
 
     def __init__(self, config: Union[str, dict], world_size: int = 1):
         if isinstance(config, str):
@@ -198,6 +224,7 @@ class DeepSpeedConfig:
         elif config is None:
             config = {}
         self._raw = dict(config)
+        self._warn_unknown_keys(config)
 
         self.train_batch_size = config.get("train_batch_size")
         self.train_micro_batch_size_per_gpu = config.get("train_micro_batch_size_per_gpu")
