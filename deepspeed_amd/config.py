@@ -260,6 +260,8 @@ class DeepSpeedConfig:
         self.csv_monitor = MonitorCSVConfig(**config.get("csv_monitor", {}))
         self.tensorboard = MonitorTensorBoardConfig(**config.get("tensorboard", {}))
         self.ulysses = UlyssesConfig(**config.get("sequence_parallel", {}))
+        self.quantize_training = dict(
+            config.get("quantize_training", {}) or {})
         tp_raw = config.get("tensor_parallel", {})
         self.tensor_parallel = TensorParallelConfig(
             autotp_size=tp_raw.get("autotp_size", tp_raw.get("tp_size", 1))

@@ -125,6 +125,8 @@ class DeepSpeedEngine(torch.nn.Module):
             self._configure_optimizer(optimizer, model_parameters)
             self._configure_lr_scheduler(lr_scheduler)
 
+        self.quantizer = self._configure_quantization()
+
         self.training_dataloader = (self.deepspeed_io(training_data)
                                     if training_data is not None else None)
 
@@ -654,8 +656,44 @@ class DeepSpeedEngine(torch.nn.Module):
             self.flops_profiler.end_profile()
             self.flops_profiler = None
 
+    def _configure_quantization(self):
+        """MoQ quantize-aware training (ref engine.py:2257): fake-int
+        quantize the 16-bit weights after each step with annealing bit
+        widths. Stage 1/2 quantizes the flat buckets; stage 0 the
+        module params; stage 3 is unsupported (sharded weights)."""
+        q = self._config.quantize_training
+        if not q or not q.get("enabled"):
+            return None
+        if self.zero_optimization_stage() == 3:
+            log_dist("quantize_training: stage 3 not supported; "
+                     "disabled", ranks=[0])
+            return None
+        from .quantize import Quantizer
+        bits = q.get("quantize_bits", {})
+        sched = q.get("schedule", {})
+        n_groups = len(getattr(self.optimizer, "buckets", [])) or \
+            sum(1 for p in self.module.parameters() if p.requires_grad)
+        return Quantizer(
+            q_groups=q.get("quantize_groups", 1),
+            q_verbose=q.get("quantize_verbose", False),
+            start_bits=bits.get("start_bits", 16),
+            target_bits=bits.get("target_bits", 8),
+            quantize_period=sched.get("quantize_period", 1000),
+            layer_num=n_groups)
+
+    def _apply_moq(self):
+        overflow = getattr(self.optimizer, "overflow", False)
+        if hasattr(self.optimizer, "buckets"):  # ZeRO 1/2 flat slabs
+            tensors = [b.flat16 for b in self.optimizer.buckets]
+        else:
+            tensors = [p for p in self.module.parameters()
+                       if p.requires_grad and p.dim() >= 2]
+        self.quantizer.quantize(tensors, overflow=overflow)
+
     def _take_model_step(self, lr_kwargs=None):
         self.optimizer.step()
+        if self.quantizer is not None:
+            self._apply_moq()
         if self.random_ltd_scheduler is not None:
             self.random_ltd_scheduler.update(self.global_steps + 1)
         from ..ops.fp8_linear import bump_fp8_version

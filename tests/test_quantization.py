@@ -177,3 +177,42 @@ def test_fp6_roundtrip_packed():
     y = fpq.dequantize(q2)
     y_ref = fpq.dequantize(q)
     assert torch.equal(y, y_ref)
+
+
+def test_moq_quantize_training_config():
+    """quantize_training config: weights fake-quantize after steps once
+    the annealing schedule reaches target bits (ref engine.py:2257)."""
+    import torch
+    import torch.distributed as tdist
+    import deepspeed_amd
+    from deepspeed_amd.comm import groups
+    groups.reset_groups()
+    if not tdist.is_initialized():
+        tdist.init_process_group("gloo",
+                                 init_method="tcp://127.0.0.1:29641",
+                                 rank=0, world_size=1)
+    torch.manual_seed(0)
+    model = torch.nn.Sequential(torch.nn.Linear(32, 32), torch.nn.Tanh(),
+                                torch.nn.Linear(32, 8))
+    engine, _, _, _ = deepspeed_amd.initialize(model=model, config={
+        "train_micro_batch_size_per_gpu": 4,
+        "optimizer": {"type": "AdamW", "params": {"lr": 1e-3}},
+        "bf16": {"enabled": True},
+        "zero_optimization": {"stage": 2},
+        "quantize_training": {
+            "enabled": True,
+            "quantize_bits": {"start_bits": 16, "target_bits": 8},
+            "schedule": {"quantize_period": 1},
+        }})
+    assert engine.quantizer is not None
+    x = torch.randn(4, 32).bfloat16()
+    y = torch.randn(4, 8).bfloat16()
+    for _ in range(4):  # period 1: bits 16 -> 8 after first steps
+        loss = (engine(x) - y).float().pow(2).mean()
+        engine.backward(loss)
+        engine.step()
+    assert engine.quantizer.bits[0] == 8
+    # int8 fake-quant leaves at most 257 distinct levels per bucket
+    for b in engine.optimizer.buckets:
+        assert b.flat16.unique().numel() <= 257, b.flat16.unique().numel()
+    assert torch.isfinite(torch.tensor(loss.item()))
