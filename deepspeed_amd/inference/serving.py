@@ -125,6 +125,8 @@ class ContinuousBatchingEngine:
         # prompt per step (legacy behavior). prefill_budget additionally
         # lets SEVERAL requests advance per step until the token budget
         # is spent (defaults to one chunk of one request).
+        if prefill_budget and not prefill_chunk:
+            prefill_chunk = prefill_budget  # budget implies chunking
         self.prefill_chunk = prefill_chunk
         self.prefill_budget = prefill_budget
         self.prefilling: List[Request] = []
