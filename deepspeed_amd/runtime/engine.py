@@ -958,10 +958,7 @@ class DeepSpeedEngine(torch.nn.Module):
         return groups.get_sequence_parallel_group()
 
     def get_model_parallel_rank(self):
-        try:
-            return dist.get_rank(groups.get_tensor_parallel_group())
-        except Exception:
-            return 0
+        return groups.get_tensor_parallel_rank()  # 0 when no TP group
 
     get_tensor_parallel_rank = get_model_parallel_rank
 
