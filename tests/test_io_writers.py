@@ -84,3 +84,23 @@ def test_fast_writer_single_buffer_aio(tmp_path):
         w.write(data[i:i + 1234])
     w.close()
     assert open(p, "rb").read() == data
+
+
+def test_fast_writer_randomized_chunks(tmp_path):
+    """Property-style: random chunk sequences across buffer boundaries
+    always reproduce the byte stream exactly."""
+    import random
+    rng = random.Random(7)
+    for trial in range(6):
+        p = str(tmp_path / f"r{trial}.bin")
+        total = rng.randrange(1, 60000)
+        data = bytes(rng.getrandbits(8) for _ in range(total))
+        w = FastFileWriter(p, buffer_bytes=8192,
+                           double_buffer=bool(trial % 2))
+        i = 0
+        while i < total:
+            n = rng.randrange(1, 5000)
+            w.write(data[i:i + n])
+            i += n
+        w.close()
+        assert open(p, "rb").read() == data, (trial, total)
