@@ -216,3 +216,23 @@ def test_moq_quantize_training_config():
     for b in engine.optimizer.buckets:
         assert b.flat16.unique().numel() <= 257, b.flat16.unique().numel()
     assert torch.isfinite(torch.tensor(loss.item()))
+
+
+def test_moq_eigenvalue_modulated_annealing():
+    """Layers with larger Hessian eigenvalues stretch their period and
+    anneal LATER (MoQ scheduling rule)."""
+    from deepspeed_amd.runtime.quantize import Quantizer
+    q = Quantizer(layer_num=2, start_bits=16, target_bits=8,
+                  quantize_period=4, q_eigenvalue=True)
+    eig = {0: 0.1, 1: 1.0}  # layer 1 is 10x more sensitive
+    import torch
+    params = [[torch.randn(8, 8)], [torch.randn(8, 8)]]
+    for _ in range(7):
+        q.quantize(params, eigenvalue_enabled=True, block_eigenvalue=eig)
+    # after 7 steps: layer 0 (period ~4.4) has halved once; layer 1
+    # (period 8) has not
+    assert q.bits[0] < 16 and q.bits[1] == 16, q.bits
+    for _ in range(10):
+        q.quantize(params, eigenvalue_enabled=True, block_eigenvalue=eig)
+    assert q.bits == [8, 8], q.bits  # both reach target eventually
+    assert not q.any_precision_switch()
