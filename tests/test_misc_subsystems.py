@@ -392,3 +392,24 @@ def test_on_device_meta_init():
     assert m.weight.dtype == torch.bfloat16
     assert torch.get_default_dtype() == torch.float32
     assert torch.zeros(1).device.type == "cpu"
+
+
+def test_clip_grad_norm_reference_compat():
+    """runtime.utils.clip_grad_norm_ clips in place and returns the
+    pre-clip global norm (ref runtime/utils.py:359)."""
+    import torch
+    from deepspeed_amd.runtime.utils import clip_grad_norm_
+    a = torch.nn.Parameter(torch.ones(4))
+    b = torch.nn.Parameter(torch.ones(3))
+    a.grad = torch.full((4,), 3.0)
+    b.grad = torch.full((3,), 4.0)
+    total = (9 * 4 + 16 * 3) ** 0.5
+    n = clip_grad_norm_([a, b], max_norm=1.0)
+    assert abs(n - total) < 1e-5
+    got = (a.grad.pow(2).sum() + b.grad.pow(2).sum()).sqrt().item()
+    assert abs(got - 1.0) < 1e-4
+    # under the norm: untouched
+    a.grad = torch.full((4,), 0.01)
+    b.grad = None
+    n2 = clip_grad_norm_([a, b], max_norm=1.0)
+    assert abs(a.grad[0].item() - 0.01) < 1e-9 and n2 < 1.0
