@@ -413,3 +413,14 @@ def test_clip_grad_norm_reference_compat():
     b.grad = None
     n2 = clip_grad_norm_([a, b], max_norm=1.0)
     assert abs(a.grad[0].item() - 0.01) < 1e-9 and n2 < 1.0
+
+
+def test_utils_groups_alias_sees_live_state():
+    """deepspeed_amd.utils.groups mirrors comm.groups including
+    MODULE-LEVEL state mutated after import (reference import path)."""
+    from deepspeed_amd.comm import groups as real
+    from deepspeed_amd.utils import groups as alias
+    assert alias.get_tensor_parallel_world_size \
+        is real.get_tensor_parallel_world_size
+    # private state resolves dynamically through __getattr__
+    assert alias._TENSOR_PARALLEL_GROUP is real._TENSOR_PARALLEL_GROUP
